@@ -1,0 +1,51 @@
+# faabric-mi355x build: C++20 core + gfx950 HIP kernels + pybind11 module.
+# Everything is compiled with hipcc (amdclang++) so HIP-using translation
+# units and host-only ones share one toolchain; device code targets gfx950
+# ONLY (MI355X/CDNA4) — no multi-arch fatbins, no CUDA paths.
+
+HIPCC ?= /opt/rocm/bin/hipcc
+GPU_ARCH ?= gfx950
+
+PY := python3
+PY_INC := $(shell $(PY) -c "import sysconfig; print(sysconfig.get_paths()['include'])")
+PYBIND_INC := $(shell $(PY) -c "import pybind11; print(pybind11.get_include())")
+EXT_SUFFIX := $(shell $(PY) -c "import sysconfig; print(sysconfig.get_config_var('EXT_SUFFIX'))")
+
+BUILD := build
+TARGET := faabric_amd/_core$(EXT_SUFFIX)
+
+CXXFLAGS := -O2 -g -std=c++20 -fPIC -Wall -Wno-unused-function \
+            -Icpp/include -I$(PY_INC) -I$(PYBIND_INC) \
+            --offload-arch=$(GPU_ARCH) -fvisibility=hidden
+LDFLAGS := -shared -L/opt/rocm/lib -lrccl -lamdhip64 -lz -pthread
+
+CPP_SRCS := $(wildcard cpp/src/*.cpp) $(wildcard cpp/bindings/*.cpp)
+HIP_SRCS := $(wildcard cpp/hip/*.hip)
+
+CPP_OBJS := $(patsubst cpp/%.cpp,$(BUILD)/%.o,$(CPP_SRCS))
+HIP_OBJS := $(patsubst cpp/%.hip,$(BUILD)/%.o,$(HIP_SRCS))
+OBJS := $(CPP_OBJS) $(HIP_OBJS)
+
+all: $(TARGET)
+
+$(BUILD)/%.o: cpp/%.cpp
+	@mkdir -p $(dir $@)
+	$(HIPCC) $(CXXFLAGS) -c $< -o $@
+
+$(BUILD)/%.o: cpp/%.hip
+	@mkdir -p $(dir $@)
+	$(HIPCC) $(CXXFLAGS) -x hip -c $< -o $@
+
+$(TARGET): $(OBJS)
+	$(HIPCC) $(OBJS) $(LDFLAGS) -o $@
+
+# C++ example embedder (reference examples/check.cpp parity)
+examples: $(TARGET)
+	$(HIPCC) -O2 -std=c++20 -Icpp/include examples/check.cpp \
+	    $(filter-out $(BUILD)/bindings/%,$(OBJS)) \
+	    -L/opt/rocm/lib -lrccl -lamdhip64 -lz -pthread -o $(BUILD)/check
+
+clean:
+	rm -rf $(BUILD) $(TARGET)
+
+.PHONY: all clean examples

@@ -1,0 +1,314 @@
+// Concurrency primitives: blocking queue with timeout, token pool, latch,
+// barrier, flag-waiter, periodic background thread.
+//
+// MI355X-native equivalents of the reference's util concurrency layer
+// (reference: include/faabric/util/queue.h:25,148,220,245,
+//  util/latch.h:11-34, src/util/barrier.cpp,
+//  transport/PointToPointBroker.h:165-168 FlagWaiter,
+//  util/PeriodicBackgroundThread.h:15-44). Re-designed, not ported:
+// std::condition_variable throughout, no third-party lock-free deps.
+#pragma once
+
+#include <chrono>
+#include <condition_variable>
+#include <deque>
+#include <functional>
+#include <memory>
+#include <mutex>
+#include <stdexcept>
+#include <thread>
+
+#include "faabricamd/util.h"
+
+namespace faabricamd {
+
+inline constexpr int DEFAULT_QUEUE_TIMEOUT_MS = 5000;
+
+class QueueTimeoutException : public FaabricException
+{
+  public:
+    using FaabricException::FaabricException;
+};
+
+template<typename T>
+class Queue
+{
+  public:
+    void enqueue(T value)
+    {
+        {
+            std::lock_guard<std::mutex> lock(mx);
+            q.emplace_back(std::move(value));
+        }
+        cv.notify_one();
+    }
+
+    T dequeue(int timeoutMs = DEFAULT_QUEUE_TIMEOUT_MS)
+    {
+        std::unique_lock<std::mutex> lock(mx);
+        if (timeoutMs <= 0) {
+            cv.wait(lock, [this] { return !q.empty(); });
+        } else if (!cv.wait_for(lock,
+                                std::chrono::milliseconds(timeoutMs),
+                                [this] { return !q.empty(); })) {
+            throw QueueTimeoutException("queue dequeue timed out");
+        }
+        T v = std::move(q.front());
+        q.pop_front();
+        return v;
+    }
+
+    bool tryDequeue(T& out)
+    {
+        std::lock_guard<std::mutex> lock(mx);
+        if (q.empty()) {
+            return false;
+        }
+        out = std::move(q.front());
+        q.pop_front();
+        return true;
+    }
+
+    size_t size()
+    {
+        std::lock_guard<std::mutex> lock(mx);
+        return q.size();
+    }
+
+    void drain()
+    {
+        std::lock_guard<std::mutex> lock(mx);
+        q.clear();
+    }
+
+  private:
+    std::mutex mx;
+    std::condition_variable cv;
+    std::deque<T> q;
+};
+
+// Pool of integer tokens (reference: util/queue.h:245 TokenPool)
+class TokenPool
+{
+  public:
+    explicit TokenPool(int nTokens)
+    {
+        for (int i = 0; i < nTokens; i++) {
+            tokens.enqueue(i);
+        }
+        nTotal = nTokens;
+    }
+
+    int getToken(int timeoutMs = DEFAULT_QUEUE_TIMEOUT_MS)
+    {
+        return tokens.dequeue(timeoutMs);
+    }
+
+    void releaseToken(int token) { tokens.enqueue(token); }
+    int size() const { return nTotal; }
+
+  private:
+    Queue<int> tokens;
+    int nTotal = 0;
+};
+
+// Countdown latch with timeout (reference: util/latch.h:11-34 — shared_ptr
+// only, wait() with timeout)
+class Latch
+{
+  public:
+    static std::shared_ptr<Latch> create(
+      int countIn,
+      int timeoutMsIn = DEFAULT_QUEUE_TIMEOUT_MS)
+    {
+        return std::make_shared<Latch>(countIn, timeoutMsIn);
+    }
+
+    Latch(int countIn, int timeoutMsIn)
+      : count(countIn)
+      , timeoutMs(timeoutMsIn)
+    {}
+
+    void wait()
+    {
+        std::unique_lock<std::mutex> lock(mx);
+        waiters++;
+        if (waiters > count) {
+            throw FaabricException("latch already used");
+        }
+        if (!cv.wait_for(lock, std::chrono::milliseconds(timeoutMs), [this] {
+                return waiters >= count;
+            })) {
+            throw QueueTimeoutException("latch wait timed out");
+        }
+        cv.notify_all();
+    }
+
+  private:
+    int count;
+    int waiters = 0;
+    int timeoutMs;
+    std::mutex mx;
+    std::condition_variable cv;
+};
+
+// Reusable barrier with an optional completion function
+// (reference: src/util/barrier.cpp)
+class Barrier
+{
+  public:
+    static std::shared_ptr<Barrier> create(
+      int count,
+      std::function<void()> completionIn = nullptr,
+      int timeoutMsIn = DEFAULT_QUEUE_TIMEOUT_MS)
+    {
+        return std::make_shared<Barrier>(count, completionIn, timeoutMsIn);
+    }
+
+    Barrier(int countIn, std::function<void()> completionIn, int timeoutMsIn)
+      : count(countIn)
+      , completion(std::move(completionIn))
+      , timeoutMs(timeoutMsIn)
+    {}
+
+    void wait()
+    {
+        std::unique_lock<std::mutex> lock(mx);
+        int phaseAtEntry = phase;
+        arrived++;
+        if (arrived == count) {
+            if (completion) {
+                completion();
+            }
+            arrived = 0;
+            phase++;
+            cv.notify_all();
+            return;
+        }
+        if (!cv.wait_for(lock, std::chrono::milliseconds(timeoutMs), [&] {
+                return phase != phaseAtEntry;
+            })) {
+            throw QueueTimeoutException("barrier wait timed out");
+        }
+    }
+
+  private:
+    int count;
+    int arrived = 0;
+    int phase = 0;
+    std::function<void()> completion;
+    int timeoutMs;
+    std::mutex mx;
+    std::condition_variable cv;
+};
+
+// Set-once flag other threads can block on, used for PTP mapping waits
+// (reference: transport/PointToPointBroker.h:165-168)
+class FlagWaiter
+{
+  public:
+    explicit FlagWaiter(int timeoutMsIn = DEFAULT_QUEUE_TIMEOUT_MS)
+      : timeoutMs(timeoutMsIn)
+    {}
+
+    void waitOnFlag()
+    {
+        std::unique_lock<std::mutex> lock(mx);
+        if (!cv.wait_for(lock, std::chrono::milliseconds(timeoutMs), [this] {
+                return flag;
+            })) {
+            throw QueueTimeoutException("flag wait timed out");
+        }
+    }
+
+    void setFlag(bool value)
+    {
+        {
+            std::lock_guard<std::mutex> lock(mx);
+            flag = value;
+        }
+        cv.notify_all();
+    }
+
+    bool isSet()
+    {
+        std::lock_guard<std::mutex> lock(mx);
+        return flag;
+    }
+
+  private:
+    int timeoutMs;
+    bool flag = false;
+    std::mutex mx;
+    std::condition_variable cv;
+};
+
+// Base class running doWork() every intervalSeconds until stopped
+// (reference: util/PeriodicBackgroundThread.h:15-44)
+class PeriodicBackgroundThread
+{
+  public:
+    virtual ~PeriodicBackgroundThread() { stop(); }
+
+    void start(int intervalSecondsIn)
+    {
+        intervalMs = intervalSecondsIn * 1000;
+        startMillis(intervalMs);
+    }
+
+    void startMillis(int intervalMsIn)
+    {
+        intervalMs = intervalMsIn;
+        stopped = false;
+        worker = std::thread([this] {
+            std::unique_lock<std::mutex> lock(mx);
+            while (!stopped) {
+                if (cv.wait_for(lock,
+                                std::chrono::milliseconds(intervalMs),
+                                [this] { return stopped; })) {
+                    break;
+                }
+                lock.unlock();
+                try {
+                    doWork();
+                } catch (const std::exception& e) {
+                    FAM_ERROR("periodic thread error: %s", e.what());
+                }
+                lock.lock();
+            }
+        });
+    }
+
+    void stop()
+    {
+        {
+            std::lock_guard<std::mutex> lock(mx);
+            if (stopped) {
+                if (worker.joinable()) {
+                    worker.join();
+                }
+                return;
+            }
+            stopped = true;
+        }
+        cv.notify_all();
+        if (worker.joinable()) {
+            worker.join();
+        }
+        tidyUp();
+    }
+
+    virtual void doWork() = 0;
+    virtual void tidyUp() {}
+
+  protected:
+    int intervalMs = 1000;
+
+  private:
+    std::thread worker;
+    std::mutex mx;
+    std::condition_variable cv;
+    bool stopped = true;
+};
+
+} // namespace faabricamd
