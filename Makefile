@@ -16,7 +16,7 @@ TARGET := faabric_amd/_core$(EXT_SUFFIX)
 
 CXXFLAGS := -O2 -g -std=c++20 -fPIC -Wall -Wno-unused-function \
             -Icpp/include -I$(PY_INC) -I$(PYBIND_INC) \
-            --offload-arch=$(GPU_ARCH) -fvisibility=hidden
+            --offload-arch=$(GPU_ARCH) -fvisibility=hidden -MMD -MP
 LDFLAGS := -shared -L/opt/rocm/lib -lrccl -lamdhip64 -lz -pthread
 
 CPP_SRCS := $(wildcard cpp/src/*.cpp) $(wildcard cpp/bindings/*.cpp)
@@ -47,5 +47,7 @@ examples: $(TARGET)
 
 clean:
 	rm -rf $(BUILD) $(TARGET)
+
+-include $(OBJS:.o=.d)
 
 .PHONY: all clean examples

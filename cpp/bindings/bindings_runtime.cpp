@@ -1,10 +1,468 @@
-// Runtime bindings: planner / scheduler / executor / MPI / state / snapshot.
-// Filled in as each subsystem lands.
+// Runtime bindings: planner / scheduler / executor / state / snapshot / ptp.
+#include <pybind11/functional.h>
 #include <pybind11/pybind11.h>
+#include <pybind11/stl.h>
+
+#include "faabricamd/executor.h"
+#include "faabricamd/planner.h"
+#include "faabricamd/ptp.h"
+#include "faabricamd/runner.h"
+#include "faabricamd/scheduler.h"
+#include "faabricamd/snapshot.h"
+#include "faabricamd/state.h"
+#include "faabricamd/util.h"
 
 namespace py = pybind11;
+using namespace faabricamd;
 
 void initRuntimeBindings(py::module_& m)
 {
-    (void)m;
+    // ---------------- config ----------------
+    m.def("set_endpoint_host", [](const std::string& host) {
+        getSystemConfig().endpointHost = host;
+    });
+    m.def("get_endpoint_host",
+          [] { return getSystemConfig().endpointHost; });
+    m.def("set_planner_host", [](const std::string& host) {
+        getSystemConfig().plannerHost = host;
+        resetPlannerClient();
+    });
+    m.def("set_batch_scheduler_mode", [](const std::string& mode) {
+        resetBatchScheduler(mode);
+    });
+
+    // ---------------- scheduling decision ----------------
+    py::class_<SchedulingDecision>(m, "SchedulingDecision")
+      .def(py::init<>())
+      .def_readwrite("app_id", &SchedulingDecision::appId)
+      .def_readwrite("group_id", &SchedulingDecision::groupId)
+      .def_readwrite("n_functions", &SchedulingDecision::nFunctions)
+      .def_readwrite("hosts", &SchedulingDecision::hosts)
+      .def_readwrite("message_ids", &SchedulingDecision::messageIds)
+      .def_readwrite("app_idxs", &SchedulingDecision::appIdxs)
+      .def_readwrite("group_idxs", &SchedulingDecision::groupIdxs)
+      .def_readwrite("mpi_ports", &SchedulingDecision::mpiPorts);
+
+    m.def("NOT_ENOUGH_SLOTS", [] { return NOT_ENOUGH_SLOTS; });
+    m.def("DO_NOT_MIGRATE", [] { return DO_NOT_MIGRATE; });
+    m.def("MUST_FREEZE", [] { return MUST_FREEZE; });
+
+    // ---------------- runtimes ----------------
+    py::class_<PlannerRuntime>(m, "PlannerRuntime")
+      .def(py::init<>())
+      .def("start",
+           &PlannerRuntime::start,
+           py::arg("with_snapshot_server") = true,
+           py::call_guard<py::gil_scoped_release>())
+      .def("shutdown",
+           &PlannerRuntime::shutdown,
+           py::call_guard<py::gil_scoped_release>());
+
+    py::class_<FaabricMain>(m, "FaabricMain")
+      .def(py::init([]() {
+          return new FaabricMain(getExecutorFactory());
+      }))
+      .def("start_background",
+           &FaabricMain::startBackground,
+           py::call_guard<py::gil_scoped_release>())
+      .def("shutdown",
+           &FaabricMain::shutdown,
+           py::call_guard<py::gil_scoped_release>());
+
+    // ---------------- function registry ----------------
+    m.def("register_function",
+          [](const std::string& user,
+             const std::string& function,
+             py::function fn) {
+              FunctionRegistry::get().registerFunction(
+                user, function, [fn](Message& msg) -> int32_t {
+                    py::gil_scoped_acquire gil;
+                    py::object result = fn(&msg);
+                    if (result.is_none()) {
+                        return 0;
+                    }
+                    return result.cast<int32_t>();
+                });
+          });
+    m.def("clear_function_registry",
+          [] { FunctionRegistry::get().clear(); });
+
+    // Native no-op / sleep functions for benchmarking the scheduling path
+    // without touching Python from executor threads
+    m.def("register_native_noop",
+          [](const std::string& user, const std::string& function) {
+              FunctionRegistry::get().registerFunction(
+                user, function, [](Message&) -> int32_t { return 0; });
+          });
+    m.def("register_native_echo",
+          [](const std::string& user, const std::string& function) {
+              FunctionRegistry::get().registerFunction(
+                user, function, [](Message& msg) -> int32_t {
+                    msg.outputData.assign(msg.inputData.begin(),
+                                          msg.inputData.end());
+                    return 0;
+                });
+          });
+    m.def("register_native_sleep",
+          [](const std::string& user,
+             const std::string& function,
+             int sleepMs) {
+              FunctionRegistry::get().registerFunction(
+                user, function, [sleepMs](Message&) -> int32_t {
+                    std::this_thread::sleep_for(
+                      std::chrono::milliseconds(sleepMs));
+                    return 0;
+                });
+          });
+
+    // ---------------- planner client ----------------
+    m.def(
+      "call_functions",
+      [](const BatchExecuteRequest& reqIn) {
+          auto req = std::make_shared<BatchExecuteRequest>(reqIn);
+          py::gil_scoped_release release;
+          return *getPlannerClient().callFunctions(req);
+      });
+    m.def(
+      "get_message_result",
+      [](int32_t appId, int32_t msgId, int timeoutMs) {
+          py::gil_scoped_release release;
+          return getPlannerClient().getMessageResult(appId, msgId, timeoutMs);
+      });
+    m.def("get_batch_results", [](int32_t appId) {
+        py::gil_scoped_release release;
+        return getPlannerClient().getBatchResults(appId);
+    });
+    m.def("get_available_hosts", [] {
+        py::gil_scoped_release release;
+        return getPlannerClient().getAvailableHosts();
+    });
+    m.def("get_num_migrations", [] {
+        py::gil_scoped_release release;
+        return getPlannerClient().getNumMigrations();
+    });
+    m.def("preload_scheduling_decision",
+          [](int32_t appId, const SchedulingDecision& decision) {
+              py::gil_scoped_release release;
+              getPlannerClient().preloadSchedulingDecision(appId, decision);
+          });
+    m.def("planner_ping", [] {
+        py::gil_scoped_release release;
+        getPlannerClient().ping();
+    });
+
+    // Direct planner access (valid in the planner process only)
+    m.def("planner_reset", [] { Planner::get().reset(); });
+    m.def("planner_flush_scheduling_state",
+          [] { Planner::get().flushSchedulingState(); });
+    m.def("planner_set_next_evicted_vms",
+          [](const std::vector<std::string>& ips) {
+              Planner::get().setNextEvictedVm(
+                { ips.begin(), ips.end() });
+          });
+    m.def("planner_set_policy",
+          [](const std::string& p) { Planner::get().setPolicy(p); });
+    m.def("planner_get_policy", [] { return Planner::get().getPolicy(); });
+    m.def("planner_num_in_flight_apps", [] {
+        return (int)Planner::get().getInFlightApps().apps.size();
+    });
+
+    // ---------------- scheduler ----------------
+    m.def("set_this_host_resources", [](int slots, int usedSlots) {
+        HostResources res;
+        res.slots = slots;
+        res.usedSlots = usedSlots;
+        Scheduler::get().setThisHostResources(res);
+    });
+    m.def("get_this_host_slots", [] {
+        return Scheduler::get().getThisHostResources().slots;
+    });
+    m.def("get_executor_count",
+          [] { return (int)Scheduler::get().getExecutorCount(); });
+    m.def("scheduler_reset", [] {
+        py::gil_scoped_release release;
+        Scheduler::get().reset();
+    });
+    m.def("get_recorded_messages",
+          [] { return Scheduler::get().getRecordedMessages(); });
+    m.def("set_test_mode", &setTestMode);
+
+    // ---------------- point-to-point ----------------
+    m.def("ptp_send",
+          [](int32_t appId,
+             int32_t groupId,
+             int32_t sendIdx,
+             int32_t recvIdx,
+             const py::bytes& data,
+             bool ordered) {
+              std::string s = data;
+              py::gil_scoped_release release;
+              getPointToPointBroker().sendMessage(appId,
+                                                  groupId,
+                                                  sendIdx,
+                                                  recvIdx,
+                                                  (const uint8_t*)s.data(),
+                                                  s.size(),
+                                                  ordered);
+          },
+          py::arg("app_id"),
+          py::arg("group_id"),
+          py::arg("send_idx"),
+          py::arg("recv_idx"),
+          py::arg("data"),
+          py::arg("ordered") = false);
+    m.def("ptp_recv",
+          [](int32_t groupId,
+             int32_t sendIdx,
+             int32_t recvIdx,
+             bool ordered,
+             int timeoutMs) {
+              std::vector<uint8_t> out;
+              {
+                  py::gil_scoped_release release;
+                  out = getPointToPointBroker().recvMessage(
+                    groupId, sendIdx, recvIdx, ordered, timeoutMs);
+              }
+              return py::bytes((const char*)out.data(), out.size());
+          },
+          py::arg("group_id"),
+          py::arg("send_idx"),
+          py::arg("recv_idx"),
+          py::arg("ordered") = false,
+          py::arg("timeout_ms") = DEFAULT_QUEUE_TIMEOUT_MS);
+    m.def("ptp_group_barrier", [](int32_t groupId, int32_t groupIdx) {
+        py::gil_scoped_release release;
+        PointToPointGroup::getOrAwaitGroup(groupId)->barrier(groupIdx);
+    });
+    m.def("ptp_group_lock",
+          [](int32_t groupId, int32_t groupIdx, bool recursive) {
+              py::gil_scoped_release release;
+              PointToPointGroup::getOrAwaitGroup(groupId)->lock(groupIdx,
+                                                                recursive);
+          },
+          py::arg("group_id"),
+          py::arg("group_idx"),
+          py::arg("recursive") = false);
+    m.def("ptp_group_unlock",
+          [](int32_t groupId, int32_t groupIdx, bool recursive) {
+              py::gil_scoped_release release;
+              PointToPointGroup::getOrAwaitGroup(groupId)->unlock(groupIdx,
+                                                                  recursive);
+          },
+          py::arg("group_id"),
+          py::arg("group_idx"),
+          py::arg("recursive") = false);
+    m.def("ptp_group_notify", [](int32_t groupId, int32_t groupIdx) {
+        py::gil_scoped_release release;
+        PointToPointGroup::getOrAwaitGroup(groupId)->notify(groupIdx);
+    });
+    m.def("ptp_setup_local_mappings", [](const SchedulingDecision& d) {
+        getPointToPointBroker().setUpLocalMappingsFromSchedulingDecision(d);
+    });
+    m.def("ptp_clear", [] { getPointToPointBroker().clear(); });
+    m.def("ptp_wait_for_mappings", [](int32_t groupId) {
+        py::gil_scoped_release release;
+        getPointToPointBroker().waitForMappingsOnThisHost(groupId);
+    });
+
+    // ---------------- state ----------------
+    py::class_<StateKeyValue, std::shared_ptr<StateKeyValue>>(m,
+                                                              "StateKeyValue")
+      .def_property_readonly("user", &StateKeyValue::getUser)
+      .def_property_readonly("key", &StateKeyValue::getKey)
+      .def_property_readonly("size", &StateKeyValue::size)
+      .def_property_readonly("is_master", &StateKeyValue::isMaster)
+      .def_property_readonly("master_host", &StateKeyValue::getMasterHost)
+      .def("get",
+           [](StateKeyValue& kv) {
+               std::vector<uint8_t> out;
+               {
+                   py::gil_scoped_release release;
+                   out = kv.get();
+               }
+               return py::bytes((const char*)out.data(), out.size());
+           })
+      .def("set",
+           [](StateKeyValue& kv, const py::bytes& data) {
+               std::string s = data;
+               py::gil_scoped_release release;
+               kv.set((const uint8_t*)s.data(), s.size());
+           })
+      .def("get_chunk",
+           [](StateKeyValue& kv, uint64_t offset, size_t len) {
+               std::vector<uint8_t> out(len);
+               {
+                   py::gil_scoped_release release;
+                   kv.getChunk(offset, out.data(), len);
+               }
+               return py::bytes((const char*)out.data(), out.size());
+           })
+      .def("set_chunk",
+           [](StateKeyValue& kv, uint64_t offset, const py::bytes& data) {
+               std::string s = data;
+               py::gil_scoped_release release;
+               kv.setChunk(offset, (const uint8_t*)s.data(), s.size());
+           })
+      .def("append",
+           [](StateKeyValue& kv, const py::bytes& data) {
+               std::string s = data;
+               py::gil_scoped_release release;
+               kv.append((const uint8_t*)s.data(), s.size());
+           })
+      .def("get_appended",
+           [](StateKeyValue& kv, size_t n) {
+               std::vector<std::vector<uint8_t>> vals;
+               {
+                   py::gil_scoped_release release;
+                   vals = kv.getAppended(n);
+               }
+               py::list out;
+               for (auto& v : vals) {
+                   out.append(py::bytes((const char*)v.data(), v.size()));
+               }
+               return out;
+           })
+      .def("clear_appended", [](StateKeyValue& kv) {
+          py::gil_scoped_release release;
+          kv.clearAppended();
+      })
+      .def("pull",
+           [](StateKeyValue& kv) {
+               py::gil_scoped_release release;
+               kv.pull();
+           })
+      .def("push_full", [](StateKeyValue& kv) {
+          py::gil_scoped_release release;
+          kv.pushFull();
+      });
+
+    m.def("state_get_kv",
+          [](const std::string& user, const std::string& key, size_t size) {
+              return State::get().getKV(user, key, size);
+          });
+    m.def("state_set_master_host",
+          [](const std::string& user,
+             const std::string& key,
+             const std::string& host) {
+              State::get().setMasterHost(user, key, host);
+          });
+    m.def("state_size",
+          [](const std::string& user, const std::string& key) {
+              py::gil_scoped_release release;
+              return State::get().getStateSize(user, key);
+          });
+    m.def("state_clear_all", [] { State::get().forceClearAll(false); });
+    m.def("state_kv_count", [] { return State::get().getKVCount(); });
+
+    // ---------------- snapshots ----------------
+    py::enum_<SnapshotDataType>(m, "SnapshotDataType")
+      .value("Raw", SnapshotDataType::Raw)
+      .value("Bool", SnapshotDataType::Bool)
+      .value("Int", SnapshotDataType::Int)
+      .value("Long", SnapshotDataType::Long)
+      .value("Float", SnapshotDataType::Float)
+      .value("Double", SnapshotDataType::Double);
+
+    py::enum_<SnapshotMergeOperation>(m, "SnapshotMergeOperation")
+      .value("Bytewise", SnapshotMergeOperation::Bytewise)
+      .value("Sum", SnapshotMergeOperation::Sum)
+      .value("Product", SnapshotMergeOperation::Product)
+      .value("Subtract", SnapshotMergeOperation::Subtract)
+      .value("Max", SnapshotMergeOperation::Max)
+      .value("Min", SnapshotMergeOperation::Min)
+      .value("XOR", SnapshotMergeOperation::XOR);
+
+    py::class_<SnapshotDiff>(m, "SnapshotDiff")
+      .def(py::init<>())
+      .def_readwrite("data_type", &SnapshotDiff::dataType)
+      .def_readwrite("operation", &SnapshotDiff::operation)
+      .def_readwrite("offset", &SnapshotDiff::offset)
+      .def_property(
+        "data",
+        [](const SnapshotDiff& d) {
+            return py::bytes((const char*)d.getData(), d.size());
+        },
+        [](SnapshotDiff& d, const py::bytes& b) {
+            std::string s = b;
+            d.dataCopy.assign(s.begin(), s.end());
+        });
+
+    py::class_<SnapshotData, std::shared_ptr<SnapshotData>>(m,
+                                                            "SnapshotData")
+      .def(py::init([](const py::bytes& data, size_t maxSize) {
+               std::string s = data;
+               std::vector<uint8_t> v(s.begin(), s.end());
+               if (maxSize == 0) {
+                   return std::make_shared<SnapshotData>(v);
+               }
+               return std::make_shared<SnapshotData>(v, maxSize);
+           }),
+           py::arg("data"),
+           py::arg("max_size") = 0)
+      .def(py::init([](size_t size) {
+          return std::make_shared<SnapshotData>(size);
+      }))
+      .def_property_readonly("size", &SnapshotData::getSize)
+      .def_property_readonly("max_size", &SnapshotData::getMaxSize)
+      .def("get_data",
+           [](SnapshotData& s) {
+               auto v = s.getDataCopy();
+               return py::bytes((const char*)v.data(), v.size());
+           })
+      .def("copy_in_data",
+           [](SnapshotData& s, const py::bytes& data, uint32_t offset) {
+               std::string str = data;
+               s.copyInData((const uint8_t*)str.data(), str.size(), offset);
+           },
+           py::arg("data"),
+           py::arg("offset") = 0)
+      .def("add_merge_region",
+           &SnapshotData::addMergeRegion,
+           py::arg("offset"),
+           py::arg("length"),
+           py::arg("data_type"),
+           py::arg("operation"))
+      .def("fill_gaps_with_bytewise_regions",
+           &SnapshotData::fillGapsWithBytewiseRegions)
+      .def("clear_merge_regions", &SnapshotData::clearMergeRegions)
+      .def("diff_with_memory",
+           [](SnapshotData& s, const py::bytes& updated) {
+               std::string str = updated;
+               return s.diffWithMemory((const uint8_t*)str.data(),
+                                       str.size());
+           })
+      .def("diff_with_dirty_regions",
+           [](SnapshotData& s,
+              const py::bytes& updated,
+              const std::vector<char>& dirtyPages) {
+               std::string str = updated;
+               return s.diffWithDirtyRegions(
+                 (const uint8_t*)str.data(), str.size(), dirtyPages);
+           })
+      .def("apply_diffs", &SnapshotData::applyDiffs)
+      .def("queue_diffs", &SnapshotData::queueDiffs)
+      .def("write_queued_diffs", &SnapshotData::writeQueuedDiffs);
+
+    m.def("snapshot_register",
+          [](const std::string& key, std::shared_ptr<SnapshotData> snap) {
+              SnapshotRegistry::get().registerSnapshot(key, snap);
+          });
+    m.def("snapshot_get", [](const std::string& key) {
+        return SnapshotRegistry::get().getSnapshot(key);
+    });
+    m.def("snapshot_exists", [](const std::string& key) {
+        return SnapshotRegistry::get().snapshotExists(key);
+    });
+    m.def("snapshot_delete", [](const std::string& key) {
+        SnapshotRegistry::get().deleteSnapshot(key);
+    });
+    m.def("snapshot_count",
+          [] { return SnapshotRegistry::get().getSnapshotCount(); });
+    m.def("snapshot_clear", [] { SnapshotRegistry::get().clear(); });
+
+    // Mock recordings
+    m.def("get_batch_requests_sent_mock", [] {
+        return getBatchRequestsSentMock();
+    });
+    m.def("clear_mocked_function_calls", &clearMockedFunctionCalls);
 }

@@ -6,6 +6,7 @@
 #include <pybind11/pybind11.h>
 #include <pybind11/stl.h>
 
+#include "faabricamd/executor.h"
 #include "faabricamd/messages.h"
 #include "faabricamd/queue.h"
 #include "faabricamd/transport.h"
@@ -169,4 +170,11 @@ PYBIND11_MODULE(_core, m)
 
     initRuntimeBindings(m);
     initOpsBindings(m);
+
+    // Python callables captured by the C++ function registry must be
+    // released while the interpreter is still alive (the registry is a
+    // C++ static, destroyed after Py_Finalize otherwise).
+    m.add_object("_registry_cleanup", py::capsule([]() {
+                     faabricamd::FunctionRegistry::get().clear();
+                 }));
 }

@@ -1,0 +1,107 @@
+// Per-host scheduler: executor pool + function-call RPC + reaper
+// (reference: include/faabric/scheduler/Scheduler.h:41,
+//  src/scheduler/Scheduler.cpp:48-448,
+//  src/scheduler/FunctionCallServer.cpp:21-95).
+#pragma once
+
+#include <memory>
+#include <mutex>
+#include <unordered_map>
+#include <vector>
+
+#include "faabricamd/executor.h"
+#include "faabricamd/messages.h"
+#include "faabricamd/transport.h"
+
+namespace faabricamd {
+
+// RPC call codes (reference: scheduler/FunctionCallApi.h:4-10)
+enum class FunctionCalls : uint8_t
+{
+    ExecuteFunctions = 1,
+    Flush = 2,
+    SetMessageResult = 3,
+};
+
+class Scheduler
+{
+  public:
+    static Scheduler& get();
+
+    // Execute a batch scheduled to this host
+    // (reference: src/scheduler/Scheduler.cpp:250)
+    void executeBatch(std::shared_ptr<BatchExecuteRequest> req);
+
+    // Resources advertised to the planner
+    HostResources getThisHostResources();
+    void setThisHostResources(const HostResources& res);
+
+    // Periodic reaping of idle warm executors
+    // (reference: src/scheduler/Scheduler.cpp:166-241)
+    int reapStaleExecutors();
+    void startReaper();
+    void stopReaper();
+
+    // Migration check entry point called periodically by long-running apps
+    // (reference: src/scheduler/Scheduler.cpp:448)
+    std::shared_ptr<PendingMigration> checkForMigrationOpportunities(
+      Message& msg,
+      int32_t overwriteNewGroupId = 0);
+
+    void flushLocally();
+    void shutdown();
+    void reset();
+
+    // Test-mode recording (reference: src/scheduler/Scheduler.cpp:263-268)
+    std::vector<Message> getRecordedMessages();
+    void clearRecordedMessages();
+
+    size_t getExecutorCount();
+
+  private:
+    Scheduler();
+
+    std::shared_ptr<Executor> claimExecutor(Message& msg);
+
+    std::mutex schedMx;
+    std::unordered_map<std::string,
+                       std::vector<std::shared_ptr<Executor>>>
+      executors;
+    HostResources overriddenResources;
+    bool resourcesOverridden = false;
+
+    std::vector<Message> recordedMessages;
+
+    class ReaperThread;
+    std::shared_ptr<ReaperThread> reaper;
+};
+
+class FunctionCallServer : public MessageEndpointServer
+{
+  public:
+    FunctionCallServer();
+    void doAsyncRecv(uint8_t code,
+                     const std::string& body,
+                     uint32_t seq) override;
+    std::string doSyncRecv(uint8_t code, const std::string& body) override;
+};
+
+class FunctionCallClient : public MessageEndpointClient
+{
+  public:
+    explicit FunctionCallClient(const std::string& host);
+    void executeFunctions(const BatchExecuteRequest& req);
+    void setMessageResult(const Message& msg);
+    void sendFlush();
+};
+
+std::shared_ptr<FunctionCallClient> getFunctionCallClient(
+  const std::string& host);
+void clearFunctionCallClients();
+
+// Mock-mode recording (reference: src/scheduler/FunctionCallClient.cpp:14-99)
+std::vector<std::pair<std::string, BatchExecuteRequest>>
+getBatchRequestsSentMock();
+void clearMockedFunctionCalls();
+
+} // namespace faabricamd

@@ -1,0 +1,165 @@
+// Distributed state KV.
+//
+// MI355X-native re-design of the reference state layer (reference:
+// include/faabric/state/State.h:23-58, StateKeyValue.h:19-166,
+// InMemoryStateKeyValue.cpp, src/state/StateServer.cpp:24-164,
+// src/state/StateClient.cpp). Master-per-key becomes owner-host (one GPU
+// per worker process); values can be HOST-resident or DEVICE-resident in
+// HBM3E (device path wired through ops.cpp); chunked lazy pull and
+// dirty-mask partial push are kept.
+#pragma once
+
+#include <map>
+#include <memory>
+#include <mutex>
+#include <string>
+#include <vector>
+
+#include "faabricamd/messages.h"
+#include "faabricamd/transport.h"
+
+namespace faabricamd {
+
+// Streaming chunk size (reference: state/StateKeyValue.h:19 — 64 KiB)
+inline constexpr size_t STATE_STREAM_CHUNK_SIZE = 64 * 1024;
+
+// (reference: state/State.h:11-21)
+enum class StateCalls : uint8_t
+{
+    Pull = 1,
+    Push = 2,
+    Size = 3,
+    Append = 4,
+    ClearAppended = 5,
+    PullAppended = 6,
+    Delete = 7,
+};
+
+class StateKeyValue
+{
+  public:
+    StateKeyValue(std::string userIn,
+                  std::string keyIn,
+                  size_t sizeIn,
+                  std::string masterHostIn);
+
+    const std::string& getUser() const { return user; }
+    const std::string& getKey() const { return key; }
+    size_t size() const { return valueSize; }
+    bool isMaster() const;
+    const std::string& getMasterHost() const { return masterHost; }
+
+    // --- whole-value ---
+    void get(uint8_t* buffer);
+    std::vector<uint8_t> get();
+    void set(const uint8_t* buffer, size_t n);
+    void set(const std::vector<uint8_t>& data);
+
+    // --- chunks ---
+    void getChunk(uint64_t offset, uint8_t* buffer, size_t len);
+    void setChunk(uint64_t offset, const uint8_t* buffer, size_t len);
+
+    // --- lazy pull / push against the master ---
+    void pull();     // fetch whole value (chunked)
+    void pushFull(); // push whole value
+    void flagDirty();
+    void flagChunkDirty(uint64_t offset, size_t len);
+    void pushPartial(); // push only dirty chunks
+
+    // --- append channel ---
+    void append(const uint8_t* data, size_t len);
+    std::vector<std::vector<uint8_t>> getAppended(size_t nValues);
+    void clearAppended();
+
+    // Direct access for zero-copy users (host path)
+    uint8_t* getDataPtr() { return value.data(); }
+
+    // Master-side servicing
+    std::vector<uint8_t> serviceChunk(uint64_t offset, size_t len);
+    void serviceSet(uint64_t offset, const uint8_t* data, size_t len);
+    void serviceAppend(const uint8_t* data, size_t len);
+    std::vector<std::vector<uint8_t>> serviceGetAppended(size_t n);
+    void serviceClearAppended();
+
+  private:
+    std::string user;
+    std::string key;
+    size_t valueSize;
+    std::string masterHost;
+
+    std::mutex kvMx;
+    std::vector<uint8_t> value;
+    std::vector<char> dirtyChunks; // one flag per STATE_STREAM_CHUNK_SIZE
+    bool fullyPulled = false;
+    std::vector<std::vector<uint8_t>> appendedValues;
+};
+
+class State
+{
+  public:
+    static State& get();
+
+    std::shared_ptr<StateKeyValue> getKV(const std::string& user,
+                                         const std::string& key,
+                                         size_t size);
+    std::shared_ptr<StateKeyValue> getKV(const std::string& user,
+                                         const std::string& key);
+    size_t getStateSize(const std::string& user, const std::string& key);
+    void deleteKV(const std::string& user, const std::string& key);
+    void deleteKVLocally(const std::string& user, const std::string& key);
+    size_t getKVCount();
+    void forceClearAll(bool global);
+
+    // Owner directory: where a key lives. Defaults to this host on first
+    // getKV; setMasterHost lets the deployment pin owner GPUs.
+    void setMasterHost(const std::string& user,
+                       const std::string& key,
+                       const std::string& host);
+    std::string getMasterHost(const std::string& user,
+                              const std::string& key);
+
+  private:
+    std::mutex mx;
+    std::map<std::string, std::shared_ptr<StateKeyValue>> kvMap;
+    std::map<std::string, std::string> masterMap;
+};
+
+class StateServer : public MessageEndpointServer
+{
+  public:
+    StateServer();
+    void doAsyncRecv(uint8_t code,
+                     const std::string& body,
+                     uint32_t seq) override;
+    std::string doSyncRecv(uint8_t code, const std::string& body) override;
+};
+
+class StateClient : public MessageEndpointClient
+{
+  public:
+    explicit StateClient(const std::string& host);
+    std::vector<uint8_t> pullChunk(const std::string& user,
+                                   const std::string& key,
+                                   uint64_t offset,
+                                   size_t len);
+    void pushChunk(const std::string& user,
+                   const std::string& key,
+                   uint64_t offset,
+                   const uint8_t* data,
+                   size_t len);
+    size_t stateSize(const std::string& user, const std::string& key);
+    void append(const std::string& user,
+                const std::string& key,
+                const uint8_t* data,
+                size_t len);
+    std::vector<std::vector<uint8_t>> pullAppended(const std::string& user,
+                                                   const std::string& key,
+                                                   size_t nValues);
+    void clearAppended(const std::string& user, const std::string& key);
+    void deleteKV(const std::string& user, const std::string& key);
+};
+
+std::shared_ptr<StateClient> getStateClient(const std::string& host);
+void clearStateClients();
+
+} // namespace faabricamd

@@ -19,6 +19,7 @@
 #include <memory>
 #include <mutex>
 #include <optional>
+#include <set>
 #include <string>
 #include <thread>
 #include <vector>
@@ -40,11 +41,16 @@ inline constexpr int PLANNER_ASYNC_PORT = 8011;
 inline constexpr int PLANNER_SYNC_PORT = 8012;
 inline constexpr int DEFAULT_MPI_BASE_PORT = 8020;
 
-// Offset added to all ports for test isolation / multi-process-per-host
-// deployments (one "host" per GPU on a node shares the IP). Set via
-// FAABRIC_PORT_OFFSET or programmatically before servers start.
+// Offset added to all ports this process's SERVERS bind, so several
+// single-GPU worker processes can share one node's IP. Set via
+// FAABRIC_PORT_OFFSET or programmatically before servers start. A host
+// advertises itself as "ip@offset"; clients parse that identity and add
+// the offset when dialling (see parseHostIdentity).
 int getPortOffset();
 void setPortOffset(int offset);
+void parseHostIdentity(const std::string& identity,
+                       std::string& ipOut,
+                       int& offsetOut);
 
 struct WireHeader
 {
@@ -146,8 +152,11 @@ class MessageEndpointServer
     void start();
     void stop();
 
-    // Subclass API (reference: MessageEndpointServer.h:64-66)
-    virtual void doAsyncRecv(uint8_t code, const std::string& body) = 0;
+    // Subclass API (reference: MessageEndpointServer.h:64-66); seq carries
+    // the frame header's sequence number (used by ordered PTP delivery)
+    virtual void doAsyncRecv(uint8_t code,
+                             const std::string& body,
+                             uint32_t seq) = 0;
     virtual std::string doSyncRecv(uint8_t code, const std::string& body) = 0;
 
   private:
@@ -163,6 +172,7 @@ class MessageEndpointServer
     std::thread syncAcceptThread;
     std::mutex connThreadsMx;
     std::vector<std::thread> connThreads;
+    std::set<int> activeConnFds; // shut down on stop() to unblock recv
     std::atomic<bool> running{ false };
 };
 
@@ -179,6 +189,7 @@ class MessageEndpointClient
 
     void asyncSend(uint8_t code, const std::string& body);
     void asyncSend(uint8_t code, const void* body, size_t len);
+    void asyncSendSeq(uint8_t code, const void* body, size_t len, uint32_t seq);
     std::string syncSend(uint8_t code, const std::string& body);
 
     const std::string& getHost() const { return host; }

@@ -1,0 +1,650 @@
+// Point-to-point broker implementation. Reference behavior:
+// src/transport/PointToPointBroker.cpp (mappings :416-530, messaging
+// :637-859, groups :142-365, postMigrationHook :910-926) — fresh design,
+// see ptp.h.
+#include "faabricamd/ptp.h"
+#include "faabricamd/util.h"
+
+#include <algorithm>
+
+namespace faabricamd {
+
+// Control-plane channels are multiplexed over the same (group, send, recv)
+// space by offsetting sendIdx; group sizes must stay below the offset.
+static constexpr int32_t SEND_OFF_BARRIER = 1 * 16384;
+static constexpr int32_t SEND_OFF_LOCK = 2 * 16384;
+static constexpr int32_t SEND_OFF_NOTIFY = 3 * 16384;
+static constexpr uint32_t NO_SEQ = 0xffffffffu;
+
+static int64_t idxKey(int32_t groupId, int32_t idx)
+{
+    return ((int64_t)groupId << 32) | (uint32_t)idx;
+}
+
+static int64_t chanKey(int32_t groupId, int32_t sendIdx, int32_t recvIdx)
+{
+    return ((int64_t)groupId << 32) | ((uint32_t)(uint16_t)sendIdx << 16) |
+           (uint32_t)(uint16_t)recvIdx;
+}
+
+// ------------------------- groups ------------------------------------------
+
+static std::mutex groupsMx;
+static std::map<int32_t, std::shared_ptr<PointToPointGroup>> groups;
+
+std::shared_ptr<PointToPointGroup> PointToPointGroup::getGroup(int32_t groupId)
+{
+    std::lock_guard<std::mutex> lock(groupsMx);
+    auto it = groups.find(groupId);
+    if (it == groups.end()) {
+        throw FaabricException("group not found: " + std::to_string(groupId));
+    }
+    return it->second;
+}
+
+std::shared_ptr<PointToPointGroup> PointToPointGroup::getOrAwaitGroup(
+  int32_t groupId)
+{
+    getPointToPointBroker().waitForMappingsOnThisHost(groupId);
+    return getGroup(groupId);
+}
+
+bool PointToPointGroup::groupExists(int32_t groupId)
+{
+    std::lock_guard<std::mutex> lock(groupsMx);
+    return groups.find(groupId) != groups.end();
+}
+
+void PointToPointGroup::addGroup(int32_t appId,
+                                 int32_t groupId,
+                                 int32_t groupSize)
+{
+    std::lock_guard<std::mutex> lock(groupsMx);
+    if (groups.find(groupId) == groups.end()) {
+        groups[groupId] =
+          std::make_shared<PointToPointGroup>(appId, groupId, groupSize);
+    }
+}
+
+void PointToPointGroup::clearGroup(int32_t groupId)
+{
+    std::lock_guard<std::mutex> lock(groupsMx);
+    groups.erase(groupId);
+}
+
+void PointToPointGroup::clear()
+{
+    std::lock_guard<std::mutex> lock(groupsMx);
+    groups.clear();
+}
+
+PointToPointGroup::PointToPointGroup(int32_t appIdIn,
+                                     int32_t groupIdIn,
+                                     int32_t sizeIn)
+  : appId(appIdIn)
+  , groupId(groupIdIn)
+  , groupSize(sizeIn)
+{}
+
+void PointToPointGroup::lock(int32_t groupIdx, bool recursive)
+{
+    auto& broker = getPointToPointBroker();
+    std::string mainHost =
+      broker.getHostForReceiver(groupId, POINT_TO_POINT_MAIN_IDX);
+    const std::string& thisHost = getSystemConfig().endpointHost;
+
+    if (mainHost == thisHost) {
+        handleLockRequest(thisHost, groupIdx, recursive);
+    } else {
+        PointToPointClient cli(mainHost);
+        cli.groupLock(appId, groupId, groupIdx, recursive);
+    }
+    // Wait for the grant message
+    broker.recvMessage(groupId, SEND_OFF_LOCK, groupIdx);
+}
+
+void PointToPointGroup::unlock(int32_t groupIdx, bool recursive)
+{
+    auto& broker = getPointToPointBroker();
+    std::string mainHost =
+      broker.getHostForReceiver(groupId, POINT_TO_POINT_MAIN_IDX);
+    const std::string& thisHost = getSystemConfig().endpointHost;
+
+    if (mainHost == thisHost) {
+        handleUnlockRequest(thisHost, groupIdx, recursive);
+    } else {
+        PointToPointClient cli(mainHost);
+        cli.groupUnlock(appId, groupId, groupIdx, recursive);
+    }
+}
+
+void PointToPointGroup::handleLockRequest(const std::string& fromHost,
+                                          int32_t groupIdx,
+                                          bool recursive)
+{
+    auto& broker = getPointToPointBroker();
+    bool grantNow = false;
+    {
+        std::lock_guard<std::mutex> lock(internalMx);
+        if (!lockHeld) {
+            lockHeld = true;
+            lockHolderIdx = groupIdx;
+            recursiveDepth = 1;
+            grantNow = true;
+        } else if (recursive && lockHolderIdx == groupIdx) {
+            recursiveDepth++;
+            grantNow = true;
+        } else {
+            lockWaiters.emplace_back(fromHost, groupIdx);
+        }
+    }
+    if (grantNow) {
+        uint8_t token = 1;
+        broker.sendMessage(
+          appId, groupId, SEND_OFF_LOCK, groupIdx, &token, 1);
+    }
+}
+
+void PointToPointGroup::handleUnlockRequest(const std::string& fromHost,
+                                            int32_t groupIdx,
+                                            bool recursive)
+{
+    (void)fromHost;
+    auto& broker = getPointToPointBroker();
+    std::pair<std::string, int32_t> next{ "", -1 };
+    {
+        std::lock_guard<std::mutex> lock(internalMx);
+        if (recursive && recursiveDepth > 1 && lockHolderIdx == groupIdx) {
+            recursiveDepth--;
+            return;
+        }
+        if (!lockWaiters.empty()) {
+            next = lockWaiters.front();
+            lockWaiters.erase(lockWaiters.begin());
+            lockHolderIdx = next.second;
+            recursiveDepth = 1;
+        } else {
+            lockHeld = false;
+            lockHolderIdx = -1;
+            recursiveDepth = 0;
+        }
+    }
+    if (next.second >= 0) {
+        uint8_t token = 1;
+        broker.sendMessage(
+          appId, groupId, SEND_OFF_LOCK, next.second, &token, 1);
+    }
+}
+
+void PointToPointGroup::localLock()
+{
+    if (!localMx.try_lock_for(
+          std::chrono::milliseconds(getSystemConfig().boundTimeout))) {
+        throw QueueTimeoutException("local group lock timeout");
+    }
+}
+
+void PointToPointGroup::localUnlock()
+{
+    localMx.unlock();
+}
+
+void PointToPointGroup::barrier(int32_t groupIdx)
+{
+    // Gather-to-main then fan-out
+    // (reference: src/transport/PointToPointBroker.cpp:317-346)
+    auto& broker = getPointToPointBroker();
+    uint8_t token = 1;
+    if (groupIdx == POINT_TO_POINT_MAIN_IDX) {
+        for (int i = 1; i < groupSize; i++) {
+            broker.recvMessage(
+              groupId, SEND_OFF_BARRIER + i, POINT_TO_POINT_MAIN_IDX);
+        }
+        for (int i = 1; i < groupSize; i++) {
+            broker.sendMessage(appId,
+                               groupId,
+                               SEND_OFF_BARRIER + POINT_TO_POINT_MAIN_IDX,
+                               i,
+                               &token,
+                               1);
+        }
+    } else {
+        broker.sendMessage(appId,
+                           groupId,
+                           SEND_OFF_BARRIER + groupIdx,
+                           POINT_TO_POINT_MAIN_IDX,
+                           &token,
+                           1);
+        broker.recvMessage(
+          groupId, SEND_OFF_BARRIER + POINT_TO_POINT_MAIN_IDX, groupIdx);
+    }
+}
+
+void PointToPointGroup::notify(int32_t groupIdx)
+{
+    // Non-main members notify; main waits for all
+    // (reference: src/transport/PointToPointBroker.cpp:348-365)
+    auto& broker = getPointToPointBroker();
+    uint8_t token = 1;
+    if (groupIdx == POINT_TO_POINT_MAIN_IDX) {
+        for (int i = 1; i < groupSize; i++) {
+            broker.recvMessage(
+              groupId, SEND_OFF_NOTIFY + i, POINT_TO_POINT_MAIN_IDX);
+        }
+    } else {
+        broker.sendMessage(appId,
+                           groupId,
+                           SEND_OFF_NOTIFY + groupIdx,
+                           POINT_TO_POINT_MAIN_IDX,
+                           &token,
+                           1);
+    }
+}
+
+// ------------------------- broker -------------------------------------------
+
+PointToPointBroker::PointToPointBroker() = default;
+
+PointToPointBroker& getPointToPointBroker()
+{
+    static PointToPointBroker broker;
+    return broker;
+}
+
+std::shared_ptr<FlagWaiter> PointToPointBroker::getFlag(int32_t groupId)
+{
+    std::lock_guard<std::mutex> lock(brokerMx);
+    auto& flag = groupFlags[groupId];
+    if (!flag) {
+        flag =
+          std::make_shared<FlagWaiter>(getSystemConfig().boundTimeout);
+    }
+    return flag;
+}
+
+std::set<std::string>
+PointToPointBroker::setUpLocalMappingsFromSchedulingDecision(
+  const SchedulingDecision& decision)
+{
+    std::set<std::string> hosts;
+    {
+        std::lock_guard<std::mutex> lock(brokerMx);
+        for (int i = 0; i < decision.nFunctions; i++) {
+            int32_t idx = decision.groupIdxs[i];
+            mappings[idxKey(decision.groupId, idx)] = decision.hosts[i];
+            mpiPorts[idxKey(decision.groupId, idx)] = decision.mpiPorts[i];
+            groupIdxs[decision.groupId].insert(idx);
+            hosts.insert(decision.hosts[i]);
+        }
+    }
+    PointToPointGroup::addGroup(
+      decision.appId, decision.groupId, decision.nFunctions);
+    getFlag(decision.groupId)->setFlag(true);
+    return hosts;
+}
+
+void PointToPointBroker::setAndSendMappingsFromSchedulingDecision(
+  const SchedulingDecision& decision)
+{
+    auto hosts = setUpLocalMappingsFromSchedulingDecision(decision);
+    sendMappingsFromSchedulingDecision(decision, hosts);
+}
+
+void PointToPointBroker::sendMappingsFromSchedulingDecision(
+  const SchedulingDecision& decision,
+  const std::set<std::string>& hostList)
+{
+    if (isMockMode()) {
+        return;
+    }
+    const std::string& thisHost = getSystemConfig().endpointHost;
+    PointToPointMappings msg = decision.toPointToPointMappings();
+    for (const auto& host : hostList) {
+        if (host == thisHost) {
+            continue;
+        }
+        PointToPointClient cli(host);
+        cli.sendMappings(msg);
+    }
+}
+
+void PointToPointBroker::waitForMappingsOnThisHost(int32_t groupId)
+{
+    getFlag(groupId)->waitOnFlag();
+}
+
+std::string PointToPointBroker::getHostForReceiver(int32_t groupId,
+                                                   int32_t recvIdx)
+{
+    std::lock_guard<std::mutex> lock(brokerMx);
+    auto it = mappings.find(idxKey(groupId, recvIdx));
+    if (it == mappings.end()) {
+        throw FaabricException("no mapping for group " +
+                               std::to_string(groupId) + " idx " +
+                               std::to_string(recvIdx));
+    }
+    return it->second;
+}
+
+int32_t PointToPointBroker::getMpiPortForReceiver(int32_t groupId,
+                                                  int32_t recvIdx)
+{
+    std::lock_guard<std::mutex> lock(brokerMx);
+    auto it = mpiPorts.find(idxKey(groupId, recvIdx));
+    if (it == mpiPorts.end()) {
+        throw FaabricException("no mpi port for group " +
+                               std::to_string(groupId) + " idx " +
+                               std::to_string(recvIdx));
+    }
+    return it->second;
+}
+
+std::set<int32_t> PointToPointBroker::getIdxsRegisteredForGroup(
+  int32_t groupId)
+{
+    std::lock_guard<std::mutex> lock(brokerMx);
+    auto it = groupIdxs.find(groupId);
+    return it == groupIdxs.end() ? std::set<int32_t>{} : it->second;
+}
+
+void PointToPointBroker::updateHostForIdx(int32_t groupId,
+                                          int32_t groupIdx,
+                                          const std::string& newHost)
+{
+    std::lock_guard<std::mutex> lock(brokerMx);
+    mappings[idxKey(groupId, groupIdx)] = newHost;
+}
+
+PointToPointBroker::Channel& PointToPointBroker::getChannel(int32_t groupId,
+                                                            int32_t sendIdx,
+                                                            int32_t recvIdx)
+{
+    std::lock_guard<std::mutex> lock(channelsMx);
+    auto& ch = channels[chanKey(groupId, sendIdx, recvIdx)];
+    if (!ch) {
+        ch = std::make_unique<Channel>();
+    }
+    return *ch;
+}
+
+std::shared_ptr<MessageEndpointClient> PointToPointBroker::getClient(
+  const std::string& host)
+{
+    std::lock_guard<std::mutex> lock(clientsMx);
+    auto& cli = clients[host];
+    if (!cli) {
+        cli = std::make_shared<MessageEndpointClient>(
+          host, POINT_TO_POINT_ASYNC_PORT, POINT_TO_POINT_SYNC_PORT);
+    }
+    return cli;
+}
+
+void PointToPointBroker::sendMessage(int32_t appId,
+                                     int32_t groupId,
+                                     int32_t sendIdx,
+                                     int32_t recvIdx,
+                                     const uint8_t* data,
+                                     size_t size,
+                                     bool mustOrderMsgs)
+{
+    uint32_t seq = NO_SEQ;
+    if (mustOrderMsgs) {
+        std::lock_guard<std::mutex> lock(sendSeqMx);
+        seq = sendSeqs[chanKey(groupId, sendIdx, recvIdx)]++;
+    }
+
+    // recvIdx addresses in control namespaces map to the base idx's host
+    int32_t hostIdx = recvIdx % 16384;
+    std::string host = getHostForReceiver(groupId, hostIdx);
+    const std::string& thisHost = getSystemConfig().endpointHost;
+
+    if (host == thisHost) {
+        PointToPointMessage msg;
+        msg.appId = appId;
+        msg.groupId = groupId;
+        msg.sendIdx = sendIdx;
+        msg.recvIdx = recvIdx;
+        msg.data.assign(data, data + size);
+        deliverRemoteMessage(msg, seq);
+    } else {
+        PointToPointMessage msg;
+        msg.appId = appId;
+        msg.groupId = groupId;
+        msg.sendIdx = sendIdx;
+        msg.recvIdx = recvIdx;
+        msg.data.assign(data, data + size);
+        auto cli = getClient(host);
+        std::string body = msg.encode();
+        // seq travels in the frame header
+        cli->asyncSendSeq((uint8_t)PointToPointCall::MESSAGE,
+                          body.data(),
+                          body.size(),
+                          seq);
+    }
+}
+
+void PointToPointBroker::deliverRemoteMessage(const PointToPointMessage& msg,
+                                              uint32_t seq)
+{
+    Channel& ch = getChannel(msg.groupId, msg.sendIdx, msg.recvIdx);
+    {
+        std::lock_guard<std::mutex> lock(ch.mx);
+        if (seq == NO_SEQ) {
+            ch.unorderedMsgs.push_back(msg.data);
+        } else {
+            ch.bufferedMsgs[seq] = msg.data;
+        }
+    }
+    ch.cv.notify_all();
+}
+
+std::vector<uint8_t> PointToPointBroker::recvMessage(int32_t groupId,
+                                                     int32_t sendIdx,
+                                                     int32_t recvIdx,
+                                                     bool mustOrderMsgs,
+                                                     int timeoutMs)
+{
+    Channel& ch = getChannel(groupId, sendIdx, recvIdx);
+    std::unique_lock<std::mutex> lock(ch.mx);
+
+    auto ready = [&]() {
+        if (mustOrderMsgs) {
+            return ch.bufferedMsgs.count(ch.nextRecvSeq) > 0;
+        }
+        return !ch.unorderedMsgs.empty() || !ch.bufferedMsgs.empty();
+    };
+
+    if (!ch.cv.wait_for(lock, std::chrono::milliseconds(timeoutMs), ready)) {
+        throw QueueTimeoutException(
+          "ptp recv timeout group " + std::to_string(groupId) + " " +
+          std::to_string(sendIdx) + "->" + std::to_string(recvIdx));
+    }
+
+    if (mustOrderMsgs) {
+        auto it = ch.bufferedMsgs.find(ch.nextRecvSeq);
+        std::vector<uint8_t> out = std::move(it->second);
+        ch.bufferedMsgs.erase(it);
+        ch.nextRecvSeq++;
+        return out;
+    }
+    if (!ch.unorderedMsgs.empty()) {
+        std::vector<uint8_t> out = std::move(ch.unorderedMsgs.front());
+        ch.unorderedMsgs.pop_front();
+        return out;
+    }
+    auto it = ch.bufferedMsgs.begin();
+    std::vector<uint8_t> out = std::move(it->second);
+    ch.bufferedMsgs.erase(it);
+    return out;
+}
+
+void PointToPointBroker::clearGroup(int32_t groupId)
+{
+    {
+        std::lock_guard<std::mutex> lock(brokerMx);
+        auto it = groupIdxs.find(groupId);
+        if (it != groupIdxs.end()) {
+            for (int32_t idx : it->second) {
+                mappings.erase(idxKey(groupId, idx));
+                mpiPorts.erase(idxKey(groupId, idx));
+            }
+            groupIdxs.erase(it);
+        }
+        groupFlags.erase(groupId);
+    }
+    {
+        std::lock_guard<std::mutex> lock(channelsMx);
+        for (auto it = channels.begin(); it != channels.end();) {
+            if ((int32_t)(it->first >> 32) == groupId) {
+                it = channels.erase(it);
+            } else {
+                ++it;
+            }
+        }
+    }
+    {
+        std::lock_guard<std::mutex> lock(sendSeqMx);
+        for (auto it = sendSeqs.begin(); it != sendSeqs.end();) {
+            if ((int32_t)(it->first >> 32) == groupId) {
+                it = sendSeqs.erase(it);
+            } else {
+                ++it;
+            }
+        }
+    }
+    PointToPointGroup::clearGroup(groupId);
+}
+
+void PointToPointBroker::clear()
+{
+    {
+        std::lock_guard<std::mutex> lock(brokerMx);
+        mappings.clear();
+        mpiPorts.clear();
+        groupIdxs.clear();
+        groupFlags.clear();
+    }
+    {
+        std::lock_guard<std::mutex> lock(channelsMx);
+        channels.clear();
+    }
+    {
+        std::lock_guard<std::mutex> lock(sendSeqMx);
+        sendSeqs.clear();
+    }
+    PointToPointGroup::clear();
+}
+
+void PointToPointBroker::postMigrationHook(int32_t groupId, int32_t groupIdx)
+{
+    auto group = PointToPointGroup::getOrAwaitGroup(groupId);
+    group->barrier(groupIdx);
+}
+
+// ------------------------- server / client ----------------------------------
+
+PointToPointServer::PointToPointServer()
+  : MessageEndpointServer(POINT_TO_POINT_ASYNC_PORT,
+                          POINT_TO_POINT_SYNC_PORT,
+                          "ptp")
+{}
+
+void PointToPointServer::doAsyncRecv(uint8_t code,
+                                     const std::string& body,
+                                     uint32_t seq)
+{
+    auto call = (PointToPointCall)code;
+    switch (call) {
+        case PointToPointCall::MESSAGE: {
+            PointToPointMessage msg = PointToPointMessage::decode(body);
+            getPointToPointBroker().deliverRemoteMessage(msg, seq);
+            break;
+        }
+        case PointToPointCall::LOCK_GROUP:
+        case PointToPointCall::LOCK_GROUP_RECURSIVE: {
+            PointToPointMessage msg = PointToPointMessage::decode(body);
+            bool recursive = call == PointToPointCall::LOCK_GROUP_RECURSIVE;
+            auto group = PointToPointGroup::getGroup(msg.groupId);
+            group->handleLockRequest("", msg.sendIdx, recursive);
+            break;
+        }
+        case PointToPointCall::UNLOCK_GROUP:
+        case PointToPointCall::UNLOCK_GROUP_RECURSIVE: {
+            PointToPointMessage msg = PointToPointMessage::decode(body);
+            bool recursive = call == PointToPointCall::UNLOCK_GROUP_RECURSIVE;
+            auto group = PointToPointGroup::getGroup(msg.groupId);
+            group->handleUnlockRequest("", msg.sendIdx, recursive);
+            break;
+        }
+        default:
+            FAM_ERROR("ptp server: bad async call %d", (int)code);
+    }
+}
+
+std::string PointToPointServer::doSyncRecv(uint8_t code,
+                                           const std::string& body)
+{
+    auto call = (PointToPointCall)code;
+    if (call == PointToPointCall::MAPPING) {
+        PointToPointMappings mappings = PointToPointMappings::decode(body);
+        auto decision = SchedulingDecision::fromPointToPointMappings(mappings);
+        getPointToPointBroker().setUpLocalMappingsFromSchedulingDecision(
+          decision);
+        return {};
+    }
+    throw FaabricException("ptp server: bad sync call " +
+                           std::to_string(code));
+}
+
+PointToPointClient::PointToPointClient(const std::string& host)
+  : MessageEndpointClient(host,
+                          POINT_TO_POINT_ASYNC_PORT,
+                          POINT_TO_POINT_SYNC_PORT)
+{}
+
+void PointToPointClient::sendMappings(const PointToPointMappings& mappings)
+{
+    syncSend((uint8_t)PointToPointCall::MAPPING, mappings.encode());
+}
+
+void PointToPointClient::sendMessage(const PointToPointMessage& msg,
+                                     uint32_t seq)
+{
+    std::string body = msg.encode();
+    asyncSendSeq((uint8_t)PointToPointCall::MESSAGE,
+                 body.data(),
+                 body.size(),
+                 seq);
+}
+
+void PointToPointClient::groupLock(int32_t appId,
+                                   int32_t groupId,
+                                   int32_t groupIdx,
+                                   bool recursive)
+{
+    PointToPointMessage msg;
+    msg.appId = appId;
+    msg.groupId = groupId;
+    msg.sendIdx = groupIdx;
+    msg.recvIdx = POINT_TO_POINT_MAIN_IDX;
+    asyncSend(recursive ? (uint8_t)PointToPointCall::LOCK_GROUP_RECURSIVE
+                        : (uint8_t)PointToPointCall::LOCK_GROUP,
+              msg.encode());
+}
+
+void PointToPointClient::groupUnlock(int32_t appId,
+                                     int32_t groupId,
+                                     int32_t groupIdx,
+                                     bool recursive)
+{
+    PointToPointMessage msg;
+    msg.appId = appId;
+    msg.groupId = groupId;
+    msg.sendIdx = groupIdx;
+    msg.recvIdx = POINT_TO_POINT_MAIN_IDX;
+    asyncSend(recursive ? (uint8_t)PointToPointCall::UNLOCK_GROUP_RECURSIVE
+                        : (uint8_t)PointToPointCall::UNLOCK_GROUP,
+              msg.encode());
+}
+
+} // namespace faabricamd

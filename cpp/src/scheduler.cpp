@@ -1,0 +1,402 @@
+// Per-host scheduler (reference behavior: src/scheduler/Scheduler.cpp
+// :250-337 executeBatch, :339-387 claimExecutor, :166-241 reaper,
+// :448-530 migration check; src/scheduler/FunctionCallServer.cpp:21-95).
+#include "faabricamd/scheduler.h"
+#include "faabricamd/planner.h"
+#include "faabricamd/ptp.h"
+#include "faabricamd/util.h"
+
+#include <hip/hip_runtime.h>
+
+namespace faabricamd {
+
+// Cached GPU count; probing HIP is cheap but not free
+static int getNumGpus()
+{
+    static int n = []() {
+        const auto& conf = getSystemConfig();
+        if (conf.overrideGpuCount >= 0) {
+            return conf.overrideGpuCount;
+        }
+        if (!conf.useGpu) {
+            return 0;
+        }
+        int count = 0;
+        if (hipGetDeviceCount(&count) != hipSuccess) {
+            return 0;
+        }
+        return count;
+    }();
+    return n;
+}
+
+class Scheduler::ReaperThread : public PeriodicBackgroundThread
+{
+  public:
+    void doWork() override { Scheduler::get().reapStaleExecutors(); }
+};
+
+Scheduler::Scheduler() = default;
+
+Scheduler& Scheduler::get()
+{
+    static Scheduler sched;
+    return sched;
+}
+
+HostResources Scheduler::getThisHostResources()
+{
+    {
+        std::lock_guard<std::mutex> lock(schedMx);
+        if (resourcesOverridden) {
+            return overriddenResources;
+        }
+    }
+    HostResources res;
+    // On an MI355X node a slot is a GPU; CPU-only hosts fall back to cores
+    int nGpus = getNumGpus();
+    res.slots = nGpus > 0 ? nGpus : getUsableCores();
+    res.usedSlots = 0;
+    return res;
+}
+
+void Scheduler::setThisHostResources(const HostResources& res)
+{
+    std::lock_guard<std::mutex> lock(schedMx);
+    overriddenResources = res;
+    resourcesOverridden = true;
+}
+
+void Scheduler::executeBatch(std::shared_ptr<BatchExecuteRequest> req)
+{
+    if (req->messages.empty()) {
+        return;
+    }
+    bool isThreads = req->type == BatchExecuteType::THREADS;
+
+    if (isTestMode()) {
+        std::lock_guard<std::mutex> lock(schedMx);
+        for (const auto& msg : req->messages) {
+            recordedMessages.push_back(msg);
+        }
+    }
+
+    if (isThreads) {
+        // One executor runs all local threads of the app
+        // (reference: src/scheduler/Scheduler.cpp:272-298)
+        auto exec = claimExecutor(req->messages[0]);
+        std::vector<int> idxs(req->messages.size());
+        for (size_t i = 0; i < idxs.size(); i++) {
+            idxs[i] = (int)i;
+        }
+        exec->executeTasks(idxs, req);
+    } else {
+        // One executor per message
+        for (size_t i = 0; i < req->messages.size(); i++) {
+            try {
+                auto exec = claimExecutor(req->messages[i]);
+                exec->executeTasks({ (int)i }, req);
+            } catch (const std::exception& e) {
+                // Failures set an error result instead of crashing the host
+                // (reference: src/scheduler/Scheduler.cpp:304-322)
+                FAM_ERROR("claiming executor failed: %s", e.what());
+                auto msg = std::make_shared<Message>(req->messages[i]);
+                msg->returnValue = 1;
+                msg->executedHost = getSystemConfig().endpointHost;
+                msg->outputData = std::string("executor claim failed: ") +
+                                  e.what();
+                getPlannerClient().setMessageResult(msg);
+            }
+        }
+    }
+}
+
+std::shared_ptr<Executor> Scheduler::claimExecutor(Message& msg)
+{
+    std::lock_guard<std::mutex> lock(schedMx);
+    std::string key = msg.user + "/" + msg.function;
+    auto& warm = executors[key];
+
+    // Warm executor reuse (reference: src/scheduler/Scheduler.cpp:339-387)
+    for (auto& e : warm) {
+        if (e->tryClaim()) {
+            return e;
+        }
+    }
+    auto exec = getExecutorFactory()->createExecutor(msg);
+    exec->claim();
+    if (getNumGpus() > 0) {
+        exec->gpuDevice = (int)((warm.size()) % (size_t)getNumGpus());
+    }
+    warm.push_back(exec);
+    return exec;
+}
+
+int Scheduler::reapStaleExecutors()
+{
+    std::lock_guard<std::mutex> lock(schedMx);
+    const auto& conf = getSystemConfig();
+    int reaped = 0;
+    for (auto& [key, list] : executors) {
+        for (auto it = list.begin(); it != list.end();) {
+            auto& exec = *it;
+            if (!exec->isClaimed() &&
+                exec->getMillisSinceLastExec() > conf.boundTimeout) {
+                exec->shutdown();
+                it = list.erase(it);
+                reaped++;
+            } else {
+                ++it;
+            }
+        }
+    }
+    return reaped;
+}
+
+void Scheduler::startReaper()
+{
+    if (!reaper) {
+        reaper = std::make_shared<ReaperThread>();
+        reaper->startMillis(getSystemConfig().boundTimeout);
+    }
+}
+
+void Scheduler::stopReaper()
+{
+    if (reaper) {
+        reaper->stop();
+        reaper = nullptr;
+    }
+}
+
+std::shared_ptr<PendingMigration> Scheduler::checkForMigrationOpportunities(
+  Message& msg,
+  int32_t overwriteNewGroupId)
+{
+    // Non-main group members wait for the new group id decided by idx 0
+    // (reference: src/scheduler/Scheduler.cpp:448-530)
+    int32_t newGroupId = 0;
+    if (msg.groupIdx == 0 && overwriteNewGroupId == 0) {
+        auto req = std::make_shared<BatchExecuteRequest>();
+        req->appId = msg.appId;
+        req->groupId = msg.groupId;
+        req->user = msg.user;
+        req->function = msg.function;
+        req->type = BatchExecuteType::MIGRATION;
+        auto decision = getPlannerClient().callFunctions(req);
+        if (decision->appId == DO_NOT_MIGRATE) {
+            newGroupId = 0;
+        } else {
+            newGroupId = decision->groupId;
+        }
+        // Tell the rest of the group
+        auto& broker = getPointToPointBroker();
+        int32_t payload = newGroupId;
+        for (int i = 1; i < msg.groupSize; i++) {
+            broker.sendMessage(msg.appId,
+                               msg.groupId,
+                               0,
+                               i,
+                               (const uint8_t*)&payload,
+                               sizeof(payload));
+        }
+    } else if (overwriteNewGroupId != 0) {
+        newGroupId = overwriteNewGroupId;
+    } else {
+        auto data = getPointToPointBroker().recvMessage(
+          msg.groupId, 0, msg.groupIdx);
+        newGroupId = *(const int32_t*)data.data();
+    }
+
+    if (newGroupId == 0) {
+        return nullptr;
+    }
+
+    // A migration is pending: find out whether THIS message moves
+    auto decision =
+      getPlannerClient().getSchedulingDecision(msg.appId);
+    std::string newHost;
+    for (int i = 0; i < decision.nFunctions; i++) {
+        if (decision.groupIdxs[i] == msg.groupIdx) {
+            newHost = decision.hosts[i];
+            break;
+        }
+    }
+    auto migration = std::make_shared<PendingMigration>();
+    migration->appId = msg.appId;
+    migration->groupId = newGroupId;
+    migration->groupIdx = msg.groupIdx;
+    migration->srcHost = getSystemConfig().endpointHost;
+    migration->dstHost = newHost;
+    if (newHost == getSystemConfig().endpointHost) {
+        // This rank stays put; update group id only
+        msg.groupId = newGroupId;
+        return nullptr;
+    }
+    return migration;
+}
+
+void Scheduler::flushLocally()
+{
+    std::lock_guard<std::mutex> lock(schedMx);
+    for (auto& [key, list] : executors) {
+        for (auto& e : list) {
+            e->flush();
+        }
+    }
+    executors.clear();
+    getExecutorFactory()->flushHost();
+}
+
+void Scheduler::shutdown()
+{
+    stopReaper();
+    std::lock_guard<std::mutex> lock(schedMx);
+    for (auto& [key, list] : executors) {
+        for (auto& e : list) {
+            e->shutdown();
+        }
+    }
+    executors.clear();
+}
+
+void Scheduler::reset()
+{
+    shutdown();
+    std::lock_guard<std::mutex> lock(schedMx);
+    recordedMessages.clear();
+    resourcesOverridden = false;
+}
+
+std::vector<Message> Scheduler::getRecordedMessages()
+{
+    std::lock_guard<std::mutex> lock(schedMx);
+    return recordedMessages;
+}
+
+void Scheduler::clearRecordedMessages()
+{
+    std::lock_guard<std::mutex> lock(schedMx);
+    recordedMessages.clear();
+}
+
+size_t Scheduler::getExecutorCount()
+{
+    std::lock_guard<std::mutex> lock(schedMx);
+    size_t n = 0;
+    for (auto& [key, list] : executors) {
+        n += list.size();
+    }
+    return n;
+}
+
+// ------------------------- RPC server / client ------------------------------
+
+FunctionCallServer::FunctionCallServer()
+  : MessageEndpointServer(FUNCTION_CALL_ASYNC_PORT,
+                          FUNCTION_CALL_SYNC_PORT,
+                          "function-call")
+{}
+
+void FunctionCallServer::doAsyncRecv(uint8_t code,
+                                     const std::string& body,
+                                     uint32_t seq)
+{
+    (void)seq;
+    switch ((FunctionCalls)code) {
+        case FunctionCalls::ExecuteFunctions: {
+            auto req = std::make_shared<BatchExecuteRequest>(
+              BatchExecuteRequest::decode(body));
+            Scheduler::get().executeBatch(req);
+            break;
+        }
+        case FunctionCalls::SetMessageResult: {
+            auto msg = std::make_shared<Message>(Message::decode(body));
+            getPlannerClient().setMessageResultLocally(msg);
+            break;
+        }
+        default:
+            FAM_ERROR("function call server: bad async code %d", (int)code);
+    }
+}
+
+std::string FunctionCallServer::doSyncRecv(uint8_t code,
+                                           const std::string& body)
+{
+    (void)body;
+    if ((FunctionCalls)code == FunctionCalls::Flush) {
+        Scheduler::get().flushLocally();
+        return {};
+    }
+    throw FaabricException("function call server: bad sync code " +
+                           std::to_string(code));
+}
+
+FunctionCallClient::FunctionCallClient(const std::string& host)
+  : MessageEndpointClient(host,
+                          FUNCTION_CALL_ASYNC_PORT,
+                          FUNCTION_CALL_SYNC_PORT)
+{}
+
+// Mock-mode recording
+static std::mutex mockMx;
+static std::vector<std::pair<std::string, BatchExecuteRequest>>
+  mockedBatchRequests;
+
+std::vector<std::pair<std::string, BatchExecuteRequest>>
+getBatchRequestsSentMock()
+{
+    std::lock_guard<std::mutex> lock(mockMx);
+    return mockedBatchRequests;
+}
+
+void clearMockedFunctionCalls()
+{
+    std::lock_guard<std::mutex> lock(mockMx);
+    mockedBatchRequests.clear();
+}
+
+void FunctionCallClient::executeFunctions(const BatchExecuteRequest& req)
+{
+    if (isMockMode()) {
+        std::lock_guard<std::mutex> lock(mockMx);
+        mockedBatchRequests.emplace_back(getHost(), req);
+        return;
+    }
+    asyncSend((uint8_t)FunctionCalls::ExecuteFunctions, req.encode());
+}
+
+void FunctionCallClient::setMessageResult(const Message& msg)
+{
+    if (isMockMode()) {
+        return;
+    }
+    asyncSend((uint8_t)FunctionCalls::SetMessageResult, msg.encode());
+}
+
+void FunctionCallClient::sendFlush()
+{
+    syncSend((uint8_t)FunctionCalls::Flush, "");
+}
+
+static std::mutex fcClientsMx;
+static std::map<std::string, std::shared_ptr<FunctionCallClient>> fcClients;
+
+std::shared_ptr<FunctionCallClient> getFunctionCallClient(
+  const std::string& host)
+{
+    std::lock_guard<std::mutex> lock(fcClientsMx);
+    auto& cli = fcClients[host];
+    if (!cli) {
+        cli = std::make_shared<FunctionCallClient>(host);
+    }
+    return cli;
+}
+
+void clearFunctionCallClients()
+{
+    std::lock_guard<std::mutex> lock(fcClientsMx);
+    fcClients.clear();
+}
+
+} // namespace faabricamd

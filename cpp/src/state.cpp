@@ -1,0 +1,552 @@
+// State KV implementation (reference behavior: src/state/StateKeyValue.cpp,
+// InMemoryStateKeyValue.cpp:1-187, StateServer.cpp:24-164,
+// StateClient.cpp). See state.h for the re-design notes.
+#include "faabricamd/state.h"
+#include "faabricamd/util.h"
+
+#include <algorithm>
+#include <cstring>
+
+namespace faabricamd {
+
+static std::string kvKeyOf(const std::string& user, const std::string& key)
+{
+    return user + "/" + key;
+}
+
+StateKeyValue::StateKeyValue(std::string userIn,
+                             std::string keyIn,
+                             size_t sizeIn,
+                             std::string masterHostIn)
+  : user(std::move(userIn))
+  , key(std::move(keyIn))
+  , valueSize(sizeIn)
+  , masterHost(std::move(masterHostIn))
+{
+    value.resize(valueSize, 0);
+    size_t nChunks =
+      (valueSize + STATE_STREAM_CHUNK_SIZE - 1) / STATE_STREAM_CHUNK_SIZE;
+    dirtyChunks.resize(std::max<size_t>(nChunks, 1), 0);
+}
+
+bool StateKeyValue::isMaster() const
+{
+    return masterHost == getSystemConfig().endpointHost;
+}
+
+void StateKeyValue::get(uint8_t* buffer)
+{
+    pull();
+    std::lock_guard<std::mutex> lock(kvMx);
+    std::memcpy(buffer, value.data(), valueSize);
+}
+
+std::vector<uint8_t> StateKeyValue::get()
+{
+    std::vector<uint8_t> out(valueSize);
+    get(out.data());
+    return out;
+}
+
+void StateKeyValue::set(const uint8_t* buffer, size_t n)
+{
+    {
+        std::lock_guard<std::mutex> lock(kvMx);
+        if (n > valueSize) {
+            throw FaabricException("state set exceeds value size");
+        }
+        std::memcpy(value.data(), buffer, n);
+        fullyPulled = true;
+    }
+    flagDirty();
+    if (!isMaster()) {
+        pushFull();
+    }
+}
+
+void StateKeyValue::set(const std::vector<uint8_t>& data)
+{
+    set(data.data(), data.size());
+}
+
+void StateKeyValue::getChunk(uint64_t offset, uint8_t* buffer, size_t len)
+{
+    if (offset + len > valueSize) {
+        throw FaabricException("state chunk read out of bounds");
+    }
+    if (!isMaster()) {
+        // Lazy chunked pull of just this range
+        auto cli = getStateClient(masterHost);
+        auto data = cli->pullChunk(user, key, offset, len);
+        std::lock_guard<std::mutex> lock(kvMx);
+        std::memcpy(value.data() + offset, data.data(), data.size());
+        std::memcpy(buffer, data.data(), len);
+        return;
+    }
+    std::lock_guard<std::mutex> lock(kvMx);
+    std::memcpy(buffer, value.data() + offset, len);
+}
+
+void StateKeyValue::setChunk(uint64_t offset, const uint8_t* buffer,
+                             size_t len)
+{
+    if (offset + len > valueSize) {
+        throw FaabricException("state chunk write out of bounds");
+    }
+    {
+        std::lock_guard<std::mutex> lock(kvMx);
+        std::memcpy(value.data() + offset, buffer, len);
+    }
+    flagChunkDirty(offset, len);
+    if (!isMaster()) {
+        auto cli = getStateClient(masterHost);
+        cli->pushChunk(user, key, offset, buffer, len);
+    }
+}
+
+void StateKeyValue::pull()
+{
+    if (isMaster()) {
+        return;
+    }
+    {
+        std::lock_guard<std::mutex> lock(kvMx);
+        if (fullyPulled) {
+            return;
+        }
+    }
+    auto cli = getStateClient(masterHost);
+    for (uint64_t off = 0; off < valueSize;
+         off += STATE_STREAM_CHUNK_SIZE) {
+        size_t len = std::min(STATE_STREAM_CHUNK_SIZE,
+                              (size_t)(valueSize - off));
+        auto data = cli->pullChunk(user, key, off, len);
+        std::lock_guard<std::mutex> lock(kvMx);
+        std::memcpy(value.data() + off, data.data(), data.size());
+    }
+    std::lock_guard<std::mutex> lock(kvMx);
+    fullyPulled = true;
+}
+
+void StateKeyValue::pushFull()
+{
+    if (isMaster()) {
+        return;
+    }
+    auto cli = getStateClient(masterHost);
+    std::lock_guard<std::mutex> lock(kvMx);
+    for (uint64_t off = 0; off < valueSize;
+         off += STATE_STREAM_CHUNK_SIZE) {
+        size_t len = std::min(STATE_STREAM_CHUNK_SIZE,
+                              (size_t)(valueSize - off));
+        cli->pushChunk(user, key, off, value.data() + off, len);
+    }
+    std::fill(dirtyChunks.begin(), dirtyChunks.end(), 0);
+}
+
+void StateKeyValue::flagDirty()
+{
+    std::lock_guard<std::mutex> lock(kvMx);
+    std::fill(dirtyChunks.begin(), dirtyChunks.end(), 1);
+}
+
+void StateKeyValue::flagChunkDirty(uint64_t offset, size_t len)
+{
+    std::lock_guard<std::mutex> lock(kvMx);
+    size_t first = offset / STATE_STREAM_CHUNK_SIZE;
+    size_t last = (offset + len - 1) / STATE_STREAM_CHUNK_SIZE;
+    for (size_t i = first; i <= last && i < dirtyChunks.size(); i++) {
+        dirtyChunks[i] = 1;
+    }
+}
+
+void StateKeyValue::pushPartial()
+{
+    if (isMaster()) {
+        std::lock_guard<std::mutex> lock(kvMx);
+        std::fill(dirtyChunks.begin(), dirtyChunks.end(), 0);
+        return;
+    }
+    auto cli = getStateClient(masterHost);
+    std::lock_guard<std::mutex> lock(kvMx);
+    for (size_t i = 0; i < dirtyChunks.size(); i++) {
+        if (dirtyChunks[i] == 0) {
+            continue;
+        }
+        uint64_t off = i * STATE_STREAM_CHUNK_SIZE;
+        size_t len = std::min(STATE_STREAM_CHUNK_SIZE,
+                              (size_t)(valueSize - off));
+        cli->pushChunk(user, key, off, value.data() + off, len);
+        dirtyChunks[i] = 0;
+    }
+}
+
+void StateKeyValue::append(const uint8_t* data, size_t len)
+{
+    if (isMaster()) {
+        serviceAppend(data, len);
+        return;
+    }
+    auto cli = getStateClient(masterHost);
+    cli->append(user, key, data, len);
+}
+
+std::vector<std::vector<uint8_t>> StateKeyValue::getAppended(size_t nValues)
+{
+    if (isMaster()) {
+        return serviceGetAppended(nValues);
+    }
+    auto cli = getStateClient(masterHost);
+    return cli->pullAppended(user, key, nValues);
+}
+
+void StateKeyValue::clearAppended()
+{
+    if (isMaster()) {
+        serviceClearAppended();
+        return;
+    }
+    auto cli = getStateClient(masterHost);
+    cli->clearAppended(user, key);
+}
+
+std::vector<uint8_t> StateKeyValue::serviceChunk(uint64_t offset, size_t len)
+{
+    std::lock_guard<std::mutex> lock(kvMx);
+    if (offset + len > valueSize) {
+        throw FaabricException("state chunk service out of bounds");
+    }
+    return { value.begin() + offset, value.begin() + offset + len };
+}
+
+void StateKeyValue::serviceSet(uint64_t offset,
+                               const uint8_t* data,
+                               size_t len)
+{
+    std::lock_guard<std::mutex> lock(kvMx);
+    if (offset + len > valueSize) {
+        throw FaabricException("state chunk service-set out of bounds");
+    }
+    std::memcpy(value.data() + offset, data, len);
+}
+
+void StateKeyValue::serviceAppend(const uint8_t* data, size_t len)
+{
+    std::lock_guard<std::mutex> lock(kvMx);
+    appendedValues.emplace_back(data, data + len);
+}
+
+std::vector<std::vector<uint8_t>> StateKeyValue::serviceGetAppended(size_t n)
+{
+    std::lock_guard<std::mutex> lock(kvMx);
+    if (n > appendedValues.size()) {
+        throw FaabricException("not enough appended values");
+    }
+    return { appendedValues.begin(), appendedValues.begin() + n };
+}
+
+void StateKeyValue::serviceClearAppended()
+{
+    std::lock_guard<std::mutex> lock(kvMx);
+    appendedValues.clear();
+}
+
+// ------------------------- State --------------------------------------------
+
+State& State::get()
+{
+    static State st;
+    return st;
+}
+
+std::shared_ptr<StateKeyValue> State::getKV(const std::string& user,
+                                            const std::string& key,
+                                            size_t size)
+{
+    std::lock_guard<std::mutex> lock(mx);
+    std::string k = kvKeyOf(user, key);
+    auto it = kvMap.find(k);
+    if (it != kvMap.end()) {
+        return it->second;
+    }
+    std::string master;
+    auto mIt = masterMap.find(k);
+    if (mIt != masterMap.end()) {
+        master = mIt->second;
+    } else {
+        master = getSystemConfig().endpointHost;
+        masterMap[k] = master;
+    }
+    auto kv = std::make_shared<StateKeyValue>(user, key, size, master);
+    kvMap[k] = kv;
+    return kv;
+}
+
+std::shared_ptr<StateKeyValue> State::getKV(const std::string& user,
+                                            const std::string& key)
+{
+    std::lock_guard<std::mutex> lock(mx);
+    auto it = kvMap.find(kvKeyOf(user, key));
+    if (it == kvMap.end()) {
+        throw FaabricException("state key not found: " +
+                               kvKeyOf(user, key));
+    }
+    return it->second;
+}
+
+size_t State::getStateSize(const std::string& user, const std::string& key)
+{
+    {
+        std::lock_guard<std::mutex> lock(mx);
+        auto it = kvMap.find(kvKeyOf(user, key));
+        if (it != kvMap.end()) {
+            return it->second->size();
+        }
+    }
+    // Ask the owner if we know one
+    std::string master = getMasterHost(user, key);
+    if (!master.empty() && master != getSystemConfig().endpointHost) {
+        return getStateClient(master)->stateSize(user, key);
+    }
+    return 0;
+}
+
+void State::deleteKV(const std::string& user, const std::string& key)
+{
+    std::string master;
+    {
+        std::lock_guard<std::mutex> lock(mx);
+        auto mIt = masterMap.find(kvKeyOf(user, key));
+        master = mIt == masterMap.end() ? "" : mIt->second;
+    }
+    if (!master.empty() && master != getSystemConfig().endpointHost) {
+        getStateClient(master)->deleteKV(user, key);
+    }
+    deleteKVLocally(user, key);
+}
+
+void State::deleteKVLocally(const std::string& user, const std::string& key)
+{
+    std::lock_guard<std::mutex> lock(mx);
+    kvMap.erase(kvKeyOf(user, key));
+    masterMap.erase(kvKeyOf(user, key));
+}
+
+size_t State::getKVCount()
+{
+    std::lock_guard<std::mutex> lock(mx);
+    return kvMap.size();
+}
+
+void State::forceClearAll(bool global)
+{
+    (void)global;
+    std::lock_guard<std::mutex> lock(mx);
+    kvMap.clear();
+    masterMap.clear();
+}
+
+void State::setMasterHost(const std::string& user,
+                          const std::string& key,
+                          const std::string& host)
+{
+    std::lock_guard<std::mutex> lock(mx);
+    masterMap[kvKeyOf(user, key)] = host;
+}
+
+std::string State::getMasterHost(const std::string& user,
+                                 const std::string& key)
+{
+    std::lock_guard<std::mutex> lock(mx);
+    auto it = masterMap.find(kvKeyOf(user, key));
+    return it == masterMap.end() ? "" : it->second;
+}
+
+// ------------------------- server -------------------------------------------
+
+StateServer::StateServer()
+  : MessageEndpointServer(STATE_ASYNC_PORT, STATE_SYNC_PORT, "state")
+{}
+
+void StateServer::doAsyncRecv(uint8_t code,
+                              const std::string& body,
+                              uint32_t seq)
+{
+    (void)code;
+    (void)body;
+    (void)seq;
+    FAM_ERROR("state server has no async calls");
+}
+
+std::string StateServer::doSyncRecv(uint8_t code, const std::string& body)
+{
+    auto& st = State::get();
+    switch ((StateCalls)code) {
+        case StateCalls::Pull: {
+            auto req = StateChunkRequest::decode(body);
+            auto kv = st.getKV(req.user, req.key);
+            StatePart part;
+            part.user = req.user;
+            part.key = req.key;
+            part.offset = req.offset;
+            part.data = kv->serviceChunk(req.offset, req.chunkSize);
+            return part.encode();
+        }
+        case StateCalls::Push: {
+            auto part = StatePart::decode(body);
+            // Create on demand so a push can establish the value
+            auto kv = st.getKV(part.user,
+                               part.key,
+                               part.offset + part.data.size());
+            if (part.offset + part.data.size() > kv->size()) {
+                throw FaabricException("push beyond registered size");
+            }
+            kv->serviceSet(part.offset, part.data.data(), part.data.size());
+            return {};
+        }
+        case StateCalls::Size: {
+            auto req = StateRequest::decode(body);
+            StateSizeResponse resp;
+            resp.user = req.user;
+            resp.key = req.key;
+            resp.stateSize = st.getStateSize(req.user, req.key);
+            return resp.encode();
+        }
+        case StateCalls::Append: {
+            auto req = StateRequest::decode(body);
+            auto kv = st.getKV(req.user, req.key, 1);
+            kv->serviceAppend(req.data.data(), req.data.size());
+            return {};
+        }
+        case StateCalls::PullAppended: {
+            auto req = StateAppendedRequest::decode(body);
+            auto kv = st.getKV(req.user, req.key);
+            StateAppendedResponse resp;
+            resp.user = req.user;
+            resp.key = req.key;
+            resp.values = kv->serviceGetAppended(req.nValues);
+            return resp.encode();
+        }
+        case StateCalls::ClearAppended: {
+            auto req = StateRequest::decode(body);
+            auto kv = st.getKV(req.user, req.key);
+            kv->serviceClearAppended();
+            return {};
+        }
+        case StateCalls::Delete: {
+            auto req = StateRequest::decode(body);
+            st.deleteKVLocally(req.user, req.key);
+            return {};
+        }
+        default:
+            throw FaabricException("state server: bad sync code " +
+                                   std::to_string(code));
+    }
+}
+
+// ------------------------- client -------------------------------------------
+
+StateClient::StateClient(const std::string& host)
+  : MessageEndpointClient(host, STATE_ASYNC_PORT, STATE_SYNC_PORT)
+{}
+
+std::vector<uint8_t> StateClient::pullChunk(const std::string& user,
+                                            const std::string& key,
+                                            uint64_t offset,
+                                            size_t len)
+{
+    StateChunkRequest req;
+    req.user = user;
+    req.key = key;
+    req.offset = offset;
+    req.chunkSize = len;
+    std::string resp = syncSend((uint8_t)StateCalls::Pull, req.encode());
+    return StatePart::decode(resp).data;
+}
+
+void StateClient::pushChunk(const std::string& user,
+                            const std::string& key,
+                            uint64_t offset,
+                            const uint8_t* data,
+                            size_t len)
+{
+    StatePart part;
+    part.user = user;
+    part.key = key;
+    part.offset = offset;
+    part.data.assign(data, data + len);
+    syncSend((uint8_t)StateCalls::Push, part.encode());
+}
+
+size_t StateClient::stateSize(const std::string& user, const std::string& key)
+{
+    StateRequest req;
+    req.user = user;
+    req.key = key;
+    std::string resp = syncSend((uint8_t)StateCalls::Size, req.encode());
+    return StateSizeResponse::decode(resp).stateSize;
+}
+
+void StateClient::append(const std::string& user,
+                         const std::string& key,
+                         const uint8_t* data,
+                         size_t len)
+{
+    StateRequest req;
+    req.user = user;
+    req.key = key;
+    req.data.assign(data, data + len);
+    syncSend((uint8_t)StateCalls::Append, req.encode());
+}
+
+std::vector<std::vector<uint8_t>> StateClient::pullAppended(
+  const std::string& user,
+  const std::string& key,
+  size_t nValues)
+{
+    StateAppendedRequest req;
+    req.user = user;
+    req.key = key;
+    req.nValues = (uint32_t)nValues;
+    std::string resp =
+      syncSend((uint8_t)StateCalls::PullAppended, req.encode());
+    return StateAppendedResponse::decode(resp).values;
+}
+
+void StateClient::clearAppended(const std::string& user,
+                                const std::string& key)
+{
+    StateRequest req;
+    req.user = user;
+    req.key = key;
+    syncSend((uint8_t)StateCalls::ClearAppended, req.encode());
+}
+
+void StateClient::deleteKV(const std::string& user, const std::string& key)
+{
+    StateRequest req;
+    req.user = user;
+    req.key = key;
+    syncSend((uint8_t)StateCalls::Delete, req.encode());
+}
+
+static std::mutex stClientsMx;
+static std::map<std::string, std::shared_ptr<StateClient>> stClients;
+
+std::shared_ptr<StateClient> getStateClient(const std::string& host)
+{
+    std::lock_guard<std::mutex> lock(stClientsMx);
+    auto& cli = stClients[host];
+    if (!cli) {
+        cli = std::make_shared<StateClient>(host);
+    }
+    return cli;
+}
+
+void clearStateClients()
+{
+    std::lock_guard<std::mutex> lock(stClientsMx);
+    stClients.clear();
+}
+
+} // namespace faabricamd

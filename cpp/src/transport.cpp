@@ -385,6 +385,13 @@ void MessageEndpointServer::stop()
     }
     asyncListener.close();
     syncListener.close();
+    {
+        // Unblock handler threads parked in recv on live connections
+        std::lock_guard<std::mutex> lock(connThreadsMx);
+        for (int fd : activeConnFds) {
+            ::shutdown(fd, SHUT_RDWR);
+        }
+    }
     if (asyncAcceptThread.joinable()) {
         asyncAcceptThread.join();
     }
@@ -423,6 +430,11 @@ void MessageEndpointServer::acceptLoop(TcpListener& listener, bool isSync)
 
 void MessageEndpointServer::connectionLoop(TcpConnection conn, bool isSync)
 {
+    {
+        std::lock_guard<std::mutex> lock(connThreadsMx);
+        activeConnFds.insert(conn.rawFd());
+    }
+    int thisFd = conn.rawFd();
     WireHeader hdr;
     std::string body;
     while (running.load()) {
@@ -438,7 +450,7 @@ void MessageEndpointServer::connectionLoop(TcpConnection conn, bool isSync)
                 std::string resp = doSyncRecv(hdr.code, body);
                 conn.sendFrame(0, resp.data(), resp.size(), hdr.seq);
             } else {
-                doAsyncRecv(hdr.code, body);
+                doAsyncRecv(hdr.code, body, hdr.seq);
             }
         } catch (const std::exception& e) {
             FAM_ERROR("%s server handler error (code %d): %s",
@@ -456,6 +468,8 @@ void MessageEndpointServer::connectionLoop(TcpConnection conn, bool isSync)
             }
         }
     }
+    std::lock_guard<std::mutex> lock(connThreadsMx);
+    activeConnFds.erase(thisFd);
 }
 
 // ------------------------- MessageEndpointClient ----------------------------
@@ -468,12 +482,31 @@ MessageEndpointClient::MessageEndpointClient(std::string hostIn,
   , syncPort(syncPortIn)
 {}
 
+void parseHostIdentity(const std::string& identity,
+                       std::string& ipOut,
+                       int& offsetOut)
+{
+    auto at = identity.find('@');
+    if (at == std::string::npos) {
+        ipOut = identity;
+        offsetOut = 0;
+    } else {
+        ipOut = identity.substr(0, at);
+        offsetOut = atoi(identity.c_str() + at + 1);
+    }
+}
+
 TcpConnection& MessageEndpointClient::ensure(bool sync)
 {
     TcpConnection& conn = sync ? syncConn : asyncConn;
     if (!conn.isOpen()) {
-        int port = (sync ? syncPort : asyncPort) + getPortOffset();
-        conn = TcpConnection::dial(host, port);
+        // Host identities may carry a port offset ("ip@offset") so several
+        // single-GPU worker processes can share one node's IP
+        std::string ip;
+        int offset = 0;
+        parseHostIdentity(host, ip, offset);
+        int port = (sync ? syncPort : asyncPort) + offset;
+        conn = TcpConnection::dial(ip, port);
     }
     return conn;
 }
@@ -487,13 +520,21 @@ void MessageEndpointClient::asyncSend(uint8_t code,
                                       const void* body,
                                       size_t len)
 {
+    asyncSendSeq(code, body, len, 0);
+}
+
+void MessageEndpointClient::asyncSendSeq(uint8_t code,
+                                         const void* body,
+                                         size_t len,
+                                         uint32_t seq)
+{
     std::lock_guard<std::mutex> lock(asyncMx);
     try {
-        ensure(false).sendFrame(code, body, len);
+        ensure(false).sendFrame(code, body, len, seq);
     } catch (const SocketClosedException&) {
         // One reconnect attempt
         asyncConn.close();
-        ensure(false).sendFrame(code, body, len);
+        ensure(false).sendFrame(code, body, len, seq);
     }
 }
 

@@ -1,0 +1,135 @@
+"""Multi-process distributed tests: one planner process (this one) plus
+worker subprocesses with distinct port offsets, covering cross-host
+scheduling, remote dispatch, and result collection (the reference's
+tests/dist/ two-container coverage, single-node form)."""
+
+import multiprocessing as mp
+import os
+import sys
+import time
+
+import pytest
+
+REPO_ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+WORKER_SLOTS = 2
+
+
+def _worker_main(port_offset, stop_event, ready_event):
+    sys.path.insert(0, REPO_ROOT)
+    from faabric_amd import _core
+    from faabric_amd.runtime import LocalRuntime
+
+    _core.set_log_level("error")
+    rt = LocalRuntime(port_offset=port_offset, slots=WORKER_SLOTS)
+    rt.start_worker()
+    _core.register_native_echo("dist", "echo")
+    _core.register_native_noop("dist", "noop")
+    _core.register_native_sleep("dist", "sleep", 200)
+    ready_event.set()
+    stop_event.wait(120)
+    rt.stop()
+
+
+@pytest.fixture(scope="module")
+def cluster():
+    from faabric_amd import _core
+    from faabric_amd.runtime import LocalRuntime
+
+    # Planner runs in this process; no worker here
+    rt = LocalRuntime(port_offset=0)
+    rt.start_planner()
+
+    ctx = mp.get_context("spawn")
+    stop = ctx.Event()
+    workers = []
+    readies = []
+    for off in (1000, 2000):
+        ready = ctx.Event()
+        p = ctx.Process(target=_worker_main, args=(off, stop, ready))
+        p.start()
+        workers.append(p)
+        readies.append(ready)
+    for r in readies:
+        assert r.wait(60), "worker failed to start"
+    # Wait for both registrations to land
+    deadline = time.monotonic() + 10
+    while time.monotonic() < deadline:
+        if len(_core.get_available_hosts()) == 2:
+            break
+        time.sleep(0.05)
+    yield rt
+    stop.set()
+    for p in workers:
+        p.join(timeout=30)
+        if p.is_alive():
+            p.terminate()
+    rt.stop()
+
+
+def test_two_hosts_registered(cluster):
+    from faabric_amd import _core
+
+    hosts = _core.get_available_hosts()
+    assert len(hosts) == 2
+    ips = sorted(h.ip for h in hosts)
+    assert ips == ["127.0.0.1@1000", "127.0.0.1@2000"]
+    assert all(h.slots == WORKER_SLOTS for h in hosts)
+
+
+def test_batch_spans_hosts(cluster):
+    from faabric_amd import _core
+    from faabric_amd.runtime import wait_for_batch
+
+    n = 2 * WORKER_SLOTS
+    ber = _core.batch_exec_factory("dist", "echo", n)
+    msgs = ber.messages
+    for m in msgs:
+        m.input_data = b"spanning"
+    ber.messages = msgs
+    decision = _core.call_functions(ber)
+    assert decision.n_functions == n
+    assert len(set(decision.hosts)) == 2
+
+    results = wait_for_batch(ber.app_id, n)
+    assert len(results) == n
+    assert all(r.output_data == "spanning" for r in results)
+    hosts_used = {r.executed_host for r in results}
+    assert hosts_used == {"127.0.0.1@1000", "127.0.0.1@2000"}
+
+
+def test_single_message_remote(cluster):
+    from faabric_amd import _core
+    from faabric_amd.runtime import wait_for_batch
+
+    ber = _core.batch_exec_factory("dist", "noop", 1)
+    _core.call_functions(ber)
+    results = wait_for_batch(ber.app_id, 1)
+    assert results[0].return_value == 0
+
+
+def test_concurrent_batches(cluster):
+    from faabric_amd import _core
+    from faabric_amd.runtime import wait_for_batch
+
+    bers = []
+    for _ in range(4):
+        ber = _core.batch_exec_factory("dist", "noop", 1)
+        _core.call_functions(ber)
+        bers.append(ber)
+    for ber in bers:
+        results = wait_for_batch(ber.app_id, 1)
+        assert results[0].return_value == 0
+
+
+def test_slots_freed_after_batch(cluster):
+    from faabric_amd import _core
+    from faabric_amd.runtime import execute_batch
+
+    # Repeatedly run full-width batches: slots must be released each time
+    n = 2 * WORKER_SLOTS
+    for _ in range(3):
+        results = execute_batch("dist", "noop", n)
+        assert len(results) == n
+    hosts = _core.get_available_hosts()
+    assert all(h.used_slots == 0 for h in hosts)

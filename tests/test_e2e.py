@@ -1,0 +1,165 @@
+"""End-to-end single-host tests: planner + worker in one process, batches
+through the full RPC path (the reference's examples/check.cpp +
+tests/dist/scheduler/test_funcs.cpp coverage, single-host part)."""
+
+import threading
+
+import pytest
+
+import faabric_amd as fa
+from faabric_amd import _core
+from faabric_amd.runtime import LocalRuntime, execute_batch
+
+SLOTS = 8
+
+
+@pytest.fixture(scope="module")
+def runtime():
+    rt = LocalRuntime(slots=SLOTS)
+    # Colocated planner+worker share the process snapshot registry, so the
+    # planner does not need its own snapshot server (port clash otherwise)
+    rt.start_planner(with_snapshot_server=False)
+    rt.start_worker()
+    _core.register_native_echo("demo", "echo")
+    _core.register_native_noop("demo", "noop")
+    yield rt
+    rt.stop()
+
+
+def test_single_function(runtime):
+    results = execute_batch("demo", "echo", 1, input_data=b"hello world")
+    assert len(results) == 1
+    assert results[0].return_value == 0
+    assert results[0].output_data == "hello world"
+    assert results[0].executed_host == runtime.identity
+
+
+def test_batch_of_functions(runtime):
+    results = execute_batch("demo", "noop", SLOTS)
+    assert len(results) == SLOTS
+    assert all(r.return_value == 0 for r in results)
+    assert sorted(r.app_idx for r in results) == list(range(SLOTS))
+
+
+def test_python_function(runtime):
+    calls = []
+
+    def handler(msg):
+        calls.append(msg.input_data)
+        msg.output_data = "py:" + msg.input_data.decode()
+        return 0
+
+    _core.register_function("demo", "pyfunc", handler)
+    results = execute_batch("demo", "pyfunc", 2, input_data=b"x")
+    assert len(results) == 2
+    assert all(r.output_data == "py:x" for r in results)
+    assert calls == [b"x", b"x"]
+
+
+def test_failing_function(runtime):
+    def bad(msg):
+        raise ValueError("boom")
+
+    _core.register_function("demo", "bad", bad)
+    results = execute_batch("demo", "bad", 1)
+    assert results[0].return_value == 1
+    assert "boom" in results[0].output_data
+
+
+def test_not_enough_slots(runtime):
+    ber = _core.batch_exec_factory("demo", "noop", SLOTS + 1)
+    decision = _core.call_functions(ber)
+    assert decision.app_id == _core.NOT_ENOUGH_SLOTS()
+
+
+def test_sequential_batches_reuse_executors(runtime):
+    results = execute_batch("demo", "noop", 4)
+    assert len(results) == 4
+    count_after_first = _core.get_executor_count()
+    for _ in range(3):
+        results = execute_batch("demo", "noop", 4)
+        assert len(results) == 4
+    # Warm executors are reused, not recreated per batch
+    assert _core.get_executor_count() == count_after_first
+
+
+def test_get_message_result(runtime):
+    ber = _core.batch_exec_factory("demo", "echo", 1)
+    msgs = ber.messages
+    msgs[0].input_data = b"direct"
+    ber.messages = msgs
+    msg_id = ber.messages[0].id
+    _core.call_functions(ber)
+    result = _core.get_message_result(ber.app_id, msg_id, 10_000)
+    assert result.output_data == "direct"
+
+
+def test_available_hosts(runtime):
+    hosts = _core.get_available_hosts()
+    assert len(hosts) == 1
+    assert hosts[0].ip == runtime.identity
+    assert hosts[0].slots == SLOTS
+
+
+def test_ptp_local_messaging(runtime):
+    decision = _core.SchedulingDecision()
+    decision.app_id = 424242
+    decision.group_id = 424243
+    decision.hosts = [runtime.identity, runtime.identity]
+    decision.message_ids = [1, 2]
+    decision.app_idxs = [0, 1]
+    decision.group_idxs = [0, 1]
+    decision.mpi_ports = [0, 0]
+    decision.n_functions = 2
+    _core.ptp_setup_local_mappings(decision)
+
+    _core.ptp_send(424242, 424243, 0, 1, b"ping", True)
+    _core.ptp_send(424242, 424243, 0, 1, b"pong", True)
+    assert _core.ptp_recv(424243, 0, 1, True) == b"ping"
+    assert _core.ptp_recv(424243, 0, 1, True) == b"pong"
+
+
+def test_ptp_group_barrier_and_lock(runtime):
+    decision = _core.SchedulingDecision()
+    decision.app_id = 555000
+    decision.group_id = 555001
+    decision.hosts = [runtime.identity] * 3
+    decision.message_ids = [1, 2, 3]
+    decision.app_idxs = [0, 1, 2]
+    decision.group_idxs = [0, 1, 2]
+    decision.mpi_ports = [0, 0, 0]
+    decision.n_functions = 3
+    _core.ptp_setup_local_mappings(decision)
+
+    passed = []
+    counter = {"v": 0}
+
+    def member(idx):
+        _core.ptp_group_barrier(555001, idx)
+        _core.ptp_group_lock(555001, idx)
+        v = counter["v"]
+        counter["v"] = v + 1
+        _core.ptp_group_unlock(555001, idx)
+        _core.ptp_group_barrier(555001, idx)
+        passed.append(idx)
+
+    threads = [threading.Thread(target=member, args=(i,)) for i in range(3)]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join(timeout=15)
+    assert sorted(passed) == [0, 1, 2]
+    assert counter["v"] == 3
+
+
+def test_state_local(runtime):
+    kv = _core.state_get_kv("demo", "statekey", 128)
+    assert kv.is_master
+    kv.set(b"a" * 128)
+    assert kv.get() == b"a" * 128
+    kv.set_chunk(10, b"BBBB")
+    data = kv.get()
+    assert data[10:14] == b"BBBB"
+    kv.append(b"v1")
+    kv.append(b"v2")
+    assert kv.get_appended(2) == [b"v1", b"v2"]
