@@ -40,10 +40,14 @@ $(TARGET): $(OBJS)
 	$(HIPCC) $(OBJS) $(LDFLAGS) -o $@
 
 # C++ example embedder (reference examples/check.cpp parity)
-examples: $(TARGET)
-	$(HIPCC) -O2 -std=c++20 -Icpp/include examples/check.cpp \
-	    $(filter-out $(BUILD)/bindings/%,$(OBJS)) \
+CORE_OBJS = $(filter-out $(BUILD)/bindings/%,$(OBJS))
+examples: $(TARGET) $(BUILD)/examples/check.o
+	$(HIPCC) $(BUILD)/examples/check.o $(CORE_OBJS) \
 	    -L/opt/rocm/lib -lrccl -lamdhip64 -lz -pthread -o $(BUILD)/check
+
+$(BUILD)/examples/%.o: examples/%.cpp
+	@mkdir -p $(dir $@)
+	$(HIPCC) $(CXXFLAGS) -c $< -o $@
 
 clean:
 	rm -rf $(BUILD) $(TARGET)
