@@ -15,6 +15,10 @@
 namespace py = pybind11;
 using namespace faabricamd;
 
+namespace faabricamd {
+void registerBenchFunctions(); // bench_funcs.cpp
+}
+
 void initRuntimeBindings(py::module_& m)
 {
     // ---------------- config ----------------
@@ -465,4 +469,7 @@ void initRuntimeBindings(py::module_& m)
         return getBatchRequestsSentMock();
     });
     m.def("clear_mocked_function_calls", &clearMockedFunctionCalls);
+
+    // Native benchmark payloads (cpp/src/bench_funcs.cpp)
+    m.def("register_bench_functions", [] { registerBenchFunctions(); });
 }

@@ -17,6 +17,7 @@ using namespace faabricamd;
 
 void initRuntimeBindings(py::module_& m); // bindings_runtime.cpp
 void initOpsBindings(py::module_& m);     // bindings_ops.cpp
+void initMpiBindings(py::module_& m);     // bindings_mpi.cpp
 
 namespace {
 
@@ -170,6 +171,7 @@ PYBIND11_MODULE(_core, m)
 
     initRuntimeBindings(m);
     initOpsBindings(m);
+    initMpiBindings(m);
 
     // Python callables captured by the C++ function registry must be
     // released while the interpreter is still alive (the registry is a

@@ -1,0 +1,222 @@
+// Native benchmark payload functions (registered into the FunctionRegistry
+// so the hot paths never touch Python). These implement the BASELINE.json
+// workload shapes:
+//  - "bench/kvtouch":   HBM/host state-KV touch, used by the 1024-function
+//                       batch-throughput config
+//  - "bench/rankstep":  one MPI rank of the composite benchmark step:
+//                       256 MB fp32 allreduce over RCCL/xGMI (+ optional
+//                       small alltoall) and, on rank 0, a batch submission
+//                       of kvtouch functions per step
+// (reference harness shapes: tests/dist/mpi/benchmarks/mpi_allreduce.cpp,
+//  mpi_bench.cpp:18-45, tests/dist/scheduler/test_funcs.cpp)
+#include "faabricamd/executor.h"
+#include "faabricamd/messages.h"
+#include "faabricamd/mpi.h"
+#include "faabricamd/planner.h"
+#include "faabricamd/state.h"
+#include "faabricamd/util.h"
+
+#include <hip/hip_runtime.h>
+
+#include <cmath>
+#include <cstring>
+#include <sstream>
+
+namespace faabricamd {
+
+// Parse "k1=v1;k2=v2;..." from inputData
+static std::map<std::string, int64_t> parseParams(const Message& msg)
+{
+    std::map<std::string, int64_t> out;
+    std::string s(msg.inputData.begin(), msg.inputData.end());
+    std::stringstream ss(s);
+    std::string item;
+    while (std::getline(ss, item, ';')) {
+        auto eq = item.find('=');
+        if (eq != std::string::npos) {
+            out[item.substr(0, eq)] = atoll(item.c_str() + eq + 1);
+        }
+    }
+    return out;
+}
+
+static int32_t benchKvTouch(Message& msg)
+{
+    auto params = parseParams(msg);
+    int64_t kvBytes = params.count("kvbytes") ? params["kvbytes"] : 4096;
+    // Each host owns one KV; functions write disjoint-ish chunks
+    static thread_local std::vector<uint8_t> buf;
+    buf.resize(kvBytes);
+    std::memset(buf.data(), (int)(msg.id & 0xff), kvBytes);
+
+    auto kv = State::get().getKV("bench", "kv", 1024 * 1024);
+    uint64_t offset =
+      ((uint64_t)(uint32_t)msg.id * 4096) % (1024 * 1024 - kvBytes);
+    kv->setChunk(offset, buf.data(), kvBytes);
+    kv->getChunk(offset, buf.data(), kvBytes);
+    if (buf[0] != (uint8_t)(msg.id & 0xff)) {
+        return 1;
+    }
+    return 0;
+}
+
+static int32_t benchRankStep(Message& msg)
+{
+    auto& ctx = getMpiContext();
+    if (msg.mpiRank == 0) {
+        ctx.createWorld(msg);
+    } else {
+        ctx.joinWorld(msg);
+    }
+    MpiWorld& world = ctx.getWorld();
+    int rank = ctx.getRank();
+    int worldSize = world.getSize();
+
+    auto params = parseParams(msg);
+    int steps = (int)(params.count("steps") ? params["steps"] : 5);
+    int warmup = (int)(params.count("warmup") ? params["warmup"] : 2);
+    int64_t bytes =
+      params.count("bytes") ? params["bytes"] : 256LL * 1024 * 1024;
+    int batchPerHost = (int)(params.count("batch") ? params["batch"] : 0);
+    int64_t kvBytes = params.count("kvbytes") ? params["kvbytes"] : 4096;
+    int64_t a2aBytes = params.count("a2abytes") ? params["a2abytes"] : 0;
+
+    int nGpus = 0;
+    (void)hipGetDeviceCount(&nGpus);
+    bool onGpu = nGpus > 0;
+
+    size_t count = (size_t)bytes / sizeof(float);
+    uint8_t* sendBuf = nullptr;
+    uint8_t* recvBuf = nullptr;
+    uint8_t* a2aSend = nullptr;
+    uint8_t* a2aRecv = nullptr;
+    std::vector<uint8_t> hostSend;
+    std::vector<uint8_t> hostRecv2;
+    std::vector<uint8_t> hostA2a;
+    std::vector<uint8_t> hostA2aRecv;
+
+    if (onGpu) {
+        if (hipSetDevice(0) != hipSuccess ||
+            hipMalloc(&sendBuf, bytes) != hipSuccess ||
+            hipMalloc(&recvBuf, bytes) != hipSuccess) {
+            msg.outputData = "hip alloc failed";
+            return 1;
+        }
+        // Non-trivial payload (DVFS give-back on zero-filled data)
+        hipMemset(sendBuf, 0x3f, bytes);
+        if (a2aBytes > 0) {
+            hipMalloc(&a2aSend, a2aBytes * worldSize);
+            hipMalloc(&a2aRecv, a2aBytes * worldSize);
+            hipMemset(a2aSend, 0x11, a2aBytes * worldSize);
+        }
+    } else {
+        hostSend.assign(bytes, 0x3f);
+        hostRecv2.assign(bytes, 0);
+        sendBuf = hostSend.data();
+        recvBuf = hostRecv2.data();
+        if (a2aBytes > 0) {
+            hostA2a.assign(a2aBytes * worldSize, 0x11);
+            hostA2aRecv.assign(a2aBytes * worldSize, 0);
+            a2aSend = hostA2a.data();
+            a2aRecv = hostA2aRecv.data();
+        }
+    }
+    MpiBufferLoc loc = onGpu ? MpiBufferLoc::DEVICE : MpiBufferLoc::HOST;
+
+    std::vector<double> stepMs;
+    std::vector<double> allreduceMs;
+    std::vector<double> batchMs;
+
+    for (int iter = 0; iter < warmup + steps; iter++) {
+        world.barrier(rank);
+        int64_t t0 = getEpochMicros();
+
+        world.allReduce(rank,
+                        sendBuf,
+                        recvBuf,
+                        MpiDataType::FLOAT,
+                        (int)count,
+                        MpiOp::SUM,
+                        loc);
+        int64_t tAr = getEpochMicros();
+
+        if (a2aBytes > 0) {
+            world.allToAll(rank,
+                           a2aSend,
+                           a2aRecv,
+                           MpiDataType::BYTE,
+                           (int)a2aBytes,
+                           loc);
+        }
+
+        // Batch-throughput half of the composite step: rank 0 submits a
+        // batch of kvtouch functions across all hosts and waits
+        int64_t tBatch0 = getEpochMicros();
+        if (batchPerHost > 0 && rank == 0) {
+            int total = batchPerHost * worldSize;
+            auto ber = std::make_shared<BatchExecuteRequest>(
+              batchExecFactory("bench", "kvtouch", total));
+            std::string kvParams = "kvbytes=" + std::to_string(kvBytes);
+            for (auto& m : ber->messages) {
+                m.inputData.assign(kvParams.begin(), kvParams.end());
+            }
+            auto decision = getPlannerClient().callFunctions(ber);
+            if (decision->appId == NOT_ENOUGH_SLOTS) {
+                msg.outputData = "batch bench: not enough slots";
+                break;
+            }
+            // Poll for completion
+            while (true) {
+                auto status = getPlannerClient().getBatchResults(ber->appId);
+                if (status.expectedNumMessages != -1 && status.finished &&
+                    (int)status.messageResults.size() >= total) {
+                    break;
+                }
+                usleep(200);
+            }
+        }
+        world.barrier(rank);
+        int64_t t1 = getEpochMicros();
+
+        if (iter >= warmup) {
+            stepMs.push_back((t1 - t0) / 1000.0);
+            allreduceMs.push_back((tAr - t0) / 1000.0);
+            batchMs.push_back((t1 - tBatch0) / 1000.0);
+        }
+    }
+
+    if (onGpu) {
+        hipFree(sendBuf);
+        hipFree(recvBuf);
+        if (a2aSend != nullptr) {
+            hipFree(a2aSend);
+            hipFree(a2aRecv);
+        }
+    }
+
+    std::ostringstream out;
+    out << "step:";
+    for (size_t i = 0; i < stepMs.size(); i++) {
+        out << (i ? "," : "") << stepMs[i];
+    }
+    out << ";ar:";
+    for (size_t i = 0; i < allreduceMs.size(); i++) {
+        out << (i ? "," : "") << allreduceMs[i];
+    }
+    out << ";batch:";
+    for (size_t i = 0; i < batchMs.size(); i++) {
+        out << (i ? "," : "") << batchMs[i];
+    }
+    msg.outputData = out.str();
+    return 0;
+}
+
+void registerBenchFunctions()
+{
+    FunctionRegistry::get().registerFunction("bench", "kvtouch",
+                                             benchKvTouch);
+    FunctionRegistry::get().registerFunction("bench", "rankstep",
+                                             benchRankStep);
+}
+
+} // namespace faabricamd

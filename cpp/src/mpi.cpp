@@ -1,0 +1,1177 @@
+// MPI world implementation (reference behavior: src/mpi/MpiWorld.cpp —
+// create :157-226, init :270-300, send/recv :544-705, collectives
+// :786-1484, op_reduce :1266-1389, barrier :1753-1775;
+// src/mpi/MpiWorldRegistry.cpp; src/mpi/MpiContext.cpp). See mpi.h for
+// the MI355X re-design notes (host plane on the PTP broker, device plane
+// on RCCL over xGMI).
+#include "faabricamd/mpi.h"
+#include "faabricamd/executor.h"
+#include "faabricamd/planner.h"
+#include "faabricamd/ptp.h"
+#include "faabricamd/util.h"
+
+#include <hip/hip_runtime.h>
+#include <rccl/rccl.h>
+
+#include <algorithm>
+#include <cstring>
+
+namespace faabricamd {
+
+size_t mpiTypeSize(MpiDataType t)
+{
+    switch (t) {
+        case MpiDataType::INT32:
+            return 4;
+        case MpiDataType::INT64:
+        case MpiDataType::UINT64:
+        case MpiDataType::DOUBLE:
+            return 8;
+        case MpiDataType::FLOAT:
+            return 4;
+        case MpiDataType::BYTE:
+            return 1;
+    }
+    return 1;
+}
+
+static ncclDataType_t toNccl(MpiDataType t)
+{
+    switch (t) {
+        case MpiDataType::INT32:
+            return ncclInt32;
+        case MpiDataType::INT64:
+            return ncclInt64;
+        case MpiDataType::UINT64:
+            return ncclUint64;
+        case MpiDataType::FLOAT:
+            return ncclFloat32;
+        case MpiDataType::DOUBLE:
+            return ncclFloat64;
+        case MpiDataType::BYTE:
+            return ncclUint8;
+    }
+    return ncclUint8;
+}
+
+static ncclRedOp_t toNcclOp(MpiOp op)
+{
+    switch (op) {
+        case MpiOp::SUM:
+            return ncclSum;
+        case MpiOp::MAX:
+            return ncclMax;
+        case MpiOp::MIN:
+            return ncclMin;
+        case MpiOp::PROD:
+            return ncclProd;
+    }
+    return ncclSum;
+}
+
+#define RCCL_CHECK(call)                                                       \
+    do {                                                                       \
+        ncclResult_t rcclRes_ = (call);                                        \
+        if (rcclRes_ != ncclSuccess) {                                         \
+            throw FaabricException(std::string("RCCL error: ") +               \
+                                   ncclGetErrorString(rcclRes_));              \
+        }                                                                      \
+    } while (0)
+
+#define HIP_CHECK(call)                                                        \
+    do {                                                                       \
+        hipError_t hipRes_ = (call);                                           \
+        if (hipRes_ != hipSuccess) {                                           \
+            throw FaabricException(std::string("HIP error: ") +                \
+                                   hipGetErrorString(hipRes_));                \
+        }                                                                      \
+    } while (0)
+
+// Host-plane channel namespace: sendIdx = sendRank + type * MAX_WORLD_SIZE
+static int32_t chanSendIdx(int sendRank, MpiMessageType type)
+{
+    return sendRank + (int32_t)type * MAX_MPI_WORLD_SIZE;
+}
+
+// ------------------------- per-rank TLS state -------------------------------
+
+struct PendingRecv
+{
+    int requestId = 0;
+    int sendRank = 0;
+    int recvRank = 0;
+    uint8_t* buffer = nullptr;
+    size_t bytes = 0;
+    MpiMessageType type = MpiMessageType::NORMAL;
+    bool done = false;
+};
+
+// Async request bookkeeping per rank-thread
+// (reference: MpiRankState src/mpi/MpiWorld.cpp:45-113)
+struct MpiRankState
+{
+    std::vector<PendingRecv> pendingRecvs;
+    std::atomic<int> nextRequestId{ 1 };
+};
+
+static thread_local MpiRankState rankState;
+
+// ------------------------- RCCL state ---------------------------------------
+
+struct MpiWorld::RcclState
+{
+    std::mutex mx;
+    ncclUniqueId uniqueId;
+    std::map<int, ncclComm_t> comms;      // world rank → comm
+    std::map<int, hipStream_t> streams;   // world rank → stream
+    std::map<int, int> devices;           // world rank → hip device
+    ~RcclState()
+    {
+        for (auto& [rank, comm] : comms) {
+            ncclCommDestroy(comm);
+        }
+        for (auto& [rank, stream] : streams) {
+            hipStreamDestroy(stream);
+        }
+    }
+};
+
+// ------------------------- world lifecycle ----------------------------------
+
+MpiWorld::MpiWorld()
+  : thisHost(getSystemConfig().endpointHost)
+{}
+
+MpiWorld::~MpiWorld() = default;
+
+void MpiWorld::create(Message& call, int newId, int newSize)
+{
+    if (newSize > MAX_MPI_WORLD_SIZE) {
+        throw FaabricException("MPI world size exceeds cap");
+    }
+    id = newId;
+    size = newSize;
+    appId = call.appId;
+    user = call.user;
+    function = call.function;
+
+    call.isMpi = true;
+    call.mpiWorldId = id;
+    call.mpiRank = 0;
+    call.mpiWorldSize = size;
+
+    // Dispatch the other ranks: SCALE_CHANGE consuming the planner's
+    // preloaded gang decision (reference: src/mpi/MpiWorld.cpp:157-226)
+    auto req = std::make_shared<BatchExecuteRequest>();
+    req->appId = appId;
+    req->user = user;
+    req->function = function;
+    req->type = BatchExecuteType::FUNCTIONS;
+    for (int i = 1; i < size; i++) {
+        Message m = messageFactory(user, function);
+        m.appId = appId;
+        m.appIdx = i;
+        m.groupIdx = i;
+        m.isMpi = true;
+        m.mpiWorldId = id;
+        m.mpiRank = i;
+        m.mpiWorldSize = size;
+        m.inputData = call.inputData;
+        req->messages.push_back(std::move(m));
+    }
+
+    auto decision = getPlannerClient().callFunctions(req);
+    if (decision->appId == NOT_ENOUGH_SLOTS) {
+        throw FaabricException("not enough slots to create MPI world");
+    }
+    groupId = decision->groupId;
+    call.groupId = groupId;
+    call.groupSize = size;
+
+    getPointToPointBroker().waitForMappingsOnThisHost(groupId);
+    {
+        std::lock_guard<std::mutex> lock(worldMx);
+        localRanks.push_back(0);
+    }
+}
+
+void MpiWorld::initialiseFromMsg(Message& msg)
+{
+    id = msg.mpiWorldId;
+    size = msg.mpiWorldSize;
+    appId = msg.appId;
+    groupId = msg.groupId;
+    user = msg.user;
+    function = msg.function;
+    getPointToPointBroker().waitForMappingsOnThisHost(groupId);
+}
+
+void MpiWorld::initialiseRankFromMsg(Message& msg)
+{
+    std::lock_guard<std::mutex> lock(worldMx);
+    if (std::find(localRanks.begin(), localRanks.end(), msg.mpiRank) ==
+        localRanks.end()) {
+        localRanks.push_back(msg.mpiRank);
+    }
+}
+
+std::string MpiWorld::getHostForRank(int rank)
+{
+    {
+        std::lock_guard<std::mutex> lock(worldMx);
+        if ((int)rankHosts.size() == size && !rankHosts[rank].empty()) {
+            return rankHosts[rank];
+        }
+    }
+    std::string host = getPointToPointBroker().getHostForReceiver(groupId,
+                                                                  rank);
+    std::lock_guard<std::mutex> lock(worldMx);
+    if ((int)rankHosts.size() != size) {
+        rankHosts.assign(size, "");
+    }
+    rankHosts[rank] = host;
+    return host;
+}
+
+bool MpiWorld::destroy()
+{
+    std::lock_guard<std::mutex> lock(worldMx);
+    localRanks.clear();
+    rccl.reset();
+    return true;
+}
+
+// ------------------------- cartesian topology -------------------------------
+
+void MpiWorld::getCartesianRank(int rank,
+                                int maxDims,
+                                const int* dims,
+                                int* periods,
+                                int* coords)
+{
+    // Row-major cartesian layout (reference: src/mpi/MpiWorld.cpp:369-450)
+    if (rank >= size) {
+        throw FaabricException("rank out of world");
+    }
+    int remainder = rank;
+    for (int d = 0; d < maxDims; d++) {
+        int stride = 1;
+        for (int e = d + 1; e < maxDims; e++) {
+            stride *= dims[e];
+        }
+        coords[d] = remainder / stride;
+        remainder = remainder % stride;
+        periods[d] = 1;
+    }
+}
+
+void MpiWorld::getRankFromCoords(int* rank, int* coords)
+{
+    // Inverse of getCartesianRank for up to 3 dims like the reference
+    // (dims derived the same way MPI_Cart_create does: near-cubic split)
+    // For our usage dims are implicit: the caller keeps them; here we
+    // assume a 1-D layout fallback when unknown
+    *rank = coords[0];
+}
+
+void MpiWorld::shiftCartesianCoords(int rank,
+                                    int direction,
+                                    int disp,
+                                    int* source,
+                                    int* destination)
+{
+    (void)direction;
+    // 1-D periodic shift
+    *source = (rank - disp + size) % size;
+    *destination = (rank + disp) % size;
+}
+
+// ------------------------- host data plane ----------------------------------
+
+void MpiWorld::hostSend(int sendRank,
+                        int recvRank,
+                        const uint8_t* buffer,
+                        size_t bytes,
+                        MpiMessageType messageType)
+{
+    getPointToPointBroker().sendMessage(appId,
+                                        groupId,
+                                        chanSendIdx(sendRank, messageType),
+                                        recvRank,
+                                        buffer,
+                                        bytes,
+                                        /*mustOrderMsgs=*/true);
+}
+
+std::vector<uint8_t> MpiWorld::hostRecv(int sendRank,
+                                        int recvRank,
+                                        size_t expectedBytes,
+                                        MpiMessageType messageType)
+{
+    auto data = getPointToPointBroker().recvMessage(
+      groupId,
+      chanSendIdx(sendRank, messageType),
+      recvRank,
+      /*mustOrderMsgs=*/true,
+      getSystemConfig().globalMessageTimeout);
+    if (expectedBytes > 0 && data.size() > expectedBytes) {
+        throw FaabricException("mpi recv buffer too small");
+    }
+    return data;
+}
+
+// ------------------------- device plane (RCCL) ------------------------------
+
+bool MpiWorld::isDeviceBuffer(const void* ptr, MpiBufferLoc loc)
+{
+    if (loc == MpiBufferLoc::HOST) {
+        return false;
+    }
+    if (loc == MpiBufferLoc::DEVICE) {
+        return true;
+    }
+    hipPointerAttribute_t attrs;
+    if (hipPointerGetAttributes(&attrs, ptr) != hipSuccess) {
+        (void)hipGetLastError(); // clear
+        return false;
+    }
+    return attrs.type == hipMemoryTypeDevice;
+}
+
+void MpiWorld::ensureRcclComm(int rank)
+{
+    {
+        std::lock_guard<std::mutex> lock(worldMx);
+        if (!rccl) {
+            rccl = std::make_shared<RcclState>();
+        }
+    }
+    {
+        std::lock_guard<std::mutex> lock(rccl->mx);
+        if (rccl->comms.count(rank) > 0) {
+            return;
+        }
+    }
+
+    // Device for this rank: each worker process owns its GPU(s); ranks
+    // local to a process spread over visible devices in local-rank order
+    int nDevices = 0;
+    HIP_CHECK(hipGetDeviceCount(&nDevices));
+    if (nDevices == 0) {
+        throw FaabricException("RCCL path requires a GPU");
+    }
+    int device;
+    {
+        std::lock_guard<std::mutex> lock(worldMx);
+        auto it = std::find(localRanks.begin(), localRanks.end(), rank);
+        int localIdx =
+          it == localRanks.end()
+            ? 0
+            : (int)std::distance(localRanks.begin(), it);
+        device = localIdx % nDevices;
+    }
+    HIP_CHECK(hipSetDevice(device));
+
+    // Bootstrap: rank 0 generates the uniqueId and ships it to every
+    // other rank over the host plane (replaces the reference's full-mesh
+    // TCP handshake, src/mpi/MpiWorld.cpp:1789-1935)
+    ncclUniqueId uid;
+    if (rank == 0) {
+        RCCL_CHECK(ncclGetUniqueId(&uid));
+        for (int i = 1; i < size; i++) {
+            hostSend(0,
+                     i,
+                     (const uint8_t*)&uid,
+                     sizeof(uid),
+                     MpiMessageType::HANDSHAKE);
+        }
+    } else {
+        auto data = hostRecv(0, rank, sizeof(uid),
+                             MpiMessageType::HANDSHAKE);
+        std::memcpy(&uid, data.data(), sizeof(uid));
+    }
+
+    ncclComm_t comm;
+    RCCL_CHECK(ncclCommInitRank(&comm, size, uid, rank));
+    hipStream_t stream;
+    HIP_CHECK(hipStreamCreateWithFlags(&stream, hipStreamNonBlocking));
+
+    std::lock_guard<std::mutex> lock(rccl->mx);
+    rccl->comms[rank] = comm;
+    rccl->streams[rank] = stream;
+    rccl->devices[rank] = device;
+}
+
+void* MpiWorld::getRcclComm(int rank)
+{
+    ensureRcclComm(rank);
+    std::lock_guard<std::mutex> lock(rccl->mx);
+    return (void*)rccl->comms.at(rank);
+}
+
+void* MpiWorld::getRankStream(int rank)
+{
+    ensureRcclComm(rank);
+    std::lock_guard<std::mutex> lock(rccl->mx);
+    return (void*)rccl->streams.at(rank);
+}
+
+// ------------------------- point-to-point -----------------------------------
+
+void MpiWorld::send(int sendRank,
+                    int recvRank,
+                    const uint8_t* buffer,
+                    MpiDataType dataType,
+                    int count,
+                    MpiMessageType messageType,
+                    MpiBufferLoc loc)
+{
+    size_t bytes = mpiTypeSize(dataType) * (size_t)count;
+    recordMsgCount(sendRank, recvRank, messageType);
+    if (isDeviceBuffer(buffer, loc)) {
+        ensureRcclComm(sendRank);
+        ncclComm_t comm;
+        hipStream_t stream;
+        {
+            std::lock_guard<std::mutex> lock(rccl->mx);
+            comm = rccl->comms.at(sendRank);
+            stream = rccl->streams.at(sendRank);
+            HIP_CHECK(hipSetDevice(rccl->devices.at(sendRank)));
+        }
+        RCCL_CHECK(ncclSend(
+          buffer, count, toNccl(dataType), recvRank, comm, stream));
+        HIP_CHECK(hipStreamSynchronize(stream));
+        return;
+    }
+    hostSend(sendRank, recvRank, buffer, bytes, messageType);
+}
+
+void MpiWorld::recv(int sendRank,
+                    int recvRank,
+                    uint8_t* buffer,
+                    MpiDataType dataType,
+                    int count,
+                    MpiMessageType messageType,
+                    MpiBufferLoc loc)
+{
+    size_t bytes = mpiTypeSize(dataType) * (size_t)count;
+    if (isDeviceBuffer(buffer, loc)) {
+        ensureRcclComm(recvRank);
+        ncclComm_t comm;
+        hipStream_t stream;
+        {
+            std::lock_guard<std::mutex> lock(rccl->mx);
+            comm = rccl->comms.at(recvRank);
+            stream = rccl->streams.at(recvRank);
+            HIP_CHECK(hipSetDevice(rccl->devices.at(recvRank)));
+        }
+        RCCL_CHECK(ncclRecv(
+          buffer, count, toNccl(dataType), sendRank, comm, stream));
+        HIP_CHECK(hipStreamSynchronize(stream));
+        return;
+    }
+
+    // Pending irecvs on this channel must drain first to preserve MPI
+    // matching order (reference: recvBatchReturnLast
+    // src/mpi/MpiWorld.cpp:1963-2030)
+    for (auto& p : rankState.pendingRecvs) {
+        if (!p.done && p.sendRank == sendRank && p.recvRank == recvRank &&
+            p.type == messageType) {
+            auto data = hostRecv(sendRank, recvRank, p.bytes, messageType);
+            std::memcpy(p.buffer, data.data(), data.size());
+            p.done = true;
+        }
+    }
+    auto data = hostRecv(sendRank, recvRank, bytes, messageType);
+    std::memcpy(buffer, data.data(), data.size());
+}
+
+int MpiWorld::isend(int sendRank,
+                    int recvRank,
+                    const uint8_t* buffer,
+                    MpiDataType dataType,
+                    int count,
+                    MpiMessageType messageType)
+{
+    // Buffered eager send: completes immediately
+    send(sendRank, recvRank, buffer, dataType, count, messageType);
+    return rankState.nextRequestId.fetch_add(1);
+}
+
+int MpiWorld::irecv(int sendRank,
+                    int recvRank,
+                    uint8_t* buffer,
+                    MpiDataType dataType,
+                    int count,
+                    MpiMessageType messageType)
+{
+    PendingRecv p;
+    p.requestId = rankState.nextRequestId.fetch_add(1);
+    p.sendRank = sendRank;
+    p.recvRank = recvRank;
+    p.buffer = buffer;
+    p.bytes = mpiTypeSize(dataType) * (size_t)count;
+    p.type = messageType;
+    rankState.pendingRecvs.push_back(p);
+    return p.requestId;
+}
+
+void MpiWorld::awaitAsyncRequest(int requestId)
+{
+    auto& pending = rankState.pendingRecvs;
+    auto target = std::find_if(
+      pending.begin(), pending.end(), [&](const PendingRecv& p) {
+          return p.requestId == requestId;
+      });
+    if (target == pending.end()) {
+        return; // isend request or already-awaited
+    }
+    if (!target->done) {
+        // Drain earlier pending recvs on the same channel first so
+        // message order per (sender, receiver, type) is preserved
+        for (auto& p : pending) {
+            if (p.done || p.sendRank != target->sendRank ||
+                p.recvRank != target->recvRank || p.type != target->type) {
+                continue;
+            }
+            auto data = hostRecv(p.sendRank, p.recvRank, p.bytes, p.type);
+            std::memcpy(p.buffer, data.data(), data.size());
+            p.done = true;
+            if (p.requestId == requestId) {
+                break;
+            }
+        }
+    }
+    pending.erase(std::remove_if(pending.begin(),
+                                 pending.end(),
+                                 [&](const PendingRecv& p) {
+                                     return p.requestId == requestId;
+                                 }),
+                  pending.end());
+}
+
+void MpiWorld::sendRecv(const uint8_t* sendBuffer,
+                        int sendCount,
+                        MpiDataType sendType,
+                        int sendToRank,
+                        uint8_t* recvBuffer,
+                        int recvCount,
+                        MpiDataType recvType,
+                        int recvFromRank,
+                        int thisRank)
+{
+    if (isDeviceBuffer(sendBuffer, MpiBufferLoc::AUTO)) {
+        ensureRcclComm(thisRank);
+        ncclComm_t comm;
+        hipStream_t stream;
+        {
+            std::lock_guard<std::mutex> lock(rccl->mx);
+            comm = rccl->comms.at(thisRank);
+            stream = rccl->streams.at(thisRank);
+            HIP_CHECK(hipSetDevice(rccl->devices.at(thisRank)));
+        }
+        RCCL_CHECK(ncclGroupStart());
+        RCCL_CHECK(ncclSend(sendBuffer,
+                            sendCount,
+                            toNccl(sendType),
+                            sendToRank,
+                            comm,
+                            stream));
+        RCCL_CHECK(ncclRecv(recvBuffer,
+                            recvCount,
+                            toNccl(recvType),
+                            recvFromRank,
+                            comm,
+                            stream));
+        RCCL_CHECK(ncclGroupEnd());
+        HIP_CHECK(hipStreamSynchronize(stream));
+        return;
+    }
+    int req = irecv(recvFromRank,
+                    thisRank,
+                    recvBuffer,
+                    recvType,
+                    recvCount,
+                    MpiMessageType::SENDRECV);
+    send(thisRank,
+         sendToRank,
+         sendBuffer,
+         sendType,
+         sendCount,
+         MpiMessageType::SENDRECV);
+    awaitAsyncRequest(req);
+}
+
+// ------------------------- op_reduce ----------------------------------------
+
+template<typename T>
+static void opReduceLoop(MpiOp op, int count, const T* in, T* inout)
+{
+    switch (op) {
+        case MpiOp::SUM:
+            for (int i = 0; i < count; i++) {
+                inout[i] += in[i];
+            }
+            break;
+        case MpiOp::MAX:
+            for (int i = 0; i < count; i++) {
+                inout[i] = std::max(inout[i], in[i]);
+            }
+            break;
+        case MpiOp::MIN:
+            for (int i = 0; i < count; i++) {
+                inout[i] = std::min(inout[i], in[i]);
+            }
+            break;
+        case MpiOp::PROD:
+            for (int i = 0; i < count; i++) {
+                inout[i] *= in[i];
+            }
+            break;
+    }
+}
+
+void MpiWorld::opReduceHost(MpiOp op,
+                            MpiDataType type,
+                            int count,
+                            const uint8_t* in,
+                            uint8_t* inout)
+{
+    switch (type) {
+        case MpiDataType::INT32:
+            opReduceLoop(op, count, (const int32_t*)in, (int32_t*)inout);
+            break;
+        case MpiDataType::INT64:
+            opReduceLoop(op, count, (const int64_t*)in, (int64_t*)inout);
+            break;
+        case MpiDataType::UINT64:
+            opReduceLoop(op, count, (const uint64_t*)in, (uint64_t*)inout);
+            break;
+        case MpiDataType::FLOAT:
+            opReduceLoop(op, count, (const float*)in, (float*)inout);
+            break;
+        case MpiDataType::DOUBLE:
+            opReduceLoop(op, count, (const double*)in, (double*)inout);
+            break;
+        case MpiDataType::BYTE:
+            opReduceLoop(op, count, (const uint8_t*)in, (uint8_t*)inout);
+            break;
+    }
+}
+
+// ------------------------- collectives --------------------------------------
+
+void MpiWorld::barrier(int thisRank)
+{
+    // All join at rank 0, rank 0 releases
+    // (reference: src/mpi/MpiWorld.cpp:1753-1775)
+    uint8_t token = 1;
+    if (thisRank == 0) {
+        for (int i = 1; i < size; i++) {
+            hostRecv(i, 0, 1, MpiMessageType::BARRIER_JOIN);
+        }
+        for (int i = 1; i < size; i++) {
+            hostSend(0, i, &token, 1, MpiMessageType::BARRIER_DONE);
+        }
+    } else {
+        hostSend(thisRank, 0, &token, 1, MpiMessageType::BARRIER_JOIN);
+        hostRecv(0, thisRank, 1, MpiMessageType::BARRIER_DONE);
+    }
+}
+
+void MpiWorld::broadcast(int rootRank,
+                         int thisRank,
+                         uint8_t* buffer,
+                         MpiDataType dataType,
+                         int count,
+                         MpiMessageType messageType,
+                         MpiBufferLoc loc)
+{
+    if (isDeviceBuffer(buffer, loc)) {
+        ensureRcclComm(thisRank);
+        ncclComm_t comm;
+        hipStream_t stream;
+        {
+            std::lock_guard<std::mutex> lock(rccl->mx);
+            comm = rccl->comms.at(thisRank);
+            stream = rccl->streams.at(thisRank);
+            HIP_CHECK(hipSetDevice(rccl->devices.at(thisRank)));
+        }
+        RCCL_CHECK(ncclBroadcast(buffer,
+                                 buffer,
+                                 count,
+                                 toNccl(dataType),
+                                 rootRank,
+                                 comm,
+                                 stream));
+        HIP_CHECK(hipStreamSynchronize(stream));
+        return;
+    }
+    size_t bytes = mpiTypeSize(dataType) * (size_t)count;
+    if (thisRank == rootRank) {
+        for (int i = 0; i < size; i++) {
+            if (i != rootRank) {
+                hostSend(rootRank, i, buffer, bytes, messageType);
+            }
+        }
+    } else {
+        auto data = hostRecv(rootRank, thisRank, bytes, messageType);
+        std::memcpy(buffer, data.data(), data.size());
+    }
+}
+
+void MpiWorld::scatter(int rootRank,
+                       int thisRank,
+                       const uint8_t* sendBuffer,
+                       uint8_t* recvBuffer,
+                       MpiDataType dataType,
+                       int count)
+{
+    size_t bytes = mpiTypeSize(dataType) * (size_t)count;
+    if (thisRank == rootRank) {
+        for (int i = 0; i < size; i++) {
+            const uint8_t* chunk = sendBuffer + (size_t)i * bytes;
+            if (i == rootRank) {
+                std::memcpy(recvBuffer, chunk, bytes);
+            } else {
+                hostSend(rootRank, i, chunk, bytes,
+                         MpiMessageType::SCATTER);
+            }
+        }
+    } else {
+        auto data =
+          hostRecv(rootRank, thisRank, bytes, MpiMessageType::SCATTER);
+        std::memcpy(recvBuffer, data.data(), data.size());
+    }
+}
+
+void MpiWorld::gather(int thisRank,
+                      int rootRank,
+                      const uint8_t* sendBuffer,
+                      uint8_t* recvBuffer,
+                      MpiDataType dataType,
+                      int count)
+{
+    size_t bytes = mpiTypeSize(dataType) * (size_t)count;
+    if (thisRank == rootRank) {
+        for (int i = 0; i < size; i++) {
+            uint8_t* dst = recvBuffer + (size_t)i * bytes;
+            if (i == rootRank) {
+                std::memcpy(dst, sendBuffer, bytes);
+            } else {
+                auto data =
+                  hostRecv(i, rootRank, bytes, MpiMessageType::GATHER);
+                std::memcpy(dst, data.data(), data.size());
+            }
+        }
+    } else {
+        hostSend(thisRank, rootRank, sendBuffer, bytes,
+                 MpiMessageType::GATHER);
+    }
+}
+
+void MpiWorld::allGather(int thisRank,
+                         const uint8_t* sendBuffer,
+                         uint8_t* recvBuffer,
+                         MpiDataType dataType,
+                         int count,
+                         MpiBufferLoc loc)
+{
+    if (isDeviceBuffer(recvBuffer, loc)) {
+        ensureRcclComm(thisRank);
+        ncclComm_t comm;
+        hipStream_t stream;
+        {
+            std::lock_guard<std::mutex> lock(rccl->mx);
+            comm = rccl->comms.at(thisRank);
+            stream = rccl->streams.at(thisRank);
+            HIP_CHECK(hipSetDevice(rccl->devices.at(thisRank)));
+        }
+        RCCL_CHECK(ncclAllGather(
+          sendBuffer, recvBuffer, count, toNccl(dataType), comm, stream));
+        HIP_CHECK(hipStreamSynchronize(stream));
+        return;
+    }
+    // gather at 0, then broadcast the full buffer (reference: :1082-1111)
+    gather(thisRank, 0, sendBuffer, recvBuffer, dataType, count);
+    broadcast(0,
+              thisRank,
+              recvBuffer,
+              dataType,
+              count * size,
+              MpiMessageType::ALLGATHER,
+              MpiBufferLoc::HOST);
+}
+
+void MpiWorld::reduce(int thisRank,
+                      int rootRank,
+                      const uint8_t* sendBuffer,
+                      uint8_t* recvBuffer,
+                      MpiDataType dataType,
+                      int count,
+                      MpiOp op,
+                      MpiBufferLoc loc)
+{
+    if (isDeviceBuffer(sendBuffer, loc)) {
+        ensureRcclComm(thisRank);
+        ncclComm_t comm;
+        hipStream_t stream;
+        {
+            std::lock_guard<std::mutex> lock(rccl->mx);
+            comm = rccl->comms.at(thisRank);
+            stream = rccl->streams.at(thisRank);
+            HIP_CHECK(hipSetDevice(rccl->devices.at(thisRank)));
+        }
+        RCCL_CHECK(ncclReduce(sendBuffer,
+                              recvBuffer,
+                              count,
+                              toNccl(dataType),
+                              toNcclOp(op),
+                              rootRank,
+                              comm,
+                              stream));
+        HIP_CHECK(hipStreamSynchronize(stream));
+        return;
+    }
+    size_t bytes = mpiTypeSize(dataType) * (size_t)count;
+    if (thisRank == rootRank) {
+        std::memcpy(recvBuffer, sendBuffer, bytes);
+        for (int i = 0; i < size; i++) {
+            if (i == rootRank) {
+                continue;
+            }
+            auto data =
+              hostRecv(i, rootRank, bytes, MpiMessageType::REDUCE);
+            opReduceHost(op, dataType, count, data.data(), recvBuffer);
+        }
+    } else {
+        hostSend(thisRank, rootRank, sendBuffer, bytes,
+                 MpiMessageType::REDUCE);
+    }
+}
+
+void MpiWorld::allReduce(int thisRank,
+                         const uint8_t* sendBuffer,
+                         uint8_t* recvBuffer,
+                         MpiDataType dataType,
+                         int count,
+                         MpiOp op,
+                         MpiBufferLoc loc)
+{
+    if (isDeviceBuffer(sendBuffer, loc)) {
+        ensureRcclComm(thisRank);
+        ncclComm_t comm;
+        hipStream_t stream;
+        {
+            std::lock_guard<std::mutex> lock(rccl->mx);
+            comm = rccl->comms.at(thisRank);
+            stream = rccl->streams.at(thisRank);
+            HIP_CHECK(hipSetDevice(rccl->devices.at(thisRank)));
+        }
+        RCCL_CHECK(ncclAllReduce(sendBuffer,
+                                 recvBuffer,
+                                 count,
+                                 toNccl(dataType),
+                                 toNcclOp(op),
+                                 comm,
+                                 stream));
+        HIP_CHECK(hipStreamSynchronize(stream));
+        return;
+    }
+    // reduce to 0 then broadcast (reference: :1251-1264)
+    reduce(thisRank, 0, sendBuffer, recvBuffer, dataType, count, op,
+           MpiBufferLoc::HOST);
+    broadcast(0,
+              thisRank,
+              recvBuffer,
+              dataType,
+              count,
+              MpiMessageType::ALLREDUCE,
+              MpiBufferLoc::HOST);
+}
+
+void MpiWorld::allToAll(int thisRank,
+                        const uint8_t* sendBuffer,
+                        uint8_t* recvBuffer,
+                        MpiDataType dataType,
+                        int count,
+                        MpiBufferLoc loc)
+{
+    size_t bytes = mpiTypeSize(dataType) * (size_t)count;
+    if (isDeviceBuffer(sendBuffer, loc)) {
+        ensureRcclComm(thisRank);
+        ncclComm_t comm;
+        hipStream_t stream;
+        {
+            std::lock_guard<std::mutex> lock(rccl->mx);
+            comm = rccl->comms.at(thisRank);
+            stream = rccl->streams.at(thisRank);
+            HIP_CHECK(hipSetDevice(rccl->devices.at(thisRank)));
+        }
+        // Pairwise xGMI exchange under one group
+        RCCL_CHECK(ncclGroupStart());
+        for (int i = 0; i < size; i++) {
+            RCCL_CHECK(ncclSend(sendBuffer + (size_t)i * bytes,
+                                count,
+                                toNccl(dataType),
+                                i,
+                                comm,
+                                stream));
+            RCCL_CHECK(ncclRecv(recvBuffer + (size_t)i * bytes,
+                                count,
+                                toNccl(dataType),
+                                i,
+                                comm,
+                                stream));
+        }
+        RCCL_CHECK(ncclGroupEnd());
+        HIP_CHECK(hipStreamSynchronize(stream));
+        return;
+    }
+    // Direct N² exchange (reference: :1433-1484)
+    for (int i = 0; i < size; i++) {
+        const uint8_t* chunk = sendBuffer + (size_t)i * bytes;
+        if (i == thisRank) {
+            std::memcpy(recvBuffer + (size_t)i * bytes, chunk, bytes);
+        } else {
+            hostSend(thisRank, i, chunk, bytes, MpiMessageType::ALLTOALL);
+        }
+    }
+    for (int i = 0; i < size; i++) {
+        if (i == thisRank) {
+            continue;
+        }
+        auto data = hostRecv(i, thisRank, bytes, MpiMessageType::ALLTOALL);
+        std::memcpy(recvBuffer + (size_t)i * bytes,
+                    data.data(),
+                    data.size());
+    }
+}
+
+void MpiWorld::scan(int thisRank,
+                    const uint8_t* sendBuffer,
+                    uint8_t* recvBuffer,
+                    MpiDataType dataType,
+                    int count,
+                    MpiOp op)
+{
+    // Inclusive prefix along a linear chain (reference: :1390-1432)
+    size_t bytes = mpiTypeSize(dataType) * (size_t)count;
+    std::memcpy(recvBuffer, sendBuffer, bytes);
+    if (thisRank > 0) {
+        auto data =
+          hostRecv(thisRank - 1, thisRank, bytes, MpiMessageType::SCAN);
+        opReduceHost(op, dataType, count, data.data(), recvBuffer);
+    }
+    if (thisRank < size - 1) {
+        hostSend(thisRank, thisRank + 1, recvBuffer, bytes,
+                 MpiMessageType::SCAN);
+    }
+}
+
+void MpiWorld::reduceScatter(int thisRank,
+                             const uint8_t* sendBuffer,
+                             uint8_t* recvBuffer,
+                             MpiDataType dataType,
+                             int recvCount,
+                             MpiOp op,
+                             MpiBufferLoc loc)
+{
+    if (isDeviceBuffer(sendBuffer, loc)) {
+        ensureRcclComm(thisRank);
+        ncclComm_t comm;
+        hipStream_t stream;
+        {
+            std::lock_guard<std::mutex> lock(rccl->mx);
+            comm = rccl->comms.at(thisRank);
+            stream = rccl->streams.at(thisRank);
+            HIP_CHECK(hipSetDevice(rccl->devices.at(thisRank)));
+        }
+        RCCL_CHECK(ncclReduceScatter(sendBuffer,
+                                     recvBuffer,
+                                     recvCount,
+                                     toNccl(dataType),
+                                     toNcclOp(op),
+                                     comm,
+                                     stream));
+        HIP_CHECK(hipStreamSynchronize(stream));
+        return;
+    }
+    // allreduce then slice (host path; the reference stubs this call)
+    size_t bytes = mpiTypeSize(dataType) * (size_t)recvCount;
+    std::vector<uint8_t> full((size_t)size * bytes);
+    allReduce(thisRank,
+              sendBuffer,
+              full.data(),
+              dataType,
+              recvCount * size,
+              op,
+              MpiBufferLoc::HOST);
+    std::memcpy(recvBuffer, full.data() + (size_t)thisRank * bytes, bytes);
+}
+
+double MpiWorld::getWTime()
+{
+    return getSecondsSinceEpoch();
+}
+
+void MpiWorld::recordMsgCount(int sendRank,
+                              int recvRank,
+                              MpiMessageType type)
+{
+    // Only recorded when the executing message asks for an exec graph
+    if (!ExecutorContext::isSet()) {
+        return;
+    }
+    Message& msg = ExecutorContext::get().getMsg();
+    if (!msg.recordExecGraph) {
+        return;
+    }
+    std::lock_guard<std::mutex> lock(statsMx);
+    auto& counts = msgCounts[sendRank];
+    counts["mpi-msgcount-torank-" + std::to_string(recvRank)] += 1;
+    counts["mpi-msgtype-" + std::to_string((int)type) + "-torank-" +
+           std::to_string(recvRank)] += 1;
+}
+
+std::map<std::string, int32_t> MpiWorld::getMsgCountDetails(int rank)
+{
+    std::lock_guard<std::mutex> lock(statsMx);
+    return msgCounts[rank];
+}
+
+void MpiWorld::prepareMigration(int thisRank)
+{
+    // No pending async requests may be in flight
+    // (reference: src/mpi/MpiWorld.cpp:2095-2132)
+    if (!rankState.pendingRecvs.empty()) {
+        throw FaabricException(
+          "cannot migrate with pending async MPI requests");
+    }
+    // Refresh the placement from the planner's new decision
+    auto decision = getPlannerClient().getSchedulingDecision(appId);
+    if (decision.nFunctions > 0) {
+        groupId = decision.groupId;
+        std::lock_guard<std::mutex> lock(worldMx);
+        rankHosts.assign(size, "");
+        for (int i = 0; i < decision.nFunctions; i++) {
+            if (decision.groupIdxs[i] < size) {
+                rankHosts[decision.groupIdxs[i]] = decision.hosts[i];
+            }
+        }
+    }
+    // Device-side communicators must be rebuilt after migration (like the
+    // reference clears TLS sockets)
+    std::lock_guard<std::mutex> lock(worldMx);
+    rccl.reset();
+}
+
+// ------------------------- registry / context -------------------------------
+
+MpiWorldRegistry& MpiWorldRegistry::get()
+{
+    static MpiWorldRegistry reg;
+    return reg;
+}
+
+MpiWorld& MpiWorldRegistry::createWorld(Message& msg, int worldId)
+{
+    std::lock_guard<std::mutex> lock(mx);
+    if (worlds.count(worldId) > 0) {
+        throw FaabricException("MPI world already exists: " +
+                               std::to_string(worldId));
+    }
+    auto world = std::make_shared<MpiWorld>();
+    worlds[worldId] = world;
+    world->create(msg, worldId, msg.mpiWorldSize);
+    return *world;
+}
+
+MpiWorld& MpiWorldRegistry::getOrInitialiseWorld(Message& msg)
+{
+    std::shared_ptr<MpiWorld> world;
+    {
+        std::lock_guard<std::mutex> lock(mx);
+        auto it = worlds.find(msg.mpiWorldId);
+        if (it == worlds.end()) {
+            world = std::make_shared<MpiWorld>();
+            worlds[msg.mpiWorldId] = world;
+            world->initialiseFromMsg(msg);
+        } else {
+            world = it->second;
+        }
+    }
+    world->initialiseRankFromMsg(msg);
+    return *world;
+}
+
+MpiWorld& MpiWorldRegistry::getWorld(int worldId)
+{
+    std::lock_guard<std::mutex> lock(mx);
+    auto it = worlds.find(worldId);
+    if (it == worlds.end()) {
+        throw FaabricException("MPI world not found: " +
+                               std::to_string(worldId));
+    }
+    return *it->second;
+}
+
+bool MpiWorldRegistry::worldExists(int worldId)
+{
+    std::lock_guard<std::mutex> lock(mx);
+    return worlds.count(worldId) > 0;
+}
+
+void MpiWorldRegistry::clearWorld(int worldId)
+{
+    std::lock_guard<std::mutex> lock(mx);
+    worlds.erase(worldId);
+}
+
+void MpiWorldRegistry::clear()
+{
+    std::lock_guard<std::mutex> lock(mx);
+    worlds.clear();
+}
+
+MpiContext::MpiContext() = default;
+
+static thread_local MpiContext threadMpiContext;
+
+MpiContext& getMpiContext()
+{
+    return threadMpiContext;
+}
+
+int MpiContext::createWorld(Message& msg)
+{
+    // (reference: src/mpi/MpiContext.cpp:14)
+    if (msg.mpiRank > 0) {
+        throw FaabricException("createWorld called by non-zero rank");
+    }
+    int worldId = msg.mpiWorldId != 0 ? msg.mpiWorldId : generateGidInt32();
+    msg.mpiWorldId = worldId;
+    MpiWorldRegistry::get().createWorld(msg, worldId);
+    isMpi = true;
+    rank = 0;
+    this->worldId = worldId;
+    return worldId;
+}
+
+void MpiContext::joinWorld(Message& msg)
+{
+    if (!msg.isMpi) {
+        throw FaabricException("joinWorld on non-MPI message");
+    }
+    isMpi = true;
+    worldId = msg.mpiWorldId;
+    rank = msg.mpiRank;
+    MpiWorldRegistry::get().getOrInitialiseWorld(msg);
+}
+
+MpiWorld& MpiContext::getWorld()
+{
+    return MpiWorldRegistry::get().getWorld(worldId);
+}
+
+} // namespace faabricamd

@@ -15,6 +15,35 @@ REPO_ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 WORKER_SLOTS = 2
 
 
+def _mpi_allreduce_fn(msg):
+    import struct
+
+    from faabric_amd import _core
+
+    world_id, rank, size = _core.mpi_init()
+    _core.mpi_barrier(rank)
+    out = _core.mpi_allreduce_bytes(
+        rank,
+        struct.pack("<i", rank + 1),
+        _core.MpiDataType.INT32,
+        _core.MpiOp.SUM,
+    )
+    (total,) = struct.unpack("<i", out)
+    expected = size * (size + 1) // 2
+    if total != expected:
+        return 1
+    # Cross-host ring exchange
+    got = _core.mpi_sendrecv_bytes(
+        rank, (rank + 1) % size, (rank - 1 + size) % size,
+        struct.pack("<i", rank * 11),
+    )
+    (val,) = struct.unpack("<i", got)
+    if val != ((rank - 1 + size) % size) * 11:
+        return 2
+    msg.output_data = f"rank {rank} host ok"
+    return 0
+
+
 def _worker_main(port_offset, stop_event, ready_event):
     sys.path.insert(0, REPO_ROOT)
     from faabric_amd import _core
@@ -26,6 +55,7 @@ def _worker_main(port_offset, stop_event, ready_event):
     _core.register_native_echo("dist", "echo")
     _core.register_native_noop("dist", "noop")
     _core.register_native_sleep("dist", "sleep", 200)
+    _core.register_function("dist", "mpi_allreduce", _mpi_allreduce_fn)
     ready_event.set()
     stop_event.wait(120)
     rt.stop()
@@ -120,6 +150,27 @@ def test_concurrent_batches(cluster):
     for ber in bers:
         results = wait_for_batch(ber.app_id, 1)
         assert results[0].return_value == 0
+
+
+def test_mpi_world_across_hosts(cluster):
+    from faabric_amd import _core
+    from faabric_amd.runtime import wait_for_batch
+
+    world_size = 2 * WORKER_SLOTS
+    ber = _core.batch_exec_factory("dist", "mpi_allreduce", 1)
+    msgs = ber.messages
+    msgs[0].is_mpi = True
+    msgs[0].mpi_world_size = world_size
+    ber.messages = msgs
+    decision = _core.call_functions(ber)
+    assert decision.app_id == ber.app_id
+
+    results = wait_for_batch(ber.app_id, world_size, timeout_ms=60_000)
+    assert len(results) == world_size
+    for r in results:
+        assert r.return_value == 0, r.output_data
+    hosts_used = {r.executed_host for r in results}
+    assert hosts_used == {"127.0.0.1@1000", "127.0.0.1@2000"}
 
 
 def test_slots_freed_after_batch(cluster):

@@ -1,0 +1,131 @@
+"""MPI world tests on the host data plane (CPU): world creation through the
+planner's gang scheduling, point-to-point and every collective
+(reference coverage: tests/test/mpi/* and tests/dist/mpi/*, single-host)."""
+
+import struct
+
+import pytest
+
+from faabric_amd import _core
+from faabric_amd.runtime import LocalRuntime, wait_for_batch
+
+WORLD_SIZE = 4
+
+
+def ints(*vals):
+    return struct.pack(f"<{len(vals)}i", *vals)
+
+
+def unints(data):
+    return list(struct.unpack(f"<{len(data) // 4}i", data))
+
+
+def mpi_payload(msg):
+    world_id, rank, size = _core.mpi_init()
+    assert size == WORLD_SIZE
+
+    # barrier
+    _core.mpi_barrier(rank)
+
+    # send/recv ring: rank r sends (r+100) to r+1
+    if rank < size - 1:
+        _core.mpi_send_bytes(rank, rank + 1, ints(rank + 100))
+    if rank > 0:
+        got = unints(_core.mpi_recv_bytes(rank - 1, rank, 4))
+        assert got == [rank - 1 + 100], f"ring recv wrong: {got}"
+
+    # allreduce SUM of [rank, 2*rank]
+    out = unints(
+        _core.mpi_allreduce_bytes(
+            rank, ints(rank, 2 * rank), _core.MpiDataType.INT32, _core.MpiOp.SUM
+        )
+    )
+    expect_sum = sum(range(size))
+    assert out == [expect_sum, 2 * expect_sum], f"allreduce wrong: {out}"
+
+    # reduce MAX to root 0
+    out = unints(
+        _core.mpi_reduce_bytes(
+            rank, 0, ints(rank * 7), _core.MpiDataType.INT32, _core.MpiOp.MAX
+        )
+    )
+    if rank == 0:
+        assert out == [7 * (size - 1)]
+
+    # broadcast from root 2
+    data = ints(999, 888) if rank == 2 else b""
+    got = _core.mpi_bcast_bytes(2, rank, data, 8)
+    assert unints(got) == [999, 888]
+
+    # scatter from root 0
+    send = ints(*range(size)) if rank == 0 else b""
+    got = unints(_core.mpi_scatter_bytes(0, rank, send, 4))
+    assert got == [rank]
+
+    # gather to root 1
+    got = _core.mpi_gather_bytes(rank, 1, ints(rank * 3), size)
+    if rank == 1:
+        assert unints(got) == [3 * i for i in range(size)]
+
+    # allgather
+    got = unints(_core.mpi_allgather_bytes(rank, ints(rank + 50), size))
+    assert got == [50 + i for i in range(size)]
+
+    # alltoall: rank r sends value r*10+j to rank j
+    send = ints(*[rank * 10 + j for j in range(size)])
+    got = unints(_core.mpi_alltoall_bytes(rank, send, size))
+    assert got == [i * 10 + rank for i in range(size)]
+
+    # scan (inclusive prefix sum)
+    got = unints(
+        _core.mpi_scan_bytes(
+            rank, ints(rank + 1), _core.MpiDataType.INT32, _core.MpiOp.SUM
+        )
+    )
+    assert got == [sum(range(1, rank + 2))]
+
+    # sendrecv around the ring
+    got = _core.mpi_sendrecv_bytes(
+        rank, (rank + 1) % size, (rank - 1 + size) % size, ints(rank)
+    )
+    assert unints(got) == [(rank - 1 + size) % size]
+
+    _core.mpi_barrier(rank)
+    msg.output_data = f"rank {rank} ok"
+    return 0
+
+
+@pytest.fixture(scope="module")
+def runtime():
+    rt = LocalRuntime(slots=WORLD_SIZE)
+    rt.start_planner(with_snapshot_server=False)
+    rt.start_worker()
+    _core.register_function("mpi", "alltests", mpi_payload)
+    yield rt
+    rt.stop()
+
+
+def submit_mpi_batch(user, func, world_size, timeout_ms=60_000):
+    ber = _core.batch_exec_factory(user, func, 1)
+    msgs = ber.messages
+    msgs[0].is_mpi = True
+    msgs[0].mpi_world_size = world_size
+    ber.messages = msgs
+    decision = _core.call_functions(ber)
+    assert decision.app_id == ber.app_id, f"schedule failed: {decision.app_id}"
+    return wait_for_batch(ber.app_id, world_size, timeout_ms)
+
+
+def test_mpi_world_all_collectives(runtime):
+    results = submit_mpi_batch("mpi", "alltests", WORLD_SIZE)
+    assert len(results) == WORLD_SIZE
+    for r in results:
+        assert r.return_value == 0, r.output_data
+    outputs = sorted(r.output_data for r in results)
+    assert outputs == sorted(f"rank {i} ok" for i in range(WORLD_SIZE))
+
+
+def test_mpi_world_reuse(runtime):
+    # A second world after the first finished: ids and state must not leak
+    results = submit_mpi_batch("mpi", "alltests", WORLD_SIZE)
+    assert all(r.return_value == 0 for r in results)
