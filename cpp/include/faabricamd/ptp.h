@@ -132,6 +132,9 @@ class PointToPointBroker
                           int32_t groupIdx,
                           const std::string& newHost);
 
+    // Drop cached outbound connections (hosts restarted)
+    void clearClients();
+
     // Post-migration: barrier + clear send-seq state
     // (reference: src/transport/PointToPointBroker.cpp:910-926)
     void postMigrationHook(int32_t groupId, int32_t groupIdx);

@@ -7,6 +7,7 @@
 #include "faabricamd/ptp.h"
 #include "faabricamd/scheduler.h"
 #include "faabricamd/snapshot.h"
+#include "faabricamd/state.h"
 #include "faabricamd/util.h"
 #include "faabricamd/wire.h"
 
@@ -51,6 +52,15 @@ bool Planner::registerHost(const Host& hostIn, bool overwrite)
 {
     if (hostIn.ip.empty() || hostIn.slots < 0) {
         return false;
+    }
+    if (overwrite) {
+        // A host (re-)registering from scratch may have restarted: cached
+        // outbound connections to it are silently stale (async frames into
+        // a dead socket do not error), so drop them all
+        clearFunctionCallClients();
+        clearSnapshotClients();
+        clearStateClients();
+        getPointToPointBroker().clearClients();
     }
     std::unique_lock lock(plannerMx);
     auto it = state.hostMap.find(hostIn.ip);

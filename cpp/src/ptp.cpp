@@ -535,6 +535,12 @@ void PointToPointBroker::clear()
     PointToPointGroup::clear();
 }
 
+void PointToPointBroker::clearClients()
+{
+    std::lock_guard<std::mutex> lock(clientsMx);
+    clients.clear();
+}
+
 void PointToPointBroker::postMigrationHook(int32_t groupId, int32_t groupIdx)
 {
     auto group = PointToPointGroup::getOrAwaitGroup(groupId);

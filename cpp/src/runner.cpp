@@ -30,6 +30,13 @@ void FaabricMain::startBackground()
              conf.endpointHost.c_str(),
              getPortOffset());
 
+    // Drop any cached outbound connections from a previous runtime in this
+    // process (tests restart runtimes in-process)
+    clearFunctionCallClients();
+    clearSnapshotClients();
+    clearStateClients();
+    getPointToPointBroker().clearClients();
+
     functionServer = std::make_unique<FunctionCallServer>();
     functionServer->start();
     snapshotServer = std::make_unique<SnapshotServer>();

@@ -190,12 +190,11 @@ static uint64_t gidKey()
 
 uint32_t generateGid()
 {
-    uint64_t c = gidCounter.fetch_add(1, std::memory_order_relaxed) + 1;
-    uint64_t h = gidKey() + 0x9e3779b97f4a7c15ULL * c;
-    h ^= h >> 33;
-    h *= 0xff51afd7ed558ccdULL;
-    h ^= h >> 33;
-    uint32_t r = (uint32_t)(h % (uint64_t)INT32_MAX);
+    // Random per-process base + atomic counter: unique within a process
+    // for 2^31 draws, distinct across processes with high probability
+    // (reference scheme: src/util/gids.cpp:16-28)
+    uint64_t c = gidCounter.fetch_add(1, std::memory_order_relaxed);
+    uint32_t r = (uint32_t)((gidKey() + c) % (uint64_t)INT32_MAX);
     return r == 0 ? 1 : r;
 }
 
