@@ -1,0 +1,121 @@
+// Device ops: HBM-resident snapshot engine on the gfx950 kernels in
+// cpp/hip/snapshot_kernels.hip (dirty-page tracking, compacted XOR diff,
+// merge application, elementwise reductions).
+//
+// This is the MI355X-native replacement for the reference's fault-driven
+// dirty tracking + CPU diff loops (reference: src/util/dirty.cpp,
+// src/util/snapshot.cpp:30-652): there is no mprotect on HBM, so dirty
+// pages come from a compare kernel against the snapshot baseline, and the
+// XOR page diff IS the shippable delta (DIFFING_MODE=xor made native).
+#pragma once
+
+#include <cstdint>
+#include <memory>
+#include <string>
+#include <vector>
+
+#include <hip/hip_runtime.h>
+
+namespace faabricamd {
+
+bool gpuAvailable();
+int gpuCount();
+
+// Raw kernel entry points (extern "C" in the .hip TU)
+extern "C" {
+hipError_t famDirtyPages(const void* snap,
+                         const void* cur,
+                         uint64_t bytes,
+                         uint32_t* flagsDev,
+                         hipStream_t stream);
+hipError_t famXorBuffer(const void* a,
+                        const void* b,
+                        void* out,
+                        uint64_t bytes,
+                        hipStream_t stream);
+hipError_t famDiffXorPages(const void* snap,
+                           const void* cur,
+                           uint64_t bytes,
+                           uint32_t* ticketDev,
+                           uint32_t* pageIdxDev,
+                           void* payloadDev,
+                           hipStream_t stream);
+hipError_t famApplyXorPages(void* snap,
+                            const uint32_t* pageIdxDev,
+                            const void* payloadDev,
+                            uint32_t nDirty,
+                            hipStream_t stream);
+hipError_t famElementwiseOp(void* inout,
+                            const void* in,
+                            uint64_t count,
+                            int dtype,
+                            int op,
+                            hipStream_t stream);
+}
+
+inline constexpr size_t DEVICE_PAGE = 4096;
+
+// An HBM-resident snapshot of a device memory region, with the
+// diff/merge pipeline. All methods are synchronous on the owned stream.
+class DeviceSnapshot
+{
+  public:
+    DeviceSnapshot(size_t bytes, int device = 0);
+    ~DeviceSnapshot();
+    DeviceSnapshot(const DeviceSnapshot&) = delete;
+
+    size_t size() const { return bytes_; }
+    int device() const { return device_; }
+    void* data() { return snap_; }
+
+    // Baseline management
+    void copyInHost(const void* hostBuf, size_t n, size_t offset = 0);
+    void copyOutHost(void* hostBuf, size_t n, size_t offset = 0) const;
+    void captureFromDevice(const void* devPtr); // snapshot := memory view
+
+    // Dirty-page bitmap of devPtr vs the snapshot (one u32 flag per 4 KiB
+    // page, copied back to the host)
+    std::vector<uint32_t> dirtyPages(const void* devPtr);
+
+    // Compacted XOR page diff of devPtr vs the snapshot. Returns the
+    // number of dirty pages; the page indices + payload stay on-device
+    // (diffPageIdx()/diffPayload()) ready to ship or apply.
+    uint32_t diffXor(const void* devPtr);
+
+    // Merge: snapshot ^= last diff (applies the compacted pages)
+    void applyLastDiff();
+    // Merge a diff produced elsewhere (e.g. shipped from a peer GPU)
+    void applyDiffPages(const uint32_t* pageIdxDev,
+                        const void* payloadDev,
+                        uint32_t nDirty);
+
+    const uint32_t* diffPageIdx() const { return pageIdx_; }
+    const void* diffPayload() const { return payload_; }
+    uint32_t lastDiffPages() const { return lastDirty_; }
+    hipStream_t stream() const { return stream_; }
+
+  private:
+    void ensureDiffBuffers();
+
+    size_t bytes_ = 0;
+    int device_ = 0;
+    uint8_t* snap_ = nullptr;
+    hipStream_t stream_ = nullptr;
+
+    // Diff scratch (lazily allocated): worst case every page dirty
+    uint32_t* ticket_ = nullptr;
+    uint32_t* pageIdx_ = nullptr;
+    uint8_t* payload_ = nullptr;
+    uint32_t lastDirty_ = 0;
+};
+
+// Elementwise op on device buffers: inout = op(inout, in)
+// dtype matches MpiDataType; op matches MpiOp (+4=sub, 5=xor)
+void deviceElementwiseOp(void* inout,
+                         const void* in,
+                         uint64_t count,
+                         int dtype,
+                         int op,
+                         hipStream_t stream = nullptr);
+
+} // namespace faabricamd

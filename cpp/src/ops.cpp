@@ -1,0 +1,175 @@
+// Device snapshot engine host-side wrappers (kernels:
+// cpp/hip/snapshot_kernels.hip). Fails loudly when no GPU is present —
+// the CPU path lives in snapshot.cpp, never silently substituted here.
+#include "faabricamd/ops.h"
+#include "faabricamd/util.h"
+
+#include <cstring>
+
+namespace faabricamd {
+
+#define OPS_HIP_CHECK(call)                                                    \
+    do {                                                                       \
+        hipError_t err_ = (call);                                              \
+        if (err_ != hipSuccess) {                                              \
+            throw FaabricException(std::string("HIP error in ops: ") +         \
+                                   hipGetErrorString(err_));                   \
+        }                                                                      \
+    } while (0)
+
+bool gpuAvailable()
+{
+    return gpuCount() > 0;
+}
+
+int gpuCount()
+{
+    static int n = []() {
+        int count = 0;
+        if (hipGetDeviceCount(&count) != hipSuccess) {
+            return 0;
+        }
+        return count;
+    }();
+    return n;
+}
+
+DeviceSnapshot::DeviceSnapshot(size_t bytes, int device)
+  : bytes_(bytes)
+  , device_(device)
+{
+    if ((bytes % DEVICE_PAGE) != 0) {
+        throw FaabricException("device snapshot size must be page-aligned");
+    }
+    if (!gpuAvailable()) {
+        throw FaabricException(
+          "DeviceSnapshot requires an MI355X GPU (none visible)");
+    }
+    OPS_HIP_CHECK(hipSetDevice(device_));
+    OPS_HIP_CHECK(hipMalloc(&snap_, bytes_));
+    OPS_HIP_CHECK(hipStreamCreateWithFlags(&stream_, hipStreamNonBlocking));
+}
+
+DeviceSnapshot::~DeviceSnapshot()
+{
+    if (snap_ != nullptr) {
+        hipFree(snap_);
+    }
+    if (ticket_ != nullptr) {
+        hipFree(ticket_);
+    }
+    if (pageIdx_ != nullptr) {
+        hipFree(pageIdx_);
+    }
+    if (payload_ != nullptr) {
+        hipFree(payload_);
+    }
+    if (stream_ != nullptr) {
+        hipStreamDestroy(stream_);
+    }
+}
+
+void DeviceSnapshot::copyInHost(const void* hostBuf, size_t n, size_t offset)
+{
+    OPS_HIP_CHECK(hipSetDevice(device_));
+    OPS_HIP_CHECK(hipMemcpyAsync(
+      snap_ + offset, hostBuf, n, hipMemcpyHostToDevice, stream_));
+    OPS_HIP_CHECK(hipStreamSynchronize(stream_));
+}
+
+void DeviceSnapshot::copyOutHost(void* hostBuf, size_t n, size_t offset) const
+{
+    OPS_HIP_CHECK(hipSetDevice(device_));
+    OPS_HIP_CHECK(hipMemcpyAsync(
+      hostBuf, snap_ + offset, n, hipMemcpyDeviceToHost, stream_));
+    OPS_HIP_CHECK(hipStreamSynchronize(stream_));
+}
+
+void DeviceSnapshot::captureFromDevice(const void* devPtr)
+{
+    OPS_HIP_CHECK(hipSetDevice(device_));
+    OPS_HIP_CHECK(hipMemcpyAsync(
+      snap_, devPtr, bytes_, hipMemcpyDeviceToDevice, stream_));
+    OPS_HIP_CHECK(hipStreamSynchronize(stream_));
+}
+
+void DeviceSnapshot::ensureDiffBuffers()
+{
+    if (ticket_ != nullptr) {
+        return;
+    }
+    size_t nPages = bytes_ / DEVICE_PAGE;
+    OPS_HIP_CHECK(hipMalloc(&ticket_, sizeof(uint32_t)));
+    OPS_HIP_CHECK(hipMalloc(&pageIdx_, nPages * sizeof(uint32_t)));
+    OPS_HIP_CHECK(hipMalloc(&payload_, bytes_));
+}
+
+std::vector<uint32_t> DeviceSnapshot::dirtyPages(const void* devPtr)
+{
+    OPS_HIP_CHECK(hipSetDevice(device_));
+    size_t nPages = bytes_ / DEVICE_PAGE;
+    uint32_t* flagsDev = nullptr;
+    OPS_HIP_CHECK(hipMalloc(&flagsDev, nPages * sizeof(uint32_t)));
+    OPS_HIP_CHECK(
+      hipMemsetAsync(flagsDev, 0, nPages * sizeof(uint32_t), stream_));
+    OPS_HIP_CHECK(famDirtyPages(snap_, devPtr, bytes_, flagsDev, stream_));
+    std::vector<uint32_t> flags(nPages);
+    OPS_HIP_CHECK(hipMemcpyAsync(flags.data(),
+                                 flagsDev,
+                                 nPages * sizeof(uint32_t),
+                                 hipMemcpyDeviceToHost,
+                                 stream_));
+    OPS_HIP_CHECK(hipStreamSynchronize(stream_));
+    hipFree(flagsDev);
+    return flags;
+}
+
+uint32_t DeviceSnapshot::diffXor(const void* devPtr)
+{
+    OPS_HIP_CHECK(hipSetDevice(device_));
+    ensureDiffBuffers();
+    OPS_HIP_CHECK(hipMemsetAsync(ticket_, 0, sizeof(uint32_t), stream_));
+    OPS_HIP_CHECK(
+      famDiffXorPages(snap_, devPtr, bytes_, ticket_, pageIdx_, payload_,
+                      stream_));
+    uint32_t nDirty = 0;
+    OPS_HIP_CHECK(hipMemcpyAsync(&nDirty,
+                                 ticket_,
+                                 sizeof(uint32_t),
+                                 hipMemcpyDeviceToHost,
+                                 stream_));
+    OPS_HIP_CHECK(hipStreamSynchronize(stream_));
+    lastDirty_ = nDirty;
+    return nDirty;
+}
+
+void DeviceSnapshot::applyLastDiff()
+{
+    applyDiffPages(pageIdx_, payload_, lastDirty_);
+}
+
+void DeviceSnapshot::applyDiffPages(const uint32_t* pageIdxDev,
+                                    const void* payloadDev,
+                                    uint32_t nDirty)
+{
+    if (nDirty == 0) {
+        return;
+    }
+    OPS_HIP_CHECK(hipSetDevice(device_));
+    OPS_HIP_CHECK(
+      famApplyXorPages(snap_, pageIdxDev, payloadDev, nDirty, stream_));
+    OPS_HIP_CHECK(hipStreamSynchronize(stream_));
+}
+
+void deviceElementwiseOp(void* inout,
+                         const void* in,
+                         uint64_t count,
+                         int dtype,
+                         int op,
+                         hipStream_t stream)
+{
+    OPS_HIP_CHECK(famElementwiseOp(inout, in, count, dtype, op, stream));
+    OPS_HIP_CHECK(hipStreamSynchronize(stream));
+}
+
+} // namespace faabricamd

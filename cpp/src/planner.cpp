@@ -305,12 +305,16 @@ std::shared_ptr<SchedulingDecision> Planner::callBatch(
             for (int i = 0; i < decision->nFunctions; i++) {
                 auto host = state.hostMap.at(decision->hosts[i]);
                 host->info.usedSlots++;
-                try {
-                    decision->mpiPorts[i] = claimHostMpiPort(host);
-                } catch (const std::exception& e) {
-                    FAM_ERROR("mpi port claim failed for app %d: %s",
-                              appId,
-                              e.what());
+                // Data-plane ports only exist for MPI ranks (the RCCL
+                // bootstrap rides the PTP plane, so the pool is small)
+                if (isMpi) {
+                    try {
+                        decision->mpiPorts[i] = claimHostMpiPort(host);
+                    } catch (const std::exception& e) {
+                        FAM_ERROR("mpi port claim failed for app %d: %s",
+                                  appId,
+                                  e.what());
+                    }
                 }
             }
 
@@ -342,11 +346,11 @@ std::shared_ptr<SchedulingDecision> Planner::callBatch(
             for (size_t i = 0; i < req->messages.size(); i++) {
                 oldReq->messages.push_back(req->messages[i]);
                 oldDec->addMessage(decision->hosts[i], req->messages[i]);
-                if (!skipClaim) {
+                if (!skipClaim && isMpi) {
                     oldDec->mpiPorts[oldDec->nFunctions - 1] =
                       claimHostMpiPort(
                         state.hostMap.at(decision->hosts[i]));
-                } else {
+                } else if (skipClaim) {
                     oldDec->mpiPorts[oldDec->nFunctions - 1] =
                       decision->mpiPorts[i];
                 }
