@@ -99,6 +99,16 @@ class Executor
     void executeTasks(std::vector<int> msgIdxs,
                       std::shared_ptr<BatchExecuteRequest> req);
 
+    // Fork-join THREADS batch over a shared memory snapshot (reference:
+    // Executor::executeThreads; call stack SURVEY §3.4). Must be called
+    // from a running task (ExecutorContext set). Snapshots this
+    // executor's memory, gang-schedules `req` as THREADS, waits for all
+    // thread results, merges the typed diffs back and re-maps the merged
+    // snapshot over this executor's memory. Returns (msgId, returnValue).
+    std::vector<std::pair<int32_t, int32_t>> executeThreads(
+      std::shared_ptr<BatchExecuteRequest> req,
+      const std::vector<SnapshotMergeRegion>& mergeRegions);
+
     // User hook: run one task. Default implementation dispatches to the
     // FunctionRegistry (reference: Executor.h:42 pure virtual)
     virtual int32_t executeTask(int threadPoolIdx,
