@@ -472,4 +472,69 @@ void initRuntimeBindings(py::module_& m)
 
     // Native benchmark payloads (cpp/src/bench_funcs.cpp)
     m.def("register_bench_functions", [] { registerBenchFunctions(); });
+
+    // ---------------- THREADS fork-join (from inside a running task) -------
+    m.def("execute_threads",
+          [](const std::string& user,
+             const std::string& function,
+             int nThreads,
+             const std::vector<std::tuple<uint32_t, size_t, int, int>>&
+               regions,
+             const py::bytes& inputData) {
+              Executor* exec = ExecutorContext::get().getExecutor();
+              std::string input = inputData;
+              std::vector<std::pair<int32_t, int32_t>> results;
+              {
+                  py::gil_scoped_release release;
+                  auto req = std::make_shared<BatchExecuteRequest>(
+                    batchExecFactory(user, function, nThreads));
+                  for (auto& m2 : req->messages) {
+                      m2.inputData.assign(input.begin(), input.end());
+                  }
+                  std::vector<SnapshotMergeRegion> mr;
+                  for (const auto& [off, len, dt, op] : regions) {
+                      mr.emplace_back(off,
+                                      len,
+                                      (SnapshotDataType)dt,
+                                      (SnapshotMergeOperation)op);
+                  }
+                  results = exec->executeThreads(req, mr);
+              }
+              py::list out;
+              for (auto& [msgId, rv] : results) {
+                  out.append(py::make_tuple(msgId, rv));
+              }
+              return out;
+          },
+          py::arg("user"),
+          py::arg("function"),
+          py::arg("n_threads"),
+          py::arg("merge_regions") =
+            std::vector<std::tuple<uint32_t, size_t, int, int>>{},
+          py::arg("input_data") = py::bytes(""));
+
+    // Executor memory access for thread bodies / parents
+    m.def("executor_set_memory_size", [](size_t n) {
+        ExecutorContext::get().getExecutor()->setMemorySize(n);
+    });
+    m.def("executor_memory_size", [] {
+        return ExecutorContext::get().getExecutor()->getMemoryView().second;
+    });
+    m.def("executor_write_memory", [](size_t offset, const py::bytes& data) {
+        std::string s = data;
+        auto [base, size] = ExecutorContext::get().getExecutor()
+                              ->getMemoryView();
+        if (offset + s.size() > size) {
+            throw FaabricException("executor memory write out of bounds");
+        }
+        std::memcpy(base + offset, s.data(), s.size());
+    });
+    m.def("executor_read_memory", [](size_t offset, size_t n) {
+        auto [base, size] = ExecutorContext::get().getExecutor()
+                              ->getMemoryView();
+        if (offset + n > size) {
+            throw FaabricException("executor memory read out of bounds");
+        }
+        return py::bytes((const char*)base + offset, n);
+    });
 }

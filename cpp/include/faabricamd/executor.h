@@ -23,6 +23,7 @@
 
 #include "faabricamd/messages.h"
 #include "faabricamd/queue.h"
+#include "faabricamd/snapshot.h"
 
 namespace faabricamd {
 

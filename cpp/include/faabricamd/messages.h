@@ -368,6 +368,7 @@ struct ThreadResultRequest
     int32_t returnValue = 0; // 3
     std::string key;         // 4
     std::vector<SnapshotDiffMsg> diffs; // 5
+    std::string executedHost; // 6 (slot accounting on the planner)
     std::string encode() const;
     static ThreadResultRequest decode(const std::string& buf);
 };

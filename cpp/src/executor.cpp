@@ -2,6 +2,7 @@
 // :38-212 pool lifecycle, :307-576 threadPoolThread, :580-590 claims).
 #include "faabricamd/executor.h"
 #include "faabricamd/planner.h"
+#include "faabricamd/snapshot.h"
 #include "faabricamd/util.h"
 
 #include <algorithm>
@@ -162,6 +163,66 @@ void Executor::executeTasks(std::vector<int> msgIdxs,
     }
 }
 
+std::vector<std::pair<int32_t, int32_t>> Executor::executeThreads(
+  std::shared_ptr<BatchExecuteRequest> req,
+  const std::vector<SnapshotMergeRegion>& mergeRegions)
+{
+    // Fork-join over a shared memory snapshot
+    // (reference: SURVEY §3.4; src/executor/Executor.cpp executeThreads)
+    Message& parentMsg = ExecutorContext::get().getMsg();
+    std::string key = getMainThreadSnapshotKey(
+      parentMsg.user, parentMsg.function, parentMsg.appId);
+
+    auto [memBase, memSize] = getMemoryView();
+    auto& reg = SnapshotRegistry::get();
+    std::shared_ptr<SnapshotData> snap;
+    if (reg.snapshotExists(key)) {
+        snap = reg.getSnapshot(key);
+        snap->copyInData(memBase, memSize, 0);
+    } else {
+        snap = std::make_shared<SnapshotData>(
+          std::vector<uint8_t>(memBase, memBase + memSize));
+        reg.registerSnapshot(key, snap);
+    }
+    snap->clearMergeRegions();
+    for (const auto& r : mergeRegions) {
+        snap->addMergeRegion(r.offset, r.length, r.dataType, r.operation);
+    }
+
+    // Thread messages scale-change onto the running app
+    req->appId = parentMsg.appId;
+    req->type = BatchExecuteType::THREADS;
+    req->snapshotKey = key;
+    for (size_t i = 0; i < req->messages.size(); i++) {
+        auto& m = req->messages[i];
+        m.appId = parentMsg.appId;
+        m.appIdx = (int32_t)i + 1;
+        m.groupIdx = (int32_t)i + 1;
+        m.snapshotKey = key;
+        m.mainHost = getSystemConfig().endpointHost;
+    }
+
+    auto decision = getPlannerClient().callFunctions(req);
+    if (decision->appId == NOT_ENOUGH_SLOTS) {
+        throw FaabricException("not enough slots for thread fork");
+    }
+
+    // Await every thread result, then merge the queued diffs and re-map
+    // the merged snapshot over this executor's memory
+    std::vector<std::pair<int32_t, int32_t>> results;
+    const auto& conf = getSystemConfig();
+    for (const auto& m : req->messages) {
+        Message result = getPlannerClient().getMessageResult(
+          m.appId, m.id, conf.globalMessageTimeout);
+        results.emplace_back(m.id, result.returnValue);
+    }
+    snap->writeQueuedDiffs();
+    auto [memBase2, memSize2] = getMemoryView();
+    snap->mapToMemory(memBase2, memSize2);
+    snap->clearMergeRegions();
+    return results;
+}
+
 int32_t Executor::executeTask(int threadPoolIdx,
                               int msgIdx,
                               std::shared_ptr<BatchExecuteRequest> req)
@@ -189,7 +250,14 @@ void Executor::setMemorySize(size_t newSize)
 
 void Executor::restore(const std::string& snapshotKey)
 {
-    (void)snapshotKey; // wired to the snapshot registry in snapshot.cpp users
+    // Copy the snapshot over this executor's memory view
+    // (reference: src/executor/Executor.cpp:640-654 — mapToMemory)
+    auto snap = SnapshotRegistry::get().getSnapshot(snapshotKey);
+    if (dummyMemory.size() < snap->getSize()) {
+        setMemorySize(snap->getSize());
+    }
+    auto [base, size] = getMemoryView();
+    snap->mapToMemory(base, size);
 }
 
 void Executor::reset(Message& msg)
@@ -263,6 +331,23 @@ void Executor::handleTaskResult(Message& msg,
 
     bool isThreads = req->type == BatchExecuteType::THREADS;
 
+    // THREADS: the last local thread diffs this executor's memory against
+    // the shared snapshot and ships the typed diffs to the main host
+    // (reference: src/executor/Executor.cpp:509-516, :684 mergeDirtyRegions)
+    std::vector<SnapshotDiff> threadDiffs;
+    if (isThreads && isLastInBatch && !req->snapshotKey.empty()) {
+        try {
+            auto snap = SnapshotRegistry::get().getSnapshot(req->snapshotKey);
+            auto [base, size] = getMemoryView();
+            // Compare-based dirty tracking: every page is a candidate and
+            // the typed merge regions refine (no mprotect on this path)
+            snap->fillGapsWithBytewiseRegions();
+            threadDiffs = snap->diffWithMemory(base, size);
+        } catch (const std::exception& e) {
+            FAM_ERROR("thread diff failed: %s", e.what());
+        }
+    }
+
     // Claim-reset-release order matters
     // (reference: src/executor/Executor.cpp:537-552)
     if (isLastInBatch) {
@@ -274,6 +359,30 @@ void Executor::handleTaskResult(Message& msg,
             }
         }
         releaseClaim();
+    }
+
+    if (isThreads && !req->snapshotKey.empty()) {
+        // Route via the snapshot channel so the main host can queue the
+        // diffs before the result lands (reference: setThreadResult
+        // src/executor/Executor.cpp:271-299)
+        const std::string& mainHost = msg.mainHost;
+        if (mainHost == conf.endpointHost || mainHost.empty()) {
+            if (!threadDiffs.empty()) {
+                try {
+                    auto snap =
+                      SnapshotRegistry::get().getSnapshot(req->snapshotKey);
+                    snap->queueDiffs(threadDiffs);
+                } catch (const std::exception& e) {
+                    FAM_ERROR("queueing thread diffs failed: %s", e.what());
+                }
+            }
+            auto resultMsg = std::make_shared<Message>(msg);
+            getPlannerClient().setMessageResult(resultMsg);
+        } else {
+            getSnapshotClient(mainHost)->pushThreadResult(
+              msg.appId, msg.id, returnValue, req->snapshotKey, threadDiffs);
+        }
+        return;
     }
 
     // Report the result to the planner

@@ -976,6 +976,7 @@ std::string ThreadResultRequest::encode() const
     for (const auto& d : diffs) {
         w.putMessage(5, encodeDiff(d));
     }
+    w.putString(6, executedHost);
     return w.take();
 }
 
@@ -992,6 +993,7 @@ ThreadResultRequest ThreadResultRequest::decode(const std::string& buf)
             case 3: m.returnValue = r.asInt32(); break;
             case 4: m.key = r.asString(); break;
             case 5: m.diffs.push_back(decodeDiff(r.asString())); break;
+            case 6: m.executedHost = r.asString(); break;
             default: r.skip(t);
         }
     }

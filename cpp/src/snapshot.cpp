@@ -686,7 +686,9 @@ std::string SnapshotServer::doSyncRecv(uint8_t code, const std::string& body)
             msg->appId = req.appId;
             msg->id = req.messageId;
             msg->returnValue = req.returnValue;
-            msg->executedHost = getSystemConfig().endpointHost;
+            msg->executedHost = req.executedHost.empty()
+                                  ? getSystemConfig().endpointHost
+                                  : req.executedHost;
             getPlannerClient().setMessageResult(msg);
             return {};
         }
@@ -780,6 +782,7 @@ void SnapshotClient::pushThreadResult(int32_t appId,
     req.messageId = messageId;
     req.returnValue = returnValue;
     req.key = key;
+    req.executedHost = getSystemConfig().endpointHost;
     for (const auto& d : diffs) {
         req.diffs.push_back(d.toMsg());
     }

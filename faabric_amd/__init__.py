@@ -12,6 +12,17 @@ Python driver used by benchmarks, tests and deployment glue.
 
 import os
 
+# torch (PyTorch-ROCm) bundles its own copies of libamdhip64/libhsa-runtime
+# and dlopens them by absolute path. If _core loads first, its DT_NEEDED
+# pulls /opt/rocm's runtime and the process ends up with TWO HSA runtimes —
+# whichever initialises second fails device discovery intermittently.
+# Importing torch first makes its libs the canonical ones (SONAME dedup),
+# so _core and torch share a single HIP runtime.
+try:  # pragma: no cover - environment dependent
+    import torch  # noqa: F401
+except Exception:  # torch genuinely absent: /opt/rocm runtime is used
+    pass
+
 # HIP runtime import is safe without a GPU; extensions are built for gfx950.
 from faabric_amd._core import (  # noqa: F401
     BatchExecuteRequest,
