@@ -3,6 +3,7 @@
 #include <pybind11/pybind11.h>
 #include <pybind11/stl.h>
 
+#include "faabricamd/endpoint.h"
 #include "faabricamd/executor.h"
 #include "faabricamd/planner.h"
 #include "faabricamd/ptp.h"
@@ -512,6 +513,47 @@ void initRuntimeBindings(py::module_& m)
           py::arg("merge_regions") =
             std::vector<std::tuple<uint32_t, size_t, int, int>>{},
           py::arg("input_data") = py::bytes(""));
+
+    // ---------------- chaining + exec graph --------------------------------
+    m.def("chain_function",
+          [](const std::string& user,
+             const std::string& function,
+             const py::bytes& input) {
+              std::string s = input;
+              std::vector<uint8_t> in(s.begin(), s.end());
+              py::gil_scoped_release release;
+              return chainFunction(user, function, in);
+          },
+          py::arg("user"),
+          py::arg("function"),
+          py::arg("input") = py::bytes(""));
+    m.def("await_chained_call", [](int32_t msgId, int timeoutMs) {
+        py::gil_scoped_release release;
+        return awaitChainedCall(msgId, timeoutMs);
+    },
+          py::arg("msg_id"),
+          py::arg("timeout_ms") = 60000);
+    m.def("get_exec_graph_json", [](int32_t appId, int32_t msgId) {
+        return getExecGraphJson(appId, msgId);
+    });
+
+    // ---------------- HTTP ops endpoint ------------------------------------
+    py::class_<PlannerEndpoint>(m, "PlannerEndpoint")
+      .def(py::init<int>(), py::arg("port") = PLANNER_HTTP_PORT)
+      .def("start",
+           &PlannerEndpoint::start,
+           py::call_guard<py::gil_scoped_release>())
+      .def("stop",
+           &PlannerEndpoint::stop,
+           py::call_guard<py::gil_scoped_release>())
+      .def("handle", [](PlannerEndpoint& ep, const std::string& body) {
+          std::pair<int, std::string> out;
+          {
+              py::gil_scoped_release release;
+              out = ep.handle(body);
+          }
+          return py::make_tuple(out.first, out.second);
+      });
 
     // Executor memory access for thread bodies / parents
     m.def("executor_set_memory_size", [](size_t n) {

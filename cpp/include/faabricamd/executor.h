@@ -182,4 +182,11 @@ class ExecutorFactory
 void setExecutorFactory(std::shared_ptr<ExecutorFactory> factory);
 std::shared_ptr<ExecutorFactory> getExecutorFactory();
 
+// Function chaining from inside a running task (reference:
+// Executor::addChainedMessage + faasm chaining host interface)
+int32_t chainFunction(const std::string& user,
+                      const std::string& function,
+                      const std::vector<uint8_t>& input);
+Message awaitChainedCall(int32_t msgId, int timeoutMs = 60000);
+
 } // namespace faabricamd

@@ -390,6 +390,45 @@ void Executor::handleTaskResult(Message& msg,
     getPlannerClient().setMessageResult(resultMsg);
 }
 
+// ----------------------------- chaining -------------------------------------
+
+int32_t chainFunction(const std::string& user,
+                      const std::string& function,
+                      const std::vector<uint8_t>& input)
+{
+    // Chained call: a SCALE_CHANGE single-message batch on the caller's
+    // app, recorded in the parent's chainedMsgIds for the exec graph
+    // (reference: Executor::addChainedMessage src/executor/Executor.cpp:656)
+    Message& parent = ExecutorContext::get().getMsg();
+    auto req = std::make_shared<BatchExecuteRequest>();
+    req->appId = parent.appId;
+    req->user = user;
+    req->function = function;
+    Message m = messageFactory(user, function);
+    m.appId = parent.appId;
+    // Chained messages need a group idx that cannot collide with gang
+    // members (broker mappings are keyed per (groupId, idx))
+    m.appIdx = parent.appIdx;
+    m.groupIdx = 1000 + (int32_t)(generateGid() % 14000);
+    m.inputData = input;
+    m.recordExecGraph = parent.recordExecGraph;
+    req->messages.push_back(m);
+
+    auto decision = getPlannerClient().callFunctions(req);
+    if (decision->appId == NOT_ENOUGH_SLOTS) {
+        throw FaabricException("not enough slots for chained call");
+    }
+    parent.chainedMsgIds.push_back(m.id);
+    return m.id;
+}
+
+Message awaitChainedCall(int32_t msgId, int timeoutMs)
+{
+    Message& parent = ExecutorContext::get().getMsg();
+    return getPlannerClient().getMessageResult(parent.appId, msgId,
+                                               timeoutMs);
+}
+
 // ----------------------------- factory -------------------------------------
 
 static std::shared_ptr<ExecutorFactory> executorFactory =
