@@ -39,6 +39,7 @@ hipError_t famDiffXorPages(const void* snap,
                            uint32_t* ticketDev,
                            uint32_t* pageIdxDev,
                            void* payloadDev,
+                           uint32_t* bitmapDev,
                            hipStream_t stream);
 hipError_t famApplyXorPages(void* snap,
                             const uint32_t* pageIdxDev,
@@ -102,10 +103,13 @@ class DeviceSnapshot
     uint8_t* snap_ = nullptr;
     hipStream_t stream_ = nullptr;
 
-    // Diff scratch (lazily allocated): worst case every page dirty
+    // Diff scratch (lazily allocated): worst case every page dirty.
+    // The payload buffer is SPARSE (indexed by page); pageIdx_ is the
+    // compacted dirty list.
     uint32_t* ticket_ = nullptr;
     uint32_t* pageIdx_ = nullptr;
     uint8_t* payload_ = nullptr;
+    uint32_t* bitmap_ = nullptr;
     uint32_t lastDirty_ = 0;
 };
 

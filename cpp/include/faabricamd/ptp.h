@@ -38,6 +38,10 @@ enum class PointToPointCall : uint8_t
 
 inline constexpr int32_t POINT_TO_POINT_MAIN_IDX = 0;
 
+// Dedicated sendIdx namespace for migration control messages so they never
+// collide with app/MPI channels (see ptp.cpp SEND_OFF_* constants)
+inline constexpr int32_t PTP_MIGRATION_CHANNEL_OFFSET = 3 * 16384 + 8192;
+
 class PointToPointBroker;
 PointToPointBroker& getPointToPointBroker();
 

@@ -64,6 +64,9 @@ DeviceSnapshot::~DeviceSnapshot()
     if (payload_ != nullptr) {
         hipFree(payload_);
     }
+    if (bitmap_ != nullptr) {
+        hipFree(bitmap_);
+    }
     if (stream_ != nullptr) {
         hipStreamDestroy(stream_);
     }
@@ -102,6 +105,7 @@ void DeviceSnapshot::ensureDiffBuffers()
     OPS_HIP_CHECK(hipMalloc(&ticket_, sizeof(uint32_t)));
     OPS_HIP_CHECK(hipMalloc(&pageIdx_, nPages * sizeof(uint32_t)));
     OPS_HIP_CHECK(hipMalloc(&payload_, bytes_));
+    OPS_HIP_CHECK(hipMalloc(&bitmap_, ((nPages + 31) / 32) * sizeof(uint32_t)));
 }
 
 std::vector<uint32_t> DeviceSnapshot::dirtyPages(const void* devPtr)
@@ -128,10 +132,12 @@ uint32_t DeviceSnapshot::diffXor(const void* devPtr)
 {
     OPS_HIP_CHECK(hipSetDevice(device_));
     ensureDiffBuffers();
+    size_t nWords = (bytes_ / DEVICE_PAGE + 31) / 32;
     OPS_HIP_CHECK(hipMemsetAsync(ticket_, 0, sizeof(uint32_t), stream_));
     OPS_HIP_CHECK(
-      famDiffXorPages(snap_, devPtr, bytes_, ticket_, pageIdx_, payload_,
-                      stream_));
+      hipMemsetAsync(bitmap_, 0, nWords * sizeof(uint32_t), stream_));
+    OPS_HIP_CHECK(famDiffXorPages(snap_, devPtr, bytes_, ticket_, pageIdx_,
+                                  payload_, bitmap_, stream_));
     uint32_t nDirty = 0;
     OPS_HIP_CHECK(hipMemcpyAsync(&nDirty,
                                  ticket_,

@@ -173,7 +173,8 @@ std::shared_ptr<PendingMigration> Scheduler::checkForMigrationOpportunities(
   Message& msg,
   int32_t overwriteNewGroupId)
 {
-    // Non-main group members wait for the new group id decided by idx 0
+    // Group idx 0 asks the planner (DIST_CHANGE); the rest of the group
+    // receive the verdict over a dedicated PTP channel
     // (reference: src/scheduler/Scheduler.cpp:448-530)
     int32_t newGroupId = 0;
     if (msg.groupIdx == 0 && overwriteNewGroupId == 0) {
@@ -186,6 +187,8 @@ std::shared_ptr<PendingMigration> Scheduler::checkForMigrationOpportunities(
         auto decision = getPlannerClient().callFunctions(req);
         if (decision->appId == DO_NOT_MIGRATE) {
             newGroupId = 0;
+        } else if (decision->appId == MUST_FREEZE) {
+            newGroupId = MUST_FREEZE;
         } else {
             newGroupId = decision->groupId;
         }
@@ -195,7 +198,7 @@ std::shared_ptr<PendingMigration> Scheduler::checkForMigrationOpportunities(
         for (int i = 1; i < msg.groupSize; i++) {
             broker.sendMessage(msg.appId,
                                msg.groupId,
-                               0,
+                               PTP_MIGRATION_CHANNEL_OFFSET,
                                i,
                                (const uint8_t*)&payload,
                                sizeof(payload));
@@ -204,7 +207,7 @@ std::shared_ptr<PendingMigration> Scheduler::checkForMigrationOpportunities(
         newGroupId = overwriteNewGroupId;
     } else {
         auto data = getPointToPointBroker().recvMessage(
-          msg.groupId, 0, msg.groupIdx);
+          msg.groupId, PTP_MIGRATION_CHANNEL_OFFSET, msg.groupIdx);
         newGroupId = *(const int32_t*)data.data();
     }
 
@@ -212,26 +215,24 @@ std::shared_ptr<PendingMigration> Scheduler::checkForMigrationOpportunities(
         return nullptr;
     }
 
-    // A migration is pending: find out whether THIS message moves
-    auto decision =
-      getPlannerClient().getSchedulingDecision(msg.appId);
-    std::string newHost;
-    for (int i = 0; i < decision.nFunctions; i++) {
-        if (decision.groupIdxs[i] == msg.groupIdx) {
-            newHost = decision.hosts[i];
-            break;
-        }
-    }
     auto migration = std::make_shared<PendingMigration>();
     migration->appId = msg.appId;
-    migration->groupId = newGroupId;
     migration->groupIdx = msg.groupIdx;
     migration->srcHost = getSystemConfig().endpointHost;
-    migration->dstHost = newHost;
-    if (newHost == getSystemConfig().endpointHost) {
-        // This rank stays put; update group id only
-        msg.groupId = newGroupId;
-        return nullptr;
+
+    if (newGroupId == MUST_FREEZE) {
+        migration->appId = MUST_FREEZE;
+        return migration;
+    }
+
+    // A migration is happening: find out where THIS message now lives
+    migration->groupId = newGroupId;
+    auto decision = getPlannerClient().getSchedulingDecision(msg.appId);
+    for (int i = 0; i < decision.nFunctions; i++) {
+        if (decision.groupIdxs[i] == msg.groupIdx) {
+            migration->dstHost = decision.hosts[i];
+            break;
+        }
     }
     return migration;
 }
