@@ -536,6 +536,18 @@ void initRuntimeBindings(py::module_& m)
     m.def("get_exec_graph_json", [](int32_t appId, int32_t msgId) {
         return getExecGraphJson(appId, msgId);
     });
+    m.def("migration_point",
+          [](const py::bytes& reentryInput) {
+              std::string s = reentryInput;
+              std::vector<uint8_t> in(s.begin(), s.end());
+              py::gil_scoped_release release;
+              return migrationPoint(in);
+          },
+          py::arg("reentry_input") = py::bytes(""));
+    m.def("MIGRATED_FUNCTION_RETURN_VALUE",
+          [] { return MIGRATED_FUNCTION_RETURN_VALUE; });
+    m.def("FROZEN_FUNCTION_RETURN_VALUE",
+          [] { return FROZEN_FUNCTION_RETURN_VALUE; });
 
     // ---------------- HTTP ops endpoint ------------------------------------
     py::class_<PlannerEndpoint>(m, "PlannerEndpoint")

@@ -189,4 +189,9 @@ int32_t chainFunction(const std::string& user,
                       const std::vector<uint8_t>& input);
 Message awaitChainedCall(int32_t msgId, int timeoutMs = 60000);
 
+// Migration / freeze point for long-running (gang) functions. Returns 0
+// to continue, or the MIGRATED/FROZEN sentinel the function must return
+// immediately (cpp/src/migration.cpp)
+int32_t migrationPoint(const std::vector<uint8_t>& reentryInput);
+
 } // namespace faabricamd

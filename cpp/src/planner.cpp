@@ -270,7 +270,11 @@ std::shared_ptr<SchedulingDecision> Planner::callBatch(
 
     // Un-freeze bookkeeping (reference: src/planner/Planner.cpp:1038-1081)
     if (state.evictedRequests.count(appId) > 0) {
-        if (isNew && isMpi) {
+        if (isNew && !isMpi) {
+            // Non-MPI app: all messages re-dispatch at once, bookkeeping
+            // is complete as soon as the new decision exists
+            state.evictedRequests.erase(appId);
+        } else if (isNew && isMpi) {
             Message first = req->messages.at(0);
             req->messages.clear();
             req->messages.push_back(first);
@@ -628,14 +632,47 @@ std::shared_ptr<BatchExecuteRequestStatus> Planner::getBatchResults(
 {
     auto status = std::make_shared<BatchExecuteRequestStatus>();
     status->appId = appId;
-    std::shared_lock lock(plannerMx);
 
-    // Frozen apps report unfinished until re-scheduled
-    if (state.evictedRequests.count(appId) > 0) {
+    // Result polling doubles as the un-freeze trigger
+    // (reference: src/planner/Planner.cpp:700-726)
+    std::shared_ptr<BatchExecuteRequest> frozenBer;
+    {
+        std::shared_lock lock(plannerMx);
+        auto evIt = state.evictedRequests.find(appId);
+        if (evIt != state.evictedRequests.end()) {
+            bool fullyFrozen = std::all_of(
+              evIt->second->messages.begin(),
+              evIt->second->messages.end(),
+              [](const Message& m) {
+                  return m.returnValue == FROZEN_FUNCTION_RETURN_VALUE;
+              });
+            if (fullyFrozen && state.inFlightReqs.count(appId) == 0) {
+                frozenBer = std::make_shared<BatchExecuteRequest>(
+                  *evIt->second);
+            }
+            if (!frozenBer) {
+                status->finished = false;
+                return status;
+            }
+        }
+    }
+    if (frozenBer) {
+        // Re-schedule as NEW; the messages keep their freeze snapshots
+        // and re-entry input
+        for (auto& m : frozenBer->messages) {
+            m.returnValue = 0;
+            m.executedHost.clear();
+            m.finishTimestamp = 0;
+        }
+        auto decision = callBatch(frozenBer);
+        if (decision->appId == NOT_ENOUGH_SLOTS) {
+            FAM_DEBUG("cannot un-freeze app %d yet: no slots", appId);
+        }
         status->finished = false;
         return status;
     }
 
+    std::shared_lock lock(plannerMx);
     auto it = state.appResults.find(appId);
     if (it == state.appResults.end()) {
         return nullptr;
