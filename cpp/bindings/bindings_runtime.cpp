@@ -17,7 +17,8 @@ namespace py = pybind11;
 using namespace faabricamd;
 
 namespace faabricamd {
-void registerBenchFunctions(); // bench_funcs.cpp
+void registerBenchFunctions();       // bench_funcs.cpp
+void registerMpiExampleFunctions();  // mpi_examples.cpp
 }
 
 void initRuntimeBindings(py::module_& m)
@@ -473,6 +474,8 @@ void initRuntimeBindings(py::module_& m)
 
     // Native benchmark payloads (cpp/src/bench_funcs.cpp)
     m.def("register_bench_functions", [] { registerBenchFunctions(); });
+    m.def("register_mpi_example_functions",
+          [] { registerMpiExampleFunctions(); });
 
     // ---------------- THREADS fork-join (from inside a running task) -------
     m.def("execute_threads",

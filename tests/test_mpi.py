@@ -129,3 +129,14 @@ def test_mpi_world_reuse(runtime):
     # A second world after the first finished: ids and state must not leak
     results = submit_mpi_batch("mpi", "alltests", WORLD_SIZE)
     assert all(r.return_value == 0 for r in results)
+
+
+def test_cpp_mpi_examples(runtime):
+    """C++ MPI programs written against the MPI_* shim (Appendix A
+    surface) run as native functions."""
+    _core.register_mpi_example_functions()
+    for func in ("allreduce", "ring", "async", "allreduce-bench"):
+        results = submit_mpi_batch("mpi-cpp", func, WORLD_SIZE)
+        assert len(results) == WORLD_SIZE
+        for r in results:
+            assert r.return_value == 0, (func, r.output_data)
