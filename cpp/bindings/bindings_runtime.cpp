@@ -12,6 +12,8 @@
 #include "faabricamd/snapshot.h"
 #include "faabricamd/state.h"
 #include "faabricamd/util.h"
+#include "faabricamd/utilextras.h"
+#include "faabricamd/dirty.h"
 
 namespace py = pybind11;
 using namespace faabricamd;
@@ -471,6 +473,55 @@ void initRuntimeBindings(py::module_& m)
         return getBatchRequestsSentMock();
     });
     m.def("clear_mocked_function_calls", &clearMockedFunctionCalls);
+
+    // ---------------- util extras ------------------------------------------
+    m.def("delta_encode",
+          [](const py::bytes& oldData,
+             const py::bytes& newData,
+             const std::string& config) {
+              std::string o = oldData;
+              std::string n = newData;
+              DeltaConfig conf = config.empty() ? DeltaConfig{}
+                                                : DeltaConfig::parse(config);
+              auto out = deltaEncode({ o.begin(), o.end() },
+                                     { n.begin(), n.end() },
+                                     conf);
+              return py::bytes((const char*)out.data(), out.size());
+          },
+          py::arg("old_data"),
+          py::arg("new_data"),
+          py::arg("config") = "");
+    m.def("delta_apply",
+          [](const py::bytes& oldData, const py::bytes& delta) {
+              std::string o = oldData;
+              std::string d = delta;
+              auto out = deltaApply({ o.begin(), o.end() },
+                                    { d.begin(), d.end() });
+              return py::bytes((const char*)out.data(), out.size());
+          });
+    m.def("prof_summary", &profSummary);
+    m.def("prof_clear", &profClear);
+    m.def("set_up_crash_handler", &setUpCrashHandler);
+    m.def("pin_thread_to_free_cpu", &pinThreadToFreeCpu);
+
+    // Self-contained native check of the segfault dirty tracker (mprotect
+    // + SIGSEGV interplay is best kept out of Python)
+    m.def("_selftest_segfault_tracker", [] {
+        py::gil_scoped_release release;
+        SegfaultDirtyTracker tracker;
+        PageAlignedBuffer buf;
+        buf.resize(16 * 4096);
+        tracker.startTracking(buf.data(), buf.size());
+        buf.data()[3 * 4096 + 5] = 42;
+        buf.data()[9 * 4096] = 7;
+        tracker.stopTracking(buf.data(), buf.size());
+        auto dirty = tracker.getDirtyPages(buf.data(), buf.size());
+        int nDirty = 0;
+        for (char c : dirty) {
+            nDirty += c != 0;
+        }
+        return nDirty == 2 && dirty[3] == 1 && dirty[9] == 1;
+    });
 
     // Native benchmark payloads (cpp/src/bench_funcs.cpp)
     m.def("register_bench_functions", [] { registerBenchFunctions(); });

@@ -1,0 +1,97 @@
+// Dirty tracking for HOST memory regions (reference:
+// include/faabric/util/dirty.h:24-225, src/util/dirty.cpp). Three modes,
+// selected by DIRTY_TRACKING_MODE:
+//  - "compare"  (default): no tracking; diffs compare against the snapshot
+//               baseline (the honest mode for HBM, where no mprotect
+//               exists — reference NoneDirtyTracker + DIFFING_MODE=xor)
+//  - "segfault": mprotect(PROT_READ) + SIGSEGV handler marking faulting
+//               pages (reference SegfaultDirtyTracker) — host arenas only
+//  - "none":    every page dirty
+// The reference's soft-PTE and userfaultfd trackers are not reproduced:
+// they are Linux-/WASM-memory-specific optimisations of the same contract
+// and /proc/self/clear_refs is not writable in many container setups.
+#pragma once
+
+#include <cstdint>
+#include <memory>
+#include <string>
+#include <vector>
+
+namespace faabricamd {
+
+class DirtyTracker
+{
+  public:
+    virtual ~DirtyTracker() = default;
+    virtual std::string getType() const = 0;
+
+    // Global channel
+    virtual void startTracking(uint8_t* region, size_t size) = 0;
+    virtual void stopTracking(uint8_t* region, size_t size) = 0;
+    virtual std::vector<char> getDirtyPages(uint8_t* region,
+                                            size_t size) = 0;
+
+    // Thread-local channel (per-task tracking inside a THREADS batch)
+    virtual void startThreadLocalTracking(uint8_t* region, size_t size) = 0;
+    virtual void stopThreadLocalTracking(uint8_t* region, size_t size) = 0;
+    virtual std::vector<char> getThreadLocalDirtyPages(uint8_t* region,
+                                                       size_t size) = 0;
+};
+
+// Every page dirty (reference: util/dirty.h:199-225)
+class NoneDirtyTracker : public DirtyTracker
+{
+  public:
+    std::string getType() const override { return "none"; }
+    void startTracking(uint8_t*, size_t) override {}
+    void stopTracking(uint8_t*, size_t) override {}
+    std::vector<char> getDirtyPages(uint8_t* region, size_t size) override;
+    void startThreadLocalTracking(uint8_t*, size_t) override {}
+    void stopThreadLocalTracking(uint8_t*, size_t) override {}
+    std::vector<char> getThreadLocalDirtyPages(uint8_t* region,
+                                               size_t size) override;
+};
+
+// mprotect + SIGSEGV (reference: src/util/dirty.cpp:136-352). Regions
+// must be page-aligned (use PageAlignedBuffer).
+class SegfaultDirtyTracker : public DirtyTracker
+{
+  public:
+    SegfaultDirtyTracker();
+    std::string getType() const override { return "segfault"; }
+    void startTracking(uint8_t* region, size_t size) override;
+    void stopTracking(uint8_t* region, size_t size) override;
+    std::vector<char> getDirtyPages(uint8_t* region, size_t size) override;
+    void startThreadLocalTracking(uint8_t* region, size_t size) override;
+    void stopThreadLocalTracking(uint8_t* region, size_t size) override;
+    std::vector<char> getThreadLocalDirtyPages(uint8_t* region,
+                                               size_t size) override;
+};
+
+std::shared_ptr<DirtyTracker> getDirtyTracker();
+void resetDirtyTracker();
+
+// OR-merge of per-thread page flags (reference: src/util/memory.cpp:15-23)
+void mergeDirtyPages(std::vector<char>& dest,
+                     const std::vector<char>& src);
+
+// Page-aligned growable host buffer for trackable executor arenas
+class PageAlignedBuffer
+{
+  public:
+    PageAlignedBuffer() = default;
+    ~PageAlignedBuffer();
+    PageAlignedBuffer(const PageAlignedBuffer&) = delete;
+
+    void resize(size_t newSize); // rounds up to whole pages, zero-fills
+    uint8_t* data() { return base; }
+    const uint8_t* data() const { return base; }
+    size_t size() const { return usedSize; }
+
+  private:
+    uint8_t* base = nullptr;
+    size_t allocSize = 0;
+    size_t usedSize = 0;
+};
+
+} // namespace faabricamd

@@ -23,6 +23,7 @@
 
 #include "faabricamd/messages.h"
 #include "faabricamd/queue.h"
+#include "faabricamd/dirty.h"
 #include "faabricamd/snapshot.h"
 
 namespace faabricamd {
@@ -164,8 +165,9 @@ class Executor
     // Batch accounting: tasks remaining in the current batch
     std::shared_ptr<std::atomic<int>> batchCounter;
 
-    // Executor-local arena for THREADS snapshots (host path)
-    std::vector<uint8_t> dummyMemory;
+    // Executor-local arena for THREADS snapshots (host path);
+    // page-aligned so the segfault dirty tracker can mprotect it
+    PageAlignedBuffer dummyMemory;
 };
 
 class ExecutorFactory

@@ -152,6 +152,20 @@ void Executor::executeTasks(std::vector<int> msgIdxs,
         restore(req->messages[msgIdxs[0]].snapshotKey);
     }
 
+    // Fault-driven dirty tracking of the restored arena (segfault mode;
+    // compare mode diffs against the snapshot instead —
+    // reference: src/executor/Executor.cpp:142-162 startTracking)
+    if (req->type == BatchExecuteType::THREADS &&
+        !req->snapshotKey.empty()) {
+        auto tracker = getDirtyTracker();
+        if (tracker->getType() == "segfault") {
+            auto [base, size] = getMemoryView();
+            if (size > 0) {
+                tracker->startTracking(base, size);
+            }
+        }
+    }
+
     std::lock_guard<std::mutex> lock(threadsMx);
     for (int msgIdx : msgIdxs) {
         int poolIdx = threadPoolSize == 0 ? 0 : msgIdx % threadPoolSize;
@@ -339,10 +353,19 @@ void Executor::handleTaskResult(Message& msg,
         try {
             auto snap = SnapshotRegistry::get().getSnapshot(req->snapshotKey);
             auto [base, size] = getMemoryView();
-            // Compare-based dirty tracking: every page is a candidate and
-            // the typed merge regions refine (no mprotect on this path)
             snap->fillGapsWithBytewiseRegions();
-            threadDiffs = snap->diffWithMemory(base, size);
+            auto tracker = getDirtyTracker();
+            if (tracker->getType() == "segfault" && size > 0) {
+                // Fault-tracked pages only
+                tracker->stopTracking(base, size);
+                auto dirty = tracker->getDirtyPages(base, size);
+                threadDiffs =
+                  snap->diffWithDirtyRegions(base, size, dirty);
+            } else {
+                // Compare mode: every page is a candidate and the typed
+                // merge regions refine (no mprotect on the HBM path)
+                threadDiffs = snap->diffWithMemory(base, size);
+            }
         } catch (const std::exception& e) {
             FAM_ERROR("thread diff failed: %s", e.what());
         }
