@@ -281,6 +281,11 @@ void initRuntimeBindings(py::module_& m)
       .def_property_readonly("size", &StateKeyValue::size)
       .def_property_readonly("is_master", &StateKeyValue::isMaster)
       .def_property_readonly("master_host", &StateKeyValue::getMasterHost)
+      .def_property_readonly("on_device", &StateKeyValue::isOnDevice)
+      .def_property_readonly("data_ptr",
+                             [](StateKeyValue& kv) {
+                                 return (uintptr_t)kv.getDataPtr();
+                             })
       .def("get",
            [](StateKeyValue& kv) {
                std::vector<uint8_t> out;
@@ -344,6 +349,17 @@ void initRuntimeBindings(py::module_& m)
           kv.pushFull();
       });
 
+    m.def("state_get_kv_device",
+          [](const std::string& user,
+             const std::string& key,
+             size_t size,
+             int device) {
+              return State::get().getKVDevice(user, key, size, device);
+          },
+          py::arg("user"),
+          py::arg("key"),
+          py::arg("size"),
+          py::arg("device") = 0);
     m.def("state_get_kv",
           [](const std::string& user, const std::string& key, size_t size) {
               return State::get().getKV(user, key, size);

@@ -41,7 +41,10 @@ class StateKeyValue
     StateKeyValue(std::string userIn,
                   std::string keyIn,
                   size_t sizeIn,
-                  std::string masterHostIn);
+                  std::string masterHostIn,
+                  bool onDeviceIn = false,
+                  int deviceIn = 0);
+    ~StateKeyValue();
 
     const std::string& getUser() const { return user; }
     const std::string& getKey() const { return key; }
@@ -71,8 +74,14 @@ class StateKeyValue
     std::vector<std::vector<uint8_t>> getAppended(size_t nValues);
     void clearAppended();
 
-    // Direct access for zero-copy users (host path)
-    uint8_t* getDataPtr() { return value.data(); }
+    // Direct access for zero-copy users: host pointer, or the HBM
+    // pointer when the value is device-resident
+    uint8_t* getDataPtr()
+    {
+        return onDevice ? devPtr : value.data();
+    }
+    bool isOnDevice() const { return onDevice; }
+    int getDevice() const { return device; }
 
     // Master-side servicing
     std::vector<uint8_t> serviceChunk(uint64_t offset, size_t len);
@@ -88,7 +97,13 @@ class StateKeyValue
     std::string masterHost;
 
     std::mutex kvMx;
-    std::vector<uint8_t> value;
+    std::vector<uint8_t> value; // host mode
+    bool onDevice = false;      // HBM mode: value lives in devPtr
+    int device = 0;
+    uint8_t* devPtr = nullptr;
+
+    void readLocal(uint64_t offset, uint8_t* out, size_t len);
+    void writeLocal(uint64_t offset, const uint8_t* data, size_t len);
     std::vector<char> dirtyChunks; // one flag per STATE_STREAM_CHUNK_SIZE
     bool fullyPulled = false;
     std::vector<std::vector<uint8_t>> appendedValues;
@@ -102,6 +117,12 @@ class State
     std::shared_ptr<StateKeyValue> getKV(const std::string& user,
                                          const std::string& key,
                                          size_t size);
+    // HBM-resident value on this host's GPU (north star: distributed
+    // state lives in the 288 GB HBM3E per GPU)
+    std::shared_ptr<StateKeyValue> getKVDevice(const std::string& user,
+                                               const std::string& key,
+                                               size_t size,
+                                               int device = 0);
     std::shared_ptr<StateKeyValue> getKV(const std::string& user,
                                          const std::string& key);
     size_t getStateSize(const std::string& user, const std::string& key);

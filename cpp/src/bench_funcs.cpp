@@ -49,7 +49,13 @@ static int32_t benchKvTouch(Message& msg)
     buf.resize(kvBytes);
     std::memset(buf.data(), (int)(msg.id & 0xff), kvBytes);
 
-    auto kv = State::get().getKV("bench", "kv", 1024 * 1024);
+    // State lives in HBM when a GPU is present (north star), host
+    // memory otherwise
+    int nGpus = 0;
+    (void)hipGetDeviceCount(&nGpus);
+    auto kv = nGpus > 0
+                ? State::get().getKVDevice("bench", "kv", 1024 * 1024)
+                : State::get().getKV("bench", "kv", 1024 * 1024);
     uint64_t offset =
       ((uint64_t)(uint32_t)msg.id * 4096) % (1024 * 1024 - kvBytes);
     kv->setChunk(offset, buf.data(), kvBytes);

@@ -687,6 +687,9 @@ void MpiWorld::broadcast(int rootRank,
                          MpiMessageType messageType,
                          MpiBufferLoc loc)
 {
+    if (size == 1 && isDeviceBuffer(buffer, loc)) {
+        return; // broadcast to self is a no-op
+    }
     if (isDeviceBuffer(buffer, loc)) {
         ensureRcclComm(thisRank);
         ncclComm_t comm;
@@ -777,6 +780,14 @@ void MpiWorld::allGather(int thisRank,
                          int count,
                          MpiBufferLoc loc)
 {
+    if (size == 1 && isDeviceBuffer(recvBuffer, loc)) {
+        size_t bytes = mpiTypeSize(dataType) * (size_t)count;
+        if (recvBuffer != sendBuffer) {
+            HIP_CHECK(hipMemcpy(recvBuffer, sendBuffer, bytes,
+                                hipMemcpyDeviceToDevice));
+        }
+        return;
+    }
     if (isDeviceBuffer(recvBuffer, loc)) {
         ensureRcclComm(thisRank);
         ncclComm_t comm;
@@ -812,6 +823,14 @@ void MpiWorld::reduce(int thisRank,
                       MpiOp op,
                       MpiBufferLoc loc)
 {
+    if (size == 1 && isDeviceBuffer(sendBuffer, loc)) {
+        size_t bytes = mpiTypeSize(dataType) * (size_t)count;
+        if (recvBuffer != sendBuffer) {
+            HIP_CHECK(hipMemcpy(recvBuffer, sendBuffer, bytes,
+                                hipMemcpyDeviceToDevice));
+        }
+        return;
+    }
     if (isDeviceBuffer(sendBuffer, loc)) {
         ensureRcclComm(thisRank);
         ncclComm_t comm;
@@ -858,6 +877,15 @@ void MpiWorld::allReduce(int thisRank,
                          MpiOp op,
                          MpiBufferLoc loc)
 {
+    if (size == 1 && isDeviceBuffer(sendBuffer, loc)) {
+        // Single-rank collective = local copy; skip the RCCL machinery
+        size_t bytes = mpiTypeSize(dataType) * (size_t)count;
+        if (recvBuffer != sendBuffer) {
+            HIP_CHECK(hipMemcpy(recvBuffer, sendBuffer, bytes,
+                                hipMemcpyDeviceToDevice));
+        }
+        return;
+    }
     if (isDeviceBuffer(sendBuffer, loc)) {
         ensureRcclComm(thisRank);
         ncclComm_t comm;
@@ -898,6 +926,13 @@ void MpiWorld::allToAll(int thisRank,
                         MpiBufferLoc loc)
 {
     size_t bytes = mpiTypeSize(dataType) * (size_t)count;
+    if (size == 1 && isDeviceBuffer(sendBuffer, loc)) {
+        if (recvBuffer != sendBuffer) {
+            HIP_CHECK(hipMemcpy(recvBuffer, sendBuffer, bytes,
+                                hipMemcpyDeviceToDevice));
+        }
+        return;
+    }
     if (isDeviceBuffer(sendBuffer, loc)) {
         ensureRcclComm(thisRank);
         ncclComm_t comm;

@@ -2,7 +2,10 @@
 // InMemoryStateKeyValue.cpp:1-187, StateServer.cpp:24-164,
 // StateClient.cpp). See state.h for the re-design notes.
 #include "faabricamd/state.h"
+#include "faabricamd/ops.h"
 #include "faabricamd/util.h"
+
+#include <hip/hip_runtime.h>
 
 #include <algorithm>
 #include <cstring>
@@ -17,16 +20,60 @@ static std::string kvKeyOf(const std::string& user, const std::string& key)
 StateKeyValue::StateKeyValue(std::string userIn,
                              std::string keyIn,
                              size_t sizeIn,
-                             std::string masterHostIn)
+                             std::string masterHostIn,
+                             bool onDeviceIn,
+                             int deviceIn)
   : user(std::move(userIn))
   , key(std::move(keyIn))
   , valueSize(sizeIn)
   , masterHost(std::move(masterHostIn))
+  , onDevice(onDeviceIn)
+  , device(deviceIn)
 {
-    value.resize(valueSize, 0);
+    if (onDevice) {
+        if (!gpuAvailable()) {
+            throw FaabricException("device state KV requires a GPU");
+        }
+        if (hipSetDevice(device) != hipSuccess ||
+            hipMalloc(&devPtr, valueSize) != hipSuccess) {
+            throw FaabricException("HBM alloc for state KV failed");
+        }
+        hipMemset(devPtr, 0, valueSize);
+    } else {
+        value.resize(valueSize, 0);
+    }
     size_t nChunks =
       (valueSize + STATE_STREAM_CHUNK_SIZE - 1) / STATE_STREAM_CHUNK_SIZE;
     dirtyChunks.resize(std::max<size_t>(nChunks, 1), 0);
+}
+
+StateKeyValue::~StateKeyValue()
+{
+    if (devPtr != nullptr) {
+        hipFree(devPtr);
+    }
+}
+
+void StateKeyValue::readLocal(uint64_t offset, uint8_t* out, size_t len)
+{
+    if (onDevice) {
+        hipSetDevice(device);
+        hipMemcpy(out, devPtr + offset, len, hipMemcpyDeviceToHost);
+    } else {
+        std::memcpy(out, value.data() + offset, len);
+    }
+}
+
+void StateKeyValue::writeLocal(uint64_t offset,
+                               const uint8_t* data,
+                               size_t len)
+{
+    if (onDevice) {
+        hipSetDevice(device);
+        hipMemcpy(devPtr + offset, data, len, hipMemcpyHostToDevice);
+    } else {
+        std::memcpy(value.data() + offset, data, len);
+    }
 }
 
 bool StateKeyValue::isMaster() const
@@ -38,7 +85,7 @@ void StateKeyValue::get(uint8_t* buffer)
 {
     pull();
     std::lock_guard<std::mutex> lock(kvMx);
-    std::memcpy(buffer, value.data(), valueSize);
+    readLocal(0, buffer, valueSize);
 }
 
 std::vector<uint8_t> StateKeyValue::get()
@@ -55,7 +102,7 @@ void StateKeyValue::set(const uint8_t* buffer, size_t n)
         if (n > valueSize) {
             throw FaabricException("state set exceeds value size");
         }
-        std::memcpy(value.data(), buffer, n);
+        writeLocal(0, buffer, n);
         fullyPulled = true;
     }
     flagDirty();
@@ -79,12 +126,12 @@ void StateKeyValue::getChunk(uint64_t offset, uint8_t* buffer, size_t len)
         auto cli = getStateClient(masterHost);
         auto data = cli->pullChunk(user, key, offset, len);
         std::lock_guard<std::mutex> lock(kvMx);
-        std::memcpy(value.data() + offset, data.data(), data.size());
+        writeLocal(offset, data.data(), data.size());
         std::memcpy(buffer, data.data(), len);
         return;
     }
     std::lock_guard<std::mutex> lock(kvMx);
-    std::memcpy(buffer, value.data() + offset, len);
+    readLocal(offset, buffer, len);
 }
 
 void StateKeyValue::setChunk(uint64_t offset, const uint8_t* buffer,
@@ -95,7 +142,7 @@ void StateKeyValue::setChunk(uint64_t offset, const uint8_t* buffer,
     }
     {
         std::lock_guard<std::mutex> lock(kvMx);
-        std::memcpy(value.data() + offset, buffer, len);
+        writeLocal(offset, buffer, len);
     }
     flagChunkDirty(offset, len);
     if (!isMaster()) {
@@ -122,7 +169,7 @@ void StateKeyValue::pull()
                               (size_t)(valueSize - off));
         auto data = cli->pullChunk(user, key, off, len);
         std::lock_guard<std::mutex> lock(kvMx);
-        std::memcpy(value.data() + off, data.data(), data.size());
+        writeLocal(off, data.data(), data.size());
     }
     std::lock_guard<std::mutex> lock(kvMx);
     fullyPulled = true;
@@ -135,11 +182,13 @@ void StateKeyValue::pushFull()
     }
     auto cli = getStateClient(masterHost);
     std::lock_guard<std::mutex> lock(kvMx);
+    std::vector<uint8_t> staging(STATE_STREAM_CHUNK_SIZE);
     for (uint64_t off = 0; off < valueSize;
          off += STATE_STREAM_CHUNK_SIZE) {
         size_t len = std::min(STATE_STREAM_CHUNK_SIZE,
                               (size_t)(valueSize - off));
-        cli->pushChunk(user, key, off, value.data() + off, len);
+        readLocal(off, staging.data(), len);
+        cli->pushChunk(user, key, off, staging.data(), len);
     }
     std::fill(dirtyChunks.begin(), dirtyChunks.end(), 0);
 }
@@ -176,7 +225,9 @@ void StateKeyValue::pushPartial()
         uint64_t off = i * STATE_STREAM_CHUNK_SIZE;
         size_t len = std::min(STATE_STREAM_CHUNK_SIZE,
                               (size_t)(valueSize - off));
-        cli->pushChunk(user, key, off, value.data() + off, len);
+        std::vector<uint8_t> staging(len);
+        readLocal(off, staging.data(), len);
+        cli->pushChunk(user, key, off, staging.data(), len);
         dirtyChunks[i] = 0;
     }
 }
@@ -216,7 +267,9 @@ std::vector<uint8_t> StateKeyValue::serviceChunk(uint64_t offset, size_t len)
     if (offset + len > valueSize) {
         throw FaabricException("state chunk service out of bounds");
     }
-    return { value.begin() + offset, value.begin() + offset + len };
+    std::vector<uint8_t> out(len);
+    readLocal(offset, out.data(), len);
+    return out;
 }
 
 void StateKeyValue::serviceSet(uint64_t offset,
@@ -227,7 +280,7 @@ void StateKeyValue::serviceSet(uint64_t offset,
     if (offset + len > valueSize) {
         throw FaabricException("state chunk service-set out of bounds");
     }
-    std::memcpy(value.data() + offset, data, len);
+    writeLocal(offset, data, len);
 }
 
 void StateKeyValue::serviceAppend(const uint8_t* data, size_t len)
@@ -278,6 +331,31 @@ std::shared_ptr<StateKeyValue> State::getKV(const std::string& user,
         masterMap[k] = master;
     }
     auto kv = std::make_shared<StateKeyValue>(user, key, size, master);
+    kvMap[k] = kv;
+    return kv;
+}
+
+std::shared_ptr<StateKeyValue> State::getKVDevice(const std::string& user,
+                                                  const std::string& key,
+                                                  size_t size,
+                                                  int device)
+{
+    std::lock_guard<std::mutex> lock(mx);
+    std::string k = kvKeyOf(user, key);
+    auto it = kvMap.find(k);
+    if (it != kvMap.end()) {
+        return it->second;
+    }
+    std::string master;
+    auto mIt = masterMap.find(k);
+    if (mIt != masterMap.end()) {
+        master = mIt->second;
+    } else {
+        master = getSystemConfig().endpointHost;
+        masterMap[k] = master;
+    }
+    auto kv = std::make_shared<StateKeyValue>(
+      user, key, size, master, /*onDevice=*/true, device);
     kvMap[k] = kv;
     return kv;
 }

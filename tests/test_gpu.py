@@ -199,3 +199,24 @@ def test_snapshot_pipeline_bench_small():
     )
     assert res["dirty_pages"] == int(res["n_pages"] * 0.25)
     assert res["diff_gbps"] > 10  # sanity floor; target is TB/s-class
+
+
+@requires_gpu
+def test_state_kv_in_hbm(runtime):
+    kv = _core.state_get_kv_device("gpu", "hbmkey", 256 * 1024)
+    assert kv.on_device
+    assert kv.is_master
+    kv.set(b"\x07" * (256 * 1024))
+    assert kv.get() == b"\x07" * (256 * 1024)
+    kv.set_chunk(4096, b"CHUNK")
+    data = kv.get_chunk(4090, 16)
+    assert data[6:11] == b"CHUNK"
+    # The HBM pointer is directly usable by kernels
+    n = 256 * 1024
+    t = torch.zeros(n, dtype=torch.uint8, device="cuda")
+    torch.cuda.synchronize()
+    _core.device_elementwise_op(
+        t.data_ptr(), kv.data_ptr, n, 5, 0
+    )  # byte sum: t += kv
+    torch.cuda.synchronize()
+    assert int(t[0]) == 7 and int(t[5000]) == 7
