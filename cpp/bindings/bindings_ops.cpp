@@ -79,12 +79,18 @@ SnapBenchResult benchSnapshotPipelineImpl(size_t bytes,
         uint32_t* pagesDev = nullptr;
         uint8_t* flipDev = nullptr;
         hipMalloc(&pagesDev, nDirtyTarget * sizeof(uint32_t));
-        hipMalloc(&flipDev, (size_t)nDirtyTarget * DEVICE_PAGE);
+        // famApplyXorPages reads the payload SPARSELY (indexed by page),
+        // so the flip buffer must span the whole region
+        if (hipMalloc(&flipDev, bytes) != hipSuccess) {
+            hipFree(pagesDev);
+            hipFree(updated);
+            throw FaabricException("hipMalloc flip buffer failed");
+        }
         hipMemcpy(pagesDev,
                   pages.data(),
                   nDirtyTarget * sizeof(uint32_t),
                   hipMemcpyHostToDevice);
-        hipMemset(flipDev, 0x5a, (size_t)nDirtyTarget * DEVICE_PAGE);
+        hipMemset(flipDev, 0x5a, bytes);
         famApplyXorPages(updated, pagesDev, flipDev, nDirtyTarget, nullptr);
         hipDeviceSynchronize();
         hipFree(pagesDev);
