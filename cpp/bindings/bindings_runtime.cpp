@@ -38,6 +38,9 @@ void initRuntimeBindings(py::module_& m)
     m.def("set_batch_scheduler_mode", [](const std::string& mode) {
         resetBatchScheduler(mode);
     });
+    m.def("decision_cache_size",
+          [] { return DecisionCache::get().size(); });
+    m.def("decision_cache_clear", [] { DecisionCache::get().clear(); });
 
     // ---------------- scheduling decision ----------------
     py::class_<SchedulingDecision>(m, "SchedulingDecision")

@@ -9,6 +9,7 @@
 #pragma once
 
 #include <map>
+#include <mutex>
 #include <memory>
 #include <set>
 #include <string>
@@ -151,6 +152,26 @@ class SpotScheduler : public BatchScheduler
       HostMap& hostMap,
       const InFlightReqs& inFlightReqs,
       const BatchExecuteRequest& req) override;
+};
+
+// Cache of past placements keyed (user, function, size), reused when the
+// caller opts in (reference: src/batch-scheduler/DecisionCache.cpp —
+// the CACHED topology hint)
+class DecisionCache
+{
+  public:
+    static DecisionCache& get();
+    std::shared_ptr<SchedulingDecision> getCachedDecision(
+      const BatchExecuteRequest& req);
+    void addCachedDecision(const BatchExecuteRequest& req,
+                           const SchedulingDecision& decision);
+    void clear();
+    size_t size();
+
+  private:
+    std::string keyOf(const BatchExecuteRequest& req);
+    std::mutex mx;
+    std::map<std::string, std::shared_ptr<SchedulingDecision>> cache;
 };
 
 std::shared_ptr<BatchScheduler> getBatchScheduler();

@@ -518,6 +518,47 @@ std::shared_ptr<SchedulingDecision> SpotScheduler::makeSchedulingDecision(
     return decision;
 }
 
+// ------------------------- decision cache ------------------------------------
+
+DecisionCache& DecisionCache::get()
+{
+    static DecisionCache cache;
+    return cache;
+}
+
+std::string DecisionCache::keyOf(const BatchExecuteRequest& req)
+{
+    return req.user + "/" + req.function + "/" +
+           std::to_string(req.messages.size());
+}
+
+std::shared_ptr<SchedulingDecision> DecisionCache::getCachedDecision(
+  const BatchExecuteRequest& req)
+{
+    std::lock_guard<std::mutex> lock(mx);
+    auto it = cache.find(keyOf(req));
+    return it == cache.end() ? nullptr : it->second;
+}
+
+void DecisionCache::addCachedDecision(const BatchExecuteRequest& req,
+                                      const SchedulingDecision& decision)
+{
+    std::lock_guard<std::mutex> lock(mx);
+    cache[keyOf(req)] = std::make_shared<SchedulingDecision>(decision);
+}
+
+void DecisionCache::clear()
+{
+    std::lock_guard<std::mutex> lock(mx);
+    cache.clear();
+}
+
+size_t DecisionCache::size()
+{
+    std::lock_guard<std::mutex> lock(mx);
+    return cache.size();
+}
+
 // ------------------------- registry -----------------------------------------
 
 static std::shared_ptr<BatchScheduler> currentScheduler;

@@ -333,6 +333,7 @@ std::shared_ptr<SchedulingDecision> Planner::callBatch(
             }
 
             state.inFlightReqs[appId] = { req, decision };
+            DecisionCache::get().addCachedDecision(*req, *decision);
             broker.setAndSendMappingsFromSchedulingDecision(*decision);
             break;
         }
