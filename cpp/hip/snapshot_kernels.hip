@@ -149,17 +149,16 @@ __global__ __launch_bounds__(FAM_KERNEL_BLOCK) void compactPageIdxKernel(
     for (u32 w0 = waveId * 64; w0 < nWords; w0 += nWaves * 64) {
         u32 word = (w0 + lane) < nWords ? bitmap[w0 + lane] : 0u;
         u32 mine = __popc(word);
-        // Exclusive prefix over the wave's 64 lane counts
-        u32 prefix = 0;
+        // Hillis-Steele inclusive scan over the wave's 64 lane counts
+        u32 scan = mine;
         for (u32 off = 1; off < 64; off <<= 1) {
-            u32 up = (u32)__shfl_up((int)mine, (int)off);
+            u32 up = (u32)__shfl_up((int)scan, (int)off);
             if (lane >= off) {
-                prefix += up;
+                scan += up;
             }
         }
-        prefix -= mine; // inclusive → exclusive
-        u32 waveTotal =
-          (u32)__shfl((int)(prefix + mine), 63);
+        u32 prefix = scan - mine; // exclusive
+        u32 waveTotal = (u32)__shfl((int)scan, 63);
         if (waveTotal == 0) {
             continue;
         }

@@ -159,6 +159,8 @@ class PlannerClient
     Message getMessageResult(const Message& msg, int timeoutMs);
     Message getMessageResult(int32_t appId, int32_t msgId, int timeoutMs);
     BatchExecuteRequestStatus getBatchResults(int32_t appId);
+    // Lightweight poll: finished flag + result count only
+    std::pair<bool, int> getBatchStatusCounts(int32_t appId);
 
     SchedulingDecision getSchedulingDecision(int32_t appId);
     void preloadSchedulingDecision(

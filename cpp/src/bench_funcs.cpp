@@ -171,11 +171,11 @@ static int32_t benchRankStep(Message& msg)
                 msg.outputData = "batch bench: not enough slots";
                 break;
             }
-            // Poll for completion
+            // Poll for completion (count-only status, cheap at 1024 msgs)
             while (true) {
-                auto status = getPlannerClient().getBatchResults(ber->appId);
-                if (status.expectedNumMessages != -1 && status.finished &&
-                    (int)status.messageResults.size() >= total) {
+                auto [finished, n] =
+                  getPlannerClient().getBatchStatusCounts(ber->appId);
+                if (finished && n >= total) {
                     break;
                 }
                 usleep(200);

@@ -852,6 +852,15 @@ std::string PlannerServer::doSyncRecv(uint8_t code, const std::string& body)
                 empty.expectedNumMessages = -1; // marker: unknown app
                 return empty.encode();
             }
+            if (req.subType == 1) {
+                // Count-only poll: skip re-serialising every result
+                BatchExecuteRequestStatus counts;
+                counts.appId = status->appId;
+                counts.finished = status->finished;
+                counts.expectedNumMessages =
+                  (int32_t)status->messageResults.size();
+                return counts.encode();
+            }
             return status->encode();
         }
         case PlannerCalls::GetSchedulingDecision: {
@@ -1033,6 +1042,20 @@ BatchExecuteRequestStatus PlannerClient::getBatchResults(int32_t appId)
     std::string resp =
       rpc.syncSend((uint8_t)PlannerCalls::GetBatchResults, req.encode());
     return BatchExecuteRequestStatus::decode(resp);
+}
+
+std::pair<bool, int> PlannerClient::getBatchStatusCounts(int32_t appId)
+{
+    BatchExecuteRequest req;
+    req.appId = appId;
+    req.subType = 1;
+    std::string resp =
+      rpc.syncSend((uint8_t)PlannerCalls::GetBatchResults, req.encode());
+    auto status = BatchExecuteRequestStatus::decode(resp);
+    if (status.expectedNumMessages == -1) {
+        return { false, -1 };
+    }
+    return { status.finished, status.expectedNumMessages };
 }
 
 SchedulingDecision PlannerClient::getSchedulingDecision(int32_t appId)
