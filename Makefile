@@ -39,11 +39,14 @@ $(BUILD)/%.o: cpp/%.hip
 $(TARGET): $(OBJS)
 	$(HIPCC) $(OBJS) $(LDFLAGS) -o $@
 
-# C++ example embedder (reference examples/check.cpp parity)
+# C++ example binaries (reference examples/ + planner_server parity)
 CORE_OBJS = $(filter-out $(BUILD)/bindings/%,$(OBJS))
-examples: $(TARGET) $(BUILD)/examples/check.o
-	$(HIPCC) $(BUILD)/examples/check.o $(CORE_OBJS) \
-	    -L/opt/rocm/lib -lrccl -lamdhip64 -lz -pthread -o $(BUILD)/check
+EXAMPLE_BINS = $(BUILD)/check $(BUILD)/planner_server $(BUILD)/server
+examples: $(TARGET) $(EXAMPLE_BINS)
+
+$(BUILD)/%: $(BUILD)/examples/%.o $(CORE_OBJS)
+	$(HIPCC) $< $(CORE_OBJS) \
+	    -L/opt/rocm/lib -lrccl -lamdhip64 -lz -pthread -o $@
 
 $(BUILD)/examples/%.o: examples/%.cpp
 	@mkdir -p $(dir $@)

@@ -783,8 +783,11 @@ void MpiWorld::allGather(int thisRank,
     if (size == 1 && isDeviceBuffer(recvBuffer, loc)) {
         size_t bytes = mpiTypeSize(dataType) * (size_t)count;
         if (recvBuffer != sendBuffer) {
+            // D2D memcpy is asynchronous to the host: synchronise so the
+            // collective keeps blocking semantics (and honest timings)
             HIP_CHECK(hipMemcpy(recvBuffer, sendBuffer, bytes,
                                 hipMemcpyDeviceToDevice));
+            HIP_CHECK(hipStreamSynchronize(nullptr));
         }
         return;
     }
@@ -826,8 +829,11 @@ void MpiWorld::reduce(int thisRank,
     if (size == 1 && isDeviceBuffer(sendBuffer, loc)) {
         size_t bytes = mpiTypeSize(dataType) * (size_t)count;
         if (recvBuffer != sendBuffer) {
+            // D2D memcpy is asynchronous to the host: synchronise so the
+            // collective keeps blocking semantics (and honest timings)
             HIP_CHECK(hipMemcpy(recvBuffer, sendBuffer, bytes,
                                 hipMemcpyDeviceToDevice));
+            HIP_CHECK(hipStreamSynchronize(nullptr));
         }
         return;
     }
@@ -881,8 +887,11 @@ void MpiWorld::allReduce(int thisRank,
         // Single-rank collective = local copy; skip the RCCL machinery
         size_t bytes = mpiTypeSize(dataType) * (size_t)count;
         if (recvBuffer != sendBuffer) {
+            // D2D memcpy is asynchronous to the host: synchronise so the
+            // collective keeps blocking semantics (and honest timings)
             HIP_CHECK(hipMemcpy(recvBuffer, sendBuffer, bytes,
                                 hipMemcpyDeviceToDevice));
+            HIP_CHECK(hipStreamSynchronize(nullptr));
         }
         return;
     }
@@ -928,8 +937,11 @@ void MpiWorld::allToAll(int thisRank,
     size_t bytes = mpiTypeSize(dataType) * (size_t)count;
     if (size == 1 && isDeviceBuffer(sendBuffer, loc)) {
         if (recvBuffer != sendBuffer) {
+            // D2D memcpy is asynchronous to the host: synchronise so the
+            // collective keeps blocking semantics (and honest timings)
             HIP_CHECK(hipMemcpy(recvBuffer, sendBuffer, bytes,
                                 hipMemcpyDeviceToDevice));
+            HIP_CHECK(hipStreamSynchronize(nullptr));
         }
         return;
     }

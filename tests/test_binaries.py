@@ -1,0 +1,104 @@
+"""Deployment parity: the standalone planner_server + worker binaries run
+as real processes and serve an EXECUTE_BATCH over the HTTP ops API
+(reference: docker-compose deployment of planner + worker containers)."""
+
+import json
+import os
+import signal
+import subprocess
+import time
+import urllib.request
+
+import pytest
+
+REPO_ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+BUILD = os.path.join(REPO_ROOT, "build")
+
+PLANNER_OFF = 7000
+WORKER_OFF = 7200
+
+pytestmark = pytest.mark.skipif(
+    not os.path.exists(os.path.join(BUILD, "planner_server")),
+    reason="example binaries not built (run `make examples`)",
+)
+
+
+def post(http_type, payload="", timeout=10):
+    body = json.dumps({"http_type": http_type, "payload": payload}).encode()
+    req = urllib.request.Request(
+        f"http://127.0.0.1:{8080 + PLANNER_OFF}/", data=body, method="POST"
+    )
+    with urllib.request.urlopen(req, timeout=timeout) as resp:
+        return resp.status, resp.read().decode()
+
+
+@pytest.fixture(scope="module")
+def binaries():
+    env_common = {**os.environ, "LOG_LEVEL": "error"}
+    planner = subprocess.Popen(
+        [os.path.join(BUILD, "planner_server")],
+        env={
+            **env_common,
+            "FAABRIC_PORT_OFFSET": str(PLANNER_OFF),
+            "ENDPOINT_HOST": f"127.0.0.1@{PLANNER_OFF}",
+        },
+        stdout=subprocess.DEVNULL,
+        stderr=subprocess.DEVNULL,
+    )
+    worker = subprocess.Popen(
+        [os.path.join(BUILD, "server")],
+        env={
+            **env_common,
+            "FAABRIC_PORT_OFFSET": str(WORKER_OFF),
+            "ENDPOINT_HOST": f"127.0.0.1@{WORKER_OFF}",
+            "PLANNER_HOST": f"127.0.0.1@{PLANNER_OFF}",
+            "OVERRIDE_CPU_COUNT": "4",
+            "FAABRIC_USE_GPU": "0",
+        },
+        stdout=subprocess.DEVNULL,
+        stderr=subprocess.DEVNULL,
+    )
+    # Wait for registration
+    deadline = time.monotonic() + 20
+    ready = False
+    while time.monotonic() < deadline:
+        try:
+            status, body = post(5)  # GET_AVAILABLE_HOSTS
+            if status == 200 and len(json.loads(body)["hosts"]) == 1:
+                ready = True
+                break
+        except Exception:
+            pass
+        time.sleep(0.1)
+    yield ready
+    for p in (worker, planner):
+        p.send_signal(signal.SIGTERM)
+    for p in (worker, planner):
+        try:
+            p.wait(timeout=10)
+        except subprocess.TimeoutExpired:
+            p.kill()
+
+
+def test_binaries_serve_batch(binaries):
+    assert binaries, "worker did not register with planner binary"
+    ber = {
+        "user": "demo",
+        "function": "anything",
+        "messages": [{}, {}, {}],
+    }
+    status, body = post(10, json.dumps(ber))  # EXECUTE_BATCH
+    assert status == 200, body
+    app_id = json.loads(body)["appId"]
+
+    deadline = time.monotonic() + 15
+    while time.monotonic() < deadline:
+        status, body = post(11, json.dumps({"appId": app_id}))
+        if status == 200:
+            parsed = json.loads(body)
+            if parsed.get("finished") and len(parsed["messageResults"]) == 3:
+                outs = [m["output_data"] for m in parsed["messageResults"]]
+                assert all("Example executor ran" in o for o in outs)
+                return
+        time.sleep(0.05)
+    pytest.fail("batch did not finish through the binaries")
