@@ -28,8 +28,11 @@ def post(http_type, payload="", timeout=10):
     req = urllib.request.Request(
         f"http://127.0.0.1:{8080 + PLANNER_OFF}/", data=body, method="POST"
     )
-    with urllib.request.urlopen(req, timeout=timeout) as resp:
-        return resp.status, resp.read().decode()
+    try:
+        with urllib.request.urlopen(req, timeout=timeout) as resp:
+            return resp.status, resp.read().decode()
+    except urllib.error.HTTPError as e:
+        return e.code, e.read().decode()
 
 
 @pytest.fixture(scope="module")
@@ -87,7 +90,14 @@ def test_binaries_serve_batch(binaries):
         "function": "anything",
         "messages": [{}, {}, {}],
     }
-    status, body = post(10, json.dumps(ber))  # EXECUTE_BATCH
+    # The worker keep-alive is half the 5s host timeout; under a loaded
+    # test machine a beat can slip, so tolerate a transient no-hosts
+    deadline = time.monotonic() + 10
+    while True:
+        status, body = post(10, json.dumps(ber))  # EXECUTE_BATCH
+        if status == 200 or time.monotonic() > deadline:
+            break
+        time.sleep(0.2)
     assert status == 200, body
     app_id = json.loads(body)["appId"]
 
