@@ -38,6 +38,75 @@ void initRuntimeBindings(py::module_& m)
     m.def("set_batch_scheduler_mode", [](const std::string& mode) {
         resetBatchScheduler(mode);
     });
+    // Pure policy evaluation for unit tests: hosts = [(ip, slots, used)],
+    // in-flight = {appId: ([msgIds], [hosts])}; returns the decision
+    m.def("test_make_scheduling_decision",
+          [](const std::string& mode,
+             const std::vector<std::tuple<std::string, int, int>>& hosts,
+             int nMessages,
+             int32_t appId,
+             bool migration,
+             const std::vector<std::pair<int32_t, std::vector<std::string>>>&
+               inFlight) {
+              std::shared_ptr<BatchScheduler> sched;
+              if (mode == "bin-pack") {
+                  sched = std::make_shared<BinPackScheduler>();
+              } else if (mode == "compact") {
+                  sched = std::make_shared<CompactScheduler>();
+              } else if (mode == "spot") {
+                  sched = std::make_shared<SpotScheduler>();
+              } else {
+                  throw FaabricException("bad mode");
+              }
+              HostMap hostMap;
+              for (const auto& [ip, slots, used] : hosts) {
+                  hostMap[ip] =
+                    std::make_shared<HostState>(ip, slots, used);
+              }
+              InFlightReqs reqs;
+              for (const auto& [ifAppId, ifHosts] : inFlight) {
+                  auto ber = std::make_shared<BatchExecuteRequest>();
+                  ber->appId = ifAppId;
+                  ber->user = "t";
+                  ber->function = "t";
+                  auto dec =
+                    std::make_shared<SchedulingDecision>(ifAppId, 1);
+                  int idx = 0;
+                  for (const auto& h : ifHosts) {
+                      Message m;
+                      m.id = ifAppId * 1000 + idx;
+                      m.appId = ifAppId;
+                      m.appIdx = idx;
+                      m.groupIdx = idx;
+                      ber->messages.push_back(m);
+                      dec->addMessage(h, m);
+                      idx++;
+                  }
+                  reqs[ifAppId] = { ber, dec };
+              }
+              BatchExecuteRequest req;
+              req.appId = appId;
+              req.user = "t";
+              req.function = "t";
+              req.type = migration ? BatchExecuteType::MIGRATION
+                                   : BatchExecuteType::FUNCTIONS;
+              for (int i = 0; i < nMessages; i++) {
+                  Message m;
+                  m.id = appId * 1000 + i;
+                  m.appId = appId;
+                  m.appIdx = i;
+                  m.groupIdx = i;
+                  req.messages.push_back(m);
+              }
+              return *sched->makeSchedulingDecision(hostMap, reqs, req);
+          },
+          py::arg("mode"),
+          py::arg("hosts"),
+          py::arg("n_messages"),
+          py::arg("app_id") = 12345,
+          py::arg("migration") = false,
+          py::arg("in_flight") =
+            std::vector<std::pair<int32_t, std::vector<std::string>>>{});
     m.def("decision_cache_size",
           [] { return DecisionCache::get().size(); });
     m.def("decision_cache_clear", [] { DecisionCache::get().clear(); });
