@@ -52,6 +52,8 @@ def parse_args():
     p.add_argument("--a2a-bytes", type=int, default=1024 * 1024,
                    help="alltoall chunk bytes per rank pair")
     p.add_argument("--kv-bytes", type=int, default=4096)
+    p.add_argument("--pp-bytes", type=int, default=64 * 1024 * 1024,
+                   help="rank0<->rank1 ping-pong bytes (config 2)")
     return p.parse_args()
 
 
@@ -118,10 +120,12 @@ def main():
         assert len(hosts) == n, f"only {hosts} registered"
 
         # Gang placement: rank i on worker i (preloaded decision)
+        # ping-pong reuses the allreduce buffers: clamp to their size
+        pp = min(args.pp_bytes, args.bytes) if n >= 2 else 0
         params = (
             f"steps={args.steps};warmup={args.warmup};bytes={args.bytes};"
             f"batch={args.batch};kvbytes={args.kv_bytes};"
-            f"a2abytes={args.a2a_bytes}"
+            f"a2abytes={args.a2a_bytes};ppbytes={pp}"
         )
         ber = _core.batch_exec_factory("bench", "rankstep", 1)
         msgs = ber.messages
@@ -165,6 +169,7 @@ def main():
 
         per_rank_step = [parse_times(r.output_data, "step") for r in results]
         per_rank_ar = [parse_times(r.output_data, "ar") for r in results]
+        rank0_pp = parse_times(results[0].output_data, "pp")
         k = min(len(s) for s in per_rank_step)
         step_ms = [max(s[i] for s in per_rank_step) for i in range(k)]
         ar_ms = [max(s[i] for s in per_rank_ar) for i in range(k)]
@@ -177,6 +182,13 @@ def main():
         ar_mean_s = (sum(ar_ms) / len(ar_ms)) / 1000.0 if ar_ms else 0.0
         algbw = args.bytes / ar_mean_s / 1e9 if ar_mean_s > 0 else 0.0
         busbw = algbw * 2 * (n - 1) / n if n > 1 else 0.0
+        pp_mean_s = (
+            (sum(rank0_pp) / len(rank0_pp)) / 1000.0
+            if rank0_pp and n >= 2
+            else 0.0
+        )
+        # Round trip moves 2 x pp_bytes through the link
+        pp_gbps = 2 * pp / pp_mean_s / 1e9 if pp_mean_s > 0 else 0.0
 
         result = {
             "metric": "MPI_Allreduce GB/s + batch-exec msgs/sec, "
@@ -202,6 +214,8 @@ def main():
                 "allreduce_algbw_gbps": round(algbw, 2),
                 "allreduce_busbw_gbps": round(busbw, 2),
                 "allreduce_ms": round(ar_mean_s * 1000.0, 3),
+                "pingpong_bytes": pp,
+                "pingpong_gbps": round(pp_gbps, 2),
                 "batch_per_host": args.batch,
                 "batch_msgs_per_sec": round(msgs_per_sec, 2),
                 "kv_bytes": args.kv_bytes,

@@ -86,6 +86,7 @@ static int32_t benchRankStep(Message& msg)
     int batchPerHost = (int)(params.count("batch") ? params["batch"] : 0);
     int64_t kvBytes = params.count("kvbytes") ? params["kvbytes"] : 4096;
     int64_t a2aBytes = params.count("a2abytes") ? params["a2abytes"] : 0;
+    int64_t ppBytes = params.count("ppbytes") ? params["ppbytes"] : 0;
 
     int nGpus = 0;
     (void)hipGetDeviceCount(&nGpus);
@@ -132,6 +133,7 @@ static int32_t benchRankStep(Message& msg)
     std::vector<double> stepMs;
     std::vector<double> allreduceMs;
     std::vector<double> batchMs;
+    std::vector<double> pingpongMs;
 
     for (int iter = 0; iter < warmup + steps; iter++) {
         world.barrier(rank);
@@ -154,6 +156,25 @@ static int32_t benchRankStep(Message& msg)
                            (int)a2aBytes,
                            loc);
         }
+
+        // Config 2: rank0<->rank1 ping-pong over xGMI (measured when the
+        // world has at least 2 ranks; uses the allreduce buffers)
+        int64_t tPp0 = getEpochMicros();
+        if (ppBytes > 0 && worldSize >= 2 && rank <= 1) {
+            int peer = 1 - rank;
+            if (rank == 0) {
+                world.send(0, peer, sendBuf, MpiDataType::BYTE,
+                           (int)ppBytes, MpiMessageType::NORMAL, loc);
+                world.recv(peer, 0, recvBuf, MpiDataType::BYTE,
+                           (int)ppBytes, MpiMessageType::NORMAL, loc);
+            } else {
+                world.recv(peer, 1, recvBuf, MpiDataType::BYTE,
+                           (int)ppBytes, MpiMessageType::NORMAL, loc);
+                world.send(1, peer, sendBuf, MpiDataType::BYTE,
+                           (int)ppBytes, MpiMessageType::NORMAL, loc);
+            }
+        }
+        int64_t tPp1 = getEpochMicros();
 
         // Batch-throughput half of the composite step: rank 0 submits a
         // batch of kvtouch functions across all hosts and waits
@@ -188,6 +209,7 @@ static int32_t benchRankStep(Message& msg)
             stepMs.push_back((t1 - t0) / 1000.0);
             allreduceMs.push_back((tAr - t0) / 1000.0);
             batchMs.push_back((t1 - tBatch0) / 1000.0);
+            pingpongMs.push_back((tPp1 - tPp0) / 1000.0);
         }
     }
 
@@ -208,6 +230,10 @@ static int32_t benchRankStep(Message& msg)
     out << ";ar:";
     for (size_t i = 0; i < allreduceMs.size(); i++) {
         out << (i ? "," : "") << allreduceMs[i];
+    }
+    out << ";pp:";
+    for (size_t i = 0; i < pingpongMs.size(); i++) {
+        out << (i ? "," : "") << pingpongMs[i];
     }
     out << ";batch:";
     for (size_t i = 0; i < batchMs.size(); i++) {
