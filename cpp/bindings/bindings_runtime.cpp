@@ -14,6 +14,7 @@
 #include "faabricamd/util.h"
 #include "faabricamd/utilextras.h"
 #include "faabricamd/dirty.h"
+#include <hip/hip_runtime.h>
 
 namespace py = pybind11;
 using namespace faabricamd;
@@ -732,5 +733,48 @@ void initRuntimeBindings(py::module_& m)
             throw FaabricException("executor memory read out of bounds");
         }
         return py::bytes((const char*)base + offset, n);
+    });
+
+    // HBM executor arena (device THREADS flow)
+    m.def("executor_set_device_memory_size", [](size_t n) {
+        py::gil_scoped_release release;
+        ExecutorContext::get().getExecutor()->setDeviceMemorySize(n);
+    });
+    m.def("executor_device_ptr", [] {
+        return (uintptr_t)ExecutorContext::get()
+          .getExecutor()
+          ->getDeviceMemoryView()
+          .first;
+    });
+    m.def("executor_device_write_memory",
+          [](size_t offset, const py::bytes& data) {
+              std::string s = data;
+              auto [base, size] = ExecutorContext::get()
+                                    .getExecutor()
+                                    ->getDeviceMemoryView();
+              if (base == nullptr || offset + s.size() > size) {
+                  throw FaabricException("device arena write out of bounds");
+              }
+              py::gil_scoped_release release;
+              if (hipMemcpy(base + offset, s.data(), s.size(),
+                            hipMemcpyHostToDevice) != hipSuccess) {
+                  throw FaabricException("device arena write failed");
+              }
+          });
+    m.def("executor_device_read_memory", [](size_t offset, size_t n) {
+        auto [base, size] =
+          ExecutorContext::get().getExecutor()->getDeviceMemoryView();
+        if (base == nullptr || offset + n > size) {
+            throw FaabricException("device arena read out of bounds");
+        }
+        std::vector<uint8_t> out(n);
+        {
+            py::gil_scoped_release release;
+            if (hipMemcpy(out.data(), base + offset, n,
+                          hipMemcpyDeviceToHost) != hipSuccess) {
+                throw FaabricException("device arena read failed");
+            }
+        }
+        return py::bytes((const char*)out.data(), out.size());
     });
 }

@@ -178,13 +178,44 @@ __global__ __launch_bounds__(FAM_KERNEL_BLOCK) void compactPageIdxKernel(
 }
 
 // ---------------------------------------------------------------------------
-// Apply XOR page diffs: snap[page] ^= payload[page] for page in pageIdx.
-// One wave per dirty page; payload is sparse (indexed by page).
+// Apply XOR page diffs: snap[page] ^= payload[...] for page in pageIdx.
+// One wave per dirty page. compact=0: payload is sparse (indexed by
+// page, the diff kernel's output); compact=1: payload is slot-compacted
+// (the shippable wire form).
 // ---------------------------------------------------------------------------
 __global__ __launch_bounds__(FAM_KERNEL_BLOCK) void applyXorPagesKernel(
   uint4* __restrict__ snap,
   const u32* __restrict__ pageIdx,
   const uint4* __restrict__ payload,
+  u32 nDirty,
+  int compact)
+{
+    const u32 lane = threadIdx.x & 63;
+    const u32 waveId =
+      (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+    const u32 nWaves = (gridDim.x * blockDim.x) >> 6;
+    const u32 vecsPerPage = FAM_PAGE / 16;
+
+    for (u32 slot = waveId; slot < nDirty; slot += nWaves) {
+        u32 page = pageIdx[slot];
+        u64 dst = (u64)page * vecsPerPage + lane;
+        u64 src = compact ? (u64)slot * vecsPerPage + lane : dst;
+#pragma unroll
+        for (int k = 0; k < 4; k++) {
+            uint4 p = payload[src + k * 64];
+            uint4 s = snap[dst + k * 64];
+            snap[dst + k * 64] =
+              make_uint4(s.x ^ p.x, s.y ^ p.y, s.z ^ p.z, s.w ^ p.w);
+        }
+    }
+}
+
+// Gather sparse per-page payloads into the compact wire form:
+// out[slot] = payload[pageIdx[slot]]
+__global__ __launch_bounds__(FAM_KERNEL_BLOCK) void gatherPagesKernel(
+  const uint4* __restrict__ payload,
+  const u32* __restrict__ pageIdx,
+  uint4* __restrict__ out,
   u32 nDirty)
 {
     const u32 lane = threadIdx.x & 63;
@@ -195,13 +226,11 @@ __global__ __launch_bounds__(FAM_KERNEL_BLOCK) void applyXorPagesKernel(
 
     for (u32 slot = waveId; slot < nDirty; slot += nWaves) {
         u32 page = pageIdx[slot];
-        u64 addr = (u64)page * vecsPerPage + lane;
+        u64 src = (u64)page * vecsPerPage + lane;
+        u64 dst = (u64)slot * vecsPerPage + lane;
 #pragma unroll
         for (int k = 0; k < 4; k++) {
-            uint4 p = payload[addr + k * 64];
-            uint4 s = snap[addr + k * 64];
-            snap[addr + k * 64] =
-              make_uint4(s.x ^ p.x, s.y ^ p.y, s.z ^ p.z, s.w ^ p.w);
+            out[dst + k * 64] = payload[src + k * 64];
         }
     }
 }
@@ -371,11 +400,12 @@ hipError_t famDiffXorPages(const void* snap,
     return hipGetLastError();
 }
 
-hipError_t famApplyXorPages(void* snap,
-                            const uint32_t* pageIdxDev,
-                            const void* payloadDev,
-                            uint32_t nDirty,
-                            hipStream_t stream)
+hipError_t famApplyXorPagesEx(void* snap,
+                              const uint32_t* pageIdxDev,
+                              const void* payloadDev,
+                              uint32_t nDirty,
+                              int compact,
+                              hipStream_t stream)
 {
     uint32_t wavesPerBlock = FAM_KERNEL_BLOCK / 64;
     uint32_t blocks = (nDirty + wavesPerBlock - 1) / wavesPerBlock;
@@ -389,6 +419,39 @@ hipError_t famApplyXorPages(void* snap,
                        (uint4*)snap,
                        pageIdxDev,
                        (const uint4*)payloadDev,
+                       nDirty,
+                       compact);
+    return hipGetLastError();
+}
+
+hipError_t famApplyXorPages(void* snap,
+                            const uint32_t* pageIdxDev,
+                            const void* payloadDev,
+                            uint32_t nDirty,
+                            hipStream_t stream)
+{
+    return famApplyXorPagesEx(snap, pageIdxDev, payloadDev, nDirty, 0,
+                              stream);
+}
+
+hipError_t famGatherPages(const void* payloadDev,
+                          const uint32_t* pageIdxDev,
+                          void* outDev,
+                          uint32_t nDirty,
+                          hipStream_t stream)
+{
+    uint32_t wavesPerBlock = FAM_KERNEL_BLOCK / 64;
+    uint32_t blocks = (nDirty + wavesPerBlock - 1) / wavesPerBlock;
+    uint32_t grid = blocks < FAM_MAX_BLOCKS ? (blocks ? blocks : 1)
+                                            : FAM_MAX_BLOCKS;
+    hipLaunchKernelGGL(gatherPagesKernel,
+                       dim3(grid),
+                       dim3(FAM_KERNEL_BLOCK),
+                       0,
+                       stream,
+                       (const uint4*)payloadDev,
+                       pageIdxDev,
+                       (uint4*)outDev,
                        nDirty);
     return hipGetLastError();
 }

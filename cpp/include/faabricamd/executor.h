@@ -118,10 +118,14 @@ class Executor
                                 std::shared_ptr<BatchExecuteRequest> req);
 
     // Memory / snapshot hooks (reference: Executor.h:50-81). The base
-    // implementation manages a host-visible arena; the GPU executor arena
-    // lives in snapshot.h's device allocations.
+    // implementation manages a host-visible arena AND an optional
+    // HBM-resident arena; once the device arena is set, THREADS fork-join
+    // snapshots/diffs run on the gfx950 kernels.
     virtual std::pair<uint8_t*, size_t> getMemoryView();
     virtual void setMemorySize(size_t newSize);
+    void setDeviceMemorySize(size_t newSize); // rounds up to 4 KiB pages
+    std::pair<uint8_t*, size_t> getDeviceMemoryView();
+    bool hasDeviceArena() const { return deviceArena != nullptr; }
     virtual void restore(const std::string& snapshotKey);
     virtual void reset(Message& msg);
     virtual void flush();
@@ -168,6 +172,10 @@ class Executor
     // Executor-local arena for THREADS snapshots (host path);
     // page-aligned so the segfault dirty tracker can mprotect it
     PageAlignedBuffer dummyMemory;
+
+    // HBM-resident arena (device THREADS path)
+    uint8_t* deviceArena = nullptr;
+    size_t deviceArenaSize = 0;
 };
 
 class ExecutorFactory

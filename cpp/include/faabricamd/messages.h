@@ -341,6 +341,7 @@ struct SnapshotPushRequest
     uint64_t maxSize = 0;  // 2
     std::vector<uint8_t> contents; // 3
     std::vector<SnapshotMergeRegionMsg> mergeRegions; // 4
+    bool onDevice = false; // 5 (HBM-resident snapshot)
     std::string encode() const;
     static SnapshotPushRequest decode(const std::string& buf);
 };

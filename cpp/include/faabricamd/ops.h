@@ -10,7 +10,9 @@
 #pragma once
 
 #include <cstdint>
+#include <map>
 #include <memory>
+#include <mutex>
 #include <string>
 #include <vector>
 
@@ -46,6 +48,17 @@ hipError_t famApplyXorPages(void* snap,
                             const void* payloadDev,
                             uint32_t nDirty,
                             hipStream_t stream);
+hipError_t famApplyXorPagesEx(void* snap,
+                              const uint32_t* pageIdxDev,
+                              const void* payloadDev,
+                              uint32_t nDirty,
+                              int compact,
+                              hipStream_t stream);
+hipError_t famGatherPages(const void* payloadDev,
+                          const uint32_t* pageIdxDev,
+                          void* outDev,
+                          uint32_t nDirty,
+                          hipStream_t stream);
 hipError_t famElementwiseOp(void* inout,
                             const void* in,
                             uint64_t count,
@@ -90,6 +103,19 @@ class DeviceSnapshot
                         const void* payloadDev,
                         uint32_t nDirty);
 
+    // Ship/receive the compact wire form: {page indices, slot-compacted
+    // 4 KiB payloads}
+    void gatherLastDiffToHost(std::vector<uint32_t>& pagesOut,
+                              std::vector<uint8_t>& payloadOut);
+    void applyCompactDiffFromHost(const std::vector<uint32_t>& pages,
+                                  const uint8_t* payload,
+                                  size_t payloadBytes);
+
+    // Queued packed diffs (wire form: [u32 n][u32 pages[n]][n*4KiB]) from
+    // remote/other-thread merges, applied at join time
+    void queuePackedDiff(std::vector<uint8_t> packed);
+    int applyQueuedPackedDiffs();
+
     const uint32_t* diffPageIdx() const { return pageIdx_; }
     const void* diffPayload() const { return payload_; }
     uint32_t lastDiffPages() const { return lastDirty_; }
@@ -111,6 +137,26 @@ class DeviceSnapshot
     uint8_t* payload_ = nullptr;
     uint32_t* bitmap_ = nullptr;
     uint32_t lastDirty_ = 0;
+
+    std::mutex queueMx_;
+    std::vector<std::vector<uint8_t>> queuedPacked_;
+};
+
+// Registry of HBM-resident snapshots (device analog of SnapshotRegistry)
+class DeviceSnapshotRegistry
+{
+  public:
+    static DeviceSnapshotRegistry& get();
+    std::shared_ptr<DeviceSnapshot> getSnapshot(const std::string& key);
+    bool snapshotExists(const std::string& key);
+    void registerSnapshot(const std::string& key,
+                          std::shared_ptr<DeviceSnapshot> snap);
+    void deleteSnapshot(const std::string& key);
+    void clear();
+
+  private:
+    std::mutex mx;
+    std::map<std::string, std::shared_ptr<DeviceSnapshot>> snapshots;
 };
 
 // Elementwise op on device buffers: inout = op(inout, in)

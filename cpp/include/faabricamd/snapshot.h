@@ -49,6 +49,9 @@ enum class SnapshotMergeOperation : int32_t
     Max = 4,
     Min = 5,
     XOR = 6,
+    // Packed page form used by the GPU THREADS flow (this project's own
+    // extension): data = [u32 n][u32 pageIdx[n]][n x 4 KiB XOR payloads]
+    XorPages = 7,
 };
 
 struct SnapshotDiff
@@ -226,6 +229,10 @@ class SnapshotClient : public MessageEndpointClient
   public:
     explicit SnapshotClient(const std::string& host);
     void pushSnapshot(const std::string& key, SnapshotData& data);
+    // HBM-resident snapshot: ships the contents and re-registers them in
+    // the destination's device registry
+    void pushDeviceSnapshot(const std::string& key, const void* hostCopy,
+                            size_t size);
     void pushSnapshotUpdate(const std::string& key,
                             const std::vector<SnapshotDiff>& diffs,
                             const std::vector<SnapshotMergeRegion>& regions);

@@ -1,11 +1,15 @@
 // Executor implementation (reference behavior: src/executor/Executor.cpp
 // :38-212 pool lifecycle, :307-576 threadPoolThread, :580-590 claims).
 #include "faabricamd/executor.h"
+#include "faabricamd/ops.h"
 #include "faabricamd/planner.h"
+
+#include <hip/hip_runtime.h>
 #include "faabricamd/snapshot.h"
 #include "faabricamd/util.h"
 
 #include <algorithm>
+#include <cstring>
 
 namespace faabricamd {
 
@@ -96,6 +100,43 @@ Executor::Executor(Message& msg)
 Executor::~Executor()
 {
     shutdown();
+    if (deviceArena != nullptr) {
+        hipFree(deviceArena);
+    }
+}
+
+void Executor::setDeviceMemorySize(size_t newSize)
+{
+    if (!gpuAvailable()) {
+        throw FaabricException("device arena requires a GPU");
+    }
+    size_t rounded =
+      (newSize + DEVICE_PAGE - 1) / DEVICE_PAGE * DEVICE_PAGE;
+    if (rounded <= deviceArenaSize) {
+        deviceArenaSize = rounded;
+        return;
+    }
+    int dev = gpuDevice >= 0 ? gpuDevice : 0;
+    if (hipSetDevice(dev) != hipSuccess) {
+        throw FaabricException("hipSetDevice failed");
+    }
+    uint8_t* fresh = nullptr;
+    if (hipMalloc(&fresh, rounded) != hipSuccess) {
+        throw FaabricException("device arena alloc failed");
+    }
+    hipMemset(fresh, 0, rounded);
+    if (deviceArena != nullptr) {
+        hipMemcpy(fresh, deviceArena, deviceArenaSize,
+                  hipMemcpyDeviceToDevice);
+        hipFree(deviceArena);
+    }
+    deviceArena = fresh;
+    deviceArenaSize = rounded;
+}
+
+std::pair<uint8_t*, size_t> Executor::getDeviceMemoryView()
+{
+    return { deviceArena, deviceArenaSize };
 }
 
 void Executor::shutdown()
@@ -187,20 +228,44 @@ std::vector<std::pair<int32_t, int32_t>> Executor::executeThreads(
     std::string key = getMainThreadSnapshotKey(
       parentMsg.user, parentMsg.function, parentMsg.appId);
 
-    auto [memBase, memSize] = getMemoryView();
-    auto& reg = SnapshotRegistry::get();
+    bool onDevice = hasDeviceArena();
     std::shared_ptr<SnapshotData> snap;
-    if (reg.snapshotExists(key)) {
-        snap = reg.getSnapshot(key);
-        snap->copyInData(memBase, memSize, 0);
+    std::shared_ptr<DeviceSnapshot> dsnap;
+    if (onDevice) {
+        // GPU fork-join: the snapshot lives in HBM; diffs are XOR page
+        // diffs from the gfx950 kernels (typed merge regions are a host
+        // feature — reject them loudly rather than silently ignoring)
+        if (!mergeRegions.empty()) {
+            throw FaabricException(
+              "typed merge regions unsupported on the device arena "
+              "(XOR page diffing applies)");
+        }
+        auto& dreg = DeviceSnapshotRegistry::get();
+        if (dreg.snapshotExists(key) &&
+            dreg.getSnapshot(key)->size() == deviceArenaSize) {
+            dsnap = dreg.getSnapshot(key);
+        } else {
+            dsnap = std::make_shared<DeviceSnapshot>(
+              deviceArenaSize, gpuDevice >= 0 ? gpuDevice : 0);
+            dreg.registerSnapshot(key, dsnap);
+        }
+        dsnap->captureFromDevice(deviceArena);
     } else {
-        snap = std::make_shared<SnapshotData>(
-          std::vector<uint8_t>(memBase, memBase + memSize));
-        reg.registerSnapshot(key, snap);
-    }
-    snap->clearMergeRegions();
-    for (const auto& r : mergeRegions) {
-        snap->addMergeRegion(r.offset, r.length, r.dataType, r.operation);
+        auto [memBase, memSize] = getMemoryView();
+        auto& reg = SnapshotRegistry::get();
+        if (reg.snapshotExists(key)) {
+            snap = reg.getSnapshot(key);
+            snap->copyInData(memBase, memSize, 0);
+        } else {
+            snap = std::make_shared<SnapshotData>(
+              std::vector<uint8_t>(memBase, memBase + memSize));
+            reg.registerSnapshot(key, snap);
+        }
+        snap->clearMergeRegions();
+        for (const auto& r : mergeRegions) {
+            snap->addMergeRegion(r.offset, r.length, r.dataType,
+                                 r.operation);
+        }
     }
 
     // Thread messages scale-change onto the running app
@@ -221,6 +286,28 @@ std::vector<std::pair<int32_t, int32_t>> Executor::executeThreads(
         throw FaabricException("not enough slots for thread fork");
     }
 
+    // Ship the snapshot to the other hosts in the (updated) decision —
+    // the planner cannot: the snapshot lives here, not in its registry
+    {
+        const std::string& thisHost = getSystemConfig().endpointHost;
+        std::vector<uint8_t> hostCopy;
+        for (const auto& host : decision->uniqueHosts()) {
+            if (host == thisHost) {
+                continue;
+            }
+            if (onDevice) {
+                if (hostCopy.empty()) {
+                    hostCopy.resize(dsnap->size());
+                    dsnap->copyOutHost(hostCopy.data(), dsnap->size());
+                }
+                getSnapshotClient(host)->pushDeviceSnapshot(
+                  key, hostCopy.data(), hostCopy.size());
+            } else {
+                getSnapshotClient(host)->pushSnapshot(key, *snap);
+            }
+        }
+    }
+
     // Await every thread result, then merge the queued diffs and re-map
     // the merged snapshot over this executor's memory
     std::vector<std::pair<int32_t, int32_t>> results;
@@ -230,10 +317,22 @@ std::vector<std::pair<int32_t, int32_t>> Executor::executeThreads(
           m.appId, m.id, conf.globalMessageTimeout);
         results.emplace_back(m.id, result.returnValue);
     }
-    snap->writeQueuedDiffs();
-    auto [memBase2, memSize2] = getMemoryView();
-    snap->mapToMemory(memBase2, memSize2);
-    snap->clearMergeRegions();
+    if (onDevice) {
+        dsnap->applyQueuedPackedDiffs();
+        hipError_t err = hipMemcpy(deviceArena,
+                                   dsnap->data(),
+                                   dsnap->size(),
+                                   hipMemcpyDeviceToDevice);
+        if (err != hipSuccess ||
+            hipStreamSynchronize(nullptr) != hipSuccess) {
+            throw FaabricException("device snapshot merge-back failed");
+        }
+    } else {
+        snap->writeQueuedDiffs();
+        auto [memBase2, memSize2] = getMemoryView();
+        snap->mapToMemory(memBase2, memSize2);
+        snap->clearMergeRegions();
+    }
     return results;
 }
 
@@ -264,14 +363,44 @@ void Executor::setMemorySize(size_t newSize)
 
 void Executor::restore(const std::string& snapshotKey)
 {
-    // Copy the snapshot over this executor's memory view
-    // (reference: src/executor/Executor.cpp:640-654 — mapToMemory)
-    auto snap = SnapshotRegistry::get().getSnapshot(snapshotKey);
-    if (dummyMemory.size() < snap->getSize()) {
-        setMemorySize(snap->getSize());
+    // Copy the snapshot over this executor's memory view (reference:
+    // src/executor/Executor.cpp:640-654 mapToMemory). The snapshot may
+    // arrive from the forking host slightly after the dispatch, so wait
+    // for it (bounded by the bound timeout); device snapshots restore
+    // into the HBM arena via a D2D copy.
+    const auto& conf = getSystemConfig();
+    int64_t deadline = getGlobalClockEpochMillis() + conf.boundTimeout;
+    while (true) {
+        if (DeviceSnapshotRegistry::get().snapshotExists(snapshotKey)) {
+            auto dsnap =
+              DeviceSnapshotRegistry::get().getSnapshot(snapshotKey);
+            if (deviceArenaSize < dsnap->size()) {
+                setDeviceMemorySize(dsnap->size());
+            }
+            hipError_t err = hipMemcpy(deviceArena,
+                                       dsnap->data(),
+                                       dsnap->size(),
+                                       hipMemcpyDeviceToDevice);
+            if (err != hipSuccess ||
+                hipStreamSynchronize(nullptr) != hipSuccess) {
+                throw FaabricException("device snapshot restore failed");
+            }
+            return;
+        }
+        if (SnapshotRegistry::get().snapshotExists(snapshotKey)) {
+            auto snap = SnapshotRegistry::get().getSnapshot(snapshotKey);
+            if (dummyMemory.size() < snap->getSize()) {
+                setMemorySize(snap->getSize());
+            }
+            auto [base, size] = getMemoryView();
+            snap->mapToMemory(base, size);
+            return;
+        }
+        if (getGlobalClockEpochMillis() >= deadline) {
+            throw FaabricException("snapshot not found: " + snapshotKey);
+        }
+        usleep(10 * 1000);
     }
-    auto [base, size] = getMemoryView();
-    snap->mapToMemory(base, size);
 }
 
 void Executor::reset(Message& msg)
@@ -349,7 +478,40 @@ void Executor::handleTaskResult(Message& msg,
     // the shared snapshot and ships the typed diffs to the main host
     // (reference: src/executor/Executor.cpp:509-516, :684 mergeDirtyRegions)
     std::vector<SnapshotDiff> threadDiffs;
-    if (isThreads && isLastInBatch && !req->snapshotKey.empty()) {
+    bool deviceThreads =
+      isThreads && !req->snapshotKey.empty() &&
+      DeviceSnapshotRegistry::get().snapshotExists(req->snapshotKey);
+    if (deviceThreads && isLastInBatch) {
+        try {
+            // GPU fork-join: diff the HBM arena against the device
+            // snapshot, ship the compact XOR page diff
+            auto dsnap =
+              DeviceSnapshotRegistry::get().getSnapshot(req->snapshotKey);
+            uint32_t nd = dsnap->diffXor(deviceArena);
+            if (nd > 0) {
+                std::vector<uint32_t> pages;
+                std::vector<uint8_t> payload;
+                dsnap->gatherLastDiffToHost(pages, payload);
+                std::vector<uint8_t> packed(4 + pages.size() * 4 +
+                                            payload.size());
+                uint32_t n = (uint32_t)pages.size();
+                std::memcpy(packed.data(), &n, 4);
+                std::memcpy(packed.data() + 4,
+                            pages.data(),
+                            pages.size() * 4);
+                std::memcpy(packed.data() + 4 + pages.size() * 4,
+                            payload.data(),
+                            payload.size());
+                threadDiffs.emplace_back(SnapshotDataType::Raw,
+                                         SnapshotMergeOperation::XorPages,
+                                         0,
+                                         packed.data(),
+                                         packed.size());
+            }
+        } catch (const std::exception& e) {
+            FAM_ERROR("device thread diff failed: %s", e.what());
+        }
+    } else if (isThreads && isLastInBatch && !req->snapshotKey.empty()) {
         try {
             auto snap = SnapshotRegistry::get().getSnapshot(req->snapshotKey);
             auto [base, size] = getMemoryView();
@@ -392,9 +554,17 @@ void Executor::handleTaskResult(Message& msg,
         if (mainHost == conf.endpointHost || mainHost.empty()) {
             if (!threadDiffs.empty()) {
                 try {
-                    auto snap =
-                      SnapshotRegistry::get().getSnapshot(req->snapshotKey);
-                    snap->queueDiffs(threadDiffs);
+                    if (deviceThreads) {
+                        auto dsnap = DeviceSnapshotRegistry::get()
+                                       .getSnapshot(req->snapshotKey);
+                        for (auto& d : threadDiffs) {
+                            dsnap->queuePackedDiff(d.dataCopy);
+                        }
+                    } else {
+                        auto snap = SnapshotRegistry::get().getSnapshot(
+                          req->snapshotKey);
+                        snap->queueDiffs(threadDiffs);
+                    }
                 } catch (const std::exception& e) {
                     FAM_ERROR("queueing thread diffs failed: %s", e.what());
                 }

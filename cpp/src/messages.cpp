@@ -888,6 +888,7 @@ std::string SnapshotPushRequest::encode() const
     for (const auto& m : mergeRegions) {
         w.putMessage(4, encodeMergeRegion(m));
     }
+    w.putBool(5, onDevice);
     return w.take();
 }
 
@@ -905,6 +906,7 @@ SnapshotPushRequest SnapshotPushRequest::decode(const std::string& buf)
             case 4:
                 m.mergeRegions.push_back(decodeMergeRegion(r.asString()));
                 break;
+            case 5: m.onDevice = r.asBool(); break;
             default: r.skip(t);
         }
     }

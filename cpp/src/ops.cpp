@@ -167,6 +167,147 @@ void DeviceSnapshot::applyDiffPages(const uint32_t* pageIdxDev,
     OPS_HIP_CHECK(hipStreamSynchronize(stream_));
 }
 
+void DeviceSnapshot::gatherLastDiffToHost(std::vector<uint32_t>& pagesOut,
+                                          std::vector<uint8_t>& payloadOut)
+{
+    OPS_HIP_CHECK(hipSetDevice(device_));
+    pagesOut.resize(lastDirty_);
+    payloadOut.resize((size_t)lastDirty_ * DEVICE_PAGE);
+    if (lastDirty_ == 0) {
+        return;
+    }
+    // Gather on-device, then one contiguous D2H copy
+    uint8_t* compactDev = nullptr;
+    OPS_HIP_CHECK(
+      hipMalloc(&compactDev, (size_t)lastDirty_ * DEVICE_PAGE));
+    OPS_HIP_CHECK(
+      famGatherPages(payload_, pageIdx_, compactDev, lastDirty_, stream_));
+    OPS_HIP_CHECK(hipMemcpyAsync(pagesOut.data(),
+                                 pageIdx_,
+                                 lastDirty_ * sizeof(uint32_t),
+                                 hipMemcpyDeviceToHost,
+                                 stream_));
+    OPS_HIP_CHECK(hipMemcpyAsync(payloadOut.data(),
+                                 compactDev,
+                                 (size_t)lastDirty_ * DEVICE_PAGE,
+                                 hipMemcpyDeviceToHost,
+                                 stream_));
+    OPS_HIP_CHECK(hipStreamSynchronize(stream_));
+    hipFree(compactDev);
+}
+
+void DeviceSnapshot::applyCompactDiffFromHost(
+  const std::vector<uint32_t>& pages,
+  const uint8_t* payload,
+  size_t payloadBytes)
+{
+    if (pages.empty()) {
+        return;
+    }
+    if (payloadBytes != pages.size() * DEVICE_PAGE) {
+        throw FaabricException("compact diff size mismatch");
+    }
+    OPS_HIP_CHECK(hipSetDevice(device_));
+    uint32_t* pagesDev = nullptr;
+    uint8_t* payloadDev = nullptr;
+    OPS_HIP_CHECK(hipMalloc(&pagesDev, pages.size() * sizeof(uint32_t)));
+    OPS_HIP_CHECK(hipMalloc(&payloadDev, payloadBytes));
+    OPS_HIP_CHECK(hipMemcpyAsync(pagesDev,
+                                 pages.data(),
+                                 pages.size() * sizeof(uint32_t),
+                                 hipMemcpyHostToDevice,
+                                 stream_));
+    OPS_HIP_CHECK(hipMemcpyAsync(payloadDev,
+                                 payload,
+                                 payloadBytes,
+                                 hipMemcpyHostToDevice,
+                                 stream_));
+    OPS_HIP_CHECK(famApplyXorPagesEx(snap_,
+                                     pagesDev,
+                                     payloadDev,
+                                     (uint32_t)pages.size(),
+                                     /*compact=*/1,
+                                     stream_));
+    OPS_HIP_CHECK(hipStreamSynchronize(stream_));
+    hipFree(pagesDev);
+    hipFree(payloadDev);
+}
+
+void DeviceSnapshot::queuePackedDiff(std::vector<uint8_t> packed)
+{
+    std::lock_guard<std::mutex> lock(queueMx_);
+    queuedPacked_.push_back(std::move(packed));
+}
+
+int DeviceSnapshot::applyQueuedPackedDiffs()
+{
+    std::vector<std::vector<uint8_t>> toApply;
+    {
+        std::lock_guard<std::mutex> lock(queueMx_);
+        toApply.swap(queuedPacked_);
+    }
+    for (const auto& packed : toApply) {
+        if (packed.size() < 4) {
+            continue;
+        }
+        uint32_t n = 0;
+        std::memcpy(&n, packed.data(), 4);
+        size_t headerBytes = 4 + (size_t)n * 4;
+        if (packed.size() < headerBytes + (size_t)n * DEVICE_PAGE) {
+            throw FaabricException("bad packed diff");
+        }
+        std::vector<uint32_t> pages(n);
+        std::memcpy(pages.data(), packed.data() + 4, (size_t)n * 4);
+        applyCompactDiffFromHost(pages,
+                                 packed.data() + headerBytes,
+                                 (size_t)n * DEVICE_PAGE);
+    }
+    return (int)toApply.size();
+}
+
+DeviceSnapshotRegistry& DeviceSnapshotRegistry::get()
+{
+    static DeviceSnapshotRegistry reg;
+    return reg;
+}
+
+std::shared_ptr<DeviceSnapshot> DeviceSnapshotRegistry::getSnapshot(
+  const std::string& key)
+{
+    std::lock_guard<std::mutex> lock(mx);
+    auto it = snapshots.find(key);
+    if (it == snapshots.end()) {
+        throw FaabricException("device snapshot not found: " + key);
+    }
+    return it->second;
+}
+
+bool DeviceSnapshotRegistry::snapshotExists(const std::string& key)
+{
+    std::lock_guard<std::mutex> lock(mx);
+    return snapshots.count(key) > 0;
+}
+
+void DeviceSnapshotRegistry::registerSnapshot(
+  const std::string& key,
+  std::shared_ptr<DeviceSnapshot> snap)
+{
+    std::lock_guard<std::mutex> lock(mx);
+    snapshots[key] = std::move(snap);
+}
+
+void DeviceSnapshotRegistry::deleteSnapshot(const std::string& key)
+{
+    std::lock_guard<std::mutex> lock(mx);
+    snapshots.erase(key);
+}
+
+void DeviceSnapshotRegistry::clear()
+{
+    std::lock_guard<std::mutex> lock(mx);
+    snapshots.clear();
+}
+
 void deviceElementwiseOp(void* inout,
                          const void* in,
                          uint64_t count,

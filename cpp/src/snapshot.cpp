@@ -4,6 +4,7 @@
 // device (HBM) path dispatches the same operations to gfx950 HIP kernels —
 // see cpp/hip/snapshot_kernels.hip and ops.cpp.
 #include "faabricamd/snapshot.h"
+#include "faabricamd/ops.h"
 #include "faabricamd/planner.h"
 #include "faabricamd/util.h"
 
@@ -487,10 +488,13 @@ std::vector<SnapshotDiff> SnapshotData::diffWithMemory(const uint8_t* updated,
 
 void SnapshotData::applyDiff(const SnapshotDiff& diff)
 {
-    if (diff.offset + diff.size() > maxSize_) {
+    if (diff.operation != SnapshotMergeOperation::XorPages &&
+        diff.offset + diff.size() > maxSize_) {
         throw FaabricException("diff out of snapshot bounds");
     }
-    size_ = std::max(size_, (size_t)diff.offset + diff.size());
+    if (diff.operation != SnapshotMergeOperation::XorPages) {
+        size_ = std::max(size_, (size_t)diff.offset + diff.size());
+    }
     uint8_t* target = data_.data() + diff.offset;
 
     switch (diff.operation) {
@@ -501,6 +505,33 @@ void SnapshotData::applyDiff(const SnapshotDiff& diff)
             const uint8_t* src = diff.getData();
             for (size_t i = 0; i < diff.size(); i++) {
                 target[i] ^= src[i];
+            }
+            break;
+        }
+        case SnapshotMergeOperation::XorPages: {
+            // Packed page form: [u32 n][u32 pages[n]][n x 4 KiB]
+            const uint8_t* src = diff.getData();
+            if (diff.size() < 4) {
+                break;
+            }
+            uint32_t n = 0;
+            std::memcpy(&n, src, 4);
+            size_t header = 4 + (size_t)n * 4;
+            if (diff.size() < header + (size_t)n * SNAPSHOT_PAGE_SIZE) {
+                throw FaabricException("bad packed page diff");
+            }
+            for (uint32_t i = 0; i < n; i++) {
+                uint32_t page = 0;
+                std::memcpy(&page, src + 4 + (size_t)i * 4, 4);
+                size_t off = (size_t)page * SNAPSHOT_PAGE_SIZE;
+                const uint8_t* payload =
+                  src + header + (size_t)i * SNAPSHOT_PAGE_SIZE;
+                size_t len =
+                  std::min(SNAPSHOT_PAGE_SIZE, maxSize_ - off);
+                size_ = std::max(size_, off + len);
+                for (size_t b = 0; b < len; b++) {
+                    data_[off + b] ^= payload[b];
+                }
             }
             break;
         }
@@ -642,6 +673,23 @@ std::string SnapshotServer::doSyncRecv(uint8_t code, const std::string& body)
     switch ((SnapshotCalls)code) {
         case SnapshotCalls::PushSnapshot: {
             auto req = SnapshotPushRequest::decode(body);
+            if (req.onDevice) {
+                // HBM-resident snapshot: land it in this host's GPU
+                if (!gpuAvailable()) {
+                    throw FaabricException(
+                      "device snapshot pushed to GPU-less host");
+                }
+                size_t rounded =
+                  (req.contents.size() + DEVICE_PAGE - 1) / DEVICE_PAGE *
+                  DEVICE_PAGE;
+                auto dsnap =
+                  std::make_shared<DeviceSnapshot>(rounded, 0);
+                dsnap->copyInHost(req.contents.data(),
+                                  req.contents.size());
+                DeviceSnapshotRegistry::get().registerSnapshot(req.key,
+                                                               dsnap);
+                return {};
+            }
             auto snap = std::make_shared<SnapshotData>(
               req.contents, std::max<size_t>(req.maxSize,
                                              req.contents.size()));
@@ -675,12 +723,26 @@ std::string SnapshotServer::doSyncRecv(uint8_t code, const std::string& body)
             // Queue the diffs onto the main-thread snapshot, then forward
             // the thread's result (reference: SnapshotServer.cpp:28-62)
             if (!req.key.empty() && !req.diffs.empty()) {
-                auto snap = SnapshotRegistry::get().getSnapshot(req.key);
-                std::vector<SnapshotDiff> diffs;
-                for (const auto& d : req.diffs) {
-                    diffs.push_back(SnapshotDiff::fromMsg(d));
+                bool isDevice =
+                  DeviceSnapshotRegistry::get().snapshotExists(req.key);
+                if (isDevice) {
+                    auto dsnap =
+                      DeviceSnapshotRegistry::get().getSnapshot(req.key);
+                    for (const auto& d : req.diffs) {
+                        if ((SnapshotMergeOperation)d.mergeOp ==
+                            SnapshotMergeOperation::XorPages) {
+                            dsnap->queuePackedDiff(d.data);
+                        }
+                    }
+                } else {
+                    auto snap =
+                      SnapshotRegistry::get().getSnapshot(req.key);
+                    std::vector<SnapshotDiff> diffs;
+                    for (const auto& d : req.diffs) {
+                        diffs.push_back(SnapshotDiff::fromMsg(d));
+                    }
+                    snap->queueDiffs(diffs);
                 }
-                snap->queueDiffs(diffs);
             }
             auto msg = std::make_shared<Message>();
             msg->appId = req.appId;
@@ -747,6 +809,24 @@ void SnapshotClient::pushSnapshot(const std::string& key, SnapshotData& data)
         m.mergeOp = (int32_t)r.operation;
         req.mergeRegions.push_back(m);
     }
+    syncSend((uint8_t)SnapshotCalls::PushSnapshot, req.encode());
+}
+
+void SnapshotClient::pushDeviceSnapshot(const std::string& key,
+                                        const void* hostCopy,
+                                        size_t size)
+{
+    if (isMockMode()) {
+        std::lock_guard<std::mutex> lock(snapMockMx);
+        mockedPushes.emplace_back(getHost(), key);
+        return;
+    }
+    SnapshotPushRequest req;
+    req.key = key;
+    req.maxSize = size;
+    req.onDevice = true;
+    req.contents.assign((const uint8_t*)hostCopy,
+                        (const uint8_t*)hostCopy + size);
     syncSend((uint8_t)SnapshotCalls::PushSnapshot, req.encode());
 }
 

@@ -220,3 +220,50 @@ def test_state_kv_in_hbm(runtime):
     )  # byte sum: t += kv
     torch.cuda.synchronize()
     assert int(t[0]) == 7 and int(t[5000]) == 7
+
+
+def _gpu_thread_body(msg):
+    # Flip one device page per thread: thread i XORs page (i) with 0x0F
+    idx = msg.group_idx
+    page = bytes([0x0F]) * 4096
+    cur = _core.executor_device_read_memory((idx - 1) * 4096, 4096)
+    mixed = bytes(a ^ b for a, b in zip(cur, page))
+    _core.executor_device_write_memory((idx - 1) * 4096, mixed)
+    return 0
+
+
+def _gpu_fork_parent(msg):
+    n_pages = 64
+    _core.executor_set_device_memory_size(n_pages * 4096)
+    base = bytes([0x30]) * (n_pages * 4096)
+    _core.executor_device_write_memory(0, base)
+
+    results = _core.execute_threads("gputhreads", "body", 3)
+    if len(results) != 3 or any(rv != 0 for _, rv in results):
+        msg.output_data = f"thread failures: {results}"
+        return 1
+
+    data = _core.executor_device_read_memory(0, n_pages * 4096)
+    for p in range(3):
+        if data[p * 4096] != 0x3F:  # 0x30 ^ 0x0F
+            msg.output_data = f"page {p} wrong: {data[p * 4096]:#x}"
+            return 2
+    if data[3 * 4096] != 0x30:
+        msg.output_data = "untouched page modified"
+        return 3
+    msg.output_data = "gpu fork-join ok"
+    return 0
+
+
+@requires_gpu
+def test_gpu_threads_fork_join(runtime):
+    """THREADS fork-join over an HBM-resident snapshot: capture, restore,
+    XOR page diff on the gfx950 kernels, packed merge-back."""
+    _core.register_function("gputhreads", "body", _gpu_thread_body)
+    _core.register_function("gputhreads", "parent", _gpu_fork_parent)
+    ber = _core.batch_exec_factory("gputhreads", "parent", 1)
+    decision = _core.call_functions(ber)
+    assert decision.app_id == ber.app_id
+    results = wait_for_batch(ber.app_id, 1, timeout_ms=60_000)
+    assert results[0].return_value == 0, results[0].output_data
+    assert results[0].output_data == "gpu fork-join ok"
