@@ -44,6 +44,45 @@ def _mpi_allreduce_fn(msg):
     return 0
 
 
+def _dist_thread_body(msg):
+    import struct
+
+    from faabric_amd import _core
+
+    idx = msg.group_idx
+    raw = _core.executor_read_memory((idx - 1) * 4, 4)
+    (v,) = struct.unpack("<i", raw)
+    _core.executor_write_memory((idx - 1) * 4, struct.pack("<i", v + idx))
+    return 0
+
+
+def _dist_fork_parent(msg):
+    import struct
+
+    from faabric_amd import _core
+
+    _core.executor_set_memory_size(8192)
+    _core.executor_write_memory(0, struct.pack("<8i", *([100] * 8)))
+    results = _core.execute_threads(
+        "dist",
+        "threadbody",
+        3,
+        merge_regions=[
+            (0, 32, int(_core.SnapshotDataType.Int.value),
+             int(_core.SnapshotMergeOperation.Sum.value)),
+        ],
+    )
+    if any(rv != 0 for _, rv in results):
+        msg.output_data = f"thread failures: {results}"
+        return 1
+    vals = struct.unpack("<8i", _core.executor_read_memory(0, 32))
+    if vals[:3] != (101, 102, 103):
+        msg.output_data = f"bad merge: {vals}"
+        return 2
+    msg.output_data = "dist fork ok"
+    return 0
+
+
 def _worker_main(port_offset, stop_event, ready_event):
     sys.path.insert(0, REPO_ROOT)
     from faabric_amd import _core
@@ -56,6 +95,8 @@ def _worker_main(port_offset, stop_event, ready_event):
     _core.register_native_noop("dist", "noop")
     _core.register_native_sleep("dist", "sleep", 200)
     _core.register_function("dist", "mpi_allreduce", _mpi_allreduce_fn)
+    _core.register_function("dist", "threadbody", _dist_thread_body)
+    _core.register_function("dist", "forkparent", _dist_fork_parent)
     ready_event.set()
     stop_event.wait(120)
     rt.stop()
@@ -171,6 +212,24 @@ def test_mpi_world_across_hosts(cluster):
         assert r.return_value == 0, r.output_data
     hosts_used = {r.executed_host for r in results}
     assert hosts_used == {"127.0.0.1@1000", "127.0.0.1@2000"}
+
+
+def test_threads_fork_across_hosts(cluster):
+    """THREADS fork-join where threads land on BOTH worker processes:
+    snapshot ships from the forking worker, diffs merge back over the
+    snapshot channel (multi-host reference flow)."""
+    from faabric_amd import _core
+    from faabric_amd.runtime import wait_for_batch
+
+    # Parent takes 1 slot on its host; 3 threads need the remaining slot
+    # there plus 2 on the other worker
+    ber = _core.batch_exec_factory("dist", "forkparent", 1)
+    decision = _core.call_functions(ber)
+    assert decision.app_id == ber.app_id
+    results = wait_for_batch(ber.app_id, 1, timeout_ms=60_000)
+    parent = [r for r in results if r.output_data][0]
+    assert parent.return_value == 0, parent.output_data
+    assert parent.output_data == "dist fork ok"
 
 
 def test_slots_freed_after_batch(cluster):
