@@ -311,6 +311,46 @@ void initRuntimeBindings(py::module_& m)
           py::arg("recv_idx"),
           py::arg("ordered") = false,
           py::arg("timeout_ms") = DEFAULT_QUEUE_TIMEOUT_MS);
+    m.def("ptp_send_device",
+          [](int32_t appId,
+             int32_t groupId,
+             int32_t sendIdx,
+             int32_t recvIdx,
+             uintptr_t devPtr,
+             size_t size,
+             bool ordered) {
+              py::gil_scoped_release release;
+              getPointToPointBroker().sendMessageDevice(
+                appId, groupId, sendIdx, recvIdx, (const void*)devPtr,
+                size, ordered);
+          },
+          py::arg("app_id"),
+          py::arg("group_id"),
+          py::arg("send_idx"),
+          py::arg("recv_idx"),
+          py::arg("dev_ptr"),
+          py::arg("size"),
+          py::arg("ordered") = false);
+    m.def("ptp_recv_device",
+          [](int32_t groupId,
+             int32_t sendIdx,
+             int32_t recvIdx,
+             uintptr_t devPtr,
+             size_t capacity,
+             bool ordered,
+             int timeoutMs) {
+              py::gil_scoped_release release;
+              return getPointToPointBroker().recvMessageDevice(
+                groupId, sendIdx, recvIdx, (void*)devPtr, capacity,
+                ordered, timeoutMs);
+          },
+          py::arg("group_id"),
+          py::arg("send_idx"),
+          py::arg("recv_idx"),
+          py::arg("dev_ptr"),
+          py::arg("capacity"),
+          py::arg("ordered") = false,
+          py::arg("timeout_ms") = DEFAULT_QUEUE_TIMEOUT_MS);
     m.def("ptp_group_barrier", [](int32_t groupId, int32_t groupIdx) {
         py::gil_scoped_release release;
         PointToPointGroup::getOrAwaitGroup(groupId)->barrier(groupIdx);

@@ -125,6 +125,25 @@ class PointToPointBroker
                                      bool mustOrderMsgs = false,
                                      int timeoutMs = DEFAULT_QUEUE_TIMEOUT_MS);
 
+    // Device (HBM) payloads: same-process receivers get a D2D copy on the
+    // broker's side stream (never staged through the host); cross-host
+    // receivers are staged D2H and ride the normal RPC plane.
+    void sendMessageDevice(int32_t appId,
+                           int32_t groupId,
+                           int32_t sendIdx,
+                           int32_t recvIdx,
+                           const void* devPtr,
+                           size_t size,
+                           bool mustOrderMsgs = false);
+    // Receive into a device buffer (H2D-copies host-plane payloads)
+    size_t recvMessageDevice(int32_t groupId,
+                             int32_t sendIdx,
+                             int32_t recvIdx,
+                             void* devPtr,
+                             size_t capacity,
+                             bool mustOrderMsgs = false,
+                             int timeoutMs = DEFAULT_QUEUE_TIMEOUT_MS);
+
     // Deliver a message that arrived over the network
     void deliverRemoteMessage(const PointToPointMessage& msg, uint32_t seq);
 
@@ -144,14 +163,33 @@ class PointToPointBroker
     void postMigrationHook(int32_t groupId, int32_t groupIdx);
 
   private:
+    struct PtpPayload
+    {
+        std::vector<uint8_t> host;
+        void* dev = nullptr; // HBM staging copy (same-process delivery)
+        size_t devSize = 0;
+    };
+
     struct Channel
     {
         std::mutex mx;
         std::condition_variable cv;
-        std::map<uint32_t, std::vector<uint8_t>> bufferedMsgs; // seq → data
-        std::deque<std::vector<uint8_t>> unorderedMsgs;
+        std::map<uint32_t, PtpPayload> bufferedMsgs; // seq → payload
+        std::deque<PtpPayload> unorderedMsgs;
         uint32_t nextRecvSeq = 0;
     };
+
+    void deliverPayload(int32_t groupId,
+                        int32_t sendIdx,
+                        int32_t recvIdx,
+                        PtpPayload payload,
+                        uint32_t seq);
+    PtpPayload recvPayload(int32_t groupId,
+                           int32_t sendIdx,
+                           int32_t recvIdx,
+                           bool mustOrderMsgs,
+                           int timeoutMs);
+    void* sideStream(); // lazily-created HIP stream for D2D staging
 
     Channel& getChannel(int32_t groupId, int32_t sendIdx, int32_t recvIdx);
 
@@ -170,6 +208,8 @@ class PointToPointBroker
 
     std::mutex clientsMx;
     std::map<std::string, std::shared_ptr<MessageEndpointClient>> clients;
+    std::mutex streamMx;
+    void* sideStream_ = nullptr;
 
     std::shared_ptr<MessageEndpointClient> getClient(const std::string& host);
 

@@ -5,6 +5,8 @@
 #include "faabricamd/ptp.h"
 #include "faabricamd/util.h"
 
+#include <hip/hip_runtime.h>
+
 #include <algorithm>
 
 namespace faabricamd {
@@ -423,26 +425,38 @@ void PointToPointBroker::sendMessage(int32_t appId,
     }
 }
 
-void PointToPointBroker::deliverRemoteMessage(const PointToPointMessage& msg,
-                                              uint32_t seq)
+void PointToPointBroker::deliverPayload(int32_t groupId,
+                                        int32_t sendIdx,
+                                        int32_t recvIdx,
+                                        PtpPayload payload,
+                                        uint32_t seq)
 {
-    Channel& ch = getChannel(msg.groupId, msg.sendIdx, msg.recvIdx);
+    Channel& ch = getChannel(groupId, sendIdx, recvIdx);
     {
         std::lock_guard<std::mutex> lock(ch.mx);
         if (seq == NO_SEQ) {
-            ch.unorderedMsgs.push_back(msg.data);
+            ch.unorderedMsgs.push_back(std::move(payload));
         } else {
-            ch.bufferedMsgs[seq] = msg.data;
+            ch.bufferedMsgs[seq] = std::move(payload);
         }
     }
     ch.cv.notify_all();
 }
 
-std::vector<uint8_t> PointToPointBroker::recvMessage(int32_t groupId,
-                                                     int32_t sendIdx,
-                                                     int32_t recvIdx,
-                                                     bool mustOrderMsgs,
-                                                     int timeoutMs)
+void PointToPointBroker::deliverRemoteMessage(const PointToPointMessage& msg,
+                                              uint32_t seq)
+{
+    PtpPayload p;
+    p.host = msg.data;
+    deliverPayload(msg.groupId, msg.sendIdx, msg.recvIdx, std::move(p), seq);
+}
+
+PointToPointBroker::PtpPayload PointToPointBroker::recvPayload(
+  int32_t groupId,
+  int32_t sendIdx,
+  int32_t recvIdx,
+  bool mustOrderMsgs,
+  int timeoutMs)
 {
     Channel& ch = getChannel(groupId, sendIdx, recvIdx);
     std::unique_lock<std::mutex> lock(ch.mx);
@@ -462,20 +476,147 @@ std::vector<uint8_t> PointToPointBroker::recvMessage(int32_t groupId,
 
     if (mustOrderMsgs) {
         auto it = ch.bufferedMsgs.find(ch.nextRecvSeq);
-        std::vector<uint8_t> out = std::move(it->second);
+        PtpPayload out = std::move(it->second);
         ch.bufferedMsgs.erase(it);
         ch.nextRecvSeq++;
         return out;
     }
     if (!ch.unorderedMsgs.empty()) {
-        std::vector<uint8_t> out = std::move(ch.unorderedMsgs.front());
+        PtpPayload out = std::move(ch.unorderedMsgs.front());
         ch.unorderedMsgs.pop_front();
         return out;
     }
     auto it = ch.bufferedMsgs.begin();
-    std::vector<uint8_t> out = std::move(it->second);
+    PtpPayload out = std::move(it->second);
     ch.bufferedMsgs.erase(it);
     return out;
+}
+
+std::vector<uint8_t> PointToPointBroker::recvMessage(int32_t groupId,
+                                                     int32_t sendIdx,
+                                                     int32_t recvIdx,
+                                                     bool mustOrderMsgs,
+                                                     int timeoutMs)
+{
+    PtpPayload p =
+      recvPayload(groupId, sendIdx, recvIdx, mustOrderMsgs, timeoutMs);
+    if (p.dev != nullptr) {
+        // Device-staged message consumed through the host API
+        std::vector<uint8_t> out(p.devSize);
+        hipMemcpy(out.data(), p.dev, p.devSize, hipMemcpyDeviceToHost);
+        hipFree(p.dev);
+        return out;
+    }
+    return std::move(p.host);
+}
+
+void* PointToPointBroker::sideStream()
+{
+    std::lock_guard<std::mutex> lock(streamMx);
+    if (sideStream_ == nullptr) {
+        hipStream_t s = nullptr;
+        if (hipStreamCreateWithFlags(&s, hipStreamNonBlocking) !=
+            hipSuccess) {
+            throw FaabricException("ptp side stream creation failed");
+        }
+        sideStream_ = (void*)s;
+    }
+    return sideStream_;
+}
+
+void PointToPointBroker::sendMessageDevice(int32_t appId,
+                                           int32_t groupId,
+                                           int32_t sendIdx,
+                                           int32_t recvIdx,
+                                           const void* devPtr,
+                                           size_t size,
+                                           bool mustOrderMsgs)
+{
+    uint32_t seq = NO_SEQ;
+    if (mustOrderMsgs) {
+        std::lock_guard<std::mutex> lock(sendSeqMx);
+        seq = sendSeqs[chanKey(groupId, sendIdx, recvIdx)]++;
+    }
+    int32_t hostIdx = recvIdx % 16384;
+    std::string host = getHostForReceiver(groupId, hostIdx);
+    const std::string& thisHost = getSystemConfig().endpointHost;
+
+    if (host == thisHost) {
+        // Same-process delivery: stage a D2D copy on the side stream so
+        // the sender's buffer is immediately reusable
+        hipStream_t s = (hipStream_t)sideStream();
+        void* staging = nullptr;
+        if (hipMalloc(&staging, size) != hipSuccess) {
+            throw FaabricException("ptp device staging alloc failed");
+        }
+        if (hipMemcpyAsync(staging, devPtr, size,
+                           hipMemcpyDeviceToDevice, s) != hipSuccess ||
+            hipStreamSynchronize(s) != hipSuccess) {
+            hipFree(staging);
+            throw FaabricException("ptp device staging copy failed");
+        }
+        PtpPayload p;
+        p.dev = staging;
+        p.devSize = size;
+        deliverPayload(groupId, sendIdx, recvIdx, std::move(p), seq);
+        return;
+    }
+
+    // Cross-host: stage D2H and ride the RPC plane
+    PointToPointMessage msg;
+    msg.appId = appId;
+    msg.groupId = groupId;
+    msg.sendIdx = sendIdx;
+    msg.recvIdx = recvIdx;
+    msg.data.resize(size);
+    hipStream_t s = (hipStream_t)sideStream();
+    if (hipMemcpyAsync(msg.data.data(), devPtr, size,
+                       hipMemcpyDeviceToHost, s) != hipSuccess ||
+        hipStreamSynchronize(s) != hipSuccess) {
+        throw FaabricException("ptp device D2H staging failed");
+    }
+    auto cli = getClient(host);
+    std::string body = msg.encode();
+    cli->asyncSendSeq((uint8_t)PointToPointCall::MESSAGE,
+                      body.data(),
+                      body.size(),
+                      seq);
+}
+
+size_t PointToPointBroker::recvMessageDevice(int32_t groupId,
+                                             int32_t sendIdx,
+                                             int32_t recvIdx,
+                                             void* devPtr,
+                                             size_t capacity,
+                                             bool mustOrderMsgs,
+                                             int timeoutMs)
+{
+    PtpPayload p =
+      recvPayload(groupId, sendIdx, recvIdx, mustOrderMsgs, timeoutMs);
+    hipStream_t s = (hipStream_t)sideStream();
+    if (p.dev != nullptr) {
+        if (p.devSize > capacity) {
+            hipFree(p.dev);
+            throw FaabricException("ptp device recv buffer too small");
+        }
+        if (hipMemcpyAsync(devPtr, p.dev, p.devSize,
+                           hipMemcpyDeviceToDevice, s) != hipSuccess ||
+            hipStreamSynchronize(s) != hipSuccess) {
+            hipFree(p.dev);
+            throw FaabricException("ptp device recv copy failed");
+        }
+        hipFree(p.dev);
+        return p.devSize;
+    }
+    if (p.host.size() > capacity) {
+        throw FaabricException("ptp device recv buffer too small");
+    }
+    if (hipMemcpyAsync(devPtr, p.host.data(), p.host.size(),
+                       hipMemcpyHostToDevice, s) != hipSuccess ||
+        hipStreamSynchronize(s) != hipSuccess) {
+        throw FaabricException("ptp device H2D failed");
+    }
+    return p.host.size();
 }
 
 void PointToPointBroker::clearGroup(int32_t groupId)

@@ -267,3 +267,42 @@ def test_gpu_threads_fork_join(runtime):
     results = wait_for_batch(ber.app_id, 1, timeout_ms=60_000)
     assert results[0].return_value == 0, results[0].output_data
     assert results[0].output_data == "gpu fork-join ok"
+
+
+@requires_gpu
+def test_ptp_device_payloads(runtime):
+    """PTP broker device payloads: same-process delivery via D2D staging
+    on the broker's side stream, ordered."""
+    decision = _core.SchedulingDecision()
+    decision.app_id = 777000
+    decision.group_id = 777001
+    decision.hosts = [runtime.identity, runtime.identity]
+    decision.message_ids = [1, 2]
+    decision.app_idxs = [0, 1]
+    decision.group_idxs = [0, 1]
+    decision.mpi_ports = [0, 0]
+    decision.n_functions = 2
+    _core.ptp_setup_local_mappings(decision)
+
+    n = 1 << 16
+    src = torch.arange(n, dtype=torch.float32, device="cuda")
+    dst = torch.zeros(n, dtype=torch.float32, device="cuda")
+    torch.cuda.synchronize()
+    _core.ptp_send_device(
+        777000, 777001, 0, 1, src.data_ptr(), n * 4, True
+    )
+    src.fill_(0)  # sender buffer reusable immediately (staged copy)
+    torch.cuda.synchronize()
+    got = _core.ptp_recv_device(777001, 0, 1, dst.data_ptr(), n * 4, True)
+    assert got == n * 4
+    torch.cuda.synchronize()
+    assert torch.equal(dst, torch.arange(n, dtype=torch.float32,
+                                         device="cuda"))
+
+    # Host-sent payload received into a device buffer
+    _core.ptp_send(777000, 777001, 0, 1, b"\x42" * 1024, False)
+    buf = torch.zeros(1024, dtype=torch.uint8, device="cuda")
+    got = _core.ptp_recv_device(777001, 0, 1, buf.data_ptr(), 1024)
+    assert got == 1024
+    torch.cuda.synchronize()
+    assert int(buf.sum()) == 0x42 * 1024
