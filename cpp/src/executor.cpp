@@ -1,6 +1,7 @@
 // Executor implementation (reference behavior: src/executor/Executor.cpp
 // :38-212 pool lifecycle, :307-576 threadPoolThread, :580-590 claims).
 #include "faabricamd/executor.h"
+#include "faabricamd/mpi.h"
 #include "faabricamd/ops.h"
 #include "faabricamd/planner.h"
 
@@ -442,6 +443,18 @@ void Executor::threadPoolThread(int poolIdx)
         ExecutorContext::set(this, task.req, task.msgIdx);
         try {
             returnValue = executeTask(poolIdx, task.msgIdx, task.req);
+            // Exec-graph detail: per-rank MPI message counters ride the
+            // result (reference: mpi/MpiWorld.h:13-18 + Executor exec
+            // graph details)
+            if (msg.isMpi && msg.recordExecGraph &&
+                MpiWorldRegistry::get().worldExists(msg.mpiWorldId)) {
+                auto counts = MpiWorldRegistry::get()
+                                .getWorld(msg.mpiWorldId)
+                                .getMsgCountDetails(msg.mpiRank);
+                for (const auto& [k, v] : counts) {
+                    msg.intExecGraphDetails[k] = v;
+                }
+            }
         } catch (const FunctionMigratedException&) {
             FAM_DEBUG("task %d migrated", msg.id);
             returnValue = MIGRATED_FUNCTION_RETURN_VALUE;

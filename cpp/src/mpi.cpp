@@ -177,6 +177,7 @@ void MpiWorld::create(Message& call, int newId, int newSize)
         m.mpiRank = i;
         m.mpiWorldSize = size;
         m.inputData = call.inputData;
+        m.recordExecGraph = call.recordExecGraph;
         req->messages.push_back(std::move(m));
     }
 

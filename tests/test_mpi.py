@@ -140,3 +140,27 @@ def test_cpp_mpi_examples(runtime):
         assert len(results) == WORLD_SIZE
         for r in results:
             assert r.return_value == 0, (func, r.output_data)
+
+
+def test_mpi_exec_graph_details(runtime):
+    """Per-rank MPI message counters ride the exec graph (reference:
+    tests/test/mpi/test_mpi_exec_graph.cpp)."""
+
+    def traced(msg):
+        world_id, rank, size = _core.mpi_init()
+        _core.mpi_send_bytes(rank, (rank + 1) % size, b"x")
+        _core.mpi_recv_bytes((rank - 1 + size) % size, rank, 1)
+        return 0
+
+    _core.register_function("mpi", "traced", traced)
+    ber = _core.batch_exec_factory("mpi", "traced", 1)
+    msgs = ber.messages
+    msgs[0].is_mpi = True
+    msgs[0].mpi_world_size = WORLD_SIZE
+    msgs[0].record_exec_graph = True
+    ber.messages = msgs
+    _core.call_functions(ber)
+    results = wait_for_batch(ber.app_id, WORLD_SIZE, 60_000)
+    rank0 = [r for r in results if r.mpi_rank == 0][0]
+    details = rank0.int_exec_graph_details
+    assert any(k.startswith("mpi-msgcount-torank-") for k in details), details
