@@ -664,7 +664,8 @@ void initRuntimeBindings(py::module_& m)
              int nThreads,
              const std::vector<std::tuple<uint32_t, size_t, int, int>>&
                regions,
-             const py::bytes& inputData) {
+             const py::bytes& inputData,
+             bool elastic) {
               Executor* exec = ExecutorContext::get().getExecutor();
               std::string input = inputData;
               std::vector<std::pair<int32_t, int32_t>> results;
@@ -672,6 +673,7 @@ void initRuntimeBindings(py::module_& m)
                   py::gil_scoped_release release;
                   auto req = std::make_shared<BatchExecuteRequest>(
                     batchExecFactory(user, function, nThreads));
+                  req->elasticScaleHint = elastic;
                   for (auto& m2 : req->messages) {
                       m2.inputData.assign(input.begin(), input.end());
                   }
@@ -695,7 +697,8 @@ void initRuntimeBindings(py::module_& m)
           py::arg("n_threads"),
           py::arg("merge_regions") =
             std::vector<std::tuple<uint32_t, size_t, int, int>>{},
-          py::arg("input_data") = py::bytes(""));
+          py::arg("input_data") = py::bytes(""),
+          py::arg("elastic") = false);
 
     // ---------------- chaining + exec graph --------------------------------
     m.def("chain_function",

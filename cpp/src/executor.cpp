@@ -223,6 +223,8 @@ std::vector<std::pair<int32_t, int32_t>> Executor::executeThreads(
   std::shared_ptr<BatchExecuteRequest> req,
   const std::vector<SnapshotMergeRegion>& mergeRegions)
 {
+    // (elasticScaleHint on req lets the planner grow the fork to every
+    // free slot on this host)
     // Fork-join over a shared memory snapshot
     // (reference: SURVEY §3.4; src/executor/Executor.cpp executeThreads)
     Message& parentMsg = ExecutorContext::get().getMsg();
@@ -310,13 +312,26 @@ std::vector<std::pair<int32_t, int32_t>> Executor::executeThreads(
     }
 
     // Await every thread result, then merge the queued diffs and re-map
-    // the merged snapshot over this executor's memory
+    // the merged snapshot over this executor's memory. The decision is
+    // the source of truth for WHICH messages to wait on: an elastic
+    // scale-up may have grown the fork beyond what was requested
+    // (reference: src/planner/Planner.cpp:832-891)
     std::vector<std::pair<int32_t, int32_t>> results;
     const auto& conf = getSystemConfig();
+    std::set<int32_t> waitIds;
     for (const auto& m : req->messages) {
+        waitIds.insert(m.id);
+    }
+    for (int i = 0; i < decision->nFunctions; i++) {
+        int32_t id = decision->messageIds[i];
+        if (id != 0 && id != parentMsg.id) {
+            waitIds.insert(id);
+        }
+    }
+    for (int32_t id : waitIds) {
         Message result = getPlannerClient().getMessageResult(
-          m.appId, m.id, conf.globalMessageTimeout);
-        results.emplace_back(m.id, result.returnValue);
+          parentMsg.appId, id, conf.globalMessageTimeout);
+        results.emplace_back(id, result.returnValue);
     }
     if (onDevice) {
         dsnap->applyQueuedPackedDiffs();

@@ -83,3 +83,41 @@ def test_threads_repeat_fork(runtime):
         _core.call_functions(ber)
         results = wait_for_batch(ber.app_id, 1, timeout_ms=30_000)
         assert results[0].return_value == 0, results[0].output_data
+
+
+def elastic_thread(msg):
+    idx = msg.group_idx
+    raw = _core.executor_read_memory(0, 4)
+    _core.executor_write_memory(
+        64 + idx * 4, struct.pack("<i", idx * 100)
+    )
+    return 0
+
+
+def elastic_parent(msg):
+    _core.executor_set_memory_size(4096)
+    _core.executor_write_memory(0, struct.pack("<i", 7))
+    results = _core.execute_threads(
+        "threads", "elastic_thread", 1, elastic=True
+    )
+    if any(rv != 0 for _, rv in results):
+        msg.output_data = f"failures: {results}"
+        return 1
+    msg.output_data = f"nthreads={len(results)}"
+    return 0
+
+
+def test_elastic_scale_up(runtime):
+    """elasticScaleHint grows a 1-thread fork to every free slot on the
+    main host (reference: src/planner/Planner.cpp:832-891)."""
+    _core.register_function("threads", "elastic_thread", elastic_thread)
+    _core.register_function("threads", "elastic_parent", elastic_parent)
+    ber = _core.batch_exec_factory("threads", "elastic_parent", 1)
+    _core.call_functions(ber)
+    results = wait_for_batch(ber.app_id, 2, timeout_ms=30_000)
+    parent = [r for r in results if r.output_data.startswith("nthreads")][0]
+    assert parent.return_value == 0
+    n = int(parent.output_data.split("=")[1])
+    # Parent occupies 1 of SLOTS slots; the fork should expand beyond the
+    # single requested thread
+    assert n > 1, parent.output_data
