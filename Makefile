@@ -41,8 +41,36 @@ $(TARGET): $(OBJS)
 
 # C++ example binaries (reference examples/ + planner_server parity)
 CORE_OBJS = $(filter-out $(BUILD)/bindings/%,$(OBJS))
-EXAMPLE_BINS = $(BUILD)/check $(BUILD)/planner_server $(BUILD)/server
+EXAMPLE_BINS = $(BUILD)/check $(BUILD)/planner_server $(BUILD)/server \
+               $(BUILD)/selftest
 examples: $(TARGET) $(EXAMPLE_BINS)
+
+# Sanitizer sweeps (reference CI parity: Address/Thread sanitised suites,
+# .github/workflows/tests.yml). Host-only compile: sanitizers don't apply
+# to device code.
+SAN_SRCS := $(CPP_SRCS_CORE) examples/selftest.cpp
+CPP_SRCS_CORE := $(wildcard cpp/src/*.cpp)
+
+asan-check:
+	@mkdir -p $(BUILD)/asan
+	$(HIPCC) -O1 -g -std=c++20 -fsanitize=address -fno-omit-frame-pointer \
+	    -Icpp/include --offload-arch=$(GPU_ARCH) \
+	    $(CPP_SRCS_CORE) cpp/hip/snapshot_kernels.hip examples/selftest.cpp \
+	    -L/opt/rocm/lib -lrccl -lamdhip64 -lz -pthread \
+	    -o $(BUILD)/asan/selftest
+	FAABRIC_PORT_OFFSET=6600 $(BUILD)/asan/selftest
+
+tsan-check:
+	@mkdir -p $(BUILD)/tsan
+	$(HIPCC) -O1 -g -std=c++20 -fsanitize=thread \
+	    -Icpp/include --offload-arch=$(GPU_ARCH) \
+	    $(CPP_SRCS_CORE) cpp/hip/snapshot_kernels.hip examples/selftest.cpp \
+	    -L/opt/rocm/lib -lrccl -lamdhip64 -lz -pthread \
+	    -o $(BUILD)/tsan/selftest
+	FAABRIC_PORT_OFFSET=6700 TSAN_OPTIONS="report_bugs=1" \
+	    $(BUILD)/tsan/selftest
+
+.PHONY: asan-check tsan-check
 
 $(BUILD)/%: $(BUILD)/examples/%.o $(CORE_OBJS)
 	$(HIPCC) $< $(CORE_OBJS) \

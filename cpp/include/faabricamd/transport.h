@@ -115,7 +115,8 @@ class TcpConnection
     int fd = -1;
 };
 
-// Listening socket
+// Listening socket. fd is atomic: close() runs from the stopping thread
+// while accept() blocks in the accept thread.
 class TcpListener
 {
   public:
@@ -127,11 +128,11 @@ class TcpListener
     // Accept one connection; returns nullopt if the listener was closed
     std::optional<TcpConnection> accept();
     void close();
-    bool isOpen() const { return fd >= 0; }
+    bool isOpen() const { return fd.load() >= 0; }
     int boundPort() const { return port_; }
 
   private:
-    int fd = -1;
+    std::atomic<int> fd{ -1 };
     int port_ = 0;
 };
 

@@ -289,39 +289,42 @@ TcpListener::~TcpListener()
 
 void TcpListener::listen(int port, int backlog)
 {
-    fd = ::socket(AF_INET, SOCK_STREAM, 0);
-    if (fd < 0) {
+    int newFd = ::socket(AF_INET, SOCK_STREAM, 0);
+    if (newFd < 0) {
         throw FaabricException("socket() failed");
     }
     int one = 1;
-    setsockopt(fd, SOL_SOCKET, SO_REUSEADDR, &one, sizeof(one));
+    setsockopt(newFd, SOL_SOCKET, SO_REUSEADDR, &one, sizeof(one));
 
     struct sockaddr_in addr;
     std::memset(&addr, 0, sizeof(addr));
     addr.sin_family = AF_INET;
     addr.sin_addr.s_addr = htonl(INADDR_ANY);
     addr.sin_port = htons((uint16_t)port);
-    if (::bind(fd, (struct sockaddr*)&addr, sizeof(addr)) != 0) {
+    if (::bind(newFd, (struct sockaddr*)&addr, sizeof(addr)) != 0) {
         int err = errno;
-        ::close(fd);
-        fd = -1;
+        ::close(newFd);
         throw FaabricException("bind port " + std::to_string(port) +
                                " failed: " + strerror(err));
     }
-    if (::listen(fd, backlog) != 0) {
+    if (::listen(newFd, backlog) != 0) {
         int err = errno;
-        ::close(fd);
-        fd = -1;
+        ::close(newFd);
         throw FaabricException("listen failed: " +
                                std::string(strerror(err)));
     }
+    fd.store(newFd);
     port_ = port;
 }
 
 std::optional<TcpConnection> TcpListener::accept()
 {
     while (true) {
-        int cfd = ::accept(fd, nullptr, nullptr);
+        int f = fd.load();
+        if (f < 0) {
+            return std::nullopt;
+        }
+        int cfd = ::accept(f, nullptr, nullptr);
         if (cfd >= 0) {
             setCommonSockOpts(cfd);
             return TcpConnection(cfd);
@@ -336,10 +339,10 @@ std::optional<TcpConnection> TcpListener::accept()
 
 void TcpListener::close()
 {
-    if (fd >= 0) {
-        ::shutdown(fd, SHUT_RDWR);
-        ::close(fd);
-        fd = -1;
+    int old = fd.exchange(-1);
+    if (old >= 0) {
+        ::shutdown(old, SHUT_RDWR);
+        ::close(old);
     }
 }
 
