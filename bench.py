@@ -54,6 +54,8 @@ def parse_args():
     p.add_argument("--kv-bytes", type=int, default=4096)
     p.add_argument("--pp-bytes", type=int, default=64 * 1024 * 1024,
                    help="rank0<->rank1 ping-pong bytes (config 2)")
+    p.add_argument("--snap-bytes", type=int, default=1024 * 1024 * 1024,
+                   help="per-rank GPU snapshot diff+merge region (config 4)")
     return p.parse_args()
 
 
@@ -125,7 +127,8 @@ def main():
         params = (
             f"steps={args.steps};warmup={args.warmup};bytes={args.bytes};"
             f"batch={args.batch};kvbytes={args.kv_bytes};"
-            f"a2abytes={args.a2a_bytes};ppbytes={pp}"
+            f"a2abytes={args.a2a_bytes};ppbytes={pp};"
+            f"snapbytes={args.snap_bytes if have_gpu else 0}"
         )
         ber = _core.batch_exec_factory("bench", "rankstep", 1)
         msgs = ber.messages
@@ -170,6 +173,11 @@ def main():
         per_rank_step = [parse_times(r.output_data, "step") for r in results]
         per_rank_ar = [parse_times(r.output_data, "ar") for r in results]
         rank0_pp = parse_times(results[0].output_data, "pp")
+        # Per-rank snapshot diff/apply GB/s (config 4; 0 on CPU)
+        snap_diff = [parse_times(r.output_data, "snapdiff") for r in results]
+        snap_apply = [parse_times(r.output_data, "snapapply") for r in results]
+        snap_diff_total = sum(v[0] for v in snap_diff if v)
+        snap_apply_total = sum(v[0] for v in snap_apply if v)
         k = min(len(s) for s in per_rank_step)
         step_ms = [max(s[i] for s in per_rank_step) for i in range(k)]
         ar_ms = [max(s[i] for s in per_rank_ar) for i in range(k)]
@@ -219,6 +227,9 @@ def main():
                 "batch_per_host": args.batch,
                 "batch_msgs_per_sec": round(msgs_per_sec, 2),
                 "kv_bytes": args.kv_bytes,
+                "snapshot_bytes_per_rank": args.snap_bytes if have_gpu else 0,
+                "snapshot_diff_gbps_total": round(snap_diff_total, 2),
+                "snapshot_apply_gbps_total": round(snap_apply_total, 2),
                 "gpu": have_gpu,
             },
         }

@@ -12,6 +12,7 @@
 #include "faabricamd/executor.h"
 #include "faabricamd/messages.h"
 #include "faabricamd/mpi.h"
+#include "faabricamd/ops.h"
 #include "faabricamd/planner.h"
 #include "faabricamd/state.h"
 #include "faabricamd/util.h"
@@ -87,6 +88,8 @@ static int32_t benchRankStep(Message& msg)
     int64_t kvBytes = params.count("kvbytes") ? params["kvbytes"] : 4096;
     int64_t a2aBytes = params.count("a2abytes") ? params["a2abytes"] : 0;
     int64_t ppBytes = params.count("ppbytes") ? params["ppbytes"] : 0;
+    int64_t snapBytes =
+      params.count("snapbytes") ? params["snapbytes"] : 0;
 
     int nGpus = 0;
     (void)hipGetDeviceCount(&nGpus);
@@ -134,6 +137,44 @@ static int32_t benchRankStep(Message& msg)
     std::vector<double> allreduceMs;
     std::vector<double> batchMs;
     std::vector<double> pingpongMs;
+
+    // Config 4 on this rank's GPU: one diff+merge pass of a snapBytes
+    // region at 25% dirty pages, measured before the timed steps
+    double snapDiffGbps = 0;
+    double snapApplyGbps = 0;
+    if (onGpu && snapBytes > 0) {
+        try {
+            DeviceSnapshot snap((size_t)snapBytes, 0);
+            uint8_t* updatedBuf = nullptr;
+            if (hipMalloc(&updatedBuf, snapBytes) == hipSuccess) {
+                (void)hipMemset(updatedBuf, 0x77, snapBytes);
+                snap.captureFromDevice(updatedBuf);
+                // Dirty 25% of pages (a contiguous quarter)
+                (void)hipMemset(updatedBuf, 0x12, snapBytes / 4);
+                (void)hipDeviceSynchronize();
+                for (int it = 0; it < 3; it++) {
+                    int64_t s0 = getEpochMicros();
+                    uint32_t nd = snap.diffXor(updatedBuf);
+                    int64_t s1 = getEpochMicros();
+                    snap.applyLastDiff();
+                    int64_t s2 = getEpochMicros();
+                    if (it == 2 && s1 > s0 && s2 > s1) {
+                        double dirtyB = (double)nd * 4096;
+                        snapDiffGbps = (2.0 * snapBytes + dirtyB) /
+                                       ((s1 - s0) / 1e6) / 1e9;
+                        snapApplyGbps =
+                          (3.0 * dirtyB) / ((s2 - s1) / 1e6) / 1e9;
+                    }
+                    // Re-dirty for the next pass
+                    (void)hipMemset(updatedBuf, 0x12 + it, snapBytes / 4);
+                    (void)hipDeviceSynchronize();
+                }
+                (void)hipFree(updatedBuf);
+            }
+        } catch (const std::exception& e) {
+            FAM_ERROR("snapshot sub-bench failed: %s", e.what());
+        }
+    }
 
     for (int iter = 0; iter < warmup + steps; iter++) {
         world.barrier(rank);
@@ -231,6 +272,7 @@ static int32_t benchRankStep(Message& msg)
     for (size_t i = 0; i < allreduceMs.size(); i++) {
         out << (i ? "," : "") << allreduceMs[i];
     }
+    out << ";snapdiff:" << snapDiffGbps << ";snapapply:" << snapApplyGbps;
     out << ";pp:";
     for (size_t i = 0; i < pingpongMs.size(); i++) {
         out << (i ? "," : "") << pingpongMs[i];

@@ -54,11 +54,30 @@ StateKeyValue::~StateKeyValue()
     }
 }
 
+// Per-thread HIP stream so concurrent chunk ops from different executors
+// pipeline through the copy engines instead of serialising on the null
+// stream
+static hipStream_t threadCopyStream()
+{
+    thread_local hipStream_t stream = []() {
+        hipStream_t s = nullptr;
+        if (hipStreamCreateWithFlags(&s, hipStreamNonBlocking) !=
+            hipSuccess) {
+            s = nullptr;
+        }
+        return s;
+    }();
+    return stream;
+}
+
 void StateKeyValue::readLocal(uint64_t offset, uint8_t* out, size_t len)
 {
     if (onDevice) {
-        hipSetDevice(device);
-        hipMemcpy(out, devPtr + offset, len, hipMemcpyDeviceToHost);
+        (void)hipSetDevice(device);
+        hipStream_t s = threadCopyStream();
+        (void)hipMemcpyAsync(out, devPtr + offset, len,
+                             hipMemcpyDeviceToHost, s);
+        (void)hipStreamSynchronize(s);
     } else {
         std::memcpy(out, value.data() + offset, len);
     }
@@ -69,8 +88,11 @@ void StateKeyValue::writeLocal(uint64_t offset,
                                size_t len)
 {
     if (onDevice) {
-        hipSetDevice(device);
-        hipMemcpy(devPtr + offset, data, len, hipMemcpyHostToDevice);
+        (void)hipSetDevice(device);
+        hipStream_t s = threadCopyStream();
+        (void)hipMemcpyAsync(devPtr + offset, data, len,
+                             hipMemcpyHostToDevice, s);
+        (void)hipStreamSynchronize(s);
     } else {
         std::memcpy(value.data() + offset, data, len);
     }
@@ -125,9 +147,17 @@ void StateKeyValue::getChunk(uint64_t offset, uint8_t* buffer, size_t len)
         // Lazy chunked pull of just this range
         auto cli = getStateClient(masterHost);
         auto data = cli->pullChunk(user, key, offset, len);
-        std::lock_guard<std::mutex> lock(kvMx);
-        writeLocal(offset, data.data(), data.size());
+        if (onDevice) {
+            writeLocal(offset, data.data(), data.size());
+        } else {
+            std::lock_guard<std::mutex> lock(kvMx);
+            writeLocal(offset, data.data(), data.size());
+        }
         std::memcpy(buffer, data.data(), len);
+        return;
+    }
+    if (onDevice) {
+        readLocal(offset, buffer, len);
         return;
     }
     std::lock_guard<std::mutex> lock(kvMx);
@@ -140,7 +170,9 @@ void StateKeyValue::setChunk(uint64_t offset, const uint8_t* buffer,
     if (offset + len > valueSize) {
         throw FaabricException("state chunk write out of bounds");
     }
-    {
+    if (onDevice) {
+        writeLocal(offset, buffer, len);
+    } else {
         std::lock_guard<std::mutex> lock(kvMx);
         writeLocal(offset, buffer, len);
     }
@@ -263,11 +295,15 @@ void StateKeyValue::clearAppended()
 
 std::vector<uint8_t> StateKeyValue::serviceChunk(uint64_t offset, size_t len)
 {
-    std::lock_guard<std::mutex> lock(kvMx);
     if (offset + len > valueSize) {
         throw FaabricException("state chunk service out of bounds");
     }
     std::vector<uint8_t> out(len);
+    if (onDevice) {
+        readLocal(offset, out.data(), len);
+        return out;
+    }
+    std::lock_guard<std::mutex> lock(kvMx);
     readLocal(offset, out.data(), len);
     return out;
 }
