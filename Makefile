@@ -42,7 +42,7 @@ $(TARGET): $(OBJS)
 # C++ example binaries (reference examples/ + planner_server parity)
 CORE_OBJS = $(filter-out $(BUILD)/bindings/%,$(OBJS))
 EXAMPLE_BINS = $(BUILD)/check $(BUILD)/planner_server $(BUILD)/server \
-               $(BUILD)/selftest
+               $(BUILD)/selftest $(BUILD)/is_app_migratable
 examples: $(TARGET) $(EXAMPLE_BINS)
 
 # Sanitizer sweeps (reference CI parity: Address/Thread sanitised suites,

@@ -172,6 +172,8 @@ def main():
 
         per_rank_step = [parse_times(r.output_data, "step") for r in results]
         per_rank_ar = [parse_times(r.output_data, "ar") for r in results]
+        per_rank_batch = [parse_times(r.output_data, "batch")
+                          for r in results]
         rank0_pp = parse_times(results[0].output_data, "pp")
         # Per-rank snapshot diff/apply GB/s (config 4; 0 on CPU)
         snap_diff = [parse_times(r.output_data, "snapdiff") for r in results]
@@ -181,6 +183,8 @@ def main():
         k = min(len(s) for s in per_rank_step)
         step_ms = [max(s[i] for s in per_rank_step) for i in range(k)]
         ar_ms = [max(s[i] for s in per_rank_ar) for i in range(k)]
+        batch_ms = [max(s[i] for s in per_rank_batch) for i in range(k)]
+        batch_mean_ms = sum(batch_ms) / len(batch_ms) if batch_ms else 0.0
 
         total_s = sum(step_ms) / 1000.0
         ms_per_step = sum(step_ms) / k
@@ -226,6 +230,7 @@ def main():
                 "pingpong_gbps": round(pp_gbps, 2),
                 "batch_per_host": args.batch,
                 "batch_msgs_per_sec": round(msgs_per_sec, 2),
+                "batch_ms": round(batch_mean_ms, 3),
                 "kv_bytes": args.kv_bytes,
                 "snapshot_bytes_per_rank": args.snap_bytes if have_gpu else 0,
                 "snapshot_diff_gbps_total": round(snap_diff_total, 2),

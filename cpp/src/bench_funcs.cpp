@@ -287,6 +287,17 @@ static int32_t benchRankStep(Message& msg)
 
 void registerBenchFunctions()
 {
+    FunctionRegistry::get().registerFunction(
+      "bench", "sleep", [](Message& msg) {
+          int ms = 1000;
+          if (!msg.inputData.empty()) {
+              ms = atoi(std::string(msg.inputData.begin(),
+                                    msg.inputData.end())
+                          .c_str());
+          }
+          std::this_thread::sleep_for(std::chrono::milliseconds(ms));
+          return 0;
+      });
     FunctionRegistry::get().registerFunction("bench", "kvtouch",
                                              benchKvTouch);
     FunctionRegistry::get().registerFunction("bench", "rankstep",

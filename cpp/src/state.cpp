@@ -70,14 +70,25 @@ static hipStream_t threadCopyStream()
     return stream;
 }
 
+static bool kvAsyncCopy()
+{
+    static bool v = getEnvVarInt("FAABRIC_KV_ASYNC_COPY", 0) != 0;
+    return v;
+}
+
 void StateKeyValue::readLocal(uint64_t offset, uint8_t* out, size_t len)
 {
     if (onDevice) {
         (void)hipSetDevice(device);
-        hipStream_t s = threadCopyStream();
-        (void)hipMemcpyAsync(out, devPtr + offset, len,
-                             hipMemcpyDeviceToHost, s);
-        (void)hipStreamSynchronize(s);
+        if (kvAsyncCopy()) {
+            hipStream_t s = threadCopyStream();
+            (void)hipMemcpyAsync(out, devPtr + offset, len,
+                                 hipMemcpyDeviceToHost, s);
+            (void)hipStreamSynchronize(s);
+        } else {
+            (void)hipMemcpy(out, devPtr + offset, len,
+                            hipMemcpyDeviceToHost);
+        }
     } else {
         std::memcpy(out, value.data() + offset, len);
     }
@@ -89,10 +100,15 @@ void StateKeyValue::writeLocal(uint64_t offset,
 {
     if (onDevice) {
         (void)hipSetDevice(device);
-        hipStream_t s = threadCopyStream();
-        (void)hipMemcpyAsync(devPtr + offset, data, len,
-                             hipMemcpyHostToDevice, s);
-        (void)hipStreamSynchronize(s);
+        if (kvAsyncCopy()) {
+            hipStream_t s = threadCopyStream();
+            (void)hipMemcpyAsync(devPtr + offset, data, len,
+                                 hipMemcpyHostToDevice, s);
+            (void)hipStreamSynchronize(s);
+        } else {
+            (void)hipMemcpy(devPtr + offset, data, len,
+                            hipMemcpyHostToDevice);
+        }
     } else {
         std::memcpy(value.data() + offset, data, len);
     }

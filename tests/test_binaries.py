@@ -2,6 +2,7 @@
 as real processes and serve an EXECUTE_BATCH over the HTTP ops API
 (reference: docker-compose deployment of planner + worker containers)."""
 
+import base64
 import json
 import os
 import signal
@@ -112,3 +113,45 @@ def test_binaries_serve_batch(binaries):
                 return
         time.sleep(0.05)
     pytest.fail("batch did not finish through the binaries")
+
+
+def test_is_app_migratable_cli(binaries):
+    """Deployment CLI parity (reference src/planner/is_app_migratable.cpp):
+    replays a DIST_CHANGE scheduling pass against the live planner state."""
+    assert binaries
+    cli = os.path.join(BUILD, "is_app_migratable")
+    if not os.path.exists(cli):
+        pytest.skip("is_app_migratable not built")
+    env = {**os.environ, "FAABRIC_PORT_OFFSET": str(PLANNER_OFF)}
+
+    # Unknown app -> exit 1
+    out = subprocess.run(
+        [cli, "999999"], env=env, capture_output=True, text=True, timeout=20
+    )
+    assert out.returncode == 1, out.stderr
+
+    # In-flight app on a single host -> "NO" (bin-pack has nowhere better)
+    ber = {
+        "user": "bench",
+        "function": "sleep",
+        "messages": [
+            {"input_data": base64.b64encode(b"3000").decode()},
+            {"input_data": base64.b64encode(b"3000").decode()},
+        ],
+    }
+    status, body = post(10, json.dumps(ber))
+    assert status == 200, body
+    app_id = json.loads(body)["appId"]
+    out = subprocess.run(
+        [cli, str(app_id)], env=env, capture_output=True, text=True,
+        timeout=20,
+    )
+    assert out.returncode == 0, out.stderr
+    assert out.stdout.strip() == "NO"
+    # Drain the batch so the module fixture tears down cleanly
+    deadline = time.monotonic() + 15
+    while time.monotonic() < deadline:
+        status, body = post(11, json.dumps({"appId": app_id}))
+        if status == 200 and json.loads(body).get("finished"):
+            break
+        time.sleep(0.2)
