@@ -9,6 +9,7 @@
 // std::condition_variable throughout, no third-party lock-free deps.
 #pragma once
 
+#include <atomic>
 #include <chrono>
 #include <condition_variable>
 #include <deque>
@@ -17,6 +18,7 @@
 #include <mutex>
 #include <stdexcept>
 #include <thread>
+#include <vector>
 
 #include "faabricamd/util.h"
 
@@ -88,6 +90,74 @@ class Queue
 };
 
 // Pool of integer tokens (reference: util/queue.h:245 TokenPool)
+// Bounded single-producer/single-consumer ring queue. Lock-free on the
+// fast path: head/tail are C++20 atomics the producer/consumer each own,
+// with acquire/release pairing on the other side's index; blocked sides
+// poll with yield so timeouts hold (std::atomic::wait has no timed form). MI355X-native
+// replacement for the reference's moodycamel-backed FixedCapacityQueue
+// (reference: include/faabric/util/queue.h:148-218) without the
+// third-party dependency.
+template<typename T>
+class FixedCapacityQueue
+{
+  public:
+    explicit FixedCapacityQueue(size_t capacityIn)
+      : cap(capacityIn + 1)
+      , ring(capacityIn + 1)
+    {
+        if (capacityIn == 0) {
+            throw FaabricException("FixedCapacityQueue capacity must be > 0");
+        }
+    }
+
+    void enqueue(T value, int timeoutMs = DEFAULT_QUEUE_TIMEOUT_MS)
+    {
+        size_t t = tail.load(std::memory_order_relaxed);
+        size_t next = (t + 1) % cap;
+        auto deadline = std::chrono::steady_clock::now() +
+                        std::chrono::milliseconds(timeoutMs);
+        // Slow path is a bounded poll: std::atomic::wait has no timed
+        // variant, and the queue must honour timeoutMs
+        while (next == head.load(std::memory_order_acquire)) {
+            if (std::chrono::steady_clock::now() > deadline) {
+                throw QueueTimeoutException("SPSC enqueue timed out (full)");
+            }
+            std::this_thread::yield();
+        }
+        ring[t] = std::move(value);
+        tail.store(next, std::memory_order_release);
+    }
+
+    T dequeue(int timeoutMs = DEFAULT_QUEUE_TIMEOUT_MS)
+    {
+        size_t h = head.load(std::memory_order_relaxed);
+        auto deadline = std::chrono::steady_clock::now() +
+                        std::chrono::milliseconds(timeoutMs);
+        while (h == tail.load(std::memory_order_acquire)) {
+            if (std::chrono::steady_clock::now() > deadline) {
+                throw QueueTimeoutException("SPSC dequeue timed out (empty)");
+            }
+            std::this_thread::yield();
+        }
+        T v = std::move(ring[h]);
+        head.store((h + 1) % cap, std::memory_order_release);
+        return v;
+    }
+
+    size_t size() const
+    {
+        size_t t = tail.load(std::memory_order_acquire);
+        size_t h = head.load(std::memory_order_acquire);
+        return (t + cap - h) % cap;
+    }
+
+  private:
+    size_t cap;
+    std::vector<T> ring;
+    alignas(64) std::atomic<size_t> head{ 0 };
+    alignas(64) std::atomic<size_t> tail{ 0 };
+};
+
 class TokenPool
 {
   public:

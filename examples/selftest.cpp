@@ -9,10 +9,12 @@
 #include <faabricamd/runner.h>
 #include <faabricamd/scheduler.h>
 #include <faabricamd/snapshot.h>
+#include <faabricamd/queue.h>
 #include <faabricamd/util.h>
 
 #include <cstdio>
 #include <cstring>
+#include <thread>
 
 using namespace faabricamd;
 
@@ -163,7 +165,32 @@ int main()
         CHECK(r.returnValue == 0);
     }
 
-    // 5. Snapshot diff/apply semantics
+    // 5. SPSC FixedCapacityQueue under a producer/consumer pair (the
+    // sanitizer builds verify the acquire/release protocol)
+    {
+        FixedCapacityQueue<int64_t> q(64);
+        int64_t sum = 0;
+        std::thread consumer([&] {
+            for (int i = 0; i < 10000; i++) {
+                sum += q.dequeue(10000);
+            }
+        });
+        for (int i = 0; i < 10000; i++) {
+            q.enqueue(i, 10000);
+        }
+        consumer.join();
+        CHECK(sum == 10000LL * 9999 / 2);
+        CHECK(q.size() == 0);
+        bool threw = false;
+        try {
+            q.dequeue(30);
+        } catch (QueueTimeoutException&) {
+            threw = true;
+        }
+        CHECK(threw);
+    }
+
+    // 6. Snapshot diff/apply semantics
     {
         std::vector<uint8_t> base(8192, 0);
         SnapshotData snap(base);
