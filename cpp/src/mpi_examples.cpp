@@ -143,7 +143,17 @@ static int32_t exampleAllReduceBench(Message& msg)
     double elapsed = MPI_Wtime() - t0;
     MPI_Barrier(MPI_COMM_WORLD);
     MPI_Finalize();
-    msg.outputData = "elapsed_s=" + std::to_string(elapsed);
+    // Reference reports workload = 4*(np-1)*sizeof(int)*Sum(n) per rep
+    // (tests/dist/mpi/benchmarks/mpi_allreduce.cpp:41-51)
+    int64_t sumN = 0;
+    for (int n : sizes) {
+        sumN += n;
+    }
+    double workload =
+      4.0 * (worldSize - 1) * sizeof(int) * (double)sumN * reps;
+    double gbps = elapsed > 0 ? workload / elapsed / 1e9 : 0.0;
+    msg.outputData = "elapsed_s=" + std::to_string(elapsed) +
+                     ";gbps=" + std::to_string(gbps);
     return 0;
 }
 
