@@ -652,6 +652,34 @@ void initRuntimeBindings(py::module_& m)
         return nDirty == 2 && dirty[3] == 1 && dirty[9] == 1;
     });
 
+    // Same check for the userfaultfd write-protect tracker; returns None
+    // (skip) when the kernel lacks uffd-wp
+    m.def("_selftest_uffd_tracker", []() -> py::object {
+        if (!UffdDirtyTracker::isAvailable()) {
+            return py::none();
+        }
+        bool ok;
+        {
+            py::gil_scoped_release release;
+            UffdDirtyTracker tracker;
+            PageAlignedBuffer buf;
+            buf.resize(16 * 4096);
+            tracker.startTracking(buf.data(), buf.size());
+            buf.data()[2 * 4096 + 11] = 42;
+            buf.data()[14 * 4096] = 7;
+            // Writes block until the poller clears WP, so they are
+            // ordered before stopTracking on this thread
+            tracker.stopTracking(buf.data(), buf.size());
+            auto dirty = tracker.getDirtyPages(buf.data(), buf.size());
+            int nDirty = 0;
+            for (char d : dirty) {
+                nDirty += d != 0;
+            }
+            ok = nDirty == 2 && dirty[2] == 1 && dirty[14] == 1;
+        }
+        return py::bool_(ok);
+    });
+
     // Native benchmark payloads (cpp/src/bench_funcs.cpp)
     m.def("register_bench_functions", [] { registerBenchFunctions(); });
     m.def("register_mpi_example_functions",

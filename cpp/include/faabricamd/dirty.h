@@ -7,9 +7,12 @@
 //  - "segfault": mprotect(PROT_READ) + SIGSEGV handler marking faulting
 //               pages (reference SegfaultDirtyTracker) — host arenas only
 //  - "none":    every page dirty
-// The reference's soft-PTE and userfaultfd trackers are not reproduced:
-// they are Linux-/WASM-memory-specific optimisations of the same contract
-// and /proc/self/clear_refs is not writable in many container setups.
+//  - "uffd":    userfaultfd write-protect mode — kernel-async fault
+//               delivery to a poller thread, no SIGSEGV involvement
+//               (reference UffdDirtyTracker, wp sub-mode)
+// The reference's soft-PTE tracker is not reproduced: this kernel ships
+// without CONFIG_MEM_SOFT_DIRTY (probed: /proc/self/clear_refs accepts
+// writes but pagemap bit 55 never sets), so it cannot be validated.
 #pragma once
 
 #include <cstdint>
@@ -66,6 +69,29 @@ class SegfaultDirtyTracker : public DirtyTracker
     void stopThreadLocalTracking(uint8_t* region, size_t size) override;
     std::vector<char> getThreadLocalDirtyPages(uint8_t* region,
                                                size_t size) override;
+};
+
+// userfaultfd write-protect tracker (reference: src/util/dirty.cpp uffd
+// modes). Global channel is exact; the thread-local channel attributes by
+// time-window (pages dirtied between start/stop on any thread) because
+// uffd faults are delivered to a poller thread, not the faulting thread —
+// overlapping same-page writes from concurrent THREADS tasks are app
+// responsibility, as in the reference.
+class UffdDirtyTracker : public DirtyTracker
+{
+  public:
+    UffdDirtyTracker(); // throws if the kernel lacks uffd-wp
+    ~UffdDirtyTracker() override;
+    std::string getType() const override { return "uffd"; }
+    void startTracking(uint8_t* region, size_t size) override;
+    void stopTracking(uint8_t* region, size_t size) override;
+    std::vector<char> getDirtyPages(uint8_t* region, size_t size) override;
+    void startThreadLocalTracking(uint8_t* region, size_t size) override;
+    void stopThreadLocalTracking(uint8_t* region, size_t size) override;
+    std::vector<char> getThreadLocalDirtyPages(uint8_t* region,
+                                               size_t size) override;
+
+    static bool isAvailable();
 };
 
 std::shared_ptr<DirtyTracker> getDirtyTracker();

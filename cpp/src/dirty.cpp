@@ -4,12 +4,21 @@
 #include "faabricamd/dirty.h"
 #include "faabricamd/util.h"
 
+#include <atomic>
+#include <cerrno>
+#include <cstdio>
 #include <csignal>
 #include <cstdlib>
 #include <cstring>
+#include <fcntl.h>
+#include <linux/userfaultfd.h>
 #include <map>
 #include <mutex>
+#include <poll.h>
+#include <sys/ioctl.h>
 #include <sys/mman.h>
+#include <sys/syscall.h>
+#include <thread>
 #include <unistd.h>
 
 namespace faabricamd {
@@ -173,6 +182,231 @@ std::vector<char> SegfaultDirtyTracker::getThreadLocalDirtyPages(
     return it->second;
 }
 
+// ------------------------- userfaultfd (wp mode) -----------------------------
+
+namespace {
+
+struct UffdRegion
+{
+    uint8_t* base = nullptr;
+    size_t size = 0;
+    std::vector<char> dirty; // exact, written only by the poller thread
+    std::mutex mx;
+};
+
+struct UffdState
+{
+    int fd = -1;
+    int stopPipe[2] = { -1, -1 };
+    std::thread poller;
+    std::mutex mx;
+    std::map<uint8_t*, std::shared_ptr<UffdRegion>> regions;
+};
+
+UffdState uffd;
+
+// Thread-local window baselines for the thread-local channel
+thread_local std::map<uint8_t*, std::vector<char>> uffdThreadBaseline;
+
+void uffdPollLoop()
+{
+    while (true) {
+        struct pollfd fds[2];
+        fds[0] = { uffd.fd, POLLIN, 0 };
+        fds[1] = { uffd.stopPipe[0], POLLIN, 0 };
+        if (::poll(fds, 2, -1) < 0) {
+            if (errno == EINTR) {
+                continue;
+            }
+            break;
+        }
+        if (fds[1].revents != 0) {
+            break; // shutdown
+        }
+        struct uffd_msg msg;
+        ssize_t n = ::read(uffd.fd, &msg, sizeof(msg));
+        if (n <= 0) {
+            continue;
+        }
+        if (msg.event != UFFD_EVENT_PAGEFAULT) {
+            continue;
+        }
+        uint8_t* addr = (uint8_t*)(uintptr_t)msg.arg.pagefault.address;
+        std::shared_ptr<UffdRegion> hit;
+        {
+            std::lock_guard<std::mutex> lock(uffd.mx);
+            for (auto& [base, region] : uffd.regions) {
+                if (addr >= base && addr < base + region->size) {
+                    hit = region;
+                    break;
+                }
+            }
+        }
+        uint8_t* pageAddr = (uint8_t*)((uintptr_t)addr & ~(TRACK_PAGE - 1));
+        if (hit) {
+            size_t page = (size_t)(pageAddr - hit->base) / TRACK_PAGE;
+            std::lock_guard<std::mutex> lock(hit->mx);
+            hit->dirty[page] = 1;
+        }
+        // Drop write-protection on the faulting page and wake the writer
+        struct uffdio_writeprotect wp;
+        wp.range.start = (uintptr_t)pageAddr;
+        wp.range.len = TRACK_PAGE;
+        wp.mode = 0; // clear WP
+        (void)::ioctl(uffd.fd, UFFDIO_WRITEPROTECT, &wp);
+        // Belt-and-braces: wake explicitly (the clear-WP wake alone has
+        // been seen to leave a second waiter asleep on this kernel)
+        struct uffdio_range wake;
+        wake.start = wp.range.start;
+        wake.len = wp.range.len;
+        (void)::ioctl(uffd.fd, UFFDIO_WAKE, &wake);
+    }
+}
+
+} // namespace
+
+bool UffdDirtyTracker::isAvailable()
+{
+    int fd = (int)syscall(SYS_userfaultfd, O_CLOEXEC | O_NONBLOCK);
+    if (fd < 0) {
+        return false;
+    }
+    struct uffdio_api api;
+    api.api = UFFD_API;
+    api.features = UFFD_FEATURE_PAGEFAULT_FLAG_WP;
+    bool ok = ::ioctl(fd, UFFDIO_API, &api) == 0 &&
+              (api.features & UFFD_FEATURE_PAGEFAULT_FLAG_WP) != 0;
+    ::close(fd);
+    return ok;
+}
+
+UffdDirtyTracker::UffdDirtyTracker()
+{
+    uffd.fd = (int)syscall(SYS_userfaultfd, O_CLOEXEC | O_NONBLOCK);
+    if (uffd.fd < 0) {
+        throw FaabricException("userfaultfd syscall unavailable");
+    }
+    struct uffdio_api api;
+    api.api = UFFD_API;
+    api.features = UFFD_FEATURE_PAGEFAULT_FLAG_WP;
+    if (::ioctl(uffd.fd, UFFDIO_API, &api) != 0 ||
+        (api.features & UFFD_FEATURE_PAGEFAULT_FLAG_WP) == 0) {
+        ::close(uffd.fd);
+        uffd.fd = -1;
+        throw FaabricException("kernel lacks uffd write-protect");
+    }
+    if (::pipe(uffd.stopPipe) != 0) {
+        ::close(uffd.fd);
+        uffd.fd = -1;
+        throw FaabricException("uffd stop pipe failed");
+    }
+    uffd.poller = std::thread(uffdPollLoop);
+}
+
+UffdDirtyTracker::~UffdDirtyTracker()
+{
+    if (uffd.fd < 0) {
+        return;
+    }
+    char b = 1;
+    (void)!::write(uffd.stopPipe[1], &b, 1);
+    if (uffd.poller.joinable()) {
+        uffd.poller.join();
+    }
+    ::close(uffd.stopPipe[0]);
+    ::close(uffd.stopPipe[1]);
+    ::close(uffd.fd);
+    uffd.fd = -1;
+    uffd.regions.clear();
+}
+
+void UffdDirtyTracker::startTracking(uint8_t* region, size_t size)
+{
+    if (((uintptr_t)region % TRACK_PAGE) != 0) {
+        throw FaabricException("uffd tracker needs page-aligned region");
+    }
+    size_t len = nPagesOf(size) * TRACK_PAGE;
+    auto tracked = std::make_shared<UffdRegion>();
+    tracked->base = region;
+    tracked->size = len;
+    tracked->dirty.assign(nPagesOf(size), 0);
+    {
+        std::lock_guard<std::mutex> lock(uffd.mx);
+        uffd.regions[region] = tracked;
+    }
+    struct uffdio_register reg;
+    reg.range.start = (uintptr_t)region;
+    reg.range.len = len;
+    reg.mode = UFFDIO_REGISTER_MODE_WP;
+    if (::ioctl(uffd.fd, UFFDIO_REGISTER, &reg) != 0) {
+        throw FaabricException("UFFDIO_REGISTER failed");
+    }
+    struct uffdio_writeprotect wp;
+    wp.range.start = (uintptr_t)region;
+    wp.range.len = len;
+    wp.mode = UFFDIO_WRITEPROTECT_MODE_WP;
+    if (::ioctl(uffd.fd, UFFDIO_WRITEPROTECT, &wp) != 0) {
+        throw FaabricException("UFFDIO_WRITEPROTECT arm failed");
+    }
+}
+
+void UffdDirtyTracker::stopTracking(uint8_t* region, size_t size)
+{
+    size_t len = nPagesOf(size) * TRACK_PAGE;
+    struct uffdio_writeprotect wp;
+    wp.range.start = (uintptr_t)region;
+    wp.range.len = len;
+    wp.mode = 0;
+    (void)::ioctl(uffd.fd, UFFDIO_WRITEPROTECT, &wp);
+    struct uffdio_range range;
+    range.start = (uintptr_t)region;
+    range.len = len;
+    (void)::ioctl(uffd.fd, UFFDIO_UNREGISTER, &range);
+}
+
+std::vector<char> UffdDirtyTracker::getDirtyPages(uint8_t* region,
+                                                  size_t size)
+{
+    std::shared_ptr<UffdRegion> tracked;
+    {
+        std::lock_guard<std::mutex> lock(uffd.mx);
+        auto it = uffd.regions.find(region);
+        if (it == uffd.regions.end()) {
+            return std::vector<char>(nPagesOf(size), 0);
+        }
+        tracked = it->second;
+    }
+    std::lock_guard<std::mutex> lock(tracked->mx);
+    return tracked->dirty;
+}
+
+void UffdDirtyTracker::startThreadLocalTracking(uint8_t* region, size_t size)
+{
+    uffdThreadBaseline[region] = getDirtyPages(region, size);
+}
+
+void UffdDirtyTracker::stopThreadLocalTracking(uint8_t* region, size_t size)
+{
+    (void)region;
+    (void)size;
+}
+
+std::vector<char> UffdDirtyTracker::getThreadLocalDirtyPages(uint8_t* region,
+                                                             size_t size)
+{
+    std::vector<char> now = getDirtyPages(region, size);
+    auto it = uffdThreadBaseline.find(region);
+    if (it == uffdThreadBaseline.end()) {
+        return now;
+    }
+    for (size_t i = 0; i < now.size() && i < it->second.size(); i++) {
+        if (it->second[i]) {
+            now[i] = 0; // dirty before this thread's window: not ours
+        }
+    }
+    return now;
+}
+
 // ------------------------- registry ------------------------------------------
 
 static std::shared_ptr<DirtyTracker> trackerInstance;
@@ -187,6 +421,14 @@ std::shared_ptr<DirtyTracker> getDirtyTracker()
     const std::string& mode = getSystemConfig().dirtyTrackingMode;
     if (mode == "segfault") {
         trackerInstance = std::make_shared<SegfaultDirtyTracker>();
+    } else if (mode == "uffd") {
+        if (UffdDirtyTracker::isAvailable()) {
+            trackerInstance = std::make_shared<UffdDirtyTracker>();
+        } else {
+            FAM_WARN("uffd-wp unavailable; falling back to segfault "
+                     "dirty tracking");
+            trackerInstance = std::make_shared<SegfaultDirtyTracker>();
+        }
     } else {
         // "compare" and "none" both report every page; "compare" relies
         // on the snapshot diff to refine

@@ -3,6 +3,7 @@ dirty tracker (reference: src/util/delta.cpp, util/timing.h,
 src/util/dirty.cpp)."""
 
 import os
+import pytest
 import random
 
 from faabric_amd import _core
@@ -51,3 +52,12 @@ def test_prof_timers():
 def test_pin_thread():
     core = _core.pin_thread_to_free_cpu()
     assert core >= 0 or os.cpu_count() is None
+
+
+def test_uffd_dirty_tracker():
+    """userfaultfd write-protect tracking (reference: src/util/dirty.cpp
+    uffd modes); skipped where the kernel lacks uffd-wp."""
+    result = _core._selftest_uffd_tracker()
+    if result is None:
+        pytest.skip("kernel lacks uffd write-protect")
+    assert result
