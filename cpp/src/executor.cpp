@@ -1,6 +1,7 @@
 // Executor implementation (reference behavior: src/executor/Executor.cpp
 // :38-212 pool lifecycle, :307-576 threadPoolThread, :580-590 claims).
 #include "faabricamd/executor.h"
+#include "faabricamd/utilextras.h"
 #include "faabricamd/mpi.h"
 #include "faabricamd/ops.h"
 #include "faabricamd/planner.h"
@@ -457,7 +458,9 @@ void Executor::threadPoolThread(int poolIdx)
         int32_t returnValue = 0;
         ExecutorContext::set(this, task.req, task.msgIdx);
         try {
+            PROF_START(exec_task)
             returnValue = executeTask(poolIdx, task.msgIdx, task.req);
+            PROF_END(exec_task)
             // Exec-graph detail: per-rank MPI message counters ride the
             // result (reference: mpi/MpiWorld.h:13-18 + Executor exec
             // graph details)
@@ -598,7 +601,9 @@ void Executor::handleTaskResult(Message& msg,
                 }
             }
             auto resultMsg = std::make_shared<Message>(msg);
+            PROF_START(result_rpc)
             getPlannerClient().setMessageResult(resultMsg);
+            PROF_END(result_rpc)
         } else {
             getSnapshotClient(mainHost)->pushThreadResult(
               msg.appId, msg.id, returnValue, req->snapshotKey, threadDiffs);
