@@ -6,10 +6,12 @@
 // slot is one GPU.
 #pragma once
 
+#include <atomic>
 #include <map>
 #include <memory>
 #include <set>
 #include <shared_mutex>
+#include <thread>
 #include <vector>
 
 #include "faabricamd/messages.h"
@@ -127,10 +129,21 @@ class PlannerServer : public MessageEndpointServer
 {
   public:
     PlannerServer();
+    ~PlannerServer() override;
     void doAsyncRecv(uint8_t code,
                      const std::string& body,
                      uint32_t seq) override;
     std::string doSyncRecv(uint8_t code, const std::string& body) override;
+
+  private:
+    // Result-ingestion pool: SetMessageResult frames from one worker
+    // connection would otherwise decode+apply serially (~17 us each, the
+    // wall-clock floor of a large batch). Results of distinct messages
+    // commute, so a few workers drain them concurrently.
+    void resultWorkerLoop();
+    Queue<std::string> resultQueue;
+    std::vector<std::thread> resultWorkers;
+    std::atomic<bool> resultWorkersStop{ false };
 };
 
 // Per-process client with local result cache; results are pushed to waiting

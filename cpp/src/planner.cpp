@@ -4,6 +4,7 @@
 // membership :166-360; src/planner/PlannerServer.cpp:22-182;
 // src/planner/PlannerClient.cpp). Fresh implementation, see planner.h.
 #include "faabricamd/planner.h"
+#include "faabricamd/utilextras.h"
 #include "faabricamd/ptp.h"
 #include "faabricamd/scheduler.h"
 #include "faabricamd/snapshot.h"
@@ -786,7 +787,53 @@ void Planner::flushHosts()
 
 PlannerServer::PlannerServer()
   : MessageEndpointServer(PLANNER_ASYNC_PORT, PLANNER_SYNC_PORT, "planner")
-{}
+{
+    const int nResultWorkers = getEnvVarInt("FAABRIC_RESULT_WORKERS", 4);
+    for (int i = 0; i < nResultWorkers; i++) {
+        resultWorkers.emplace_back([this] { resultWorkerLoop(); });
+    }
+}
+
+PlannerServer::~PlannerServer()
+{
+    resultWorkersStop.store(true);
+    for (size_t i = 0; i < resultWorkers.size(); i++) {
+        resultQueue.enqueue(std::string());
+    }
+    for (auto& t : resultWorkers) {
+        if (t.joinable()) {
+            t.join();
+        }
+    }
+}
+
+void PlannerServer::resultWorkerLoop()
+{
+    while (true) {
+        std::string body;
+        try {
+            body = resultQueue.dequeue(0);
+        } catch (...) {
+            continue;
+        }
+        if (resultWorkersStop.load()) {
+            return;
+        }
+        if (body.empty()) {
+            continue;
+        }
+        try {
+            PROF_START(result_decode)
+            auto msg = std::make_shared<Message>(Message::decode(body));
+            PROF_END(result_decode)
+            PROF_START(result_core)
+            Planner::get().setMessageResult(msg);
+            PROF_END(result_core)
+        } catch (const std::exception& e) {
+            FAM_ERROR("result ingestion failed: %s", e.what());
+        }
+    }
+}
 
 void PlannerServer::doAsyncRecv(uint8_t code,
                                 const std::string& body,
@@ -794,8 +841,12 @@ void PlannerServer::doAsyncRecv(uint8_t code,
 {
     (void)seq;
     if ((PlannerCalls)code == PlannerCalls::SetMessageResult) {
-        auto msg = std::make_shared<Message>(Message::decode(body));
-        Planner::get().setMessageResult(msg);
+        if (resultWorkers.empty()) {
+            auto msg = std::make_shared<Message>(Message::decode(body));
+            Planner::get().setMessageResult(msg);
+        } else {
+            resultQueue.enqueue(body);
+        }
         return;
     }
     FAM_ERROR("planner server: bad async code %d", (int)code);
