@@ -66,6 +66,11 @@ struct MPI_Status
     int bytesSize;
 };
 
+typedef int MPI_Group;
+#define MPI_GROUP_NULL (-1)
+#define MPI_BOTTOM ((void*)0)
+#define MPI_COMM_NULL ((MPI_Comm) nullptr)
+
 #define MPI_STATUS_IGNORE ((MPI_Status*)nullptr)
 #define MPI_STATUSES_IGNORE ((MPI_Status*)nullptr)
 #define MPI_IN_PLACE ((void*)-1)
@@ -169,6 +174,12 @@ int MPI_Irecv(void* buf,
               MPI_Comm comm,
               MPI_Request* request);
 int MPI_Wait(MPI_Request* request, MPI_Status* status);
+int MPI_Waitall(int count, MPI_Request* requests, MPI_Status* statuses);
+int MPI_Waitany(int count,
+                MPI_Request* requests,
+                int* index,
+                MPI_Status* status);
+int MPI_Request_free(MPI_Request* request);
 int MPI_Probe(int source, int tag, MPI_Comm comm, MPI_Status* status);
 int MPI_Get_count(const MPI_Status* status,
                   MPI_Datatype datatype,
@@ -236,6 +247,99 @@ int MPI_Alltoall(const void* sendbuf,
                  int recvcount,
                  MPI_Datatype recvtype,
                  MPI_Comm comm);
+
+int MPI_Gatherv(const void* sendbuf,
+                int sendcount,
+                MPI_Datatype sendtype,
+                void* recvbuf,
+                const int* recvcounts,
+                const int* displs,
+                MPI_Datatype recvtype,
+                int root,
+                MPI_Comm comm);
+int MPI_Allgatherv(const void* sendbuf,
+                   int sendcount,
+                   MPI_Datatype sendtype,
+                   void* recvbuf,
+                   const int* recvcounts,
+                   const int* displs,
+                   MPI_Datatype recvtype,
+                   MPI_Comm comm);
+int MPI_Alltoallv(const void* sendbuf,
+                  const int* sendcounts,
+                  const int* sdispls,
+                  MPI_Datatype sendtype,
+                  void* recvbuf,
+                  const int* recvcounts,
+                  const int* rdispls,
+                  MPI_Datatype recvtype,
+                  MPI_Comm comm);
+
+// --- communicator / group management ---
+int MPI_Comm_dup(MPI_Comm comm, MPI_Comm* newcomm);
+int MPI_Comm_free(MPI_Comm* comm);
+int MPI_Comm_split(MPI_Comm comm, int color, int key, MPI_Comm* newcomm);
+int MPI_Comm_split_type(MPI_Comm comm,
+                        int split_type,
+                        int key,
+                        MPI_Info info,
+                        MPI_Comm* newcomm);
+int MPI_Comm_create(MPI_Comm comm, MPI_Group group, MPI_Comm* newcomm);
+int MPI_Comm_create_group(MPI_Comm comm,
+                          MPI_Group group,
+                          int tag,
+                          MPI_Comm* newcomm);
+int MPI_Comm_group(MPI_Comm comm, MPI_Group* group);
+int MPI_Group_incl(MPI_Group group,
+                   int n,
+                   const int* ranks,
+                   MPI_Group* newgroup);
+int MPI_Group_free(MPI_Group* group);
+int MPI_Op_create(void* user_fn, int commute, MPI_Op* op);
+int MPI_Op_free(MPI_Op* op);
+
+// --- one-sided (RMA) — declared for API parity; the runtime rejects
+// them at call time, matching the reference where RMA exists only in
+// the header (reference include/faabric/mpi/mpi.h) ---
+int MPI_Win_create(void* base,
+                   MPI_Aint size,
+                   int disp_unit,
+                   MPI_Info info,
+                   MPI_Comm comm,
+                   MPI_Win* win);
+int MPI_Win_allocate_shared(MPI_Aint size,
+                            int disp_unit,
+                            MPI_Info info,
+                            MPI_Comm comm,
+                            void* baseptr,
+                            MPI_Win* win);
+int MPI_Win_shared_query(MPI_Win win,
+                         int rank,
+                         MPI_Aint* size,
+                         int* disp_unit,
+                         void* baseptr);
+int MPI_Win_get_attr(MPI_Win win,
+                     int win_keyval,
+                     void* attribute_val,
+                     int* flag);
+int MPI_Win_fence(int assert_arg, MPI_Win win);
+int MPI_Win_free(MPI_Win* win);
+int MPI_Get(void* origin_addr,
+            int origin_count,
+            MPI_Datatype origin_datatype,
+            int target_rank,
+            MPI_Aint target_disp,
+            int target_count,
+            MPI_Datatype target_datatype,
+            MPI_Win win);
+int MPI_Put(const void* origin_addr,
+            int origin_count,
+            MPI_Datatype origin_datatype,
+            int target_rank,
+            MPI_Aint target_disp,
+            int target_count,
+            MPI_Datatype target_datatype,
+            MPI_Win win);
 
 // --- cartesian topology ---
 int MPI_Cart_create(MPI_Comm old_comm,

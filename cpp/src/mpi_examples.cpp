@@ -157,12 +157,115 @@ static int32_t exampleAllReduceBench(Message& msg)
     return 0;
 }
 
+
+// v-collectives + request-array coverage: Gatherv/Allgatherv/Alltoallv/
+// Waitall/Comm_dup (reference parity: tests/dist/mpi/examples gatherv,
+// alltoall programs)
+static int32_t exampleVCollectives(Message& msg)
+{
+    MPI_Init(nullptr, nullptr);
+    int rank;
+    int n;
+    MPI_Comm_rank(MPI_COMM_WORLD, &rank);
+    MPI_Comm_size(MPI_COMM_WORLD, &n);
+
+    // Gatherv: rank r contributes (r+1) ints of value r
+    std::vector<int> counts(n);
+    std::vector<int> displs(n);
+    int total = 0;
+    for (int r = 0; r < n; r++) {
+        counts[r] = r + 1;
+        displs[r] = total;
+        total += counts[r];
+    }
+    std::vector<int> mine(rank + 1, rank);
+    std::vector<int> gathered(total, -1);
+    MPI_Gatherv(mine.data(), rank + 1, MPI_INT, gathered.data(),
+                counts.data(), displs.data(), MPI_INT, 0, MPI_COMM_WORLD);
+    if (rank == 0) {
+        for (int r = 0; r < n; r++) {
+            for (int i = 0; i < counts[r]; i++) {
+                if (gathered[displs[r] + i] != r) {
+                    msg.outputData = "gatherv mismatch";
+                    return 1;
+                }
+            }
+        }
+    }
+    MPI_Barrier(MPI_COMM_WORLD);
+
+    // Allgatherv of the same shape
+    std::vector<int> all(total, -1);
+    MPI_Allgatherv(mine.data(), rank + 1, MPI_INT, all.data(), counts.data(),
+                   displs.data(), MPI_INT, MPI_COMM_WORLD);
+    for (int r = 0; r < n; r++) {
+        for (int i = 0; i < counts[r]; i++) {
+            if (all[displs[r] + i] != r) {
+                msg.outputData = "allgatherv mismatch";
+                return 2;
+            }
+        }
+    }
+
+    // Alltoallv: rank r sends one int (r*100+dest) to each dest
+    std::vector<int> ones(n, 1);
+    std::vector<int> offs(n);
+    for (int r = 0; r < n; r++) {
+        offs[r] = r;
+    }
+    std::vector<int> sendv(n);
+    for (int d = 0; d < n; d++) {
+        sendv[d] = rank * 100 + d;
+    }
+    std::vector<int> recvv(n, -1);
+    MPI_Alltoallv(sendv.data(), ones.data(), offs.data(), MPI_INT,
+                  recvv.data(), ones.data(), offs.data(), MPI_INT,
+                  MPI_COMM_WORLD);
+    for (int s = 0; s < n; s++) {
+        if (recvv[s] != s * 100 + rank) {
+            msg.outputData = "alltoallv mismatch";
+            return 3;
+        }
+    }
+
+    // Waitall over a ring of isend/irecv
+    int next = (rank + 1) % n;
+    int prev = (rank + n - 1) % n;
+    int out = rank;
+    int in = -1;
+    MPI_Request reqs[2];
+    MPI_Irecv(&in, 1, MPI_INT, prev, 0, MPI_COMM_WORLD, &reqs[0]);
+    MPI_Isend(&out, 1, MPI_INT, next, 0, MPI_COMM_WORLD, &reqs[1]);
+    MPI_Waitall(2, reqs, MPI_STATUSES_IGNORE);
+    if (in != prev) {
+        msg.outputData = "waitall ring mismatch";
+        return 4;
+    }
+
+    // Comm_dup / split_type degenerate paths
+    MPI_Comm dup = nullptr;
+    MPI_Comm_dup(MPI_COMM_WORLD, &dup);
+    int dupRank = -1;
+    MPI_Comm_rank(dup, &dupRank);
+    if (dupRank != rank) {
+        msg.outputData = "comm_dup mismatch";
+        return 5;
+    }
+    MPI_Comm_free(&dup);
+
+    MPI_Barrier(MPI_COMM_WORLD);
+    MPI_Finalize();
+    msg.outputData = "vcollectives ok rank " + std::to_string(rank);
+    return 0;
+}
+
 void registerMpiExampleFunctions()
 {
     auto& reg = FunctionRegistry::get();
     reg.registerFunction("mpi-cpp", "allreduce", exampleAllReduce);
     reg.registerFunction("mpi-cpp", "ring", exampleRing);
     reg.registerFunction("mpi-cpp", "async", exampleAsync);
+    reg.registerFunction("mpi-cpp", "vcollectives", exampleVCollectives);
     reg.registerFunction("mpi-cpp", "allreduce-bench",
                          exampleAllReduceBench);
 }

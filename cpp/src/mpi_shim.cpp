@@ -655,3 +655,331 @@ int MPI_Free_mem(void* base)
     free(base);
     return MPI_SUCCESS;
 }
+
+int MPI_Waitall(int count, MPI_Request* requests, MPI_Status* statuses)
+{
+    for (int i = 0; i < count; i++) {
+        MPI_Wait(&requests[i],
+                 statuses == MPI_STATUSES_IGNORE ? MPI_STATUS_IGNORE
+                                                 : &statuses[i]);
+    }
+    return MPI_SUCCESS;
+}
+
+int MPI_Waitany(int count, MPI_Request* requests, int* index, MPI_Status* status)
+{
+    // Async requests complete in channel order here, so waiting on the
+    // first outstanding request is a valid "any" (reference semantics:
+    // recvBatchReturnLast drains in order)
+    for (int i = 0; i < count; i++) {
+        if (requests[i] >= 0) {
+            MPI_Wait(&requests[i], status);
+            requests[i] = -1;
+            *index = i;
+            return MPI_SUCCESS;
+        }
+    }
+    *index = MPI_UNDEFINED;
+    return MPI_SUCCESS;
+}
+
+int MPI_Request_free(MPI_Request* request)
+{
+    // Matching must stay aligned, so a freed request is drained rather
+    // than abandoned
+    if (request != nullptr && *request >= 0) {
+        MPI_Wait(request, MPI_STATUS_IGNORE);
+        *request = -1;
+    }
+    return MPI_SUCCESS;
+}
+
+int MPI_Gatherv(const void* sendbuf,
+                int sendcount,
+                MPI_Datatype sendtype,
+                void* recvbuf,
+                const int* recvcounts,
+                const int* displs,
+                MPI_Datatype recvtype,
+                int root,
+                MPI_Comm comm)
+{
+    (void)comm;
+    int rank = thisRank();
+    int size = world().getSize();
+    if (rank == root) {
+        for (int r = 0; r < size; r++) {
+            uint8_t* dst =
+              (uint8_t*)recvbuf + (size_t)displs[r] * recvtype->size;
+            if (r == rank) {
+                const uint8_t* src =
+                  sendbuf == MPI_IN_PLACE ? dst : (const uint8_t*)sendbuf;
+                if (src != dst) {
+                    memcpy(dst, src, (size_t)recvcounts[r] * recvtype->size);
+                }
+            } else {
+                world().recv(r, rank, dst, toType(recvtype),
+                             toCount(recvtype, recvcounts[r]),
+                             MpiMessageType::GATHER);
+            }
+        }
+    } else {
+        world().send(rank, root, (const uint8_t*)sendbuf, toType(sendtype),
+                     toCount(sendtype, sendcount), MpiMessageType::GATHER);
+    }
+    return MPI_SUCCESS;
+}
+
+int MPI_Allgatherv(const void* sendbuf,
+                   int sendcount,
+                   MPI_Datatype sendtype,
+                   void* recvbuf,
+                   const int* recvcounts,
+                   const int* displs,
+                   MPI_Datatype recvtype,
+                   MPI_Comm comm)
+{
+    int size = world().getSize();
+    int rc = MPI_Gatherv(sendbuf, sendcount, sendtype, recvbuf, recvcounts,
+                         displs, recvtype, 0, comm);
+    if (rc != MPI_SUCCESS) {
+        return rc;
+    }
+    // Broadcast the fully-gathered buffer (extent = max over ranks)
+    size_t extent = 0;
+    for (int r = 0; r < size; r++) {
+        extent = std::max(
+          extent, (size_t)displs[r] + (size_t)recvcounts[r]);
+    }
+    world().broadcast(0, thisRank(), (uint8_t*)recvbuf, toType(recvtype),
+                      toCount(recvtype, (int)extent),
+                      MpiMessageType::BROADCAST);
+    return MPI_SUCCESS;
+}
+
+int MPI_Alltoallv(const void* sendbuf,
+                  const int* sendcounts,
+                  const int* sdispls,
+                  MPI_Datatype sendtype,
+                  void* recvbuf,
+                  const int* recvcounts,
+                  const int* rdispls,
+                  MPI_Datatype recvtype,
+                  MPI_Comm comm)
+{
+    (void)comm;
+    int rank = thisRank();
+    int size = world().getSize();
+    // Post all irecvs, then send, then drain (pairwise exchange; the
+    // reference's allToAll is also direct N^2 sends)
+    std::vector<int> reqs;
+    reqs.reserve(size);
+    for (int r = 0; r < size; r++) {
+        if (r == rank) {
+            continue;
+        }
+        uint8_t* dst = (uint8_t*)recvbuf + (size_t)rdispls[r] * recvtype->size;
+        reqs.push_back(world().irecv(r, rank, dst, toType(recvtype),
+                                     toCount(recvtype, recvcounts[r]),
+                                     MpiMessageType::ALLTOALL));
+    }
+    for (int r = 0; r < size; r++) {
+        const uint8_t* src =
+          (const uint8_t*)sendbuf + (size_t)sdispls[r] * sendtype->size;
+        if (r == rank) {
+            uint8_t* dst =
+              (uint8_t*)recvbuf + (size_t)rdispls[r] * recvtype->size;
+            memcpy(dst, src, (size_t)recvcounts[r] * recvtype->size);
+            continue;
+        }
+        world().send(rank, r, src, toType(sendtype),
+                     toCount(sendtype, sendcounts[r]),
+                     MpiMessageType::ALLTOALL);
+    }
+    for (int req : reqs) {
+        world().awaitAsyncRequest(req);
+    }
+    return MPI_SUCCESS;
+}
+
+int MPI_Comm_dup(MPI_Comm comm, MPI_Comm* newcomm)
+{
+    *newcomm = comm;
+    return MPI_SUCCESS;
+}
+
+int MPI_Comm_free(MPI_Comm* comm)
+{
+    *comm = MPI_COMM_NULL;
+    return MPI_SUCCESS;
+}
+
+int MPI_Comm_split(MPI_Comm comm, int color, int key, MPI_Comm* newcomm)
+{
+    (void)key;
+    // Single-color split (the degenerate dup) is supported; true
+    // sub-communicators are not part of the reference contract either
+    if (color == 0) {
+        *newcomm = comm;
+        return MPI_SUCCESS;
+    }
+    NOT_IMPLEMENTED("MPI_Comm_split with color != 0");
+}
+
+int MPI_Comm_split_type(MPI_Comm comm,
+                        int split_type,
+                        int key,
+                        MPI_Info info,
+                        MPI_Comm* newcomm)
+{
+    (void)split_type;
+    (void)key;
+    (void)info;
+    // Single node: every rank shares the node, so the split is a dup
+    *newcomm = comm;
+    return MPI_SUCCESS;
+}
+
+int MPI_Comm_create(MPI_Comm comm, MPI_Group group, MPI_Comm* newcomm)
+{
+    (void)comm;
+    (void)group;
+    (void)newcomm;
+    NOT_IMPLEMENTED("MPI_Comm_create");
+}
+
+int MPI_Comm_create_group(MPI_Comm comm,
+                          MPI_Group group,
+                          int tag,
+                          MPI_Comm* newcomm)
+{
+    (void)comm;
+    (void)group;
+    (void)tag;
+    (void)newcomm;
+    NOT_IMPLEMENTED("MPI_Comm_create_group");
+}
+
+int MPI_Comm_group(MPI_Comm comm, MPI_Group* group)
+{
+    (void)comm;
+    *group = 0;
+    return MPI_SUCCESS;
+}
+
+int MPI_Group_incl(MPI_Group group, int n, const int* ranks, MPI_Group* newgroup)
+{
+    (void)group;
+    (void)n;
+    (void)ranks;
+    (void)newgroup;
+    NOT_IMPLEMENTED("MPI_Group_incl");
+}
+
+int MPI_Group_free(MPI_Group* group)
+{
+    *group = MPI_GROUP_NULL;
+    return MPI_SUCCESS;
+}
+
+int MPI_Op_create(void* user_fn, int commute, MPI_Op* op)
+{
+    (void)user_fn;
+    (void)commute;
+    (void)op;
+    NOT_IMPLEMENTED("MPI_Op_create");
+}
+
+int MPI_Op_free(MPI_Op* op)
+{
+    *op = MPI_OP_NULL;
+    return MPI_SUCCESS;
+}
+
+// One-sided RMA: header-only in the reference too (no MpiWorld
+// implementation, reference src/mpi/MpiWorld.cpp) — reject loudly
+int MPI_Win_create(void* base,
+                   MPI_Aint size,
+                   int disp_unit,
+                   MPI_Info info,
+                   MPI_Comm comm,
+                   MPI_Win* win)
+{
+    (void)base; (void)size; (void)disp_unit; (void)info; (void)comm;
+    (void)win;
+    NOT_IMPLEMENTED("MPI_Win_create");
+}
+
+int MPI_Win_allocate_shared(MPI_Aint size,
+                            int disp_unit,
+                            MPI_Info info,
+                            MPI_Comm comm,
+                            void* baseptr,
+                            MPI_Win* win)
+{
+    (void)size; (void)disp_unit; (void)info; (void)comm; (void)baseptr;
+    (void)win;
+    NOT_IMPLEMENTED("MPI_Win_allocate_shared");
+}
+
+int MPI_Win_shared_query(MPI_Win win,
+                         int rank,
+                         MPI_Aint* size,
+                         int* disp_unit,
+                         void* baseptr)
+{
+    (void)win; (void)rank; (void)size; (void)disp_unit; (void)baseptr;
+    NOT_IMPLEMENTED("MPI_Win_shared_query");
+}
+
+int MPI_Win_get_attr(MPI_Win win,
+                     int win_keyval,
+                     void* attribute_val,
+                     int* flag)
+{
+    (void)win; (void)win_keyval; (void)attribute_val; (void)flag;
+    NOT_IMPLEMENTED("MPI_Win_get_attr");
+}
+
+int MPI_Win_fence(int assert_arg, MPI_Win win)
+{
+    (void)assert_arg;
+    (void)win;
+    NOT_IMPLEMENTED("MPI_Win_fence");
+}
+
+int MPI_Win_free(MPI_Win* win)
+{
+    (void)win;
+    NOT_IMPLEMENTED("MPI_Win_free");
+}
+
+int MPI_Get(void* origin_addr,
+            int origin_count,
+            MPI_Datatype origin_datatype,
+            int target_rank,
+            MPI_Aint target_disp,
+            int target_count,
+            MPI_Datatype target_datatype,
+            MPI_Win win)
+{
+    (void)origin_addr; (void)origin_count; (void)origin_datatype;
+    (void)target_rank; (void)target_disp; (void)target_count;
+    (void)target_datatype; (void)win;
+    NOT_IMPLEMENTED("MPI_Get");
+}
+
+int MPI_Put(const void* origin_addr,
+            int origin_count,
+            MPI_Datatype origin_datatype,
+            int target_rank,
+            MPI_Aint target_disp,
+            int target_count,
+            MPI_Datatype target_datatype,
+            MPI_Win win)
+{
+    (void)origin_addr; (void)origin_count; (void)origin_datatype;
+    (void)target_rank; (void)target_disp; (void)target_count;
+    (void)target_datatype; (void)win;
+    NOT_IMPLEMENTED("MPI_Put");
+}

@@ -135,7 +135,8 @@ def test_cpp_mpi_examples(runtime):
     """C++ MPI programs written against the MPI_* shim (Appendix A
     surface) run as native functions."""
     _core.register_mpi_example_functions()
-    for func in ("allreduce", "ring", "async", "allreduce-bench"):
+    for func in ("allreduce", "ring", "async", "allreduce-bench",
+                 "vcollectives"):
         results = submit_mpi_batch("mpi-cpp", func, WORLD_SIZE)
         assert len(results) == WORLD_SIZE
         for r in results:
