@@ -1,3 +1,9 @@
+"""RCCL bootstrap probe: 2 MPI ranks on ONE GPU. Expected outcome on a
+1-GPU box is a clean `RCCL error: invalid usage` (duplicate device) from
+both ranks, which proves the cross-process ncclUniqueId handshake and
+RCCL rendezvous work and that comm-init failures propagate as loud task
+failures instead of hangs. On a multi-GPU node each rank binds its own
+device and init succeeds (driver scale run)."""
 import os, sys, multiprocessing as mp
 sys.path.insert(0, "/root/repo")
 
