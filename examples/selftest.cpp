@@ -132,7 +132,7 @@ int main()
     }
 
     // 3. MPI world, all example programs (host data plane)
-    for (const char* fn : { "allreduce", "ring", "async" }) {
+    for (const char* fn : { "allreduce", "ring", "async", "vcollectives" }) {
         auto ber = std::make_shared<BatchExecuteRequest>(
           batchExecFactory("mpi-cpp", fn, 1));
         ber->messages[0].isMpi = true;
