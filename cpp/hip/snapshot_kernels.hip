@@ -90,6 +90,33 @@ __global__ __launch_bounds__(FAM_KERNEL_BLOCK) void xorBufferKernel(
 //
 // Consumers (apply / ship) address the payload via pageIdx[slot] → page.
 // ---------------------------------------------------------------------------
+typedef unsigned int nt_u32v4 __attribute__((ext_vector_type(4)));
+
+template<bool NT>
+__device__ inline uint4 ldVec(const uint4* p)
+{
+    if constexpr (NT) {
+        nt_u32v4 v =
+          __builtin_nontemporal_load((const nt_u32v4*)p);
+        return make_uint4(v.x, v.y, v.z, v.w);
+    }
+    return *p;
+}
+
+template<bool NT>
+__device__ inline void stVec(uint4* p, uint4 v)
+{
+    if constexpr (NT) {
+        nt_u32v4 w = { v.x, v.y, v.z, v.w };
+        __builtin_nontemporal_store(w, (nt_u32v4*)p);
+    } else {
+        *p = v;
+    }
+}
+
+// The 4 GiB-scale working sets never fit in L2 (32 MiB/XCD), so the NT
+// instantiation streams through with slc-tagged loads/stores
+template<bool NT>
 __global__ __launch_bounds__(FAM_KERNEL_BLOCK) void diffXorPagesKernel(
   const uint4* __restrict__ snap,
   const uint4* __restrict__ cur,
@@ -105,14 +132,14 @@ __global__ __launch_bounds__(FAM_KERNEL_BLOCK) void diffXorPagesKernel(
 
     for (u32 page = waveId; page < nPages; page += nWaves) {
         u64 base = (u64)page * vecsPerPage + lane;
-        uint4 s0 = snap[base];
-        uint4 s1 = snap[base + 64];
-        uint4 s2 = snap[base + 128];
-        uint4 s3 = snap[base + 192];
-        uint4 c0 = cur[base];
-        uint4 c1 = cur[base + 64];
-        uint4 c2 = cur[base + 128];
-        uint4 c3 = cur[base + 192];
+        uint4 s0 = ldVec<NT>(&snap[base]);
+        uint4 s1 = ldVec<NT>(&snap[base + 64]);
+        uint4 s2 = ldVec<NT>(&snap[base + 128]);
+        uint4 s3 = ldVec<NT>(&snap[base + 192]);
+        uint4 c0 = ldVec<NT>(&cur[base]);
+        uint4 c1 = ldVec<NT>(&cur[base + 64]);
+        uint4 c2 = ldVec<NT>(&cur[base + 128]);
+        uint4 c3 = ldVec<NT>(&cur[base + 192]);
         bool diff = neq16(s0, c0) || neq16(s1, c1) || neq16(s2, c2) ||
                     neq16(s3, c3);
         u64 mask = __ballot(diff);
@@ -122,14 +149,14 @@ __global__ __launch_bounds__(FAM_KERNEL_BLOCK) void diffXorPagesKernel(
         if (lane == 0) {
             atomicOr(&bitmap[page >> 5], 1u << (page & 31));
         }
-        payloadOut[base] =
-          make_uint4(s0.x ^ c0.x, s0.y ^ c0.y, s0.z ^ c0.z, s0.w ^ c0.w);
-        payloadOut[base + 64] =
-          make_uint4(s1.x ^ c1.x, s1.y ^ c1.y, s1.z ^ c1.z, s1.w ^ c1.w);
-        payloadOut[base + 128] =
-          make_uint4(s2.x ^ c2.x, s2.y ^ c2.y, s2.z ^ c2.z, s2.w ^ c2.w);
-        payloadOut[base + 192] =
-          make_uint4(s3.x ^ c3.x, s3.y ^ c3.y, s3.z ^ c3.z, s3.w ^ c3.w);
+        stVec<NT>(&payloadOut[base],
+          make_uint4(s0.x ^ c0.x, s0.y ^ c0.y, s0.z ^ c0.z, s0.w ^ c0.w));
+        stVec<NT>(&payloadOut[base + 64],
+          make_uint4(s1.x ^ c1.x, s1.y ^ c1.y, s1.z ^ c1.z, s1.w ^ c1.w));
+        stVec<NT>(&payloadOut[base + 128],
+          make_uint4(s2.x ^ c2.x, s2.y ^ c2.y, s2.z ^ c2.z, s2.w ^ c2.w));
+        stVec<NT>(&payloadOut[base + 192],
+          make_uint4(s3.x ^ c3.x, s3.y ^ c3.y, s3.z ^ c3.z, s3.w ^ c3.w));
     }
 }
 
@@ -183,6 +210,7 @@ __global__ __launch_bounds__(FAM_KERNEL_BLOCK) void compactPageIdxKernel(
 // page, the diff kernel's output); compact=1: payload is slot-compacted
 // (the shippable wire form).
 // ---------------------------------------------------------------------------
+template<bool NT>
 __global__ __launch_bounds__(FAM_KERNEL_BLOCK) void applyXorPagesKernel(
   uint4* __restrict__ snap,
   const u32* __restrict__ pageIdx,
@@ -202,10 +230,10 @@ __global__ __launch_bounds__(FAM_KERNEL_BLOCK) void applyXorPagesKernel(
         u64 src = compact ? (u64)slot * vecsPerPage + lane : dst;
 #pragma unroll
         for (int k = 0; k < 4; k++) {
-            uint4 p = payload[src + k * 64];
-            uint4 s = snap[dst + k * 64];
-            snap[dst + k * 64] =
-              make_uint4(s.x ^ p.x, s.y ^ p.y, s.z ^ p.z, s.w ^ p.w);
+            uint4 p = ldVec<NT>(&payload[src + k * 64]);
+            uint4 s = ldVec<NT>(&snap[dst + k * 64]);
+            stVec<NT>(&snap[dst + k * 64],
+              make_uint4(s.x ^ p.x, s.y ^ p.y, s.z ^ p.z, s.w ^ p.w));
         }
     }
 }
@@ -365,16 +393,33 @@ hipError_t famDiffXorPages(const void* snap,
     uint32_t blocks = (nPages + wavesPerBlock - 1) / wavesPerBlock;
     uint32_t grid = blocks < FAM_MAX_BLOCKS ? (blocks ? blocks : 1)
                                             : FAM_MAX_BLOCKS;
-    hipLaunchKernelGGL(diffXorPagesKernel,
-                       dim3(grid),
-                       dim3(FAM_KERNEL_BLOCK),
-                       0,
-                       stream,
-                       (const uint4*)snap,
-                       (const uint4*)cur,
-                       nPages,
-                       bitmapDev,
-                       (uint4*)payloadDev);
+    static const bool useNt = []() {
+        const char* e = getenv("FAM_DIFF_NT");
+        return e == nullptr || e[0] != '0'; // default on
+    }();
+    if (useNt) {
+        hipLaunchKernelGGL(diffXorPagesKernel<true>,
+                           dim3(grid),
+                           dim3(FAM_KERNEL_BLOCK),
+                           0,
+                           stream,
+                           (const uint4*)snap,
+                           (const uint4*)cur,
+                           nPages,
+                           bitmapDev,
+                           (uint4*)payloadDev);
+    } else {
+        hipLaunchKernelGGL(diffXorPagesKernel<false>,
+                           dim3(grid),
+                           dim3(FAM_KERNEL_BLOCK),
+                           0,
+                           stream,
+                           (const uint4*)snap,
+                           (const uint4*)cur,
+                           nPages,
+                           bitmapDev,
+                           (uint4*)payloadDev);
+    }
     hipError_t err = hipGetLastError();
     if (err != hipSuccess) {
         return err;
@@ -411,7 +456,13 @@ hipError_t famApplyXorPagesEx(void* snap,
     uint32_t blocks = (nDirty + wavesPerBlock - 1) / wavesPerBlock;
     uint32_t grid = blocks < FAM_MAX_BLOCKS ? (blocks ? blocks : 1)
                                             : FAM_MAX_BLOCKS;
-    hipLaunchKernelGGL(applyXorPagesKernel,
+    static const bool useNtApply = []() {
+        const char* e = getenv("FAM_APPLY_NT");
+        return e == nullptr || e[0] != '0'; // default on
+    }();
+    auto* kern =
+      useNtApply ? applyXorPagesKernel<true> : applyXorPagesKernel<false>;
+    hipLaunchKernelGGL(kern,
                        dim3(grid),
                        dim3(FAM_KERNEL_BLOCK),
                        0,
