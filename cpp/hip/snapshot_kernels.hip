@@ -40,56 +40,6 @@ __device__ __forceinline__ bool neq16(const uint4& a, const uint4& b)
 // Dirty-page bitmap: flags[p] = 1 where any byte of 4 KiB page p differs.
 // Each block iteration covers one page: 256 lanes × 16 B.
 // ---------------------------------------------------------------------------
-__global__ __launch_bounds__(FAM_KERNEL_BLOCK) void dirtyPagesKernel(
-  const uint4* __restrict__ snap,
-  const uint4* __restrict__ cur,
-  u32 nPages,
-  u32* __restrict__ flags)
-{
-    const u32 vecsPerPage = FAM_PAGE / 16; // 256
-    for (u32 page = blockIdx.x; page < nPages; page += gridDim.x) {
-        u64 v = (u64)page * vecsPerPage + threadIdx.x;
-        bool diff = neq16(snap[v], cur[v]);
-        u64 mask = __ballot(diff);
-        if ((threadIdx.x & 63) == 0 && mask != 0) {
-            atomicOr(&flags[page], 1u);
-        }
-    }
-}
-
-// ---------------------------------------------------------------------------
-// XOR delta over a whole buffer: out = a ^ b (16 B/lane, grid-stride)
-// ---------------------------------------------------------------------------
-__global__ __launch_bounds__(FAM_KERNEL_BLOCK) void xorBufferKernel(
-  const uint4* __restrict__ a,
-  const uint4* __restrict__ b,
-  uint4* __restrict__ out,
-  u64 nVec)
-{
-    for (u64 i = (u64)blockIdx.x * blockDim.x + threadIdx.x; i < nVec;
-         i += (u64)gridDim.x * blockDim.x) {
-        uint4 x = a[i];
-        uint4 y = b[i];
-        out[i] = make_uint4(x.x ^ y.x, x.y ^ y.y, x.z ^ y.z, x.w ^ y.w);
-    }
-}
-
-// ---------------------------------------------------------------------------
-// Page diff, two-kernel pipeline:
-//
-//  (A) diffXorPagesKernel — one WAVE per page (64 lanes × 4 × 16 B = 4 KiB):
-//      compare, and for dirty pages write the XOR payload SPARSELY at the
-//      page's own slot plus one bit in a page bitmap. The bitmap atomics
-//      are distributed over nPages/32 words, so there is no contention —
-//      a single global ticket counter measured 1.07 TB/s at 100% dirty
-//      (≈11 ns/atomic on one word, the "dequeue" price), the bitmap form
-//      removes that serialization entirely.
-//  (B) compactPageIdxKernel — scan the bitmap (2048 pages per wave
-//      iteration) and emit the compacted dirty-page index list with ONE
-//      ticket atomic per wave-iteration.
-//
-// Consumers (apply / ship) address the payload via pageIdx[slot] → page.
-// ---------------------------------------------------------------------------
 typedef unsigned int nt_u32v4 __attribute__((ext_vector_type(4)));
 
 template<bool NT>
@@ -114,6 +64,57 @@ __device__ inline void stVec(uint4* p, uint4 v)
     }
 }
 
+__global__ __launch_bounds__(FAM_KERNEL_BLOCK) void dirtyPagesKernel(
+  const uint4* __restrict__ snap,
+  const uint4* __restrict__ cur,
+  u32 nPages,
+  u32* __restrict__ flags)
+{
+    const u32 vecsPerPage = FAM_PAGE / 16; // 256
+    for (u32 page = blockIdx.x; page < nPages; page += gridDim.x) {
+        u64 v = (u64)page * vecsPerPage + threadIdx.x;
+        bool diff = neq16(ldVec<true>(&snap[v]), ldVec<true>(&cur[v]));
+        u64 mask = __ballot(diff);
+        if ((threadIdx.x & 63) == 0 && mask != 0) {
+            atomicOr(&flags[page], 1u);
+        }
+    }
+}
+
+// ---------------------------------------------------------------------------
+// XOR delta over a whole buffer: out = a ^ b (16 B/lane, grid-stride)
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(FAM_KERNEL_BLOCK) void xorBufferKernel(
+  const uint4* __restrict__ a,
+  const uint4* __restrict__ b,
+  uint4* __restrict__ out,
+  u64 nVec)
+{
+    for (u64 i = (u64)blockIdx.x * blockDim.x + threadIdx.x; i < nVec;
+         i += (u64)gridDim.x * blockDim.x) {
+        uint4 x = ldVec<true>(&a[i]);
+        uint4 y = ldVec<true>(&b[i]);
+        stVec<true>(&out[i],
+                    make_uint4(x.x ^ y.x, x.y ^ y.y, x.z ^ y.z, x.w ^ y.w));
+    }
+}
+
+// ---------------------------------------------------------------------------
+// Page diff, two-kernel pipeline:
+//
+//  (A) diffXorPagesKernel — one WAVE per page (64 lanes × 4 × 16 B = 4 KiB):
+//      compare, and for dirty pages write the XOR payload SPARSELY at the
+//      page's own slot plus one bit in a page bitmap. The bitmap atomics
+//      are distributed over nPages/32 words, so there is no contention —
+//      a single global ticket counter measured 1.07 TB/s at 100% dirty
+//      (≈11 ns/atomic on one word, the "dequeue" price), the bitmap form
+//      removes that serialization entirely.
+//  (B) compactPageIdxKernel — scan the bitmap (2048 pages per wave
+//      iteration) and emit the compacted dirty-page index list with ONE
+//      ticket atomic per wave-iteration.
+//
+// Consumers (apply / ship) address the payload via pageIdx[slot] → page.
+// ---------------------------------------------------------------------------
 // The 4 GiB-scale working sets never fit in L2 (32 MiB/XCD), so the NT
 // instantiation streams through with slc-tagged loads/stores
 template<bool NT>
@@ -258,7 +259,8 @@ __global__ __launch_bounds__(FAM_KERNEL_BLOCK) void gatherPagesKernel(
         u64 dst = (u64)slot * vecsPerPage + lane;
 #pragma unroll
         for (int k = 0; k < 4; k++) {
-            out[dst + k * 64] = payload[src + k * 64];
+            stVec<true>(&out[dst + k * 64],
+                        ldVec<true>(&payload[src + k * 64]));
         }
     }
 }
