@@ -175,6 +175,8 @@ def main():
         per_rank_batch = [parse_times(r.output_data, "batch")
                           for r in results]
         rank0_pp = parse_times(results[0].output_data, "pp")
+        per_rank_ring = [parse_times(r.output_data, "ring")
+                         for r in results]
         # Per-rank snapshot diff/apply GB/s (config 4; 0 on CPU)
         snap_diff = [parse_times(r.output_data, "snapdiff") for r in results]
         snap_apply = [parse_times(r.output_data, "snapapply") for r in results]
@@ -201,6 +203,12 @@ def main():
         )
         # Round trip moves 2 x pp_bytes through the link
         pp_gbps = 2 * pp / pp_mean_s / 1e9 if pp_mean_s > 0 else 0.0
+        # Ring step: slowest rank's sendRecv of pp bytes each way
+        ring_ms = [max(s[i] for s in per_rank_ring if s)
+                   for i in range(k)] if n >= 2 and pp > 0 else []
+        ring_mean_s = (sum(ring_ms) / len(ring_ms)) / 1000.0 if ring_ms \
+            else 0.0
+        ring_gbps = 2 * pp / ring_mean_s / 1e9 if ring_mean_s > 0 else 0.0
 
         result = {
             "metric": "MPI_Allreduce GB/s + batch-exec msgs/sec, "
@@ -228,6 +236,7 @@ def main():
                 "allreduce_ms": round(ar_mean_s * 1000.0, 3),
                 "pingpong_bytes": pp,
                 "pingpong_gbps": round(pp_gbps, 2),
+                "ring_exchange_gbps_per_rank": round(ring_gbps, 2),
                 "batch_per_host": args.batch,
                 "batch_msgs_per_sec": round(msgs_per_sec, 2),
                 "batch_ms": round(batch_mean_ms, 3),

@@ -137,6 +137,7 @@ static int32_t benchRankStep(Message& msg)
     std::vector<double> allreduceMs;
     std::vector<double> batchMs;
     std::vector<double> pingpongMs;
+    std::vector<double> ringMs;
 
     // Config 4 on this rank's GPU: one diff+merge pass of a snapBytes
     // region at 25% dirty pages, measured before the timed steps
@@ -217,6 +218,19 @@ static int32_t benchRankStep(Message& msg)
         }
         int64_t tPp1 = getEpochMicros();
 
+        // Ring exchange over xGMI (the sendRecv ring is the
+        // primitive a ring-attention/sequence-parallel layer would use):
+        // every rank sendRecvs ppBytes to next/from prev in one step
+        int64_t tRing0 = getEpochMicros();
+        if (ppBytes > 0 && worldSize >= 2) {
+            int next = (rank + 1) % worldSize;
+            int prev = (rank + worldSize - 1) % worldSize;
+            world.sendRecv(sendBuf, (int)ppBytes, MpiDataType::BYTE, next,
+                           recvBuf, (int)ppBytes, MpiDataType::BYTE, prev,
+                           rank);
+        }
+        int64_t tRing1 = getEpochMicros();
+
         // Batch-throughput half of the composite step: rank 0 submits a
         // batch of kvtouch functions across all hosts and waits
         int64_t tBatch0 = getEpochMicros();
@@ -251,6 +265,7 @@ static int32_t benchRankStep(Message& msg)
             allreduceMs.push_back((tAr - t0) / 1000.0);
             batchMs.push_back((t1 - tBatch0) / 1000.0);
             pingpongMs.push_back((tPp1 - tPp0) / 1000.0);
+            ringMs.push_back((tRing1 - tRing0) / 1000.0);
         }
     }
 
@@ -273,6 +288,10 @@ static int32_t benchRankStep(Message& msg)
         out << (i ? "," : "") << allreduceMs[i];
     }
     out << ";snapdiff:" << snapDiffGbps << ";snapapply:" << snapApplyGbps;
+    out << ";ring:";
+    for (size_t i = 0; i < ringMs.size(); i++) {
+        out << (i ? "," : "") << ringMs[i];
+    }
     out << ";pp:";
     for (size_t i = 0; i < pingpongMs.size(); i++) {
         out << (i ? "," : "") << pingpongMs[i];
