@@ -228,6 +228,27 @@ void initOpsBindings(py::module_& m)
                py::gil_scoped_release release;
                return s.diffXor((const void*)ptr);
            })
+      .def("gather_last_diff",
+           [](DeviceSnapshot& s) {
+               std::vector<uint32_t> pages;
+               std::vector<uint8_t> payload;
+               {
+                   py::gil_scoped_release release;
+                   s.gatherLastDiffToHost(pages, payload);
+               }
+               return py::make_tuple(
+                 pages,
+                 py::bytes((const char*)payload.data(), payload.size()));
+           })
+      .def("apply_compact_diff",
+           [](DeviceSnapshot& s,
+              const std::vector<uint32_t>& pages,
+              const py::bytes& payload) {
+               std::string data = payload;
+               py::gil_scoped_release release;
+               s.applyCompactDiffFromHost(
+                 pages, (const uint8_t*)data.data(), data.size());
+           })
       .def("apply_last_diff", [](DeviceSnapshot& s) {
           py::gil_scoped_release release;
           s.applyLastDiff();

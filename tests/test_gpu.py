@@ -306,3 +306,39 @@ def test_ptp_device_payloads(runtime):
     assert got == 1024
     torch.cuda.synchronize()
     assert int(buf.sum()) == 0x42 * 1024
+
+
+@pytest.mark.gpu
+def test_device_snapshot_wire_diff_ship():
+    """Peer-GPU snapshot shipping contract: diff on snapshot A, gather the
+    compact wire form ({pages, slot-compacted 4KiB payloads}) to the host,
+    apply it onto an identical snapshot B (reference parity:
+    SnapshotClient::pushSnapshotUpdate shipping diffs between hosts,
+    src/snapshot/SnapshotClient.cpp)."""
+    import torch
+
+    from faabric_amd import _core
+
+    n = 64 * 4096
+    base = torch.randint(0, 256, (n,), dtype=torch.uint8, device="cuda")
+    snap_a = _core.DeviceSnapshot(n, 0)
+    snap_b = _core.DeviceSnapshot(n, 0)
+    snap_a.capture_from_ptr(base.data_ptr())
+    snap_b.capture_from_ptr(base.data_ptr())
+
+    # Mutate a few pages
+    updated = base.clone()
+    updated[5 * 4096 + 3] ^= 0xFF
+    updated[17 * 4096 : 18 * 4096] = 0xAB
+    updated[63 * 4096 + 4095] ^= 0x01
+    torch.cuda.synchronize()
+
+    nd = snap_a.diff_xor(updated.data_ptr())
+    assert nd == 3, nd
+    pages, payload = snap_a.gather_last_diff()
+    assert sorted(pages) == [5, 17, 63]
+    assert len(payload) == 3 * 4096
+
+    snap_b.apply_compact_diff(pages, payload)
+    out = snap_b.copy_out_host(n)
+    assert out == updated.cpu().numpy().tobytes()
