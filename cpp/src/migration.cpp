@@ -3,6 +3,7 @@
 // mpiMigrationPoint — in Faasm this glue lives in the WASM host layer;
 // here it is part of the runtime so any registered function can call it).
 #include "faabricamd/executor.h"
+#include <hip/hip_runtime.h>
 #include "faabricamd/mpi.h"
 #include "faabricamd/planner.h"
 #include "faabricamd/ptp.h"
@@ -35,15 +36,34 @@ int32_t migrationPoint(const std::vector<uint8_t>& reentryInput)
         call.inputData = reentryInput;
         call.snapshotKey = snapKey;
 
-        auto [base, size] = exec->getMemoryView();
-        auto snap = std::make_shared<SnapshotData>(
-          std::vector<uint8_t>(base, base + size));
-        SnapshotRegistry::get().registerSnapshot(snapKey, snap);
-        try {
-            getSnapshotClient(getSystemConfig().plannerHost)
-              ->pushSnapshot(snapKey, *snap);
-        } catch (const std::exception& e) {
-            FAM_ERROR("freeze snapshot push failed: %s", e.what());
+        if (exec->hasDeviceArena()) {
+            // GPU-resident function: freeze the HBM arena. It travels as
+            // bytes and lands as a DeviceSnapshot wherever the app thaws
+            // (snapshot.cpp PushSnapshot onDevice path)
+            auto [dbase, dsize] = exec->getDeviceMemoryView();
+            std::vector<uint8_t> hostCopy(dsize);
+            if (hipMemcpy(hostCopy.data(), dbase, dsize,
+                          hipMemcpyDeviceToHost) != hipSuccess) {
+                FAM_ERROR("freeze: device arena copy-out failed");
+            }
+            try {
+                getSnapshotClient(getSystemConfig().plannerHost)
+                  ->pushDeviceSnapshot(snapKey, hostCopy.data(), dsize);
+            } catch (const std::exception& e) {
+                FAM_ERROR("freeze device snapshot push failed: %s",
+                          e.what());
+            }
+        } else {
+            auto [base, size] = exec->getMemoryView();
+            auto snap = std::make_shared<SnapshotData>(
+              std::vector<uint8_t>(base, base + size));
+            SnapshotRegistry::get().registerSnapshot(snapKey, snap);
+            try {
+                getSnapshotClient(getSystemConfig().plannerHost)
+                  ->pushSnapshot(snapKey, *snap);
+            } catch (const std::exception& e) {
+                FAM_ERROR("freeze snapshot push failed: %s", e.what());
+            }
         }
         if (call.isMpi &&
             MpiWorldRegistry::get().worldExists(call.mpiWorldId)) {
@@ -92,13 +112,28 @@ int32_t migrationPoint(const std::vector<uint8_t>& reentryInput)
     }
 
     std::string snapKey = "migration_" + std::to_string(msg.id);
-    auto [base, size] = exec->getMemoryView();
-    if (size > 0) {
-        auto snap = std::make_shared<SnapshotData>(
-          std::vector<uint8_t>(base, base + size));
-        SnapshotRegistry::get().registerSnapshot(snapKey, snap);
-        getSnapshotClient(migration->dstHost)->pushSnapshot(snapKey, *snap);
+    if (exec->hasDeviceArena()) {
+        // Ship the HBM arena; the destination lands it in ITS GPU and
+        // restore() D2D-copies it into the fresh executor's arena
+        auto [dbase, dsize] = exec->getDeviceMemoryView();
+        std::vector<uint8_t> hostCopy(dsize);
+        if (hipMemcpy(hostCopy.data(), dbase, dsize,
+                      hipMemcpyDeviceToHost) != hipSuccess) {
+            throw FaabricException("migration: device copy-out failed");
+        }
+        getSnapshotClient(migration->dstHost)
+          ->pushDeviceSnapshot(snapKey, hostCopy.data(), dsize);
         msg.snapshotKey = snapKey;
+    } else {
+        auto [base, size] = exec->getMemoryView();
+        if (size > 0) {
+            auto snap = std::make_shared<SnapshotData>(
+              std::vector<uint8_t>(base, base + size));
+            SnapshotRegistry::get().registerSnapshot(snapKey, snap);
+            getSnapshotClient(migration->dstHost)
+              ->pushSnapshot(snapKey, *snap);
+            msg.snapshotKey = snapKey;
+        }
     }
 
     FAM_INFO("migrating %s idx %d from %s to %s",

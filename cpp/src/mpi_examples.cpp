@@ -259,6 +259,38 @@ static int32_t exampleVCollectives(Message& msg)
     return 0;
 }
 
+
+// Small-message latency harness: 1000 x 8-int allreduces (reference:
+// tests/dist/mpi/benchmarks/mpi_bench.cpp:18-23, mpi_allreduce.cpp:88-100)
+static int32_t exampleAllReduceSmallBench(Message& msg)
+{
+    MPI_Init(nullptr, nullptr);
+    int rank;
+    int worldSize;
+    MPI_Comm_rank(MPI_COMM_WORLD, &rank);
+    MPI_Comm_size(MPI_COMM_WORLD, &worldSize);
+
+    const int reps = 1000;
+    int input[8];
+    int output[8];
+    for (int i = 0; i < 8; i++) {
+        input[i] = rank + i;
+    }
+    MPI_Barrier(MPI_COMM_WORLD);
+    double t0 = MPI_Wtime();
+    for (int r = 0; r < reps; r++) {
+        MPI_Allreduce(input, output, 8, MPI_INT, MPI_SUM, MPI_COMM_WORLD);
+    }
+    double elapsed = MPI_Wtime() - t0;
+    MPI_Barrier(MPI_COMM_WORLD);
+    MPI_Finalize();
+    double usPer = elapsed * 1e6 / reps;
+    double perSec = reps / elapsed;
+    msg.outputData = "us_per_op=" + std::to_string(usPer) +
+                     ";ops_per_sec=" + std::to_string(perSec);
+    return 0;
+}
+
 void registerMpiExampleFunctions()
 {
     auto& reg = FunctionRegistry::get();
@@ -266,6 +298,8 @@ void registerMpiExampleFunctions()
     reg.registerFunction("mpi-cpp", "ring", exampleRing);
     reg.registerFunction("mpi-cpp", "async", exampleAsync);
     reg.registerFunction("mpi-cpp", "vcollectives", exampleVCollectives);
+    reg.registerFunction("mpi-cpp", "allreduce-small-bench",
+                         exampleAllReduceSmallBench);
     reg.registerFunction("mpi-cpp", "allreduce-bench",
                          exampleAllReduceBench);
 }
