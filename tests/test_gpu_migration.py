@@ -51,7 +51,8 @@ def _worker_main(port_offset, stop_event, ready_event):
     from faabric_amd.runtime import LocalRuntime
 
     _core.set_log_level("error")
-    rt = LocalRuntime(port_offset=port_offset, slots=WORKER_SLOTS)
+    rt = LocalRuntime(port_offset=port_offset, planner_port_offset=5200,
+                      slots=WORKER_SLOTS)
     rt.start_worker()
     _core.register_native_sleep("gmig", "blocker", 600)
     _core.register_function("gmig", "worker", _gpu_mig_fn)
