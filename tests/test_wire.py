@@ -106,3 +106,70 @@ def test_ptp_mappings_roundtrip():
     assert len(dec.mappings) == 3
     assert dec.mappings[2].host == "10.0.0.2"
     assert dec.mappings[2].mpi_port == 8022
+
+
+# ---- property-based fuzz (hypothesis): arbitrary field values must
+# round-trip through the hand-written protobuf codec ----
+
+from hypothesis import given, settings, strategies as st  # noqa: E402
+
+_i32 = st.integers(min_value=-(2**31), max_value=2**31 - 1)
+_u31 = st.integers(min_value=0, max_value=2**31 - 1)
+_i64 = st.integers(min_value=-(2**63), max_value=2**63 - 1)
+_text = st.text(max_size=40)
+_blob = st.binary(max_size=256)
+
+
+@settings(max_examples=80, deadline=None)
+@given(
+    user=_text,
+    function=_text,
+    input_data=_blob,
+    output_data=_text,
+    return_value=_i32,
+    group_id=_i32,
+    group_idx=_i32,
+    mpi_rank=_i32,
+    timestamp=_i64,
+    chained=st.lists(_u31, max_size=6),
+    int_details=st.dictionaries(_text, _i32, max_size=4),
+    str_details=st.dictionaries(_text, _text, max_size=4),
+)
+def test_message_fuzz_roundtrip(user, function, input_data, output_data,
+                                return_value, group_id, group_idx,
+                                mpi_rank, timestamp, chained, int_details,
+                                str_details):
+    m = fa.message_factory(user, function)
+    m.input_data = input_data
+    m.output_data = output_data
+    m.return_value = return_value
+    m.group_id = group_id
+    m.group_idx = group_idx
+    m.mpi_rank = mpi_rank
+    m.finish_timestamp = timestamp
+    m.chained_msg_ids = chained
+    m.int_exec_graph_details = int_details
+    m.exec_graph_details = str_details
+
+    dec = fa.Message.decode(m.encode())
+    assert dec.user == user
+    assert dec.function == function
+    assert dec.input_data == input_data
+    assert dec.output_data == output_data
+    assert dec.return_value == return_value
+    assert dec.group_id == group_id
+    assert dec.group_idx == group_idx
+    assert dec.mpi_rank == mpi_rank
+    assert dec.finish_timestamp == timestamp
+    assert dec.chained_msg_ids == chained
+    assert dec.int_exec_graph_details == int_details
+    assert dec.exec_graph_details == str_details
+
+
+@settings(max_examples=40, deadline=None)
+@given(old=st.binary(max_size=20000), new=st.binary(max_size=20000))
+def test_delta_codec_fuzz(old, new):
+    from faabric_amd import _core
+
+    delta = _core.delta_encode(old, new)
+    assert _core.delta_apply(old, delta) == new
