@@ -54,8 +54,9 @@ def parse_args():
     p.add_argument("--kv-bytes", type=int, default=4096)
     p.add_argument("--pp-bytes", type=int, default=64 * 1024 * 1024,
                    help="rank0<->rank1 ping-pong bytes (config 2)")
-    p.add_argument("--snap-bytes", type=int, default=1024 * 1024 * 1024,
-                   help="per-rank GPU snapshot diff+merge region (config 4)")
+    p.add_argument("--snap-bytes", type=int, default=4 * 1024 * 1024 * 1024,
+                   help="per-rank GPU snapshot diff+merge region (config 4: "
+                        "4 GB random-byte region per GPU)")
     return p.parse_args()
 
 
