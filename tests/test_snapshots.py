@@ -145,3 +145,51 @@ def test_registry_lifecycle():
     assert got.get_data() == b"x" * 64
     _core.snapshot_delete("t/reg_1")
     assert not _core.snapshot_exists("t/reg_1")
+
+
+# ---- property test: typed merges vs a pure-python model (reference
+# semantics: util/snapshot.h calculateDiffValue/applyDiffValue — Sum/
+# Subtract/Product ship the delta, Max/Min ship the value) ----
+
+from hypothesis import given, settings, strategies as st  # noqa: E402
+
+_vals = st.lists(
+    st.integers(min_value=-(2**30), max_value=2**30 - 1),
+    min_size=8, max_size=8,
+)
+
+
+@settings(max_examples=60, deadline=None)
+@given(base=_vals, up1=_vals, up2=_vals,
+       op=st.sampled_from(["sum", "max", "min"]))
+def test_int_merge_matrix_vs_model(base, up1, up2, op):
+    opmap = {"sum": Sum, "max": Max, "min": Min}
+    raw = struct.pack("<8i", *base)
+    snap = make_snap(raw)
+    snap.add_merge_region(0, 32, Int, opmap[op])
+
+    diffs = [snap.diff_with_memory(struct.pack("<8i", *u))
+             for u in (up1, up2)]
+    for d in diffs:
+        snap.queue_diffs(d)
+    snap.write_queued_diffs()
+    got = list(struct.unpack("<8i", snap.get_data()))
+
+    # Pure-python model of the reference semantics
+    expect = list(base)
+    for i in range(8):
+        if op == "sum":
+            # Each thread contributes its delta; overflow wraps at i32
+            v = base[i] + (up1[i] - base[i]) + (up2[i] - base[i])
+            v = (v + 2**31) % 2**32 - 2**31
+            expect[i] = v
+        elif op == "max":
+            # Only values differing from base ship; merge keeps the max
+            for u in (up1, up2):
+                if u[i] != base[i] and u[i] > expect[i]:
+                    expect[i] = u[i]
+        elif op == "min":
+            for u in (up1, up2):
+                if u[i] != base[i] and u[i] < expect[i]:
+                    expect[i] = u[i]
+    assert got == expect, (op, base, up1, up2)
