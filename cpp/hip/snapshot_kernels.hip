@@ -99,6 +99,21 @@ __global__ __launch_bounds__(FAM_KERNEL_BLOCK) void xorBufferKernel(
     }
 }
 
+// Plain device-to-device copy, nontemporal both sides: hipMemcpy D2D
+// measures ~1.9 TB/s on MI355X while a grid-strided NT copy approaches
+// the 2-unit HBM bound (~3.15 TB/s) — used for the single-rank
+// collective short-circuits (MpiWorld)
+__global__ __launch_bounds__(FAM_KERNEL_BLOCK) void copyBufferKernel(
+  const uint4* __restrict__ src,
+  uint4* __restrict__ dst,
+  u64 nVec)
+{
+    for (u64 i = (u64)blockIdx.x * blockDim.x + threadIdx.x; i < nVec;
+         i += (u64)gridDim.x * blockDim.x) {
+        stVec<true>(&dst[i], ldVec<true>(&src[i]));
+    }
+}
+
 // ---------------------------------------------------------------------------
 // Page diff, two-kernel pipeline:
 //
@@ -359,6 +374,28 @@ hipError_t famDirtyPages(const void* snap,
                        (const uint4*)cur,
                        nPages,
                        flagsDev);
+    return hipGetLastError();
+}
+
+hipError_t famCopyBuffer(const void* src,
+                         void* dst,
+                         uint64_t bytes,
+                         hipStream_t stream)
+{
+    if ((bytes % 16) != 0) {
+        // Tail not vectorisable: fall back to the runtime copy
+        return hipMemcpyAsync(dst, src, bytes, hipMemcpyDeviceToDevice,
+                              stream);
+    }
+    uint64_t nVec = bytes / 16;
+    hipLaunchKernelGGL(copyBufferKernel,
+                       dim3(gridFor(nVec)),
+                       dim3(FAM_KERNEL_BLOCK),
+                       0,
+                       stream,
+                       (const uint4*)src,
+                       (uint4*)dst,
+                       nVec);
     return hipGetLastError();
 }
 

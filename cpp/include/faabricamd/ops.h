@@ -30,6 +30,12 @@ hipError_t famDirtyPages(const void* snap,
                          uint64_t bytes,
                          uint32_t* flagsDev,
                          hipStream_t stream);
+// NT device-to-device copy (beats hipMemcpy D2D for large HBM buffers)
+hipError_t famCopyBuffer(const void* src,
+                         void* dst,
+                         uint64_t bytes,
+                         hipStream_t stream);
+
 hipError_t famXorBuffer(const void* a,
                         const void* b,
                         void* out,

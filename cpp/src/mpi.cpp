@@ -5,6 +5,7 @@
 // the MI355X re-design notes (host plane on the PTP broker, device plane
 // on RCCL over xGMI).
 #include "faabricamd/mpi.h"
+#include "faabricamd/ops.h"
 #include "faabricamd/executor.h"
 #include "faabricamd/planner.h"
 #include "faabricamd/ptp.h"
@@ -784,10 +785,10 @@ void MpiWorld::allGather(int thisRank,
     if (size == 1 && isDeviceBuffer(recvBuffer, loc)) {
         size_t bytes = mpiTypeSize(dataType) * (size_t)count;
         if (recvBuffer != sendBuffer) {
-            // D2D memcpy is asynchronous to the host: synchronise so the
-            // collective keeps blocking semantics (and honest timings)
-            HIP_CHECK(hipMemcpy(recvBuffer, sendBuffer, bytes,
-                                hipMemcpyDeviceToDevice));
+            // NT copy kernel beats hipMemcpy D2D for HBM buffers; sync
+            // so the collective keeps blocking semantics (honest timings)
+            HIP_CHECK(famCopyBuffer(sendBuffer, recvBuffer, bytes,
+                                    nullptr));
             HIP_CHECK(hipStreamSynchronize(nullptr));
         }
         return;
@@ -830,10 +831,10 @@ void MpiWorld::reduce(int thisRank,
     if (size == 1 && isDeviceBuffer(sendBuffer, loc)) {
         size_t bytes = mpiTypeSize(dataType) * (size_t)count;
         if (recvBuffer != sendBuffer) {
-            // D2D memcpy is asynchronous to the host: synchronise so the
-            // collective keeps blocking semantics (and honest timings)
-            HIP_CHECK(hipMemcpy(recvBuffer, sendBuffer, bytes,
-                                hipMemcpyDeviceToDevice));
+            // NT copy kernel beats hipMemcpy D2D for HBM buffers; sync
+            // so the collective keeps blocking semantics (honest timings)
+            HIP_CHECK(famCopyBuffer(sendBuffer, recvBuffer, bytes,
+                                    nullptr));
             HIP_CHECK(hipStreamSynchronize(nullptr));
         }
         return;
@@ -888,10 +889,10 @@ void MpiWorld::allReduce(int thisRank,
         // Single-rank collective = local copy; skip the RCCL machinery
         size_t bytes = mpiTypeSize(dataType) * (size_t)count;
         if (recvBuffer != sendBuffer) {
-            // D2D memcpy is asynchronous to the host: synchronise so the
-            // collective keeps blocking semantics (and honest timings)
-            HIP_CHECK(hipMemcpy(recvBuffer, sendBuffer, bytes,
-                                hipMemcpyDeviceToDevice));
+            // NT copy kernel beats hipMemcpy D2D for HBM buffers; sync
+            // so the collective keeps blocking semantics (honest timings)
+            HIP_CHECK(famCopyBuffer(sendBuffer, recvBuffer, bytes,
+                                    nullptr));
             HIP_CHECK(hipStreamSynchronize(nullptr));
         }
         return;
@@ -938,10 +939,10 @@ void MpiWorld::allToAll(int thisRank,
     size_t bytes = mpiTypeSize(dataType) * (size_t)count;
     if (size == 1 && isDeviceBuffer(sendBuffer, loc)) {
         if (recvBuffer != sendBuffer) {
-            // D2D memcpy is asynchronous to the host: synchronise so the
-            // collective keeps blocking semantics (and honest timings)
-            HIP_CHECK(hipMemcpy(recvBuffer, sendBuffer, bytes,
-                                hipMemcpyDeviceToDevice));
+            // NT copy kernel beats hipMemcpy D2D for HBM buffers; sync
+            // so the collective keeps blocking semantics (honest timings)
+            HIP_CHECK(famCopyBuffer(sendBuffer, recvBuffer, bytes,
+                                    nullptr));
             HIP_CHECK(hipStreamSynchronize(nullptr));
         }
         return;
