@@ -265,6 +265,31 @@ void initOpsBindings(py::module_& m)
                 (void*)inout, (const void*)in, count, dtype, op);
           });
 
+    // Direct famCopyBuffer bandwidth: returns GB/s (2-unit traffic)
+    m.def("bench_copy", [](size_t bytes, int iters) {
+        py::gil_scoped_release release;
+        uint8_t* src = nullptr;
+        uint8_t* dst = nullptr;
+        if (hipMalloc(&src, bytes) != hipSuccess ||
+            hipMalloc(&dst, bytes) != hipSuccess) {
+            throw FaabricException("bench_copy alloc failed");
+        }
+        (void)hipMemset(src, 0x5a, bytes);
+        (void)hipDeviceSynchronize();
+        // warmup
+        (void)famCopyBuffer(src, dst, bytes, nullptr);
+        (void)hipDeviceSynchronize();
+        int64_t t0 = getEpochMicros();
+        for (int i = 0; i < iters; i++) {
+            (void)famCopyBuffer(src, dst, bytes, nullptr);
+        }
+        (void)hipDeviceSynchronize();
+        int64_t t1 = getEpochMicros();
+        (void)hipFree(src);
+        (void)hipFree(dst);
+        return 2.0 * bytes * iters / ((t1 - t0) / 1e6) / 1e9;
+    }, py::arg("bytes"), py::arg("iters") = 20);
+
     m.def("bench_snapshot_pipeline",
           &benchSnapshotPipeline,
           py::arg("bytes"),

@@ -108,8 +108,17 @@ __global__ __launch_bounds__(FAM_KERNEL_BLOCK) void copyBufferKernel(
   uint4* __restrict__ dst,
   u64 nVec)
 {
-    for (u64 i = (u64)blockIdx.x * blockDim.x + threadIdx.x; i < nVec;
-         i += (u64)gridDim.x * blockDim.x) {
+    // 2 vecs per lane per iteration: the second load issues while the
+    // first store's cacheline drains
+    u64 stride = (u64)gridDim.x * blockDim.x;
+    u64 i = (u64)blockIdx.x * blockDim.x + threadIdx.x;
+    for (; i + stride < nVec; i += 2 * stride) {
+        uint4 a = ldVec<true>(&src[i]);
+        uint4 b = ldVec<true>(&src[i + stride]);
+        stVec<true>(&dst[i], a);
+        stVec<true>(&dst[i + stride], b);
+    }
+    for (; i < nVec; i += stride) {
         stVec<true>(&dst[i], ldVec<true>(&src[i]));
     }
 }
@@ -388,8 +397,16 @@ hipError_t famCopyBuffer(const void* src,
                               stream);
     }
     uint64_t nVec = bytes / 16;
+    static const u32 copyGrid = []() {
+        const char* e = getenv("FAM_COPY_GRID");
+        return e ? (u32)atoi(e) : 0;
+    }();
+    u32 grid = copyGrid ? copyGrid : gridFor(nVec);
+    if ((u64)grid * FAM_KERNEL_BLOCK > nVec && nVec > 0) {
+        grid = gridFor(nVec);
+    }
     hipLaunchKernelGGL(copyBufferKernel,
-                       dim3(gridFor(nVec)),
+                       dim3(grid),
                        dim3(FAM_KERNEL_BLOCK),
                        0,
                        stream,
