@@ -304,6 +304,22 @@ void initMpiBindings(py::module_& m)
                                 op,
                                 MpiBufferLoc::AUTO);
           });
+    m.def("mpi_scan_ptr",
+          [](int rank,
+             uintptr_t sendPtr,
+             uintptr_t recvPtr,
+             int count,
+             MpiDataType dtype,
+             MpiOp op) {
+              py::gil_scoped_release release;
+              world().scan(rank,
+                           (const uint8_t*)sendPtr,
+                           (uint8_t*)recvPtr,
+                           dtype,
+                           count,
+                           op,
+                           MpiBufferLoc::AUTO);
+          });
     m.def("mpi_reduce_ptr",
           [](int rank,
              int root,

@@ -204,7 +204,8 @@ class MpiWorld
               uint8_t* recvBuffer,
               MpiDataType dataType,
               int count,
-              MpiOp op);
+              MpiOp op,
+              MpiBufferLoc loc = MpiBufferLoc::AUTO);
     void reduceScatter(int thisRank,
                        const uint8_t* sendBuffer,
                        uint8_t* recvBuffer,

@@ -1002,10 +1002,43 @@ void MpiWorld::scan(int thisRank,
                     uint8_t* recvBuffer,
                     MpiDataType dataType,
                     int count,
-                    MpiOp op)
+                    MpiOp op,
+                    MpiBufferLoc loc)
 {
-    // Inclusive prefix along a linear chain (reference: :1390-1432)
     size_t bytes = mpiTypeSize(dataType) * (size_t)count;
+    if (isDeviceBuffer(sendBuffer, loc)) {
+        // Device chain: RCCL p2p hop + fused elementwise combine on the
+        // GPU (the SURVEY worklist's "GPU prefix + p2p chain" mapping for
+        // the reference's linear scan, src/mpi/MpiWorld.cpp:1390-1432)
+        ensureRcclComm(thisRank);
+        ncclComm_t comm;
+        hipStream_t stream;
+        {
+            std::lock_guard<std::mutex> lock(rccl->mx);
+            comm = rccl->comms.at(thisRank);
+            stream = rccl->streams.at(thisRank);
+            HIP_CHECK(hipSetDevice(rccl->devices.at(thisRank)));
+        }
+        if (recvBuffer != sendBuffer) {
+            HIP_CHECK(famCopyBuffer(sendBuffer, recvBuffer, bytes, stream));
+        }
+        if (thisRank > 0) {
+            uint8_t* tmp = nullptr;
+            HIP_CHECK(hipMallocAsync((void**)&tmp, bytes, stream));
+            RCCL_CHECK(ncclRecv(tmp, count, toNccl(dataType), thisRank - 1,
+                                comm, stream));
+            HIP_CHECK(famElementwiseOp(recvBuffer, tmp, (uint64_t)count,
+                                       (int)dataType, (int)op, stream));
+            HIP_CHECK(hipFreeAsync(tmp, stream));
+        }
+        if (thisRank < size - 1) {
+            RCCL_CHECK(ncclSend(recvBuffer, count, toNccl(dataType),
+                                thisRank + 1, comm, stream));
+        }
+        HIP_CHECK(hipStreamSynchronize(stream));
+        return;
+    }
+    // Inclusive prefix along a linear chain (reference: :1390-1432)
     std::memcpy(recvBuffer, sendBuffer, bytes);
     if (thisRank > 0) {
         auto data =

@@ -342,3 +342,28 @@ def test_device_snapshot_wire_diff_ship():
     snap_b.apply_compact_diff(pages, payload)
     out = snap_b.copy_out_host(n)
     assert out == updated.cpu().numpy().tobytes()
+
+
+def _scan_1rank_fn(msg):
+    world_id, rank, size = _core.mpi_init()
+    n = 1 << 18
+    send = torch.arange(n, dtype=torch.float32, device="cuda")
+    recv = torch.zeros(n, dtype=torch.float32, device="cuda")
+    torch.cuda.synchronize()
+    _core.mpi_scan_ptr(rank, send.data_ptr(), recv.data_ptr(), n,
+                       _core.MpiDataType.FLOAT, _core.MpiOp.SUM)
+    torch.cuda.synchronize()
+    if not torch.equal(recv, send):
+        return 1
+    msg.output_data = "scan ok"
+    return 0
+
+
+@requires_gpu
+def test_device_scan_single_rank(runtime):
+    """Device-buffer MPI_Scan (RCCL chain + fused elementwise combine);
+    the 1-rank world exercises the NT device copy path."""
+    _core.register_function("gpu", "scan1", _scan_1rank_fn)
+    results = submit_mpi("gpu", "scan1", 1)
+    assert results[0].return_value == 0, results[0].output_data
+    assert results[0].output_data == "scan ok"
