@@ -90,8 +90,22 @@ __global__ __launch_bounds__(FAM_KERNEL_BLOCK) void xorBufferKernel(
   uint4* __restrict__ out,
   u64 nVec)
 {
-    for (u64 i = (u64)blockIdx.x * blockDim.x + threadIdx.x; i < nVec;
-         i += (u64)gridDim.x * blockDim.x) {
+    // 2-way unrolled like copyBufferKernel: overlaps the second pair of
+    // loads with the first store's drain
+    u64 stride = (u64)gridDim.x * blockDim.x;
+    u64 i = (u64)blockIdx.x * blockDim.x + threadIdx.x;
+    for (; i + stride < nVec; i += 2 * stride) {
+        uint4 x0 = ldVec<true>(&a[i]);
+        uint4 y0 = ldVec<true>(&b[i]);
+        uint4 x1 = ldVec<true>(&a[i + stride]);
+        uint4 y1 = ldVec<true>(&b[i + stride]);
+        stVec<true>(&out[i], make_uint4(x0.x ^ y0.x, x0.y ^ y0.y,
+                                        x0.z ^ y0.z, x0.w ^ y0.w));
+        stVec<true>(&out[i + stride],
+                    make_uint4(x1.x ^ y1.x, x1.y ^ y1.y, x1.z ^ y1.z,
+                               x1.w ^ y1.w));
+    }
+    for (; i < nVec; i += stride) {
         uint4 x = ldVec<true>(&a[i]);
         uint4 y = ldVec<true>(&b[i]);
         stVec<true>(&out[i],
