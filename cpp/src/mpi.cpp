@@ -1233,7 +1233,16 @@ int MpiContext::createWorld(Message& msg)
     }
     int worldId = msg.mpiWorldId != 0 ? msg.mpiWorldId : generateGidInt32();
     msg.mpiWorldId = worldId;
-    MpiWorldRegistry::get().createWorld(msg, worldId);
+    // A migrated or unfrozen rank 0 re-enters with a snapshotKey (and the
+    // world may already live on this host): JOIN the existing cluster-wide
+    // world instead of re-creating it (reference: re-entry goes through
+    // getOrInitialiseWorld, src/mpi/MpiWorldRegistry.cpp)
+    if (!msg.snapshotKey.empty() ||
+        MpiWorldRegistry::get().worldExists(worldId)) {
+        MpiWorldRegistry::get().getOrInitialiseWorld(msg);
+    } else {
+        MpiWorldRegistry::get().createWorld(msg, worldId);
+    }
     isMpi = true;
     rank = 0;
     this->worldId = worldId;

@@ -291,6 +291,58 @@ static int32_t exampleAllReduceSmallBench(Message& msg)
     return 0;
 }
 
+
+// MPI world live-migration payload (reference:
+// tests/dist/mpi/examples/mpi_migration.cpp — rounds of collectives with
+// a migration point between them; re-entry skips the pre-migration round)
+static int32_t exampleMigrate(Message& msg)
+{
+    bool resumed = std::string(msg.inputData.begin(),
+                               msg.inputData.end()) == "resumed";
+    MPI_Init(nullptr, nullptr);
+    int rank;
+    int worldSize;
+    MPI_Comm_rank(MPI_COMM_WORLD, &rank);
+    MPI_Comm_size(MPI_COMM_WORLD, &worldSize);
+
+    std::vector<int> input(64, rank + 1);
+    std::vector<int> output(64, 0);
+    int expected = worldSize * (worldSize + 1) / 2;
+
+    if (!resumed) {
+        MPI_Allreduce(input.data(), output.data(), 64, MPI_INT, MPI_SUM,
+                      MPI_COMM_WORLD);
+        for (int v : output) {
+            if (v != expected) {
+                msg.outputData = "pre-migration allreduce mismatch";
+                return 1;
+            }
+        }
+        // The whole group checks for a migration opportunity together;
+        // a migrating rank unwinds here and re-enters with "resumed"
+        int32_t rc = migrationPoint(
+          std::vector<uint8_t>{ 'r', 'e', 's', 'u', 'm', 'e', 'd' });
+        if (rc != 0) {
+            return rc;
+        }
+    }
+
+    // Post-migration round over the re-built world/groupId
+    output.assign(64, 0);
+    MPI_Allreduce(input.data(), output.data(), 64, MPI_INT, MPI_SUM,
+                  MPI_COMM_WORLD);
+    for (int v : output) {
+        if (v != expected) {
+            msg.outputData = "post-migration allreduce mismatch";
+            return 2;
+        }
+    }
+    MPI_Barrier(MPI_COMM_WORLD);
+    MPI_Finalize();
+    msg.outputData = resumed ? "migrated+rejoined" : "stayed+continued";
+    return 0;
+}
+
 void registerMpiExampleFunctions()
 {
     auto& reg = FunctionRegistry::get();
@@ -298,6 +350,7 @@ void registerMpiExampleFunctions()
     reg.registerFunction("mpi-cpp", "ring", exampleRing);
     reg.registerFunction("mpi-cpp", "async", exampleAsync);
     reg.registerFunction("mpi-cpp", "vcollectives", exampleVCollectives);
+    reg.registerFunction("mpi-cpp", "migrate", exampleMigrate);
     reg.registerFunction("mpi-cpp", "allreduce-small-bench",
                          exampleAllReduceSmallBench);
     reg.registerFunction("mpi-cpp", "allreduce-bench",

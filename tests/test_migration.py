@@ -63,6 +63,7 @@ def _worker_main(port_offset, stop_event, ready_event):
     _core.register_native_sleep("mig", "blocker", 600)
     _core.register_function("mig", "worker", _mig_worker_fn)
     _core.register_function("mig", "freezer", _freeze_worker_fn)
+    _core.register_mpi_example_functions()
     ready_event.set()
     stop_event.wait(180)
     rt.stop()
@@ -192,3 +193,48 @@ def test_spot_freeze_and_unfreeze(cluster):
     finally:
         _core.planner_set_policy("bin-pack")
         _core.planner_set_next_evicted_vms([])
+
+
+def test_mpi_world_migration(cluster):
+    """An MPI rank live-migrates between workers mid-app: the group learns
+    the new groupId over the migration PTP channel, the world's rank-host
+    map and RCCL comms are rebuilt (MpiWorld::prepareMigration), and the
+    post-migration allreduce still reduces correctly (reference:
+    tests/dist/mpi test_migration + examples/mpi_migration.cpp)."""
+    from faabric_amd import _core
+    from faabric_amd.runtime import wait_for_batch
+
+    w1, w2 = idents()
+    # One blocker on each host forces the 2-rank world to split 1+1; the
+    # blockers exit before the migration check fires, freeing w? slots so
+    # bin-pack consolidates the app onto one host
+    blockers = submit_pinned("mig", "blocker", [w1, w2], group=False)
+
+    n = 2
+    ber = _core.batch_exec_factory("mpi-cpp", "migrate", 1)
+    msgs = ber.messages
+    msgs[0].is_mpi = True
+    msgs[0].mpi_world_size = n
+    ber.messages = msgs
+    decision = _core.SchedulingDecision()
+    decision.app_id = ber.app_id
+    for i, h in enumerate([w1, w2]):
+        decision.hosts = decision.hosts + [h]
+        decision.message_ids = decision.message_ids + [0]
+        decision.app_idxs = decision.app_idxs + [i]
+        decision.group_idxs = decision.group_idxs + [i]
+        decision.mpi_ports = decision.mpi_ports + [0]
+    decision.n_functions = n
+    _core.preload_scheduling_decision(ber.app_id, decision)
+    sched = _core.call_functions(ber)
+    assert sched.app_id == ber.app_id
+
+    results = wait_for_batch(ber.app_id, n, timeout_ms=90_000)
+    assert all(r.return_value == 0 for r in results), [
+        (r.mpi_rank, r.return_value, r.output_data) for r in results
+    ]
+    outputs = sorted(r.output_data for r in results)
+    assert outputs == ["migrated+rejoined", "stayed+continued"], outputs
+    assert len({r.executed_host for r in results}) == 1
+
+    wait_for_batch(blockers.app_id, 2, timeout_ms=30_000)
