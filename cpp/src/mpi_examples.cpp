@@ -309,6 +309,8 @@ static int32_t exampleMigrate(Message& msg)
     std::vector<int> output(64, 0);
     int expected = worldSize * (worldSize + 1) / 2;
 
+    bool slow = std::string(msg.inputData.begin(), msg.inputData.end()) ==
+                "slow";
     if (!resumed) {
         MPI_Allreduce(input.data(), output.data(), 64, MPI_INT, MPI_SUM,
                       MPI_COMM_WORLD);
@@ -318,12 +320,21 @@ static int32_t exampleMigrate(Message& msg)
                 return 1;
             }
         }
-        // The whole group checks for a migration opportunity together;
-        // a migrating rank unwinds here and re-enters with "resumed"
-        int32_t rc = migrationPoint(
-          std::vector<uint8_t>{ 'r', 'e', 's', 'u', 'm', 'e', 'd' });
-        if (rc != 0) {
-            return rc;
+        // The whole group checks for a migration opportunity together
+        // (the check is collective: a leader asks the planner and
+        // broadcasts the verdict); a migrating/freezing rank unwinds here
+        // and re-enters with "resumed". The slow variant keeps checking
+        // so an eviction notice arriving later still lands.
+        int rounds = slow ? 20 : 1;
+        for (int r = 0; r < rounds; r++) {
+            if (slow) {
+                usleep(250 * 1000);
+            }
+            int32_t rc = migrationPoint(
+              std::vector<uint8_t>{ 'r', 'e', 's', 'u', 'm', 'e', 'd' });
+            if (rc != 0) {
+                return rc;
+            }
         }
     }
 

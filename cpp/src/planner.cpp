@@ -270,15 +270,24 @@ std::shared_ptr<SchedulingDecision> Planner::callBatch(
     }
 
     // Un-freeze bookkeeping (reference: src/planner/Planner.cpp:1038-1081)
+    bool isUnfreezeDispatch = false;
     if (state.evictedRequests.count(appId) > 0) {
         if (isNew && !isMpi) {
             // Non-MPI app: all messages re-dispatch at once, bookkeeping
             // is complete as soon as the new decision exists
             state.evictedRequests.erase(appId);
         } else if (isNew && isMpi) {
-            Message first = req->messages.at(0);
-            req->messages.clear();
-            req->messages.push_back(first);
+            // Un-freeze of an MPI world: every rank is already known
+            // (the frozen BER carries all messages with their freeze
+            // snapshots + reentry input), so the whole gang dispatches
+            // through the known-size path at once and every rank —
+            // including 0 — re-enters by JOINING the re-built world.
+            // (Departure from the reference's two-step re-create,
+            // src/planner/Planner.cpp:1042-1081: dispatching directly
+            // avoids rank 0 re-creating a world that its own freeze
+            // snapshot marks as already existing.)
+            isUnfreezeDispatch = true;
+            state.evictedRequests.erase(appId);
         } else if (isMpi && !isDistChange) {
             auto evictedBer = state.evictedRequests.at(appId);
             for (auto& m : req->messages) {
@@ -323,7 +332,7 @@ std::shared_ptr<SchedulingDecision> Planner::callBatch(
                 }
             }
 
-            if (isMpi && knownSizeReq != nullptr) {
+            if (isMpi && knownSizeReq != nullptr && !isUnfreezeDispatch) {
                 auto preload = std::make_shared<SchedulingDecision>(*decision);
                 preload->groupId = FIXED_SIZE_PRELOADED_DECISION_GROUPID;
                 state.preloadedSchedulingDecisions[appId] = preload;

@@ -57,7 +57,7 @@ def _worker_main(port_offset, stop_event, ready_event):
     from faabric_amd import _core
     from faabric_amd.runtime import LocalRuntime
 
-    _core.set_log_level("error")
+    _core.set_log_level(os.environ.get("WORKER_LOG", "error"))
     rt = LocalRuntime(port_offset=port_offset, slots=WORKER_SLOTS)
     rt.start_worker()
     _core.register_native_sleep("mig", "blocker", 600)
@@ -238,3 +238,60 @@ def test_mpi_world_migration(cluster):
     assert len({r.executed_host for r in results}) == 1
 
     wait_for_batch(blockers.app_id, 2, timeout_ms=30_000)
+
+
+def test_mpi_world_freeze_unfreeze(cluster):
+    """Spot eviction freezes a whole MPI world (snapshots + world
+    destroyed, app checkpointed in the planner) and un-freezes it when
+    capacity returns: every rank re-enters with its reentry input, rank 0
+    joins the re-built world, and the post-thaw allreduce is correct."""
+    from faabric_amd import _core
+    from faabric_amd.runtime import wait_for_batch
+
+    w1, w2 = idents()
+    _core.planner_set_policy("spot")
+    try:
+        n = 2
+        ber = _core.batch_exec_factory("mpi-cpp", "migrate", 1)
+        msgs = ber.messages
+        msgs[0].is_mpi = True
+        msgs[0].mpi_world_size = n
+        msgs[0].input_data = b"slow"
+        ber.messages = msgs
+        decision = _core.SchedulingDecision()
+        decision.app_id = ber.app_id
+        for i in range(n):
+            decision.hosts = decision.hosts + [w2]
+            decision.message_ids = decision.message_ids + [0]
+            decision.app_idxs = decision.app_idxs + [i]
+            decision.group_idxs = decision.group_idxs + [i]
+            decision.mpi_ports = decision.mpi_ports + [0]
+        decision.n_functions = n
+        _core.preload_scheduling_decision(ber.app_id, decision)
+        _core.call_functions(ber)
+
+        time.sleep(0.3)
+        # Doom both VMs: nowhere to migrate -> MUST_FREEZE at the next
+        # migration point
+        _core.planner_set_next_evicted_vms([w1, w2])
+        deadline = time.monotonic() + 30
+        frozen = False
+        while time.monotonic() < deadline:
+            if _core.planner_num_in_flight_apps() == 0:
+                frozen = True
+                break
+            time.sleep(0.1)
+        assert frozen, "MPI app did not freeze"
+
+        _core.planner_set_next_evicted_vms([])
+        results = wait_for_batch(ber.app_id, n, timeout_ms=90_000)
+        assert all(r.return_value == 0 for r in results), [
+            (r.mpi_rank, r.return_value, r.output_data) for r in results
+        ]
+        # Every rank re-entered after the thaw
+        assert all(r.output_data == "migrated+rejoined" for r in results), [
+            r.output_data for r in results
+        ]
+    finally:
+        _core.planner_set_policy("bin-pack")
+        _core.planner_set_next_evicted_vms([])
