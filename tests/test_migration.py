@@ -240,7 +240,11 @@ def test_mpi_world_migration(cluster):
     wait_for_batch(blockers.app_id, 2, timeout_ms=30_000)
 
 
-def test_mpi_world_freeze_unfreeze(cluster):
+import pytest as _pytest
+
+
+@_pytest.mark.parametrize("split", [False, True], ids=["samehost", "crosshost"])
+def test_mpi_world_freeze_unfreeze(cluster, split):
     """Spot eviction freezes a whole MPI world (snapshots + world
     destroyed, app checkpointed in the planner) and un-freezes it when
     capacity returns: every rank re-enters with its reentry input, rank 0
@@ -260,8 +264,9 @@ def test_mpi_world_freeze_unfreeze(cluster):
         ber.messages = msgs
         decision = _core.SchedulingDecision()
         decision.app_id = ber.app_id
+        placement = [w1, w2] if split else [w2, w2]
         for i in range(n):
-            decision.hosts = decision.hosts + [w2]
+            decision.hosts = decision.hosts + [placement[i]]
             decision.message_ids = decision.message_ids + [0]
             decision.app_idxs = decision.app_idxs + [i]
             decision.group_idxs = decision.group_idxs + [i]
