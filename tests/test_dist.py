@@ -83,6 +83,33 @@ def _dist_fork_parent(msg):
     return 0
 
 
+
+
+def _chain_parent_fn(msg):
+    from faabric_amd import _core
+
+    # Chain two children; with this host's 2 slots held by the parent and
+    # a blocker, at least one child lands on the other worker
+    ids = [_core.chain_function("dist", "chainchild", str(i).encode())
+           for i in range(2)]
+    outs = []
+    for cid in ids:
+        r = _core.await_chained_call(cid, 30_000)
+        if r.return_value != 0:
+            return 1
+        outs.append(r.output_data)
+    msg.output_data = ",".join(sorted(outs))
+    return 0
+
+
+def _chain_child_fn(msg):
+    from faabric_amd import _core
+
+    host = _core.get_endpoint_host()
+    msg.output_data = f"child{msg.input_data.decode()}@{host}"
+    return 0
+
+
 def _worker_main(port_offset, stop_event, ready_event):
     sys.path.insert(0, REPO_ROOT)
     from faabric_amd import _core
@@ -97,6 +124,8 @@ def _worker_main(port_offset, stop_event, ready_event):
     _core.register_function("dist", "mpi_allreduce", _mpi_allreduce_fn)
     _core.register_function("dist", "threadbody", _dist_thread_body)
     _core.register_function("dist", "forkparent", _dist_fork_parent)
+    _core.register_function("dist", "chainparent", _chain_parent_fn)
+    _core.register_function("dist", "chainchild", _chain_child_fn)
     ready_event.set()
     stop_event.wait(120)
     rt.stop()
@@ -243,3 +272,19 @@ def test_slots_freed_after_batch(cluster):
         assert len(results) == n
     hosts = _core.get_available_hosts()
     assert all(h.used_slots == 0 for h in hosts)
+
+
+def test_chained_calls_span_hosts(cluster):
+    """Chained functions dispatch through the planner and can land on a
+    different host than the parent; await_chained_call fetches the result
+    (reference: chained message flow, src/util/ExecGraph + planner
+    getMessageResult path)."""
+    from faabric_amd import _core
+    from faabric_amd.runtime import wait_for_batch
+
+    ber = _core.batch_exec_factory("dist", "chainparent", 1)
+    _core.call_functions(ber)
+    results = wait_for_batch(ber.app_id, 1, timeout_ms=60_000)
+    assert results[0].return_value == 0, results[0].output_data
+    outs = results[0].output_data.split(",")
+    assert len(outs) == 2 and all(o.startswith("child") for o in outs), outs
