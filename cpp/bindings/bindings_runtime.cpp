@@ -457,6 +457,10 @@ void initRuntimeBindings(py::module_& m)
                py::gil_scoped_release release;
                kv.pull();
            })
+      .def("push_partial", [](StateKeyValue& kv) {
+          py::gil_scoped_release release;
+          kv.pushPartial();
+      })
       .def("push_full", [](StateKeyValue& kv) {
           py::gil_scoped_release release;
           kv.pushFull();

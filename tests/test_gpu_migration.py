@@ -45,6 +45,36 @@ def _gpu_mig_fn(msg):
     return 0
 
 
+def _hbm_state_owner_fn(msg):
+    from faabric_amd import _core
+
+    kv = _core.state_get_kv_device("gpu", "xhost", 128 * 1024)
+    kv.set(bytes([0xC3]) * (128 * 1024))
+    kv.set_chunk(8192, b"OWNED")
+    msg.output_data = "owner ok"
+    return 0
+
+
+def _hbm_state_reader_fn(msg):
+    from faabric_amd import _core
+
+    master = msg.input_data.decode()
+    _core.state_set_master_host("gpu", "xhost", master)
+    kv = _core.state_get_kv_device("gpu", "xhost", 128 * 1024)
+    if kv.is_master:
+        msg.output_data = "reader unexpectedly master"
+        return 1
+    chunk = kv.get_chunk(8190, 16)
+    if chunk[2:7] != b"OWNED" or chunk[0] != 0xC3:
+        msg.output_data = f"bad chunk: {chunk!r}"
+        return 2
+    # Push a write back to the owner
+    kv.set_chunk(4096, b"FROMB")
+    kv.push_partial()
+    msg.output_data = "reader ok"
+    return 0
+
+
 def _worker_main(port_offset, stop_event, ready_event):
     sys.path.insert(0, REPO_ROOT)
     from faabric_amd import _core
@@ -56,6 +86,8 @@ def _worker_main(port_offset, stop_event, ready_event):
     rt.start_worker()
     _core.register_native_sleep("gmig", "blocker", 600)
     _core.register_function("gmig", "worker", _gpu_mig_fn)
+    _core.register_function("gmig", "stateowner", _hbm_state_owner_fn)
+    _core.register_function("gmig", "statereader", _hbm_state_reader_fn)
     ready_event.set()
     stop_event.wait(180)
     rt.stop()
@@ -140,3 +172,43 @@ def test_gpu_arena_migrates(cluster):
     assert len({r.executed_host for r in results}) == 1
 
     wait_for_batch(blockers.app_id, n, timeout_ms=30_000)
+
+
+def _submit_one(user, func, host, input_data=b""):
+    from faabric_amd import _core
+
+    ber = _core.batch_exec_factory(user, func, 1)
+    msgs = ber.messages
+    msgs[0].input_data = input_data
+    ber.messages = msgs
+    d = _core.SchedulingDecision()
+    d.app_id = ber.app_id
+    d.hosts = [host]
+    d.message_ids = [0]
+    d.app_idxs = [0]
+    d.group_idxs = [0]
+    d.mpi_ports = [0]
+    d.n_functions = 1
+    _core.preload_scheduling_decision(ber.app_id, d)
+    _core.call_functions(ber)
+    return ber
+
+
+def test_hbm_state_cross_worker(cluster):
+    """Distributed state with the value resident in the owner worker's
+    HBM: a second worker lazily pulls chunks over the state RPC into its
+    own GPU and pushes partial writes back (reference: two-backend
+    StateKeyValue pull/push, src/state/StateKeyValue.cpp — HBM-resident
+    re-design)."""
+    from faabric_amd import _core
+    from faabric_amd.runtime import wait_for_batch
+
+    w1, w2 = [f"127.0.0.1@{o}" for o in OFFSETS]
+    owner = _submit_one("gmig", "stateowner", w1)
+    r = wait_for_batch(owner.app_id, 1, timeout_ms=60_000)
+    assert r[0].return_value == 0, r[0].output_data
+
+    reader = _submit_one("gmig", "statereader", w2, w1.encode())
+    r = wait_for_batch(reader.app_id, 1, timeout_ms=60_000)
+    assert r[0].return_value == 0, r[0].output_data
+    assert r[0].output_data == "reader ok"
