@@ -10,6 +10,7 @@ byte-crunching loops (faabric ``src/util/snapshot.cpp``,
 from faabric_amd._core import (  # noqa: F401
     DeviceSnapshot,
     bench_snapshot_pipeline,
+    bench_copy,
     delta_apply,
     delta_encode,
     device_elementwise_op,
