@@ -85,6 +85,30 @@ def _dist_fork_parent(msg):
 
 
 
+def _append_owner_fn(msg):
+    from faabric_amd import _core
+
+    kv = _core.state_get_kv("dist", "alog", 64)
+    kv.append(b"owner-entry")
+    msg.output_data = "appended"
+    return 0
+
+
+def _append_reader_fn(msg):
+    from faabric_amd import _core
+
+    master = msg.input_data.decode()
+    _core.state_set_master_host("dist", "alog", master)
+    kv = _core.state_get_kv("dist", "alog", 64)
+    kv.append(b"reader-entry")
+    entries = kv.get_appended(2)
+    if list(entries) != [b"owner-entry", b"reader-entry"]:
+        msg.output_data = f"bad log: {entries!r}"
+        return 1
+    msg.output_data = "append ok"
+    return 0
+
+
 def _chain_parent_fn(msg):
     from faabric_amd import _core
 
@@ -125,6 +149,8 @@ def _worker_main(port_offset, stop_event, ready_event):
     _core.register_function("dist", "threadbody", _dist_thread_body)
     _core.register_function("dist", "forkparent", _dist_fork_parent)
     _core.register_function("dist", "chainparent", _chain_parent_fn)
+    _core.register_function("dist", "appendowner", _append_owner_fn)
+    _core.register_function("dist", "appendreader", _append_reader_fn)
     _core.register_function("dist", "chainchild", _chain_child_fn)
     ready_event.set()
     stop_event.wait(120)
@@ -283,8 +309,49 @@ def test_chained_calls_span_hosts(cluster):
     from faabric_amd.runtime import wait_for_batch
 
     ber = _core.batch_exec_factory("dist", "chainparent", 1)
+    parent_id = ber.messages[0].id
     _core.call_functions(ber)
-    results = wait_for_batch(ber.app_id, 1, timeout_ms=60_000)
-    assert results[0].return_value == 0, results[0].output_data
-    outs = results[0].output_data.split(",")
+    # The chained children join the same app: wait for all 3 results and
+    # pick the parent by message id
+    results = wait_for_batch(ber.app_id, 3, timeout_ms=60_000)
+    parent = [r for r in results if r.id == parent_id][0]
+    assert parent.return_value == 0, parent.output_data
+    outs = parent.output_data.split(",")
     assert len(outs) == 2 and all(o.startswith("child") for o in outs), outs
+    children = [r for r in results if r.id != parent_id]
+    assert all(c.return_value == 0 for c in children)
+
+
+def test_state_append_cross_host(cluster):
+    """The append channel is ordered across hosts: a non-master worker's
+    append lands on the master and pullAppended returns the global log
+    (reference: src/state/StateServer.cpp Append/PullAppended)."""
+    from faabric_amd import _core
+    from faabric_amd.runtime import wait_for_batch
+
+    hosts = sorted(h.ip for h in _core.get_available_hosts())
+    w1, w2 = hosts
+
+    ber = _core.batch_exec_factory("dist", "appendowner", 1)
+    d = _core.SchedulingDecision()
+    d.app_id = ber.app_id
+    d.hosts = [w1]; d.message_ids = [0]; d.app_idxs = [0]
+    d.group_idxs = [0]; d.mpi_ports = [0]; d.n_functions = 1
+    _core.preload_scheduling_decision(ber.app_id, d)
+    _core.call_functions(ber)
+    r = wait_for_batch(ber.app_id, 1, timeout_ms=30_000)
+    assert r[0].return_value == 0, r[0].output_data
+
+    ber = _core.batch_exec_factory("dist", "appendreader", 1)
+    msgs = ber.messages
+    msgs[0].input_data = w1.encode()
+    ber.messages = msgs
+    d = _core.SchedulingDecision()
+    d.app_id = ber.app_id
+    d.hosts = [w2]; d.message_ids = [0]; d.app_idxs = [0]
+    d.group_idxs = [0]; d.mpi_ports = [0]; d.n_functions = 1
+    _core.preload_scheduling_decision(ber.app_id, d)
+    _core.call_functions(ber)
+    r = wait_for_batch(ber.app_id, 1, timeout_ms=30_000)
+    assert r[0].return_value == 0, r[0].output_data
+    assert r[0].output_data == "append ok"
