@@ -136,6 +136,8 @@ def test_ptp_group_barrier_and_lock(runtime):
 
     def member(idx):
         _core.ptp_group_barrier(555001, idx)
+        # notify: non-main members signal, main waits for all
+        _core.ptp_group_notify(555001, idx)
         _core.ptp_group_lock(555001, idx)
         v = counter["v"]
         counter["v"] = v + 1
