@@ -40,8 +40,11 @@ class ExampleExecutorFactory : public ExecutorFactory
 
 int main()
 {
-    getSystemConfig().endpointHost = "127.0.0.1";
-    getSystemConfig().plannerHost = "127.0.0.1";
+    // Identity carries the port offset so multiple deployments share an IP
+    std::string ident =
+      "127.0.0.1@" + std::to_string(getPortOffset());
+    getSystemConfig().endpointHost = ident;
+    getSystemConfig().plannerHost = ident;
 
     HostResources res;
     res.slots = 4;
