@@ -155,3 +155,19 @@ def test_is_app_migratable_cli(binaries):
         if status == 200 and json.loads(body).get("finished"):
             break
         time.sleep(0.2)
+
+
+def test_check_binary():
+    """The minimal C++ embedder check binary (BASELINE config 1) runs
+    end-to-end under a port offset."""
+    check = os.path.join(BUILD, "check")
+    if not os.path.exists(check):
+        pytest.skip("check not built")
+    out = subprocess.run(
+        [check],
+        env={**os.environ, "FAABRIC_PORT_OFFSET": "7600",
+             "LOG_LEVEL": "error"},
+        capture_output=True, text=True, timeout=60,
+    )
+    assert out.returncode == 0, out.stderr[-500:]
+    assert "CHECK OK" in out.stdout
