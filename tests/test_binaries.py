@@ -171,3 +171,19 @@ def test_check_binary():
     )
     assert out.returncode == 0, out.stderr[-500:]
     assert "CHECK OK" in out.stdout
+
+
+def test_selftest_binary():
+    """The C++ integration sweep (batches, THREADS merge, MPI world,
+    chaining, SPSC queue, snapshot semantics) — same binary the
+    sanitizer targets run."""
+    selftest = os.path.join(BUILD, "selftest")
+    if not os.path.exists(selftest):
+        pytest.skip("selftest not built")
+    out = subprocess.run(
+        [selftest],
+        env={**os.environ, "FAABRIC_PORT_OFFSET": "7700"},
+        capture_output=True, text=True, timeout=120,
+    )
+    assert out.returncode == 0, out.stderr[-500:]
+    assert "SELFTEST OK" in out.stdout
