@@ -165,3 +165,23 @@ def test_mpi_exec_graph_details(runtime):
     rank0 = [r for r in results if r.mpi_rank == 0][0]
     details = rank0.int_exec_graph_details
     assert any(k.startswith("mpi-msgcount-torank-") for k in details), details
+
+
+def test_concurrent_mpi_worlds(runtime):
+    """Two MPI worlds run concurrently in one worker without channel or
+    registry interference (reference: tests/test/mpi multi-world cases)."""
+    bers = []
+    for _ in range(2):
+        ber = _core.batch_exec_factory("mpi-cpp", "vcollectives", 1)
+        msgs = ber.messages
+        msgs[0].is_mpi = True
+        msgs[0].mpi_world_size = 2
+        ber.messages = msgs
+        d = _core.call_functions(ber)
+        assert d.app_id == ber.app_id, f"schedule failed: {d.app_id}"
+        bers.append(ber)
+    for ber in bers:
+        results = wait_for_batch(ber.app_id, 2, 60_000)
+        assert all(r.return_value == 0 for r in results), [
+            (r.mpi_rank, r.return_value, r.output_data) for r in results
+        ]
