@@ -166,6 +166,14 @@ class Executor
     std::atomic<bool> claimed{ false };
     std::atomic<int64_t> lastExecMs{ 0 };
 
+    // Fork-join snapshot lifecycle: the main-thread snapshot (keyed
+    // "<user>/<function>_<appId>") is kept for same-app repeat forks and
+    // deleted — locally and on the hosts it was shipped to — when a fork
+    // for a DIFFERENT app starts or the executor is destroyed
+    std::string lastForkSnapshotKey;
+    std::vector<std::string> lastForkRemoteHosts;
+    void cleanupForkSnapshot();
+
     // Batch accounting: tasks remaining in the current batch
     std::shared_ptr<std::atomic<int>> batchCounter;
 

@@ -99,8 +99,30 @@ Executor::Executor(Message& msg)
     lastExecMs = getGlobalClockEpochMillis();
 }
 
+void Executor::cleanupForkSnapshot()
+{
+    if (lastForkSnapshotKey.empty()) {
+        return;
+    }
+    DeviceSnapshotRegistry::get().deleteSnapshot(lastForkSnapshotKey);
+    SnapshotRegistry::get().deleteSnapshot(lastForkSnapshotKey);
+    for (const auto& host : lastForkRemoteHosts) {
+        try {
+            getSnapshotClient(host)->deleteSnapshot(lastForkSnapshotKey);
+        } catch (const std::exception& e) {
+            FAM_DEBUG("remote fork-snapshot delete failed: %s", e.what());
+        }
+    }
+    lastForkSnapshotKey.clear();
+    lastForkRemoteHosts.clear();
+}
+
 Executor::~Executor()
 {
+    try {
+        cleanupForkSnapshot();
+    } catch (...) {
+    }
     shutdown();
     if (deviceArena != nullptr) {
         hipFree(deviceArena);
@@ -233,6 +255,16 @@ std::vector<std::pair<int32_t, int32_t>> Executor::executeThreads(
       parentMsg.user, parentMsg.function, parentMsg.appId);
 
     bool onDevice = hasDeviceArena();
+    // A fork for a new app supersedes the previous app's snapshot
+    {
+        Message& pm = ExecutorContext::get().getMsg();
+        std::string newKey = getMainThreadSnapshotKey(
+          pm.user, pm.function, pm.appId);
+        if (!lastForkSnapshotKey.empty() &&
+            lastForkSnapshotKey != newKey) {
+            cleanupForkSnapshot();
+        }
+    }
     std::shared_ptr<SnapshotData> snap;
     std::shared_ptr<DeviceSnapshot> dsnap;
     if (onDevice) {
@@ -292,12 +324,18 @@ std::vector<std::pair<int32_t, int32_t>> Executor::executeThreads(
 
     // Ship the snapshot to the other hosts in the (updated) decision —
     // the planner cannot: the snapshot lives here, not in its registry
+    lastForkSnapshotKey = key;
     {
         const std::string& thisHost = getSystemConfig().endpointHost;
         std::vector<uint8_t> hostCopy;
         for (const auto& host : decision->uniqueHosts()) {
             if (host == thisHost) {
                 continue;
+            }
+            if (std::find(lastForkRemoteHosts.begin(),
+                          lastForkRemoteHosts.end(),
+                          host) == lastForkRemoteHosts.end()) {
+                lastForkRemoteHosts.push_back(host);
             }
             if (onDevice) {
                 if (hostCopy.empty()) {

@@ -663,6 +663,8 @@ void SnapshotServer::doAsyncRecv(uint8_t code,
     if ((SnapshotCalls)code == SnapshotCalls::DeleteSnapshot) {
         auto req = SnapshotDeleteRequest::decode(body);
         SnapshotRegistry::get().deleteSnapshot(req.key);
+        // Device snapshots shipped here land in the device registry
+        DeviceSnapshotRegistry::get().deleteSnapshot(req.key);
         return;
     }
     FAM_ERROR("snapshot server: bad async code %d", (int)code);
