@@ -81,9 +81,9 @@ def main(cycles=25):
     torch.cuda.synchronize()
     free1, _ = torch.cuda.mem_get_info()
     drift_mb = (free0 - free1) / (1 << 20)
-    # State KVs (3 x 1 MiB) stay resident by design; anything well beyond
-    # that is a leak
-    assert drift_mb < 256, f"HBM leak: {drift_mb:.0f} MiB drift"
+    # State KVs (3 x 1 MiB) + allocator pools settle around ~32 MiB;
+    # a per-cycle leak shows as linear growth well beyond that
+    assert drift_mb < 64, f"HBM leak: {drift_mb:.0f} MiB drift"
     print(f"GPU SOAK OK: {cycles} cycles, HBM drift {drift_mb:+.1f} MiB")
     rt.stop()
 
