@@ -98,7 +98,9 @@ class MpiWorld
     int getId() const { return id; }
     int getSize() const { return size; }
 
-    bool destroy(); // true when no local ranks remain
+    bool destroy();
+    // Returns true when this was the last local rank (world reclaimable)
+    bool rankFinished(int rank); // true when no local ranks remain
 
     // --- cartesian topology (reference: :369-543) ---
     void getCartesianRank(int rank,

@@ -543,6 +543,18 @@ void Executor::handleTaskResult(Message& msg,
 
     bool isThreads = req->type == BatchExecuteType::THREADS;
 
+    // Last local MPI rank done: release this host's world state (RCCL
+    // comms, streams, queues) and drop the registry entry
+    if (msg.isMpi && returnValue != MIGRATED_FUNCTION_RETURN_VALUE &&
+        returnValue != FROZEN_FUNCTION_RETURN_VALUE &&
+        MpiWorldRegistry::get().worldExists(msg.mpiWorldId)) {
+        if (MpiWorldRegistry::get()
+              .getWorld(msg.mpiWorldId)
+              .rankFinished(msg.mpiRank)) {
+            MpiWorldRegistry::get().clearWorld(msg.mpiWorldId);
+        }
+    }
+
     // THREADS: the last local thread diffs this executor's memory against
     // the shared snapshot and ships the typed diffs to the main host
     // (reference: src/executor/Executor.cpp:509-516, :684 mergeDirtyRegions)

@@ -243,6 +243,24 @@ bool MpiWorld::destroy()
     return true;
 }
 
+bool MpiWorld::rankFinished(int rank)
+{
+    // Called by the executor when a rank's function completes; once the
+    // last LOCAL rank is done this host's world state (incl. RCCL comms
+    // and streams) is released and the caller clears the registry entry
+    // (reference: executor-side MPI cleanup)
+    std::lock_guard<std::mutex> lock(worldMx);
+    auto it = std::find(localRanks.begin(), localRanks.end(), rank);
+    if (it != localRanks.end()) {
+        localRanks.erase(it);
+    }
+    if (!localRanks.empty()) {
+        return false;
+    }
+    rccl.reset();
+    return true;
+}
+
 // ------------------------- cartesian topology -------------------------------
 
 void MpiWorld::getCartesianRank(int rank,

@@ -15,6 +15,18 @@ from faabric_amd import _core  # noqa: E402
 from faabric_amd.runtime import LocalRuntime, wait_for_batch  # noqa: E402
 
 
+def _rccl_world_fn(msg):
+    world_id, rank, size = _core.mpi_init()
+    n = 1 << 16
+    send = torch.full((n,), 3.0, dtype=torch.float32, device="cuda")
+    recv = torch.zeros(n, dtype=torch.float32, device="cuda")
+    torch.cuda.synchronize()
+    _core.mpi_allreduce_ptr(rank, send.data_ptr(), recv.data_ptr(), n,
+                            _core.MpiDataType.FLOAT, _core.MpiOp.SUM)
+    torch.cuda.synchronize()
+    return 0 if torch.equal(recv, send) else 1
+
+
 def _fork_parent(msg):
     _core.executor_set_device_memory_size(1 << 20)
     _core.executor_device_write_memory(0, bytes(64))
@@ -38,6 +50,7 @@ def main(cycles=25):
     rt.start_worker()
     _core.register_function("gsoak", "forkparent", _fork_parent)
     _core.register_function("gsoak", "forkchild", _fork_child)
+    _core.register_function("gsoak", "rcclworld", _rccl_world_fn)
 
     # Warm up allocators/pools before baselining free HBM
     for _ in range(3):
@@ -59,7 +72,18 @@ def main(cycles=25):
         kv.set(bytes([i % 251]) * (1 << 20))
         assert kv.get_chunk(0, 1) == bytes([i % 251])
 
-        # 3. Raw snapshot cycle
+        # 3. RCCL world churn: a fresh device-plane world per cycle; its
+        # communicator and stream must be reclaimed when the rank finishes
+        ber = _core.batch_exec_factory("gsoak", "rcclworld", 1)
+        msgs = ber.messages
+        msgs[0].is_mpi = True
+        msgs[0].mpi_world_size = 1
+        ber.messages = msgs
+        _core.call_functions(ber)
+        rs = wait_for_batch(ber.app_id, 1, 60_000)
+        assert rs[0].return_value == 0, rs[0].output_data
+
+        # 4. Raw snapshot cycle
         snap = _core.DeviceSnapshot(1 << 22, 0)
         t = torch.full((1 << 22,), i % 127, dtype=torch.uint8,
                        device="cuda")
