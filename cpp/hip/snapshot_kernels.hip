@@ -343,9 +343,19 @@ __global__ __launch_bounds__(FAM_KERNEL_BLOCK) void elementwiseOpKernel(
   u64 n,
   int op)
 {
-    for (u64 i = (u64)blockIdx.x * blockDim.x + threadIdx.x; i < n;
-         i += (u64)gridDim.x * blockDim.x) {
-        inout[i] = famApply<T>(op, inout[i], in[i]);
+    // `in` is read once (NT); `inout` is RMW and keeps cached accesses
+    // (same finding as the apply kernel: streaming the RMW side regresses)
+    u64 stride = (u64)gridDim.x * blockDim.x;
+    u64 i = (u64)blockIdx.x * blockDim.x + threadIdx.x;
+    for (; i + stride < n; i += 2 * stride) {
+        T a0 = __builtin_nontemporal_load(&in[i]);
+        T a1 = __builtin_nontemporal_load(&in[i + stride]);
+        inout[i] = famApply<T>(op, inout[i], a0);
+        inout[i + stride] = famApply<T>(op, inout[i + stride], a1);
+    }
+    for (; i < n; i += stride) {
+        inout[i] = famApply<T>(op, inout[i],
+                               __builtin_nontemporal_load(&in[i]));
     }
 }
 
