@@ -175,9 +175,10 @@ def test_diff_apply_roundtrip_vs_torch():
         (4, 0, lambda a, b: a + b),  # double sum
     ],
 )
-def test_elementwise_op_vs_torch(dtype, op, torch_fn):
+@pytest.mark.parametrize("n", [1 << 20, 1_000_003, 513])
+def test_elementwise_op_vs_torch(dtype, op, torch_fn, n):
+    """Odd sizes exercise the unrolled kernels' tail loops."""
     torch_dtype = {0: torch.int32, 3: torch.float32, 4: torch.float64}[dtype]
-    n = 1 << 20
     torch.manual_seed(2)
     if torch_dtype == torch.int32:
         a = torch.randint(-1000, 1000, (n,), dtype=torch_dtype, device="cuda")
