@@ -685,6 +685,10 @@ void initRuntimeBindings(py::module_& m)
     });
 
     // Native benchmark payloads (cpp/src/bench_funcs.cpp)
+    m.def("reap_stale_executors", [] {
+        py::gil_scoped_release release;
+        return Scheduler::get().reapStaleExecutors();
+    });
     m.def("register_bench_functions", [] { registerBenchFunctions(); });
     m.def("register_mpi_example_functions",
           [] { registerMpiExampleFunctions(); });

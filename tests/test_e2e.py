@@ -165,3 +165,31 @@ def test_state_local(runtime):
     kv.append(b"v1")
     kv.append(b"v2")
     assert kv.get_appended(2) == [b"v1", b"v2"]
+
+
+def test_executor_reaper(runtime):
+    """Idle warm executors are reaped after the bound timeout (reference:
+    Scheduler::reapStaleExecutors, src/scheduler/Scheduler.cpp:166)."""
+    import time as _t
+
+    from faabric_amd import _core
+
+    from faabric_amd.runtime import wait_for_batch
+
+    ber = _core.batch_exec_factory("demo", "echo", 4)
+    _core.call_functions(ber)
+    wait_for_batch(ber.app_id, 4, 30_000)
+    assert _core.get_executor_count() >= 4
+
+    # Not yet stale
+    assert _core.reap_stale_executors() == 0
+
+    # The reaper reads the live config; shrink the bound and retry
+    _core.set_bound_timeout(50)
+    try:
+        _t.sleep(0.3)
+        reaped = _core.reap_stale_executors()
+        assert reaped >= 4, reaped
+        assert _core.get_executor_count() == 0
+    finally:
+        _core.set_bound_timeout(30_000)
