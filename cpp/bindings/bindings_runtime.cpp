@@ -685,6 +685,8 @@ void initRuntimeBindings(py::module_& m)
     });
 
     // Native benchmark payloads (cpp/src/bench_funcs.cpp)
+    m.def("set_bound_timeout",
+          [](int ms) { getSystemConfig().boundTimeout = ms; });
     m.def("reap_stale_executors", [] {
         py::gil_scoped_release release;
         return Scheduler::get().reapStaleExecutors();
