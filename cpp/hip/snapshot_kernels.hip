@@ -395,6 +395,10 @@ hipError_t famDirtyPages(const void* snap,
                          uint32_t* flagsDev,
                          hipStream_t stream)
 {
+    if ((bytes % FAM_PAGE) != 0) {
+        // A silently-floored tail would be missed dirty data
+        return hipErrorInvalidValue;
+    }
     uint32_t nPages = (uint32_t)(bytes / FAM_PAGE);
     uint32_t grid = nPages < FAM_MAX_BLOCKS ? (nPages ? nPages : 1)
                                             : FAM_MAX_BLOCKS;
@@ -446,6 +450,10 @@ hipError_t famXorBuffer(const void* a,
                         uint64_t bytes,
                         hipStream_t stream)
 {
+    if ((bytes % 16) != 0) {
+        // No scalar tail path: reject rather than silently drop bytes
+        return hipErrorInvalidValue;
+    }
     uint64_t nVec = bytes / 16;
     hipLaunchKernelGGL(xorBufferKernel,
                        dim3(gridFor(nVec)),
@@ -468,6 +476,9 @@ hipError_t famDiffXorPages(const void* snap,
                            uint32_t* bitmapDev, // nPages/32 words, zeroed
                            hipStream_t stream)
 {
+    if ((bytes % FAM_PAGE) != 0) {
+        return hipErrorInvalidValue;
+    }
     uint32_t nPages = (uint32_t)(bytes / FAM_PAGE);
     uint32_t wavesPerBlock = FAM_KERNEL_BLOCK / 64;
     uint32_t blocks = (nPages + wavesPerBlock - 1) / wavesPerBlock;

@@ -103,9 +103,19 @@ class MpiWorld
     bool rankFinished(int rank); // true when no local ranks remain
 
     // --- cartesian topology (reference: :369-543) ---
+    // N-dim row-major periodic grids; dims stored on the world the first
+    // time the app supplies them (Cart_create or Cart_get), defaulting to
+    // a 1-D {size} layout like the reference's pre-Cart state.
+    void setCartesianDims(int ndims, const int* dims);
     void getCartesianRank(int rank,
                           int maxDims,
                           const int* dims,
+                          int* periods,
+                          int* coords);
+    // Cart_get form: outputs the stored dims + this rank's coords
+    void getCartesianGrid(int rank,
+                          int maxDims,
+                          int* dims,
                           int* periods,
                           int* coords);
     void getRankFromCoords(int* rank, int* coords);
@@ -245,6 +255,10 @@ class MpiWorld
     std::mutex worldMx;
     std::vector<std::string> rankHosts;
     std::vector<int> localRanks;
+
+    // Cartesian grid dims (row-major, periodic); empty = 1-D {size}
+    std::mutex cartMx;
+    std::vector<int> cartDims;
 
     // Exec-graph counters: rank → (peer,type) counts
     std::mutex statsMx;

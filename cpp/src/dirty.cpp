@@ -59,6 +59,35 @@ struct TrackedRegion
 std::mutex regionsMx;
 std::map<uint8_t*, std::shared_ptr<TrackedRegion>> regions;
 
+// Immutable region list for the SIGSEGV handler: a std::mutex is not
+// async-signal-safe (a fault while another thread holds regionsMx in
+// startTracking/getDirtyPages would deadlock), so the handler reads an
+// atomically-swapped snapshot guarded by a hazard counter instead.
+struct RegionList
+{
+    std::vector<std::shared_ptr<TrackedRegion>> items;
+};
+std::atomic<RegionList*> activeRegions{ nullptr };
+std::atomic<int> handlersActive{ 0 };
+
+// Rebuild the handler snapshot from `regions`; caller holds regionsMx.
+void publishRegionList()
+{
+    auto* next = new RegionList;
+    next->items.reserve(regions.size());
+    for (auto& [base, region] : regions) {
+        next->items.push_back(region);
+    }
+    RegionList* old =
+      activeRegions.exchange(next, std::memory_order_acq_rel);
+    if (old != nullptr) {
+        // Wait for in-flight handlers before reclaiming the old list
+        while (handlersActive.load(std::memory_order_acquire) != 0) {
+        }
+        delete old;
+    }
+}
+
 // Thread-local dirty channel: region base → flags
 thread_local std::map<uint8_t*, std::vector<char>> threadDirty;
 thread_local bool threadTrackingOn = false;
@@ -69,20 +98,20 @@ bool handlerInstalled = false;
 void segvHandler(int sig, siginfo_t* info, void* ucontext)
 {
     uint8_t* addr = (uint8_t*)info->si_addr;
-    std::shared_ptr<TrackedRegion> hit;
-    {
-        // NOTE: not strictly async-signal-safe; matches the reference's
-        // pragmatic approach (faults only come from tracked regions while
-        // tracking is active)
-        std::lock_guard<std::mutex> lock(regionsMx);
-        for (auto& [base, region] : regions) {
-            if (addr >= base && addr < base + region->size) {
-                hit = region;
+    handlersActive.fetch_add(1, std::memory_order_acq_rel);
+    RegionList* list = activeRegions.load(std::memory_order_acquire);
+    TrackedRegion* hit = nullptr;
+    if (list != nullptr) {
+        for (auto& region : list->items) {
+            if (addr >= region->base &&
+                addr < region->base + region->size) {
+                hit = region.get();
                 break;
             }
         }
     }
-    if (!hit) {
+    if (hit == nullptr) {
+        handlersActive.fetch_sub(1, std::memory_order_acq_rel);
         // Not ours: restore the previous handler and re-raise
         sigaction(SIGSEGV, &oldSegvAction, nullptr);
         raise(sig);
@@ -91,6 +120,8 @@ void segvHandler(int sig, siginfo_t* info, void* ucontext)
     size_t page = (size_t)(addr - hit->base) / TRACK_PAGE;
     hit->globalDirty[page] = 1;
     if (threadTrackingOn) {
+        // threadDirty is this thread's own map; the entry is preallocated
+        // by startThreadLocalTracking so the common path does not allocate
         auto& flags = threadDirty[hit->base];
         if (flags.size() <= page) {
             flags.resize(nPagesOf(hit->size), 0);
@@ -100,6 +131,7 @@ void segvHandler(int sig, siginfo_t* info, void* ucontext)
     // Re-enable the page for writing
     mprotect(hit->base + page * TRACK_PAGE, TRACK_PAGE,
              PROT_READ | PROT_WRITE);
+    handlersActive.fetch_sub(1, std::memory_order_acq_rel);
     (void)ucontext;
 }
 
@@ -135,6 +167,7 @@ void SegfaultDirtyTracker::startTracking(uint8_t* region, size_t size)
     {
         std::lock_guard<std::mutex> lock(regionsMx);
         regions[region] = tracked;
+        publishRegionList();
     }
     mprotect(region, nPagesOf(size) * TRACK_PAGE, PROT_READ);
 }

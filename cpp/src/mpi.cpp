@@ -263,6 +263,39 @@ bool MpiWorld::rankFinished(int rank)
 
 // ------------------------- cartesian topology -------------------------------
 
+void MpiWorld::setCartesianDims(int ndims, const int* dims)
+{
+    // Validate the grid covers the world exactly, then store it. The
+    // reference stores dims on first use and requires dims[0]*dims[1] ==
+    // size (src/mpi/MpiWorld.cpp:379-395); we accept any N-dim row-major
+    // grid with the same product rule.
+    long product = 1;
+    for (int d = 0; d < ndims; d++) {
+        if (dims[d] <= 0) {
+            throw FaabricException("cartesian dim must be positive");
+        }
+        product *= dims[d];
+    }
+    if (product != size) {
+        throw FaabricException(
+          "product of cartesian dims does not equal world size");
+    }
+    std::lock_guard<std::mutex> lock(cartMx);
+    cartDims.assign(dims, dims + ndims);
+}
+
+// Stored dims, defaulting to a 1-D {size} layout before any Cart call
+static std::vector<int> effectiveDims(std::mutex& mx,
+                                      const std::vector<int>& stored,
+                                      int size)
+{
+    std::lock_guard<std::mutex> lock(mx);
+    if (stored.empty()) {
+        return { size };
+    }
+    return stored;
+}
+
 void MpiWorld::getCartesianRank(int rank,
                                 int maxDims,
                                 const int* dims,
@@ -273,6 +306,7 @@ void MpiWorld::getCartesianRank(int rank,
     if (rank >= size) {
         throw FaabricException("rank out of world");
     }
+    setCartesianDims(maxDims, dims);
     int remainder = rank;
     for (int d = 0; d < maxDims; d++) {
         int stride = 1;
@@ -281,17 +315,46 @@ void MpiWorld::getCartesianRank(int rank,
         }
         coords[d] = remainder / stride;
         remainder = remainder % stride;
+        periods[d] = 1; // always periodic, like the reference (LAMMPS)
+    }
+}
+
+void MpiWorld::getCartesianGrid(int rank,
+                                int maxDims,
+                                int* dims,
+                                int* periods,
+                                int* coords)
+{
+    std::vector<int> eff = effectiveDims(cartMx, cartDims, size);
+    int remainder = rank;
+    for (int d = 0; d < maxDims; d++) {
+        int dim = d < (int)eff.size() ? eff[d] : 1;
+        dims[d] = dim;
+        int stride = 1;
+        for (int e = d + 1; e < (int)eff.size(); e++) {
+            stride *= eff[e];
+        }
+        if (d < (int)eff.size()) {
+            coords[d] = remainder / stride;
+            remainder = remainder % stride;
+        } else {
+            coords[d] = 0;
+        }
         periods[d] = 1;
     }
 }
 
 void MpiWorld::getRankFromCoords(int* rank, int* coords)
 {
-    // Inverse of getCartesianRank for up to 3 dims like the reference
-    // (dims derived the same way MPI_Cart_create does: near-cubic split)
-    // For our usage dims are implicit: the caller keeps them; here we
-    // assume a 1-D layout fallback when unknown
-    *rank = coords[0];
+    // Row-major fold using the stored dims (inverse of getCartesianRank).
+    // Reference: *rank = coords[1] + coords[0]*dims[1] for its 2-D grids
+    // (src/mpi/MpiWorld.cpp:422-438); this is the N-dim generalisation.
+    std::vector<int> eff = effectiveDims(cartMx, cartDims, size);
+    int r = 0;
+    for (size_t d = 0; d < eff.size(); d++) {
+        r = r * eff[d] + coords[d];
+    }
+    *rank = r;
 }
 
 void MpiWorld::shiftCartesianCoords(int rank,
@@ -300,10 +363,50 @@ void MpiWorld::shiftCartesianCoords(int rank,
                                     int* source,
                                     int* destination)
 {
-    (void)direction;
-    // 1-D periodic shift
-    *source = (rank - disp + size) % size;
-    *destination = (rank + disp) % size;
+    // source = the rank that reaches me moving disp units in direction;
+    // destination = the rank I reach. Periodic in every dimension
+    // (reference: src/mpi/MpiWorld.cpp:440-490).
+    std::vector<int> eff = effectiveDims(cartMx, cartDims, size);
+    int ndims = (int)eff.size();
+
+    // Decompose my rank into row-major coords
+    std::vector<int> coords(ndims);
+    int remainder = rank;
+    for (int d = 0; d < ndims; d++) {
+        int stride = 1;
+        for (int e = d + 1; e < ndims; e++) {
+            stride *= eff[e];
+        }
+        coords[d] = remainder / stride;
+        remainder = remainder % stride;
+    }
+
+    auto fold = [&](const std::vector<int>& c) {
+        int r = 0;
+        for (int d = 0; d < ndims; d++) {
+            r = r * eff[d] + c[d];
+        }
+        return r;
+    };
+
+    if (direction < 0 || direction >= ndims) {
+        // Unused dimension: single process there, periodicity lands on self
+        *source = rank;
+        *destination = rank;
+        return;
+    }
+
+    int dim = eff[direction];
+    // Normalise disp into [0, dim) so negative displacements wrap
+    int step = ((disp % dim) + dim) % dim;
+
+    std::vector<int> fwd = coords;
+    fwd[direction] = (coords[direction] + step) % dim;
+    *destination = fold(fwd);
+
+    std::vector<int> bwd = coords;
+    bwd[direction] = (coords[direction] - step + dim) % dim;
+    *source = fold(bwd);
 }
 
 // ------------------------- host data plane ----------------------------------
@@ -755,7 +858,9 @@ void MpiWorld::scatter(int rootRank,
         for (int i = 0; i < size; i++) {
             const uint8_t* chunk = sendBuffer + (size_t)i * bytes;
             if (i == rootRank) {
-                std::memcpy(recvBuffer, chunk, bytes);
+                if (recvBuffer != chunk) {
+                    std::memcpy(recvBuffer, chunk, bytes);
+                }
             } else {
                 hostSend(rootRank, i, chunk, bytes,
                          MpiMessageType::SCATTER);
@@ -780,7 +885,9 @@ void MpiWorld::gather(int thisRank,
         for (int i = 0; i < size; i++) {
             uint8_t* dst = recvBuffer + (size_t)i * bytes;
             if (i == rootRank) {
-                std::memcpy(dst, sendBuffer, bytes);
+                if (dst != sendBuffer) {
+                    std::memcpy(dst, sendBuffer, bytes);
+                }
             } else {
                 auto data =
                   hostRecv(i, rootRank, bytes, MpiMessageType::GATHER);

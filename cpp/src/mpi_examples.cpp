@@ -354,6 +354,161 @@ static int32_t exampleMigrate(Message& msg)
     return 0;
 }
 
+// 2-D cartesian topology: 2x2 grid over 4 ranks, row-major, periodic.
+// Mirrors the reference's LAMMPS-shaped usage (src/mpi/MpiWorld.cpp:369-490)
+static int32_t exampleCartesian(Message& msg)
+{
+    MPI_Init(nullptr, nullptr);
+    int rank;
+    int worldSize;
+    MPI_Comm_rank(MPI_COMM_WORLD, &rank);
+    MPI_Comm_size(MPI_COMM_WORLD, &worldSize);
+    if (worldSize != 4) {
+        msg.outputData = "cartesian example needs world size 4";
+        MPI_Finalize();
+        return 1;
+    }
+
+    int dims[2] = { 2, 2 };
+    int periods[2] = { 1, 1 };
+    MPI_Comm cart;
+    MPI_Cart_create(MPI_COMM_WORLD, 2, dims, periods, 0, &cart);
+
+    // Row-major: rank = row*2 + col
+    int myCoords[2] = { rank / 2, rank % 2 };
+    int gotRank = -1;
+    MPI_Cart_rank(cart, myCoords, &gotRank);
+    if (gotRank != rank) {
+        msg.outputData = "cart_rank mismatch";
+        return 1;
+    }
+
+    int gDims[2], gPeriods[2], gCoords[2];
+    MPI_Cart_get(cart, 2, gDims, gPeriods, gCoords);
+    if (gDims[0] != 2 || gDims[1] != 2 || gCoords[0] != myCoords[0] ||
+        gCoords[1] != myCoords[1]) {
+        msg.outputData = "cart_get mismatch";
+        return 1;
+    }
+
+    // Shift along rows (direction 0): with 2 rows, +1 wraps to the other
+    // row, same column
+    int src = -1;
+    int dst = -1;
+    MPI_Cart_shift(cart, 0, 1, &src, &dst);
+    int expect = ((myCoords[0] + 1) % 2) * 2 + myCoords[1];
+    if (dst != expect || src != expect) {
+        msg.outputData = "cart_shift dir0 mismatch";
+        return 1;
+    }
+    // Shift along columns (direction 1)
+    MPI_Cart_shift(cart, 1, 1, &src, &dst);
+    expect = myCoords[0] * 2 + ((myCoords[1] + 1) % 2);
+    if (dst != expect || src != expect) {
+        msg.outputData = "cart_shift dir1 mismatch";
+        return 1;
+    }
+    // Negative displacement must wrap too
+    MPI_Cart_shift(cart, 1, -1, &src, &dst);
+    if (dst != expect || src != expect) {
+        msg.outputData = "cart_shift negative disp mismatch";
+        return 1;
+    }
+    // Unused dimension: single proc, periodic lands on self
+    MPI_Cart_shift(cart, 2, 1, &src, &dst);
+    if (dst != rank || src != rank) {
+        msg.outputData = "cart_shift unused dim mismatch";
+        return 1;
+    }
+
+    MPI_Barrier(MPI_COMM_WORLD);
+    MPI_Finalize();
+    msg.outputData = "cartesian ok";
+    return 0;
+}
+
+// MPI_IN_PLACE across the reduction + gather collectives
+static int32_t exampleInPlace(Message& msg)
+{
+    MPI_Init(nullptr, nullptr);
+    int rank;
+    int worldSize;
+    MPI_Comm_rank(MPI_COMM_WORLD, &rank);
+    MPI_Comm_size(MPI_COMM_WORLD, &worldSize);
+    int sum = worldSize * (worldSize + 1) / 2;
+
+    // Allreduce in place
+    std::vector<int> buf(16, rank + 1);
+    MPI_Allreduce(MPI_IN_PLACE, buf.data(), 16, MPI_INT, MPI_SUM,
+                  MPI_COMM_WORLD);
+    for (int v : buf) {
+        if (v != sum) {
+            msg.outputData = "in-place allreduce mismatch";
+            return 1;
+        }
+    }
+
+    // Reduce in place at root
+    std::vector<int> rbuf(8, rank + 1);
+    if (rank == 0) {
+        MPI_Reduce(MPI_IN_PLACE, rbuf.data(), 8, MPI_INT, MPI_SUM, 0,
+                   MPI_COMM_WORLD);
+        for (int v : rbuf) {
+            if (v != sum) {
+                msg.outputData = "in-place reduce mismatch";
+                return 1;
+            }
+        }
+    } else {
+        MPI_Reduce(rbuf.data(), nullptr, 8, MPI_INT, MPI_SUM, 0,
+                   MPI_COMM_WORLD);
+    }
+
+    // Allgather in place: contribution pre-placed at my slot
+    std::vector<int> gbuf(worldSize, -1);
+    gbuf[rank] = rank * 7;
+    MPI_Allgather(MPI_IN_PLACE, 0, MPI_DATATYPE_NULL, gbuf.data(), 1,
+                  MPI_INT, MPI_COMM_WORLD);
+    for (int r = 0; r < worldSize; r++) {
+        if (gbuf[r] != r * 7) {
+            msg.outputData = "in-place allgather mismatch";
+            return 1;
+        }
+    }
+
+    // Gather in place at root
+    std::vector<int> g2(worldSize, -1);
+    if (rank == 0) {
+        g2[0] = 100;
+        MPI_Gather(MPI_IN_PLACE, 0, MPI_DATATYPE_NULL, g2.data(), 1,
+                   MPI_INT, 0, MPI_COMM_WORLD);
+        for (int r = 0; r < worldSize; r++) {
+            int expect = r == 0 ? 100 : r + 100;
+            if (g2[r] != expect) {
+                msg.outputData = "in-place gather mismatch";
+                return 1;
+            }
+        }
+    } else {
+        int mine = rank + 100;
+        MPI_Gather(&mine, 1, MPI_INT, nullptr, 1, MPI_INT, 0,
+                   MPI_COMM_WORLD);
+    }
+
+    // Scan in place
+    int scanv = rank + 1;
+    MPI_Scan(MPI_IN_PLACE, &scanv, 1, MPI_INT, MPI_SUM, MPI_COMM_WORLD);
+    if (scanv != (rank + 1) * (rank + 2) / 2) {
+        msg.outputData = "in-place scan mismatch";
+        return 1;
+    }
+
+    MPI_Barrier(MPI_COMM_WORLD);
+    MPI_Finalize();
+    msg.outputData = "in-place ok";
+    return 0;
+}
+
 void registerMpiExampleFunctions()
 {
     auto& reg = FunctionRegistry::get();
@@ -362,6 +517,8 @@ void registerMpiExampleFunctions()
     reg.registerFunction("mpi-cpp", "async", exampleAsync);
     reg.registerFunction("mpi-cpp", "vcollectives", exampleVCollectives);
     reg.registerFunction("mpi-cpp", "migrate", exampleMigrate);
+    reg.registerFunction("mpi-cpp", "cartesian", exampleCartesian);
+    reg.registerFunction("mpi-cpp", "inplace", exampleInPlace);
     reg.registerFunction("mpi-cpp", "allreduce-small-bench",
                          exampleAllReduceSmallBench);
     reg.registerFunction("mpi-cpp", "allreduce-bench",

@@ -421,6 +421,12 @@ int MPI_Scatter(const void* sendbuf,
     (void)recvcount;
     (void)recvtype;
     (void)comm;
+    if (recvbuf == MPI_IN_PLACE && thisRank() == root) {
+        // Root's chunk stays where it already is in sendbuf; MpiWorld's
+        // scatter skips the aliased self-copy.
+        recvbuf = (uint8_t*)sendbuf +
+                  (size_t)root * (size_t)sendcount * sendtype->size;
+    }
     world().scatter(root,
                     thisRank(),
                     (const uint8_t*)sendbuf,
@@ -442,6 +448,14 @@ int MPI_Gather(const void* sendbuf,
     (void)recvcount;
     (void)recvtype;
     (void)comm;
+    if (sendbuf == MPI_IN_PLACE && thisRank() == root) {
+        // Root's contribution is already in place in recvbuf; sendtype /
+        // sendcount are ignored with MPI_IN_PLACE (MPI-2.2).
+        sendbuf = (uint8_t*)recvbuf +
+                  (size_t)root * (size_t)recvcount * recvtype->size;
+        sendtype = recvtype;
+        sendcount = recvcount;
+    }
     world().gather(thisRank(),
                    root,
                    (const uint8_t*)sendbuf,
@@ -462,6 +476,12 @@ int MPI_Allgather(const void* sendbuf,
     (void)recvcount;
     (void)recvtype;
     (void)comm;
+    if (sendbuf == MPI_IN_PLACE) {
+        sendbuf = (uint8_t*)recvbuf +
+                  (size_t)thisRank() * (size_t)recvcount * recvtype->size;
+        sendtype = recvtype;
+        sendcount = recvcount;
+    }
     world().allGather(thisRank(),
                       (const uint8_t*)sendbuf,
                       (uint8_t*)recvbuf,
@@ -479,6 +499,14 @@ int MPI_Reduce(const void* sendbuf,
                MPI_Comm comm)
 {
     (void)comm;
+    std::vector<uint8_t> inPlace;
+    if (sendbuf == MPI_IN_PLACE && thisRank() == root) {
+        // Input is taken from recvbuf; stage a copy so the reduction
+        // never aliases its accumulation target.
+        size_t bytes = (size_t)count * datatype->size;
+        inPlace.assign((uint8_t*)recvbuf, (uint8_t*)recvbuf + bytes);
+        sendbuf = inPlace.data();
+    }
     world().reduce(thisRank(),
                    root,
                    (const uint8_t*)sendbuf,
@@ -497,6 +525,12 @@ int MPI_Allreduce(const void* sendbuf,
                   MPI_Comm comm)
 {
     (void)comm;
+    std::vector<uint8_t> inPlace;
+    if (sendbuf == MPI_IN_PLACE) {
+        size_t bytes = (size_t)count * datatype->size;
+        inPlace.assign((uint8_t*)recvbuf, (uint8_t*)recvbuf + bytes);
+        sendbuf = inPlace.data();
+    }
     world().allReduce(thisRank(),
                       (const uint8_t*)sendbuf,
                       (uint8_t*)recvbuf,
@@ -514,6 +548,14 @@ int MPI_Reduce_scatter(const void* sendbuf,
                        MPI_Comm comm)
 {
     (void)comm;
+    std::vector<uint8_t> inPlace;
+    if (sendbuf == MPI_IN_PLACE) {
+        // Full input lives in recvbuf (must hold all size*count elems)
+        size_t bytes =
+          (size_t)recvcounts[0] * world().getSize() * datatype->size;
+        inPlace.assign((uint8_t*)recvbuf, (uint8_t*)recvbuf + bytes);
+        sendbuf = inPlace.data();
+    }
     // Uniform counts only (like RCCL's reduce-scatter)
     world().reduceScatter(thisRank(),
                           (const uint8_t*)sendbuf,
@@ -532,6 +574,12 @@ int MPI_Scan(const void* sendbuf,
              MPI_Comm comm)
 {
     (void)comm;
+    std::vector<uint8_t> inPlace;
+    if (sendbuf == MPI_IN_PLACE) {
+        size_t bytes = (size_t)count * datatype->size;
+        inPlace.assign((uint8_t*)recvbuf, (uint8_t*)recvbuf + bytes);
+        sendbuf = inPlace.data();
+    }
     world().scan(thisRank(),
                  (const uint8_t*)sendbuf,
                  (uint8_t*)recvbuf,
@@ -552,6 +600,15 @@ int MPI_Alltoall(const void* sendbuf,
     (void)recvcount;
     (void)recvtype;
     (void)comm;
+    std::vector<uint8_t> inPlace;
+    if (sendbuf == MPI_IN_PLACE) {
+        size_t bytes =
+          (size_t)recvcount * world().getSize() * recvtype->size;
+        inPlace.assign((uint8_t*)recvbuf, (uint8_t*)recvbuf + bytes);
+        sendbuf = inPlace.data();
+        sendtype = recvtype;
+        sendcount = recvcount;
+    }
     world().allToAll(thisRank(),
                      (const uint8_t*)sendbuf,
                      (uint8_t*)recvbuf,
@@ -569,10 +626,9 @@ int MPI_Cart_create(MPI_Comm old_comm,
                     int reorder,
                     MPI_Comm* comm)
 {
-    (void)ndims;
-    (void)dims;
     (void)periods;
     (void)reorder;
+    world().setCartesianDims(ndims, dims);
     *comm = old_comm;
     return MPI_SUCCESS;
 }
@@ -591,12 +647,7 @@ int MPI_Cart_get(MPI_Comm comm,
                  int* coords)
 {
     (void)comm;
-    // 1-D layout like shiftCartesianCoords
-    for (int i = 0; i < maxdims; i++) {
-        dims[i] = i == 0 ? world().getSize() : 1;
-        periods[i] = 1;
-        coords[i] = i == 0 ? thisRank() : 0;
-    }
+    world().getCartesianGrid(thisRank(), maxdims, dims, periods, coords);
     return MPI_SUCCESS;
 }
 

@@ -207,6 +207,14 @@ void DeviceSnapshot::applyCompactDiffFromHost(
     if (payloadBytes != pages.size() * DEVICE_PAGE) {
         throw FaabricException("compact diff size mismatch");
     }
+    // Page indices come off the wire; an out-of-range index would make
+    // the XOR kernel write past the snapshot allocation in HBM.
+    const size_t nPages = (bytes_ + DEVICE_PAGE - 1) / DEVICE_PAGE;
+    for (uint32_t p : pages) {
+        if ((size_t)p >= nPages) {
+            throw FaabricException("compact diff page out of range");
+        }
+    }
     OPS_HIP_CHECK(hipSetDevice(device_));
     uint32_t* pagesDev = nullptr;
     uint8_t* payloadDev = nullptr;

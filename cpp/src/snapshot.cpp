@@ -524,6 +524,12 @@ void SnapshotData::applyDiff(const SnapshotDiff& diff)
                 uint32_t page = 0;
                 std::memcpy(&page, src + 4 + (size_t)i * 4, 4);
                 size_t off = (size_t)page * SNAPSHOT_PAGE_SIZE;
+                // Page indices arrive off the wire (ThreadResult RPC);
+                // an index past the snapshot would make maxSize_ - off
+                // underflow into a heap OOB write. Reject the diff.
+                if (off >= maxSize_) {
+                    throw FaabricException("packed page diff out of range");
+                }
                 const uint8_t* payload =
                   src + header + (size_t)i * SNAPSHOT_PAGE_SIZE;
                 size_t len =

@@ -136,7 +136,8 @@ def test_cpp_mpi_examples(runtime):
     surface) run as native functions."""
     _core.register_mpi_example_functions()
     for func in ("allreduce", "ring", "async", "allreduce-bench",
-                 "allreduce-small-bench", "vcollectives"):
+                 "allreduce-small-bench", "vcollectives", "cartesian",
+                 "inplace"):
         results = submit_mpi_batch("mpi-cpp", func, WORLD_SIZE)
         assert len(results) == WORLD_SIZE
         for r in results:
