@@ -5,6 +5,7 @@
 
 #include "faabricamd/endpoint.h"
 #include "faabricamd/executor.h"
+#include "faabricamd/hipipc.h"
 #include "faabricamd/planner.h"
 #include "faabricamd/ptp.h"
 #include "faabricamd/runner.h"
@@ -381,6 +382,29 @@ void initRuntimeBindings(py::module_& m)
         getPointToPointBroker().setUpLocalMappingsFromSchedulingDecision(d);
     });
     m.def("ptp_clear", [] { getPointToPointBroker().clear(); });
+    // Standalone PTP server (multi-process PTP/IPC tests run it without
+    // the rest of the worker fabric)
+    py::class_<PointToPointServer>(m, "PointToPointServerHandle")
+      .def(py::init<>())
+      .def("start",
+           [](PointToPointServer& s) {
+               py::gil_scoped_release release;
+               s.start();
+           })
+      .def("stop", [](PointToPointServer& s) {
+          py::gil_scoped_release release;
+          s.stop();
+      });
+    m.def("ipc_shipped", [] {
+        return py::make_tuple(IpcSender::get().shippedSegments(),
+                              IpcSender::get().shippedBytes());
+    });
+    m.def("ipc_available",
+          [](const std::string& host) {
+              py::gil_scoped_release release;
+              return IpcSender::get().available(host);
+          },
+          py::arg("host"));
     m.def("ptp_wait_for_mappings", [](int32_t groupId) {
         py::gil_scoped_release release;
         getPointToPointBroker().waitForMappingsOnThisHost(groupId);

@@ -34,6 +34,10 @@ enum class PointToPointCall : uint8_t
     LOCK_GROUP_RECURSIVE = 3,
     UNLOCK_GROUP = 4,
     UNLOCK_GROUP_RECURSIVE = 5,
+    // HIP-IPC peer transport between same-node workers (hipipc.h)
+    IPC_ARENA = 6,   // sync: body = sender host, reply = IpcArenaInfo
+    MESSAGE_IPC = 7, // async: IpcPtpMessage (payload already in arena)
+    IPC_ACK = 8,     // async: IpcAck (receiver freed a segment)
 };
 
 inline constexpr int32_t POINT_TO_POINT_MAIN_IDX = 0;
@@ -147,6 +151,20 @@ class PointToPointBroker
     // Deliver a message that arrived over the network
     void deliverRemoteMessage(const PointToPointMessage& msg, uint32_t seq);
 
+    // Deliver a MESSAGE_IPC control message (payload already in our arena)
+    void deliverIpcMessage(int32_t groupId,
+                           int32_t sendIdx,
+                           int32_t recvIdx,
+                           const std::string& senderHost,
+                           uint64_t offset,
+                           uint64_t size,
+                           uint32_t seq);
+
+    // Ack a consumed arena segment back to its sender
+    void sendIpcAck(const std::string& senderHost,
+                    uint64_t offset,
+                    uint64_t size);
+
     // Group cleanup after an app finishes / migrates
     void clearGroup(int32_t groupId);
     void clear();
@@ -168,6 +186,12 @@ class PointToPointBroker
         std::vector<uint8_t> host;
         void* dev = nullptr; // HBM staging copy (same-process delivery)
         size_t devSize = 0;
+        // HIP-IPC segment in OUR arena (cross-process same-node sender);
+        // consumed by copying out of the arena then acking the segment
+        bool isArena = false;
+        std::string arenaHost; // sender host (names the arena)
+        uint64_t arenaOff = 0;
+        uint64_t arenaSize = 0;
     };
 
     struct Channel

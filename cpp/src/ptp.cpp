@@ -3,6 +3,7 @@
 // :637-859, groups :142-365, postMigrationHook :910-926) — fresh design,
 // see ptp.h.
 #include "faabricamd/ptp.h"
+#include "faabricamd/hipipc.h"
 #include "faabricamd/util.h"
 
 #include <hip/hip_runtime.h>
@@ -451,6 +452,44 @@ void PointToPointBroker::deliverRemoteMessage(const PointToPointMessage& msg,
     deliverPayload(msg.groupId, msg.sendIdx, msg.recvIdx, std::move(p), seq);
 }
 
+void PointToPointBroker::deliverIpcMessage(int32_t groupId,
+                                           int32_t sendIdx,
+                                           int32_t recvIdx,
+                                           const std::string& senderHost,
+                                           uint64_t offset,
+                                           uint64_t size,
+                                           uint32_t seq)
+{
+    PtpPayload p;
+    p.isArena = true;
+    p.arenaHost = senderHost;
+    p.arenaOff = offset;
+    p.arenaSize = size;
+    deliverPayload(groupId, sendIdx, recvIdx, std::move(p), seq);
+}
+
+void PointToPointBroker::sendIpcAck(const std::string& senderHost,
+                                    uint64_t offset,
+                                    uint64_t size)
+{
+    IpcAck ack;
+    ack.receiverHost = getSystemConfig().endpointHost;
+    ack.offset = offset;
+    ack.size = size;
+    getClient(senderHost)
+      ->asyncSend((uint8_t)PointToPointCall::IPC_ACK, ack.encode());
+}
+
+// Declared in hipipc.cpp: arena-handle exchange rides the PTP sync plane
+IpcArenaInfo fetchIpcArenaFromHost(const std::string& host)
+{
+    PointToPointClient cli(host);
+    std::string reply =
+      cli.syncSend((uint8_t)PointToPointCall::IPC_ARENA,
+                   getSystemConfig().endpointHost);
+    return IpcArenaInfo::decode(reply);
+}
+
 PointToPointBroker::PtpPayload PointToPointBroker::recvPayload(
   int32_t groupId,
   int32_t sendIdx,
@@ -500,6 +539,14 @@ std::vector<uint8_t> PointToPointBroker::recvMessage(int32_t groupId,
 {
     PtpPayload p =
       recvPayload(groupId, sendIdx, recvIdx, mustOrderMsgs, timeoutMs);
+    if (p.isArena) {
+        // HIP-IPC segment consumed through the host API
+        std::vector<uint8_t> out(p.arenaSize);
+        IpcReceiver::get().copyToHost(p.arenaHost, p.arenaOff,
+                                      out.data(), p.arenaSize);
+        sendIpcAck(p.arenaHost, p.arenaOff, p.arenaSize);
+        return out;
+    }
     if (p.dev != nullptr) {
         // Device-staged message consumed through the host API
         std::vector<uint8_t> out(p.devSize);
@@ -567,7 +614,36 @@ void PointToPointBroker::sendMessageDevice(int32_t appId,
         return;
     }
 
-    // Cross-host: stage D2H and ride the RPC plane
+    // Same-node different worker process: ship the payload straight into
+    // the receiver's HBM arena over xGMI (HIP IPC), then send a small
+    // control message. No D2H in the hot path.
+    if (isSameNodeDifferentWorker(thisHost, host) &&
+        IpcSender::get().available(host)) {
+        try {
+            uint64_t off = IpcSender::get().ship(host, devPtr, size);
+            IpcPtpMessage ctl;
+            ctl.appId = appId;
+            ctl.groupId = groupId;
+            ctl.sendIdx = sendIdx;
+            ctl.recvIdx = recvIdx;
+            ctl.senderHost = thisHost;
+            ctl.offset = off;
+            ctl.size = size;
+            auto cli = getClient(host);
+            std::string body = ctl.encode();
+            cli->asyncSendSeq((uint8_t)PointToPointCall::MESSAGE_IPC,
+                              body.data(),
+                              body.size(),
+                              seq);
+            return;
+        } catch (const std::exception& e) {
+            FAM_WARN("ptp: ipc ship to %s failed (%s); falling back",
+                     host.c_str(),
+                     e.what());
+        }
+    }
+
+    // Cross-node (or IPC unavailable): stage D2H and ride the RPC plane
     PointToPointMessage msg;
     msg.appId = appId;
     msg.groupId = groupId;
@@ -598,6 +674,16 @@ size_t PointToPointBroker::recvMessageDevice(int32_t groupId,
 {
     PtpPayload p =
       recvPayload(groupId, sendIdx, recvIdx, mustOrderMsgs, timeoutMs);
+    if (p.isArena) {
+        // HIP-IPC segment: local D2D out of our arena, then ack
+        if (p.arenaSize > capacity) {
+            throw FaabricException("ptp device recv buffer too small");
+        }
+        IpcReceiver::get().copyToDevice(p.arenaHost, p.arenaOff, devPtr,
+                                        p.arenaSize);
+        sendIpcAck(p.arenaHost, p.arenaOff, p.arenaSize);
+        return p.arenaSize;
+    }
     hipStream_t s = (hipStream_t)sideStream();
     if (p.dev != nullptr) {
         if (p.devSize > capacity) {
@@ -732,6 +818,22 @@ void PointToPointServer::doAsyncRecv(uint8_t code,
             group->handleUnlockRequest("", msg.sendIdx, recursive);
             break;
         }
+        case PointToPointCall::MESSAGE_IPC: {
+            IpcPtpMessage m = IpcPtpMessage::decode(body);
+            getPointToPointBroker().deliverIpcMessage(m.groupId,
+                                                      m.sendIdx,
+                                                      m.recvIdx,
+                                                      m.senderHost,
+                                                      m.offset,
+                                                      m.size,
+                                                      seq);
+            break;
+        }
+        case PointToPointCall::IPC_ACK: {
+            IpcAck ack = IpcAck::decode(body);
+            IpcSender::get().onAck(ack.receiverHost, ack.offset, ack.size);
+            break;
+        }
         default:
             FAM_ERROR("ptp server: bad async call %d", (int)code);
     }
@@ -747,6 +849,10 @@ std::string PointToPointServer::doSyncRecv(uint8_t code,
         getPointToPointBroker().setUpLocalMappingsFromSchedulingDecision(
           decision);
         return {};
+    }
+    if (call == PointToPointCall::IPC_ARENA) {
+        // body = requesting sender's host identity
+        return IpcReceiver::get().arenaFor(body).encode();
     }
     throw FaabricException("ptp server: bad sync call " +
                            std::to_string(code));
