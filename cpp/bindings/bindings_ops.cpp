@@ -185,6 +185,22 @@ void initOpsBindings(py::module_& m)
 {
     m.def("gpu_available", &gpuAvailable);
     m.def("gpu_count", &gpuCount);
+    m.def("device_snapshot_exists", [](const std::string& key) {
+        return DeviceSnapshotRegistry::get().snapshotExists(key);
+    });
+    m.def("device_snapshot_read",
+          [](const std::string& key, size_t offset, size_t len) {
+              auto snap = DeviceSnapshotRegistry::get().getSnapshot(key);
+              std::vector<uint8_t> out(len);
+              {
+                  py::gil_scoped_release release;
+                  snap->copyOutHost(out.data(), len, offset);
+              }
+              return py::bytes((const char*)out.data(), out.size());
+          },
+          py::arg("key"),
+          py::arg("offset"),
+          py::arg("len"));
 
     py::class_<DeviceSnapshot>(m, "DeviceSnapshot")
       .def(py::init<size_t, int>(), py::arg("bytes"), py::arg("device") = 0)

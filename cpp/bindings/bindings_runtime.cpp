@@ -382,7 +382,7 @@ void initRuntimeBindings(py::module_& m)
         getPointToPointBroker().setUpLocalMappingsFromSchedulingDecision(d);
     });
     m.def("ptp_clear", [] { getPointToPointBroker().clear(); });
-    // Standalone PTP server (multi-process PTP/IPC tests run it without
+    // Standalone servers (multi-process PTP/IPC tests run these without
     // the rest of the worker fabric)
     py::class_<PointToPointServer>(m, "PointToPointServerHandle")
       .def(py::init<>())
@@ -395,6 +395,41 @@ void initRuntimeBindings(py::module_& m)
           py::gil_scoped_release release;
           s.stop();
       });
+    py::class_<StateServer>(m, "StateServerHandle")
+      .def(py::init<>())
+      .def("start",
+           [](StateServer& s) {
+               py::gil_scoped_release release;
+               s.start();
+           })
+      .def("stop", [](StateServer& s) {
+          py::gil_scoped_release release;
+          s.stop();
+      });
+    py::class_<SnapshotServer>(m, "SnapshotServerHandle")
+      .def(py::init<>())
+      .def("start",
+           [](SnapshotServer& s) {
+               py::gil_scoped_release release;
+               s.start();
+           })
+      .def("stop", [](SnapshotServer& s) {
+          py::gil_scoped_release release;
+          s.stop();
+      });
+    m.def("snapshot_push_device_from_ptr",
+          [](const std::string& host,
+             const std::string& key,
+             uintptr_t devPtr,
+             size_t size) {
+              py::gil_scoped_release release;
+              getSnapshotClient(host)->pushDeviceSnapshotFromDevice(
+                key, (const void*)devPtr, size);
+          },
+          py::arg("host"),
+          py::arg("key"),
+          py::arg("dev_ptr"),
+          py::arg("size"));
     m.def("ipc_shipped", [] {
         return py::make_tuple(IpcSender::get().shippedSegments(),
                               IpcSender::get().shippedBytes());

@@ -56,6 +56,22 @@ struct IpcPtpMessage
     static IpcPtpMessage decode(const std::string& buf);
 };
 
+// One shipped chunk of a larger value (state KV pull/push, device
+// snapshot ship). Values larger than the arena stream as a sequence of
+// chunks, each acked as it is consumed so the ring recycles.
+struct IpcChunk
+{
+    std::string user;      // 1 (state namespace; empty for snapshots)
+    std::string key;       // 2
+    uint64_t valOffset = 0; // 3: offset in the destination value/snapshot
+    uint64_t ipcOffset = 0; // 4: segment in the receiver's arena
+    uint64_t len = 0;       // 5
+    std::string srcHost;    // 6: names the arena (the shipping side)
+    uint64_t totalSize = 0; // 7: full value size (create-on-first)
+    std::string encode() const;
+    static IpcChunk decode(const std::string& buf);
+};
+
 // Segment ack (async IPC_ACK call, receiver -> sender)
 struct IpcAck
 {
@@ -130,6 +146,10 @@ class IpcSender
     // Segments/bytes shipped so far (tests assert the IPC path ran)
     uint64_t shippedSegments() const;
     uint64_t shippedBytes() const;
+
+    // Arena capacity at targetHost (0 = unavailable); bulk transfers
+    // pick their chunk size from this
+    uint64_t peerCapacity(const std::string& targetHost);
 
     void clear();
     ~IpcSender();

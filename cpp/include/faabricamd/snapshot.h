@@ -212,6 +212,9 @@ enum class SnapshotCalls : uint8_t
     PushSnapshotUpdate = 2,
     DeleteSnapshot = 3,
     ThreadResult = 4,
+    // Chunk of a device snapshot shipped via HIP IPC (same-node workers):
+    // payload is already in our arena, body is an IpcChunk
+    PushSnapshotIpc = 5,
 };
 
 class SnapshotServer : public MessageEndpointServer
@@ -233,6 +236,11 @@ class SnapshotClient : public MessageEndpointClient
     // the destination's device registry
     void pushDeviceSnapshot(const std::string& key, const void* hostCopy,
                             size_t size);
+    // Same, straight from HBM: same-node workers stream arena chunks
+    // over xGMI (no D2H); falls back to a host copy + pushDeviceSnapshot
+    void pushDeviceSnapshotFromDevice(const std::string& key,
+                                      const void* devPtr,
+                                      size_t size);
     void pushSnapshotUpdate(const std::string& key,
                             const std::vector<SnapshotDiff>& diffs,
                             const std::vector<SnapshotMergeRegion>& regions);

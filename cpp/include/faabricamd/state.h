@@ -15,6 +15,7 @@
 #include <string>
 #include <vector>
 
+#include "faabricamd/hipipc.h"
 #include "faabricamd/messages.h"
 #include "faabricamd/transport.h"
 
@@ -33,6 +34,10 @@ enum class StateCalls : uint8_t
     ClearAppended = 5,
     PullAppended = 6,
     Delete = 7,
+    // HIP-IPC bulk paths for device-resident values between same-node
+    // workers (hipipc.h): payload rides the arena, RPC carries segments
+    PullIpc = 8,
+    PushIpc = 9,
 };
 
 class StateKeyValue
@@ -86,6 +91,15 @@ class StateKeyValue
     // Master-side servicing
     std::vector<uint8_t> serviceChunk(uint64_t offset, size_t len);
     void serviceSet(uint64_t offset, const uint8_t* data, size_t len);
+    // IPC variants: ship a range into dstHost's arena / absorb a shipped
+    // segment from our arena (master side of PullIpc/PushIpc)
+    uint64_t serviceChunkIpc(const std::string& dstHost,
+                             uint64_t offset,
+                             size_t len);
+    void serviceSetIpc(const std::string& srcHost,
+                       uint64_t ipcOffset,
+                       uint64_t valOffset,
+                       size_t len);
     void serviceAppend(const uint8_t* data, size_t len);
     std::vector<std::vector<uint8_t>> serviceGetAppended(size_t n);
     void serviceClearAppended();
@@ -104,6 +118,10 @@ class StateKeyValue
 
     void readLocal(uint64_t offset, uint8_t* out, size_t len);
     void writeLocal(uint64_t offset, const uint8_t* data, size_t len);
+    // True when bulk pull/push to the master should ride HIP IPC
+    bool useIpcToMaster();
+    void pullRangeIpc(uint64_t offset, size_t len);
+    void pushRangeIpc(uint64_t offset, size_t len);
     std::vector<char> dirtyChunks; // one flag per STATE_STREAM_CHUNK_SIZE
     bool fullyPulled = false;
     std::vector<std::vector<uint8_t>> appendedValues;
@@ -168,6 +186,13 @@ class StateClient : public MessageEndpointClient
                    uint64_t offset,
                    const uint8_t* data,
                    size_t len);
+    // IPC bulk paths (payload in the arena; see StateCalls::PullIpc)
+    IpcChunk pullChunkIpc(const std::string& user,
+                          const std::string& key,
+                          uint64_t offset,
+                          size_t len,
+                          const std::string& requesterHost);
+    void pushChunkIpc(const IpcChunk& chunk);
     size_t stateSize(const std::string& user, const std::string& key);
     void append(const std::string& user,
                 const std::string& key,

@@ -327,7 +327,6 @@ std::vector<std::pair<int32_t, int32_t>> Executor::executeThreads(
     lastForkSnapshotKey = key;
     {
         const std::string& thisHost = getSystemConfig().endpointHost;
-        std::vector<uint8_t> hostCopy;
         for (const auto& host : decision->uniqueHosts()) {
             if (host == thisHost) {
                 continue;
@@ -338,12 +337,10 @@ std::vector<std::pair<int32_t, int32_t>> Executor::executeThreads(
                 lastForkRemoteHosts.push_back(host);
             }
             if (onDevice) {
-                if (hostCopy.empty()) {
-                    hostCopy.resize(dsnap->size());
-                    dsnap->copyOutHost(hostCopy.data(), dsnap->size());
-                }
-                getSnapshotClient(host)->pushDeviceSnapshot(
-                  key, hostCopy.data(), hostCopy.size());
+                // Same-node workers stream this over xGMI (HIP IPC);
+                // only cross-node peers fall back to a host copy
+                getSnapshotClient(host)->pushDeviceSnapshotFromDevice(
+                  key, dsnap->data(), dsnap->size());
             } else {
                 getSnapshotClient(host)->pushSnapshot(key, *snap);
             }

@@ -120,6 +120,40 @@ IpcPtpMessage IpcPtpMessage::decode(const std::string& buf)
     return m;
 }
 
+std::string IpcChunk::encode() const
+{
+    PbWriter w;
+    w.putString(1, user);
+    w.putString(2, key);
+    w.putUInt64(3, valOffset);
+    w.putUInt64(4, ipcOffset);
+    w.putUInt64(5, len);
+    w.putString(6, srcHost);
+    w.putUInt64(7, totalSize);
+    return w.take();
+}
+
+IpcChunk IpcChunk::decode(const std::string& buf)
+{
+    IpcChunk m;
+    PbReader r(buf);
+    uint32_t f;
+    WireType t;
+    while (r.next(f, t)) {
+        switch (f) {
+            case 1: m.user = r.asString(); break;
+            case 2: m.key = r.asString(); break;
+            case 3: m.valOffset = r.asUInt64(); break;
+            case 4: m.ipcOffset = r.asUInt64(); break;
+            case 5: m.len = r.asUInt64(); break;
+            case 6: m.srcHost = r.asString(); break;
+            case 7: m.totalSize = r.asUInt64(); break;
+            default: r.skip(t);
+        }
+    }
+    return m;
+}
+
 std::string IpcAck::encode() const
 {
     PbWriter w;
@@ -462,6 +496,12 @@ uint64_t IpcSender::shippedSegments() const
 uint64_t IpcSender::shippedBytes() const
 {
     return bytesShipped.load(std::memory_order_relaxed);
+}
+
+uint64_t IpcSender::peerCapacity(const std::string& targetHost)
+{
+    auto peer = ensurePeer(targetHost);
+    return peer->ok ? peer->cap : 0;
 }
 
 uint64_t IpcSender::ship(const std::string& targetHost,

@@ -37,18 +37,13 @@ int32_t migrationPoint(const std::vector<uint8_t>& reentryInput)
         call.snapshotKey = snapKey;
 
         if (exec->hasDeviceArena()) {
-            // GPU-resident function: freeze the HBM arena. It travels as
-            // bytes and lands as a DeviceSnapshot wherever the app thaws
-            // (snapshot.cpp PushSnapshot onDevice path)
+            // GPU-resident function: freeze the HBM arena. Same-node it
+            // streams over xGMI (HIP IPC); otherwise it travels as bytes
+            // and lands as a DeviceSnapshot wherever the app thaws
             auto [dbase, dsize] = exec->getDeviceMemoryView();
-            std::vector<uint8_t> hostCopy(dsize);
-            if (hipMemcpy(hostCopy.data(), dbase, dsize,
-                          hipMemcpyDeviceToHost) != hipSuccess) {
-                FAM_ERROR("freeze: device arena copy-out failed");
-            }
             try {
                 getSnapshotClient(getSystemConfig().plannerHost)
-                  ->pushDeviceSnapshot(snapKey, hostCopy.data(), dsize);
+                  ->pushDeviceSnapshotFromDevice(snapKey, dbase, dsize);
             } catch (const std::exception& e) {
                 FAM_ERROR("freeze device snapshot push failed: %s",
                           e.what());
@@ -114,15 +109,11 @@ int32_t migrationPoint(const std::vector<uint8_t>& reentryInput)
     std::string snapKey = "migration_" + std::to_string(msg.id);
     if (exec->hasDeviceArena()) {
         // Ship the HBM arena; the destination lands it in ITS GPU and
-        // restore() D2D-copies it into the fresh executor's arena
+        // restore() D2D-copies it into the fresh executor's arena.
+        // Same-node destinations stream over xGMI (HIP IPC)
         auto [dbase, dsize] = exec->getDeviceMemoryView();
-        std::vector<uint8_t> hostCopy(dsize);
-        if (hipMemcpy(hostCopy.data(), dbase, dsize,
-                      hipMemcpyDeviceToHost) != hipSuccess) {
-            throw FaabricException("migration: device copy-out failed");
-        }
         getSnapshotClient(migration->dstHost)
-          ->pushDeviceSnapshot(snapKey, hostCopy.data(), dsize);
+          ->pushDeviceSnapshotFromDevice(snapKey, dbase, dsize);
         msg.snapshotKey = snapKey;
     } else {
         auto [base, size] = exec->getMemoryView();

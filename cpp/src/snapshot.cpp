@@ -4,8 +4,10 @@
 // device (HBM) path dispatches the same operations to gfx950 HIP kernels —
 // see cpp/hip/snapshot_kernels.hip and ops.cpp.
 #include "faabricamd/snapshot.h"
+#include "faabricamd/hipipc.h"
 #include "faabricamd/ops.h"
 #include "faabricamd/planner.h"
+#include "faabricamd/ptp.h"
 #include "faabricamd/util.h"
 
 #include <algorithm>
@@ -710,6 +712,40 @@ std::string SnapshotServer::doSyncRecv(uint8_t code, const std::string& body)
             SnapshotRegistry::get().registerSnapshot(req.key, snap);
             return {};
         }
+        case SnapshotCalls::PushSnapshotIpc: {
+            // Streamed device snapshot: land the arena segment straight
+            // in HBM (create the snapshot on the first chunk)
+            auto c = IpcChunk::decode(body);
+            if (!gpuAvailable()) {
+                throw FaabricException(
+                  "device snapshot pushed to GPU-less host");
+            }
+            auto& reg = DeviceSnapshotRegistry::get();
+            std::shared_ptr<DeviceSnapshot> dsnap;
+            if (reg.snapshotExists(c.key)) {
+                dsnap = reg.getSnapshot(c.key);
+                if (dsnap->size() <
+                    (c.totalSize + DEVICE_PAGE - 1) / DEVICE_PAGE *
+                      DEVICE_PAGE) {
+                    dsnap.reset();
+                }
+            }
+            if (!dsnap) {
+                size_t rounded = (c.totalSize + DEVICE_PAGE - 1) /
+                                 DEVICE_PAGE * DEVICE_PAGE;
+                dsnap = std::make_shared<DeviceSnapshot>(rounded, 0);
+                reg.registerSnapshot(c.key, dsnap);
+            }
+            if (c.valOffset + c.len > dsnap->size()) {
+                throw FaabricException("snapshot ipc chunk out of range");
+            }
+            IpcReceiver::get().copyToDevice(
+              c.srcHost, c.ipcOffset,
+              (uint8_t*)dsnap->data() + c.valOffset, c.len);
+            getPointToPointBroker().sendIpcAck(c.srcHost, c.ipcOffset,
+                                               c.len);
+            return {};
+        }
         case SnapshotCalls::PushSnapshotUpdate: {
             auto req = SnapshotUpdateRequest::decode(body);
             auto snap = SnapshotRegistry::get().getSnapshot(req.key);
@@ -836,6 +872,52 @@ void SnapshotClient::pushDeviceSnapshot(const std::string& key,
     req.contents.assign((const uint8_t*)hostCopy,
                         (const uint8_t*)hostCopy + size);
     syncSend((uint8_t)SnapshotCalls::PushSnapshot, req.encode());
+}
+
+void SnapshotClient::pushDeviceSnapshotFromDevice(const std::string& key,
+                                                  const void* devPtr,
+                                                  size_t size)
+{
+    if (isMockMode()) {
+        std::lock_guard<std::mutex> lock(snapMockMx);
+        mockedPushes.emplace_back(getHost(), key);
+        return;
+    }
+    const std::string& thisHost = getSystemConfig().endpointHost;
+    if (isSameNodeDifferentWorker(thisHost, getHost()) &&
+        IpcSender::get().available(getHost())) {
+        try {
+            uint64_t cap = IpcSender::get().peerCapacity(getHost());
+            uint64_t chunk = std::max<uint64_t>(cap / 2, 4096);
+            uint64_t off = 0;
+            while (off < size) {
+                uint64_t n = std::min(chunk, (uint64_t)size - off);
+                uint64_t ipcOff = IpcSender::get().ship(
+                  getHost(), (const uint8_t*)devPtr + off, n);
+                IpcChunk c;
+                c.key = key;
+                c.valOffset = off;
+                c.ipcOffset = ipcOff;
+                c.len = n;
+                c.srcHost = thisHost;
+                c.totalSize = size;
+                syncSend((uint8_t)SnapshotCalls::PushSnapshotIpc,
+                         c.encode());
+                off += n;
+            }
+            return;
+        } catch (const std::exception& e) {
+            FAM_WARN("snapshot ipc ship failed (%s); falling back",
+                     e.what());
+        }
+    }
+    // Fallback: stage through the host and ride the RPC plane
+    std::vector<uint8_t> hostCopy(size);
+    if (hipMemcpy(hostCopy.data(), devPtr, size, hipMemcpyDeviceToHost) !=
+        hipSuccess) {
+        throw FaabricException("device snapshot copy-out failed");
+    }
+    pushDeviceSnapshot(key, hostCopy.data(), size);
 }
 
 void SnapshotClient::pushSnapshotUpdate(

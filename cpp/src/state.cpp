@@ -3,6 +3,7 @@
 // StateClient.cpp). See state.h for the re-design notes.
 #include "faabricamd/state.h"
 #include "faabricamd/ops.h"
+#include "faabricamd/ptp.h"
 #include "faabricamd/util.h"
 
 #include <hip/hip_runtime.h>
@@ -119,6 +120,74 @@ bool StateKeyValue::isMaster() const
     return masterHost == getSystemConfig().endpointHost;
 }
 
+bool StateKeyValue::useIpcToMaster()
+{
+    // Bulk device-value traffic to a same-node worker rides HIP IPC;
+    // host values and cross-node masters keep the RPC plane
+    if (!onDevice || isMaster()) {
+        return false;
+    }
+    const std::string& thisHost = getSystemConfig().endpointHost;
+    return isSameNodeDifferentWorker(thisHost, masterHost) &&
+           IpcSender::get().available(masterHost);
+}
+
+void StateKeyValue::pullRangeIpc(uint64_t offset, size_t len)
+{
+    // Master ships straight out of its value into OUR arena; we land it
+    // in HBM with a local D2D and ack the segment. Chunk size adapts to
+    // whatever the master could ship (clamped to its view of our arena).
+    const std::string& thisHost = getSystemConfig().endpointHost;
+    auto cli = getStateClient(masterHost);
+    uint64_t off = offset;
+    uint64_t end = offset + len;
+    while (off < end) {
+        IpcChunk got = cli->pullChunkIpc(user, key, off, end - off,
+                                         thisHost);
+        if (got.len == 0 || off + got.len > end) {
+            throw FaabricException("state ipc pull returned bad chunk");
+        }
+        (void)hipSetDevice(device);
+        IpcReceiver::get().copyToDevice(got.srcHost, got.ipcOffset,
+                                        devPtr + off, got.len);
+        getPointToPointBroker().sendIpcAck(got.srcHost, got.ipcOffset,
+                                           got.len);
+        off += got.len;
+    }
+}
+
+void StateKeyValue::pushRangeIpc(uint64_t offset, size_t len)
+{
+    // Ship from our HBM into the master's arena, then a small RPC tells
+    // it where; the master lands + acks. Chunked to half the arena so
+    // consecutive chunks can overlap ship and land.
+    const std::string& thisHost = getSystemConfig().endpointHost;
+    uint64_t cap = IpcSender::get().peerCapacity(masterHost);
+    if (cap == 0) {
+        throw FaabricException("state ipc push without peer arena");
+    }
+    uint64_t chunk = std::max<uint64_t>(cap / 2, DEVICE_PAGE);
+    auto cli = getStateClient(masterHost);
+    uint64_t off = offset;
+    uint64_t end = offset + len;
+    while (off < end) {
+        uint64_t n = std::min(chunk, end - off);
+        (void)hipSetDevice(device);
+        uint64_t ipcOff =
+          IpcSender::get().ship(masterHost, devPtr + off, n);
+        IpcChunk c;
+        c.user = user;
+        c.key = key;
+        c.valOffset = off;
+        c.ipcOffset = ipcOff;
+        c.len = n;
+        c.srcHost = thisHost;
+        c.totalSize = valueSize;
+        cli->pushChunkIpc(c);
+        off += n;
+    }
+}
+
 void StateKeyValue::get(uint8_t* buffer)
 {
     pull();
@@ -160,6 +229,16 @@ void StateKeyValue::getChunk(uint64_t offset, uint8_t* buffer, size_t len)
         throw FaabricException("state chunk read out of bounds");
     }
     if (!isMaster()) {
+        if (useIpcToMaster()) {
+            try {
+                pullRangeIpc(offset, len);
+                readLocal(offset, buffer, len);
+                return;
+            } catch (const std::exception& e) {
+                FAM_WARN("state ipc chunk pull failed (%s); falling back",
+                         e.what());
+            }
+        }
         // Lazy chunked pull of just this range
         auto cli = getStateClient(masterHost);
         auto data = cli->pullChunk(user, key, offset, len);
@@ -194,6 +273,15 @@ void StateKeyValue::setChunk(uint64_t offset, const uint8_t* buffer,
     }
     flagChunkDirty(offset, len);
     if (!isMaster()) {
+        if (useIpcToMaster()) {
+            try {
+                pushRangeIpc(offset, len);
+                return;
+            } catch (const std::exception& e) {
+                FAM_WARN("state ipc chunk push failed (%s); falling back",
+                         e.what());
+            }
+        }
         auto cli = getStateClient(masterHost);
         cli->pushChunk(user, key, offset, buffer, len);
     }
@@ -208,6 +296,17 @@ void StateKeyValue::pull()
         std::lock_guard<std::mutex> lock(kvMx);
         if (fullyPulled) {
             return;
+        }
+    }
+    if (useIpcToMaster()) {
+        try {
+            pullRangeIpc(0, valueSize);
+            std::lock_guard<std::mutex> lock(kvMx);
+            fullyPulled = true;
+            return;
+        } catch (const std::exception& e) {
+            FAM_WARN("state ipc pull failed (%s); falling back",
+                     e.what());
         }
     }
     auto cli = getStateClient(masterHost);
@@ -227,6 +326,17 @@ void StateKeyValue::pushFull()
 {
     if (isMaster()) {
         return;
+    }
+    if (useIpcToMaster()) {
+        try {
+            pushRangeIpc(0, valueSize);
+            std::lock_guard<std::mutex> lock(kvMx);
+            std::fill(dirtyChunks.begin(), dirtyChunks.end(), 0);
+            return;
+        } catch (const std::exception& e) {
+            FAM_WARN("state ipc push failed (%s); falling back",
+                     e.what());
+        }
     }
     auto cli = getStateClient(masterHost);
     std::lock_guard<std::mutex> lock(kvMx);
@@ -263,6 +373,34 @@ void StateKeyValue::pushPartial()
         std::lock_guard<std::mutex> lock(kvMx);
         std::fill(dirtyChunks.begin(), dirtyChunks.end(), 0);
         return;
+    }
+    if (useIpcToMaster()) {
+        try {
+            // Coalesce contiguous dirty 64 KiB chunks into single ships
+            std::lock_guard<std::mutex> lock(kvMx);
+            size_t i = 0;
+            while (i < dirtyChunks.size()) {
+                if (dirtyChunks[i] == 0) {
+                    i++;
+                    continue;
+                }
+                size_t j = i;
+                while (j < dirtyChunks.size() && dirtyChunks[j] != 0) {
+                    j++;
+                }
+                uint64_t off = i * STATE_STREAM_CHUNK_SIZE;
+                size_t len = std::min((j - i) * STATE_STREAM_CHUNK_SIZE,
+                                      (size_t)(valueSize - off));
+                pushRangeIpc(off, len);
+                std::fill(dirtyChunks.begin() + i,
+                          dirtyChunks.begin() + j, 0);
+                i = j;
+            }
+            return;
+        } catch (const std::exception& e) {
+            FAM_WARN("state ipc partial push failed (%s); falling back",
+                     e.what());
+        }
     }
     auto cli = getStateClient(masterHost);
     std::lock_guard<std::mutex> lock(kvMx);
@@ -333,6 +471,42 @@ void StateKeyValue::serviceSet(uint64_t offset,
         throw FaabricException("state chunk service-set out of bounds");
     }
     writeLocal(offset, data, len);
+}
+
+uint64_t StateKeyValue::serviceChunkIpc(const std::string& dstHost,
+                                        uint64_t offset,
+                                        size_t len)
+{
+    if (offset + len > valueSize) {
+        throw FaabricException("state ipc chunk out of bounds");
+    }
+    if (onDevice) {
+        (void)hipSetDevice(device);
+        return IpcSender::get().ship(dstHost, devPtr + offset, len);
+    }
+    std::lock_guard<std::mutex> lock(kvMx);
+    return IpcSender::get().shipFromHost(dstHost, value.data() + offset,
+                                         len);
+}
+
+void StateKeyValue::serviceSetIpc(const std::string& srcHost,
+                                  uint64_t ipcOffset,
+                                  uint64_t valOffset,
+                                  size_t len)
+{
+    if (valOffset + len > valueSize) {
+        throw FaabricException("state ipc set out of bounds");
+    }
+    if (onDevice) {
+        (void)hipSetDevice(device);
+        IpcReceiver::get().copyToDevice(srcHost, ipcOffset,
+                                        devPtr + valOffset, len);
+    } else {
+        std::lock_guard<std::mutex> lock(kvMx);
+        IpcReceiver::get().copyToHost(srcHost, ipcOffset,
+                                      value.data() + valOffset, len);
+    }
+    getPointToPointBroker().sendIpcAck(srcHost, ipcOffset, len);
 }
 
 void StateKeyValue::serviceAppend(const uint8_t* data, size_t len)
@@ -542,6 +716,35 @@ std::string StateServer::doSyncRecv(uint8_t code, const std::string& body)
             resp.stateSize = st.getStateSize(req.user, req.key);
             return resp.encode();
         }
+        case StateCalls::PullIpc: {
+            // Requester asks us (the master) to ship a range into ITS
+            // arena; reply names the shipped segment. Clamp to the
+            // requester's arena so a big pull streams in pieces.
+            auto req = IpcChunk::decode(body);
+            auto kv = st.getKV(req.user, req.key);
+            uint64_t cap = IpcSender::get().peerCapacity(req.srcHost);
+            if (cap == 0) {
+                throw FaabricException("no ipc arena at requester");
+            }
+            uint64_t len =
+              std::min(req.len, std::max<uint64_t>(cap / 2, DEVICE_PAGE));
+            IpcChunk out;
+            out.user = req.user;
+            out.key = req.key;
+            out.valOffset = req.valOffset;
+            out.ipcOffset =
+              kv->serviceChunkIpc(req.srcHost, req.valOffset, len);
+            out.len = len;
+            out.srcHost = getSystemConfig().endpointHost;
+            return out.encode();
+        }
+        case StateCalls::PushIpc: {
+            auto req = IpcChunk::decode(body);
+            auto kv = st.getKV(req.user, req.key, req.totalSize);
+            kv->serviceSetIpc(req.srcHost, req.ipcOffset, req.valOffset,
+                              req.len);
+            return {};
+        }
         case StateCalls::Append: {
             auto req = StateRequest::decode(body);
             auto kv = st.getKV(req.user, req.key, 1);
@@ -592,6 +795,28 @@ std::vector<uint8_t> StateClient::pullChunk(const std::string& user,
     req.chunkSize = len;
     std::string resp = syncSend((uint8_t)StateCalls::Pull, req.encode());
     return StatePart::decode(resp).data;
+}
+
+IpcChunk StateClient::pullChunkIpc(const std::string& user,
+                                   const std::string& key,
+                                   uint64_t offset,
+                                   size_t len,
+                                   const std::string& requesterHost)
+{
+    IpcChunk req;
+    req.user = user;
+    req.key = key;
+    req.valOffset = offset;
+    req.len = len;
+    req.srcHost = requesterHost; // who the master should ship to
+    std::string resp =
+      syncSend((uint8_t)StateCalls::PullIpc, req.encode());
+    return IpcChunk::decode(resp);
+}
+
+void StateClient::pushChunkIpc(const IpcChunk& chunk)
+{
+    syncSend((uint8_t)StateCalls::PushIpc, chunk.encode());
 }
 
 void StateClient::pushChunk(const std::string& user,

@@ -158,3 +158,210 @@ def test_ipc_ring_recycles_under_pressure():
     """Total shipped bytes (24 x 1 MiB) far exceed a 4 MiB arena: acks
     must recycle segments or the sender would stall and time out."""
     _run_pair(n_msgs=24, msg_elems=(1 << 20) // 4, arena_mb=4)
+
+
+# ---------------------------------------------------------------------------
+# State KV: device values pulled/pushed between same-node workers over IPC
+# ---------------------------------------------------------------------------
+
+KV_BYTES = 8 << 20  # spans multiple IPC chunks at a 4 MiB arena
+
+
+def _state_master_proc(ready, done, result_q):
+    sys.path.insert(0, REPO_ROOT)
+    os.environ["FAABRIC_IPC_ARENA_MB"] = "4"
+    import torch as t  # noqa: F401 (brings up the GPU runtime)
+
+    from faabric_amd import _core
+
+    try:
+        _core.set_log_level("error")
+        _core.set_port_offset(5100)
+        _core.set_endpoint_host(HOST_A)
+        ptp = _core.PointToPointServerHandle()
+        ptp.start()
+        state = _core.StateServerHandle()
+        state.start()
+        _core.state_set_master_host("ipc", "kv", HOST_A)
+        kv = _core.state_get_kv_device("ipc", "kv", KV_BYTES)
+        kv.set(bytes([0xA5]) * KV_BYTES)
+        ready.set()
+        assert done.wait(120), "worker never finished"
+        # Worker pushed a modified middle range; verify it landed in HBM
+        mid = kv.get_chunk(KV_BYTES // 2, 1 << 20)
+        ok = mid == bytes([0x3C]) * (1 << 20)
+        head = kv.get_chunk(0, 64)
+        ok = ok and head == bytes([0xA5]) * 64
+        result_q.put((ok, "master verify"))
+        state.stop()
+        ptp.stop()
+    except Exception as e:
+        result_q.put((False, repr(e)))
+        ready.set()
+
+
+def _state_worker_proc(ready, done, result_q):
+    sys.path.insert(0, REPO_ROOT)
+    os.environ["FAABRIC_IPC_ARENA_MB"] = "4"
+    import torch as t  # noqa: F401
+
+    from faabric_amd import _core
+
+    try:
+        _core.set_log_level("error")
+        _core.set_port_offset(5200)
+        _core.set_endpoint_host(HOST_B)
+        ptp = _core.PointToPointServerHandle()
+        ptp.start()
+        state = _core.StateServerHandle()
+        state.start()
+        assert ready.wait(120), "master never came up"
+        _core.state_set_master_host("ipc", "kv", HOST_A)
+        kv = _core.state_get_kv_device("ipc", "kv", KV_BYTES)
+        kv.pull()  # whole 8 MiB through the 4 MiB arena in chunks
+        got = kv.get_chunk(0, KV_BYTES)
+        ok = got == bytes([0xA5]) * KV_BYTES
+        segs0, _ = _core.ipc_shipped()
+        # Modify the middle 1 MiB and push just that range
+        kv.set_chunk(KV_BYTES // 2, bytes([0x3C]) * (1 << 20))
+        result_q.put((ok, f"pull verify, pushed; local segs={segs0}"))
+        done.set()
+        state.stop()
+        ptp.stop()
+    except Exception as e:
+        result_q.put((False, repr(e)))
+        done.set()
+
+
+@requires_gpu
+def test_state_device_kv_pull_push_over_ipc():
+    """A device KV owned by worker A is pulled by worker B (8 MiB value
+    through a 4 MiB arena) and a dirty range pushed back, all over HIP
+    IPC with RPC only carrying segment descriptors."""
+    ctx = mp.get_context("spawn")
+    ready = ctx.Event()
+    done = ctx.Event()
+    master_q = ctx.Queue()
+    worker_q = ctx.Queue()
+    master = ctx.Process(
+        target=_state_master_proc, args=(ready, done, master_q)
+    )
+    worker = ctx.Process(
+        target=_state_worker_proc, args=(ready, done, worker_q)
+    )
+    master.start()
+    worker.start()
+    try:
+        worker_ok, worker_detail = worker_q.get(timeout=180)
+        master_ok, master_detail = master_q.get(timeout=180)
+    finally:
+        done.set()
+        worker.join(timeout=30)
+        master.join(timeout=30)
+        for p in (worker, master):
+            if p.is_alive():
+                p.terminate()
+    assert worker_ok, f"worker: {worker_detail}"
+    assert master_ok, f"master: {master_detail}"
+
+
+# ---------------------------------------------------------------------------
+# Device snapshots streamed between workers over IPC
+# ---------------------------------------------------------------------------
+
+SNAP_BYTES = 8 << 20
+
+
+def _snap_receiver_proc(ready, done, result_q):
+    sys.path.insert(0, REPO_ROOT)
+    os.environ["FAABRIC_IPC_ARENA_MB"] = "4"
+    import torch as t  # noqa: F401
+
+    from faabric_amd import _core
+
+    try:
+        _core.set_log_level("error")
+        _core.set_port_offset(5200)
+        _core.set_endpoint_host(HOST_B)
+        ptp = _core.PointToPointServerHandle()
+        ptp.start()
+        snap = _core.SnapshotServerHandle()
+        snap.start()
+        ready.set()
+        assert done.wait(120), "sender never finished"
+        ok = _core.device_snapshot_exists("ipcsnap")
+        detail = "snapshot missing"
+        if ok:
+            head = _core.device_snapshot_read("ipcsnap", 0, 256)
+            tail = _core.device_snapshot_read(
+                "ipcsnap", SNAP_BYTES - 256, 256
+            )
+            ok = head == bytes(range(256)) and tail == bytes(
+                reversed(range(256))
+            )
+            detail = "content mismatch" if not ok else ""
+        result_q.put((ok, detail))
+        snap.stop()
+        ptp.stop()
+    except Exception as e:
+        result_q.put((False, repr(e)))
+        ready.set()
+
+
+def _snap_sender_proc(ready, done, result_q):
+    sys.path.insert(0, REPO_ROOT)
+    os.environ["FAABRIC_IPC_ARENA_MB"] = "4"
+    import torch as t
+
+    from faabric_amd import _core
+
+    try:
+        _core.set_log_level("error")
+        _core.set_port_offset(5100)
+        _core.set_endpoint_host(HOST_A)
+        ptp = _core.PointToPointServerHandle()
+        ptp.start()
+        assert ready.wait(120), "receiver never came up"
+
+        buf = t.zeros(SNAP_BYTES, dtype=t.uint8, device="cuda")
+        buf[:256] = t.arange(256, dtype=t.uint8)
+        buf[-256:] = t.arange(255, -1, -1, dtype=t.uint8)
+        t.cuda.synchronize()
+        _core.snapshot_push_device_from_ptr(
+            HOST_B, "ipcsnap", buf.data_ptr(), SNAP_BYTES
+        )
+        segs, nbytes = _core.ipc_shipped()
+        ok = segs >= 2 and nbytes == SNAP_BYTES  # chunked through arena
+        result_q.put((ok, f"segs={segs} bytes={nbytes}"))
+        done.set()
+        ptp.stop()
+    except Exception as e:
+        result_q.put((False, repr(e)))
+        done.set()
+
+
+@requires_gpu
+def test_device_snapshot_streams_over_ipc():
+    """An 8 MiB HBM snapshot ships worker-to-worker through a 4 MiB IPC
+    arena in acked chunks; the receiver's registry holds the bytes."""
+    ctx = mp.get_context("spawn")
+    ready = ctx.Event()
+    done = ctx.Event()
+    recv_q = ctx.Queue()
+    send_q = ctx.Queue()
+    recv = ctx.Process(target=_snap_receiver_proc, args=(ready, done, recv_q))
+    send = ctx.Process(target=_snap_sender_proc, args=(ready, done, send_q))
+    recv.start()
+    send.start()
+    try:
+        send_ok, send_detail = send_q.get(timeout=180)
+        recv_ok, recv_detail = recv_q.get(timeout=180)
+    finally:
+        done.set()
+        send.join(timeout=30)
+        recv.join(timeout=30)
+        for p in (send, recv):
+            if p.is_alive():
+                p.terminate()
+    assert send_ok, f"sender: {send_detail}"
+    assert recv_ok, f"receiver: {recv_detail}"
