@@ -281,6 +281,39 @@ class MpiWorld
     void ensureRcclComm(int rank);
     bool isDeviceBuffer(const void* ptr, MpiBufferLoc loc);
 
+    // --- device fallback plane (PTP/HIP-IPC) ---
+    // RCCL requires one distinct GPU per rank; when ranks share a device
+    // (oversubscribed worlds, 1-GPU multi-worker nodes) the device data
+    // plane falls back to PTP device messages — same-process staged D2D,
+    // cross-process HIP-IPC arenas — with gfx950 elementwise kernels
+    // doing the reductions. FAABRIC_DEVICE_PLANE=rccl|ptp overrides.
+    bool rcclUsable(int rank);
+    bool rcclBroken = false; // guarded by worldMx
+    void devSend(int sendRank,
+                 int recvRank,
+                 const void* devPtr,
+                 size_t bytes,
+                 MpiMessageType type);
+    void devRecv(int sendRank,
+                 int recvRank,
+                 void* devPtr,
+                 size_t bytes,
+                 MpiMessageType type);
+    void* fbStream_ = nullptr; // lazily-created fallback stream
+    void* fallbackStream();
+    void deviceReduceFallback(int thisRank,
+                              int rootRank,
+                              const uint8_t* sendBuffer,
+                              uint8_t* recvBuffer,
+                              MpiDataType dataType,
+                              int count,
+                              MpiOp op);
+    void deviceBroadcastFallback(int rootRank,
+                                 int thisRank,
+                                 uint8_t* buffer,
+                                 size_t bytes,
+                                 MpiMessageType type);
+
     void opReduceHost(MpiOp op,
                       MpiDataType type,
                       int count,

@@ -240,6 +240,10 @@ bool MpiWorld::destroy()
     std::lock_guard<std::mutex> lock(worldMx);
     localRanks.clear();
     rccl.reset();
+    if (fbStream_ != nullptr) {
+        (void)hipStreamDestroy((hipStream_t)fbStream_);
+        fbStream_ = nullptr;
+    }
     return true;
 }
 
@@ -258,6 +262,10 @@ bool MpiWorld::rankFinished(int rank)
         return false;
     }
     rccl.reset();
+    if (fbStream_ != nullptr) {
+        (void)hipStreamDestroy((hipStream_t)fbStream_);
+        fbStream_ = nullptr;
+    }
     return true;
 }
 
@@ -525,6 +533,147 @@ void MpiWorld::ensureRcclComm(int rank)
     rccl->devices[rank] = device;
 }
 
+bool MpiWorld::rcclUsable(int rank)
+{
+    static const int forced = []() {
+        const char* e = getenv("FAABRIC_DEVICE_PLANE");
+        if (e == nullptr) {
+            return 0; // auto
+        }
+        if (strcmp(e, "rccl") == 0) {
+            return 1;
+        }
+        if (strcmp(e, "ptp") == 0) {
+            return -1;
+        }
+        return 0;
+    }();
+    if (forced == -1) {
+        return false;
+    }
+    {
+        std::lock_guard<std::mutex> lock(worldMx);
+        if (rcclBroken) {
+            return false;
+        }
+    }
+    if (forced == 1) {
+        ensureRcclComm(rank); // throw loudly if forced and unavailable
+        return true;
+    }
+    try {
+        ensureRcclComm(rank);
+        return true;
+    } catch (const std::exception& e) {
+        // Typical cause: ranks sharing one GPU — RCCL refuses duplicate
+        // devices. The PTP/IPC device plane takes over for this world.
+        FAM_WARN("world %d: RCCL unavailable for rank %d (%s); using the "
+                 "PTP/IPC device plane",
+                 id,
+                 rank,
+                 e.what());
+        std::lock_guard<std::mutex> lock(worldMx);
+        rcclBroken = true;
+        return false;
+    }
+}
+
+void* MpiWorld::fallbackStream()
+{
+    std::lock_guard<std::mutex> lock(worldMx);
+    if (fbStream_ == nullptr) {
+        hipStream_t s = nullptr;
+        HIP_CHECK(hipStreamCreateWithFlags(&s, hipStreamNonBlocking));
+        fbStream_ = (void*)s;
+    }
+    return fbStream_;
+}
+
+void MpiWorld::devSend(int sendRank,
+                       int recvRank,
+                       const void* devPtr,
+                       size_t bytes,
+                       MpiMessageType type)
+{
+    getPointToPointBroker().sendMessageDevice(appId,
+                                              groupId,
+                                              chanSendIdx(sendRank, type),
+                                              recvRank,
+                                              devPtr,
+                                              bytes,
+                                              /*mustOrderMsgs=*/true);
+}
+
+void MpiWorld::devRecv(int sendRank,
+                       int recvRank,
+                       void* devPtr,
+                       size_t bytes,
+                       MpiMessageType type)
+{
+    size_t got = getPointToPointBroker().recvMessageDevice(
+      groupId,
+      chanSendIdx(sendRank, type),
+      recvRank,
+      devPtr,
+      bytes,
+      /*mustOrderMsgs=*/true,
+      getSystemConfig().globalMessageTimeout);
+    if (got != bytes) {
+        throw FaabricException("device-plane recv size mismatch");
+    }
+}
+
+void MpiWorld::deviceReduceFallback(int thisRank,
+                                    int rootRank,
+                                    const uint8_t* sendBuffer,
+                                    uint8_t* recvBuffer,
+                                    MpiDataType dataType,
+                                    int count,
+                                    MpiOp op)
+{
+    size_t bytes = mpiTypeSize(dataType) * (size_t)count;
+    if (thisRank != rootRank) {
+        devSend(thisRank, rootRank, sendBuffer, bytes,
+                MpiMessageType::REDUCE);
+        return;
+    }
+    hipStream_t s = (hipStream_t)fallbackStream();
+    if (recvBuffer != sendBuffer) {
+        HIP_CHECK(famCopyBuffer(sendBuffer, recvBuffer, bytes, s));
+        HIP_CHECK(hipStreamSynchronize(s));
+    }
+    uint8_t* scratch = nullptr;
+    HIP_CHECK(hipMallocAsync((void**)&scratch, bytes, s));
+    for (int i = 0; i < size; i++) {
+        if (i == rootRank) {
+            continue;
+        }
+        devRecv(i, rootRank, scratch, bytes, MpiMessageType::REDUCE);
+        // Fused on-GPU combine (gfx950 elementwise kernel); syncs
+        deviceElementwiseOp(recvBuffer, scratch, (uint64_t)count,
+                            (int)dataType, (int)op, s);
+    }
+    HIP_CHECK(hipFreeAsync(scratch, s));
+    HIP_CHECK(hipStreamSynchronize(s));
+}
+
+void MpiWorld::deviceBroadcastFallback(int rootRank,
+                                       int thisRank,
+                                       uint8_t* buffer,
+                                       size_t bytes,
+                                       MpiMessageType type)
+{
+    if (thisRank == rootRank) {
+        for (int i = 0; i < size; i++) {
+            if (i != rootRank) {
+                devSend(rootRank, i, buffer, bytes, type);
+            }
+        }
+    } else {
+        devRecv(rootRank, thisRank, buffer, bytes, type);
+    }
+}
+
 void* MpiWorld::getRcclComm(int rank)
 {
     ensureRcclComm(rank);
@@ -552,7 +701,10 @@ void MpiWorld::send(int sendRank,
     size_t bytes = mpiTypeSize(dataType) * (size_t)count;
     recordMsgCount(sendRank, recvRank, messageType);
     if (isDeviceBuffer(buffer, loc)) {
-        ensureRcclComm(sendRank);
+        if (!rcclUsable(sendRank)) {
+            devSend(sendRank, recvRank, buffer, bytes, messageType);
+            return;
+        }
         ncclComm_t comm;
         hipStream_t stream;
         {
@@ -579,7 +731,10 @@ void MpiWorld::recv(int sendRank,
 {
     size_t bytes = mpiTypeSize(dataType) * (size_t)count;
     if (isDeviceBuffer(buffer, loc)) {
-        ensureRcclComm(recvRank);
+        if (!rcclUsable(recvRank)) {
+            devRecv(sendRank, recvRank, buffer, bytes, messageType);
+            return;
+        }
         ncclComm_t comm;
         hipStream_t stream;
         {
@@ -684,7 +839,17 @@ void MpiWorld::sendRecv(const uint8_t* sendBuffer,
                         int thisRank)
 {
     if (isDeviceBuffer(sendBuffer, MpiBufferLoc::AUTO)) {
-        ensureRcclComm(thisRank);
+        if (!rcclUsable(thisRank)) {
+            // Buffered device sends never rendezvous, so send-then-recv
+            // cannot deadlock in a ring
+            devSend(thisRank, sendToRank, sendBuffer,
+                    mpiTypeSize(sendType) * (size_t)sendCount,
+                    MpiMessageType::SENDRECV);
+            devRecv(recvFromRank, thisRank, recvBuffer,
+                    mpiTypeSize(recvType) * (size_t)recvCount,
+                    MpiMessageType::SENDRECV);
+            return;
+        }
         ncclComm_t comm;
         hipStream_t stream;
         {
@@ -814,7 +979,12 @@ void MpiWorld::broadcast(int rootRank,
         return; // broadcast to self is a no-op
     }
     if (isDeviceBuffer(buffer, loc)) {
-        ensureRcclComm(thisRank);
+        if (!rcclUsable(thisRank)) {
+            deviceBroadcastFallback(rootRank, thisRank, buffer,
+                                    mpiTypeSize(dataType) * (size_t)count,
+                                    messageType);
+            return;
+        }
         ncclComm_t comm;
         hipStream_t stream;
         {
@@ -919,7 +1089,30 @@ void MpiWorld::allGather(int thisRank,
         return;
     }
     if (isDeviceBuffer(recvBuffer, loc)) {
-        ensureRcclComm(thisRank);
+        if (!rcclUsable(thisRank)) {
+            // Gather at 0 (each rank's chunk lands straight in 0's
+            // recvBuffer slot), then broadcast the full device buffer
+            size_t bytes = mpiTypeSize(dataType) * (size_t)count;
+            if (thisRank == 0) {
+                hipStream_t s = (hipStream_t)fallbackStream();
+                if (recvBuffer != sendBuffer) {
+                    HIP_CHECK(famCopyBuffer(sendBuffer, recvBuffer,
+                                            bytes, s));
+                    HIP_CHECK(hipStreamSynchronize(s));
+                }
+                for (int i = 1; i < size; i++) {
+                    devRecv(i, 0, recvBuffer + (size_t)i * bytes, bytes,
+                            MpiMessageType::ALLGATHER);
+                }
+            } else {
+                devSend(thisRank, 0, sendBuffer, bytes,
+                        MpiMessageType::ALLGATHER);
+            }
+            deviceBroadcastFallback(0, thisRank, recvBuffer,
+                                    bytes * (size_t)size,
+                                    MpiMessageType::ALLGATHER);
+            return;
+        }
         ncclComm_t comm;
         hipStream_t stream;
         {
@@ -965,7 +1158,11 @@ void MpiWorld::reduce(int thisRank,
         return;
     }
     if (isDeviceBuffer(sendBuffer, loc)) {
-        ensureRcclComm(thisRank);
+        if (!rcclUsable(thisRank)) {
+            deviceReduceFallback(thisRank, rootRank, sendBuffer,
+                                 recvBuffer, dataType, count, op);
+            return;
+        }
         ncclComm_t comm;
         hipStream_t stream;
         {
@@ -1023,7 +1220,15 @@ void MpiWorld::allReduce(int thisRank,
         return;
     }
     if (isDeviceBuffer(sendBuffer, loc)) {
-        ensureRcclComm(thisRank);
+        if (!rcclUsable(thisRank)) {
+            // reduce to 0 then broadcast, all on the device plane
+            deviceReduceFallback(thisRank, 0, sendBuffer, recvBuffer,
+                                 dataType, count, op);
+            deviceBroadcastFallback(0, thisRank, recvBuffer,
+                                    mpiTypeSize(dataType) * (size_t)count,
+                                    MpiMessageType::ALLREDUCE);
+            return;
+        }
         ncclComm_t comm;
         hipStream_t stream;
         {
@@ -1073,7 +1278,31 @@ void MpiWorld::allToAll(int thisRank,
         return;
     }
     if (isDeviceBuffer(sendBuffer, loc)) {
-        ensureRcclComm(thisRank);
+        if (!rcclUsable(thisRank)) {
+            // Direct pairwise exchange; buffered sends first, then recvs
+            hipStream_t s = (hipStream_t)fallbackStream();
+            for (int i = 0; i < size; i++) {
+                const uint8_t* chunk = sendBuffer + (size_t)i * bytes;
+                if (i == thisRank) {
+                    uint8_t* dst = recvBuffer + (size_t)i * bytes;
+                    if (dst != chunk) {
+                        HIP_CHECK(famCopyBuffer(chunk, dst, bytes, s));
+                        HIP_CHECK(hipStreamSynchronize(s));
+                    }
+                } else {
+                    devSend(thisRank, i, chunk, bytes,
+                            MpiMessageType::ALLTOALL);
+                }
+            }
+            for (int i = 0; i < size; i++) {
+                if (i == thisRank) {
+                    continue;
+                }
+                devRecv(i, thisRank, recvBuffer + (size_t)i * bytes,
+                        bytes, MpiMessageType::ALLTOALL);
+            }
+            return;
+        }
         ncclComm_t comm;
         hipStream_t stream;
         {
@@ -1132,10 +1361,34 @@ void MpiWorld::scan(int thisRank,
 {
     size_t bytes = mpiTypeSize(dataType) * (size_t)count;
     if (isDeviceBuffer(sendBuffer, loc)) {
+        if (!rcclUsable(thisRank)) {
+            // Linear chain over the PTP device plane with the same
+            // fused elementwise combine
+            hipStream_t s = (hipStream_t)fallbackStream();
+            if (recvBuffer != sendBuffer) {
+                HIP_CHECK(famCopyBuffer(sendBuffer, recvBuffer, bytes,
+                                        s));
+                HIP_CHECK(hipStreamSynchronize(s));
+            }
+            if (thisRank > 0) {
+                uint8_t* tmp = nullptr;
+                HIP_CHECK(hipMallocAsync((void**)&tmp, bytes, s));
+                devRecv(thisRank - 1, thisRank, tmp, bytes,
+                        MpiMessageType::SCAN);
+                deviceElementwiseOp(recvBuffer, tmp, (uint64_t)count,
+                                    (int)dataType, (int)op, s);
+                HIP_CHECK(hipFreeAsync(tmp, s));
+                HIP_CHECK(hipStreamSynchronize(s));
+            }
+            if (thisRank < size - 1) {
+                devSend(thisRank, thisRank + 1, recvBuffer, bytes,
+                        MpiMessageType::SCAN);
+            }
+            return;
+        }
         // Device chain: RCCL p2p hop + fused elementwise combine on the
         // GPU (the SURVEY worklist's "GPU prefix + p2p chain" mapping for
         // the reference's linear scan, src/mpi/MpiWorld.cpp:1390-1432)
-        ensureRcclComm(thisRank);
         ncclComm_t comm;
         hipStream_t stream;
         {
@@ -1185,7 +1438,34 @@ void MpiWorld::reduceScatter(int thisRank,
                              MpiBufferLoc loc)
 {
     if (isDeviceBuffer(sendBuffer, loc)) {
-        ensureRcclComm(thisRank);
+        if (!rcclUsable(thisRank)) {
+            // Reduce the full buffer at 0, then scatter each rank its
+            // slice over the device plane
+            size_t sliceBytes = mpiTypeSize(dataType) * (size_t)recvCount;
+            size_t fullBytes = sliceBytes * (size_t)size;
+            hipStream_t s = (hipStream_t)fallbackStream();
+            if (thisRank == 0) {
+                uint8_t* full = nullptr;
+                HIP_CHECK(hipMallocAsync((void**)&full, fullBytes, s));
+                HIP_CHECK(hipStreamSynchronize(s));
+                deviceReduceFallback(0, 0, sendBuffer, full, dataType,
+                                     recvCount * size, op);
+                HIP_CHECK(famCopyBuffer(full, recvBuffer, sliceBytes, s));
+                HIP_CHECK(hipStreamSynchronize(s));
+                for (int i = 1; i < size; i++) {
+                    devSend(0, i, full + (size_t)i * sliceBytes,
+                            sliceBytes, MpiMessageType::SCATTER);
+                }
+                HIP_CHECK(hipFreeAsync(full, s));
+                HIP_CHECK(hipStreamSynchronize(s));
+            } else {
+                deviceReduceFallback(thisRank, 0, sendBuffer, nullptr,
+                                     dataType, recvCount * size, op);
+                devRecv(0, thisRank, recvBuffer, sliceBytes,
+                        MpiMessageType::SCATTER);
+            }
+            return;
+        }
         ncclComm_t comm;
         hipStream_t stream;
         {
