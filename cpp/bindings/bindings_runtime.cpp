@@ -523,7 +523,15 @@ void initRuntimeBindings(py::module_& m)
       .def("push_full", [](StateKeyValue& kv) {
           py::gil_scoped_release release;
           kv.pushFull();
+      })
+      .def("sync", [](StateKeyValue& kv) {
+          py::gil_scoped_release release;
+          kv.sync();
       });
+    m.def("state_sync_all", [] {
+        py::gil_scoped_release release;
+        State::get().syncAll();
+    });
 
     m.def("state_get_kv_device",
           [](const std::string& user,

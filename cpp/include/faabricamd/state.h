@@ -80,11 +80,9 @@ class StateKeyValue
     void clearAppended();
 
     // Direct access for zero-copy users: host pointer, or the HBM
-    // pointer when the value is device-resident
-    uint8_t* getDataPtr()
-    {
-        return onDevice ? devPtr : value.data();
-    }
+    // pointer when the value is device-resident (drains pending mirror
+    // writes and invalidates the mirror — the caller may write HBM)
+    uint8_t* getDataPtr();
     bool isOnDevice() const { return onDevice; }
     int getDevice() const { return device; }
 
@@ -115,6 +113,29 @@ class StateKeyValue
     bool onDevice = false;      // HBM mode: value lives in devPtr
     int device = 0;
     uint8_t* devPtr = nullptr;
+
+    // Pinned host mirror for device values (group-commit write-through):
+    // chunk writes memcpy into the mirror and enqueue an async H2D on
+    // the KV stream with NO per-op sync — the ~12 us host-visible
+    // latency of a synchronous 4 KiB HIP copy was the measured floor of
+    // the batch path (BASELINE.md config-5 history). Reads are served
+    // from the mirror (filled D2H on first touch). sync() is the
+    // durability point; direct devPtr uses flush/invalidate around it.
+    std::mutex mirrorMx;
+    uint8_t* mirror = nullptr;
+    std::vector<char> mirrorValid; // per 4 KiB page
+    void* kvStream = nullptr;
+    bool mirrorFailed = false;
+    bool mirrorUsable();
+    void mirrorFill(uint64_t offset, size_t len); // D2H fill, holds mirrorMx
+    void mirrorFlushLocked();                     // drain pending H2D
+    void mirrorInvalidate(uint64_t offset, size_t len);
+
+  public:
+    // Drain pending device writes (group-commit durability point)
+    void sync();
+
+  private:
 
     void readLocal(uint64_t offset, uint8_t* out, size_t len);
     void writeLocal(uint64_t offset, const uint8_t* data, size_t len);
@@ -148,6 +169,8 @@ class State
     void deleteKVLocally(const std::string& user, const std::string& key);
     size_t getKVCount();
     void forceClearAll(bool global);
+    // Drain every KV's pending device writes (group-commit point)
+    void syncAll();
 
     // Owner directory: where a key lives. Defaults to this host on first
     // getKV; setMasterHost lets the deployment pin owner GPUs.

@@ -257,6 +257,11 @@ static int32_t benchRankStep(Message& msg)
                 usleep(200);
             }
         }
+        world.barrier(rank); // every host's batch functions returned
+        // Group-commit: drain this host's pending HBM KV writes inside
+        // the timed region, then re-sync the world so t1 covers every
+        // host's durability point
+        State::get().syncAll();
         world.barrier(rank);
         int64_t t1 = getEpochMicros();
 

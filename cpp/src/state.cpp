@@ -50,9 +50,121 @@ StateKeyValue::StateKeyValue(std::string userIn,
 
 StateKeyValue::~StateKeyValue()
 {
+    if (kvStream != nullptr) {
+        (void)hipStreamSynchronize((hipStream_t)kvStream);
+        (void)hipStreamDestroy((hipStream_t)kvStream);
+    }
+    if (mirror != nullptr) {
+        (void)hipHostFree(mirror);
+    }
     if (devPtr != nullptr) {
         hipFree(devPtr);
     }
+}
+
+// ------------------------- pinned mirror ------------------------------------
+
+static constexpr size_t MIRROR_PAGE = 4096;
+
+static size_t mirrorMaxBytes()
+{
+    static const size_t v =
+      (size_t)getEnvVarInt("FAABRIC_KV_MIRROR_MAX_MB", 64) * 1024 * 1024;
+    return v;
+}
+
+bool StateKeyValue::mirrorUsable()
+{
+    if (!onDevice || mirrorFailed || valueSize > mirrorMaxBytes() ||
+        valueSize == 0) {
+        return false;
+    }
+    if (mirror != nullptr) {
+        return true;
+    }
+    std::lock_guard<std::mutex> lock(mirrorMx);
+    if (mirror != nullptr) {
+        return true;
+    }
+    if (mirrorFailed) {
+        return false;
+    }
+    void* pinned = nullptr;
+    hipStream_t s = nullptr;
+    if (hipSetDevice(device) != hipSuccess ||
+        hipHostMalloc(&pinned, valueSize, hipHostMallocDefault) !=
+          hipSuccess) {
+        mirrorFailed = true;
+        return false;
+    }
+    if (hipStreamCreateWithFlags(&s, hipStreamNonBlocking) != hipSuccess) {
+        (void)hipHostFree(pinned);
+        mirrorFailed = true;
+        return false;
+    }
+    mirrorValid.assign((valueSize + MIRROR_PAGE - 1) / MIRROR_PAGE, 0);
+    kvStream = (void*)s;
+    mirror = (uint8_t*)pinned;
+    return true;
+}
+
+void StateKeyValue::mirrorFill(uint64_t offset, size_t len)
+{
+    // Caller holds mirrorMx. Fill whole pages covering [offset, +len)
+    // whose mirror copy is stale. The D2H rides the KV stream, so it is
+    // ordered after any pending write-through H2Ds.
+    size_t firstPage = offset / MIRROR_PAGE;
+    size_t lastPage = (offset + len - 1) / MIRROR_PAGE;
+    size_t run = 0;
+    for (size_t p = firstPage; p <= lastPage + 1; p++) {
+        bool stale = p <= lastPage && mirrorValid[p] == 0;
+        if (stale) {
+            run++;
+            continue;
+        }
+        if (run > 0) {
+            size_t start = (p - run) * MIRROR_PAGE;
+            size_t n = std::min(run * MIRROR_PAGE, valueSize - start);
+            (void)hipMemcpyAsync(mirror + start, devPtr + start, n,
+                                 hipMemcpyDeviceToHost,
+                                 (hipStream_t)kvStream);
+            for (size_t q = p - run; q < p; q++) {
+                mirrorValid[q] = 1;
+            }
+            run = 0;
+        }
+    }
+    (void)hipStreamSynchronize((hipStream_t)kvStream);
+}
+
+void StateKeyValue::mirrorFlushLocked()
+{
+    if (kvStream != nullptr) {
+        (void)hipStreamSynchronize((hipStream_t)kvStream);
+    }
+}
+
+void StateKeyValue::mirrorInvalidate(uint64_t offset, size_t len)
+{
+    if (mirror == nullptr || len == 0) {
+        return;
+    }
+    std::lock_guard<std::mutex> lock(mirrorMx);
+    size_t firstPage = offset / MIRROR_PAGE;
+    size_t lastPage = (offset + len - 1) / MIRROR_PAGE;
+    for (size_t p = firstPage; p <= lastPage && p < mirrorValid.size();
+         p++) {
+        mirrorValid[p] = 0;
+    }
+}
+
+void StateKeyValue::sync()
+{
+    if (mirror == nullptr) {
+        return;
+    }
+    std::lock_guard<std::mutex> lock(mirrorMx);
+    mirrorFlushLocked();
 }
 
 // Per-thread HIP stream so concurrent chunk ops from different executors
@@ -80,6 +192,23 @@ static bool kvAsyncCopy()
 void StateKeyValue::readLocal(uint64_t offset, uint8_t* out, size_t len)
 {
     if (onDevice) {
+        if (mirrorUsable()) {
+            std::lock_guard<std::mutex> lock(mirrorMx);
+            bool allValid = true;
+            for (size_t p = offset / MIRROR_PAGE;
+                 p <= (offset + len - 1) / MIRROR_PAGE; p++) {
+                if (mirrorValid[p] == 0) {
+                    allValid = false;
+                    break;
+                }
+            }
+            if (!allValid) {
+                (void)hipSetDevice(device);
+                mirrorFill(offset, len);
+            }
+            std::memcpy(out, mirror + offset, len);
+            return;
+        }
         (void)hipSetDevice(device);
         if (kvAsyncCopy()) {
             hipStream_t s = threadCopyStream();
@@ -100,6 +229,39 @@ void StateKeyValue::writeLocal(uint64_t offset,
                                size_t len)
 {
     if (onDevice) {
+        if (mirrorUsable()) {
+            // Group-commit write-through: memcpy into the pinned mirror,
+            // enqueue the H2D with no per-op sync (stream order keeps
+            // HBM newest-wins), mark the pages readable. sync() is the
+            // durability point.
+            std::lock_guard<std::mutex> lock(mirrorMx);
+            (void)hipSetDevice(device);
+            // A partially-written invalid boundary page must be filled
+            // from HBM first, or its untouched bytes would read stale
+            size_t firstPage = offset / MIRROR_PAGE;
+            size_t lastPage = (offset + len - 1) / MIRROR_PAGE;
+            auto coversPage = [&](size_t p) {
+                uint64_t start = p * MIRROR_PAGE;
+                uint64_t end =
+                  std::min<uint64_t>(start + MIRROR_PAGE, valueSize);
+                return offset <= start && offset + len >= end;
+            };
+            if (mirrorValid[firstPage] == 0 && !coversPage(firstPage)) {
+                mirrorFill(firstPage * MIRROR_PAGE, 1);
+            }
+            if (lastPage != firstPage && mirrorValid[lastPage] == 0 &&
+                !coversPage(lastPage)) {
+                mirrorFill(lastPage * MIRROR_PAGE, 1);
+            }
+            std::memcpy(mirror + offset, data, len);
+            for (size_t p = firstPage; p <= lastPage; p++) {
+                mirrorValid[p] = 1;
+            }
+            (void)hipMemcpyAsync(devPtr + offset, mirror + offset, len,
+                                 hipMemcpyHostToDevice,
+                                 (hipStream_t)kvStream);
+            return;
+        }
         (void)hipSetDevice(device);
         if (kvAsyncCopy()) {
             hipStream_t s = threadCopyStream();
@@ -113,6 +275,18 @@ void StateKeyValue::writeLocal(uint64_t offset,
     } else {
         std::memcpy(value.data() + offset, data, len);
     }
+}
+
+uint8_t* StateKeyValue::getDataPtr()
+{
+    if (!onDevice) {
+        return value.data();
+    }
+    if (mirror != nullptr) {
+        sync();
+        mirrorInvalidate(0, valueSize);
+    }
+    return devPtr;
 }
 
 bool StateKeyValue::isMaster() const
@@ -148,8 +322,10 @@ void StateKeyValue::pullRangeIpc(uint64_t offset, size_t len)
             throw FaabricException("state ipc pull returned bad chunk");
         }
         (void)hipSetDevice(device);
+        sync(); // pending mirror writes must not land on top of the pull
         IpcReceiver::get().copyToDevice(got.srcHost, got.ipcOffset,
                                         devPtr + off, got.len);
+        mirrorInvalidate(off, got.len);
         getPointToPointBroker().sendIpcAck(got.srcHost, got.ipcOffset,
                                            got.len);
         off += got.len;
@@ -170,6 +346,7 @@ void StateKeyValue::pushRangeIpc(uint64_t offset, size_t len)
     auto cli = getStateClient(masterHost);
     uint64_t off = offset;
     uint64_t end = offset + len;
+    sync(); // HBM must be current before shipping out of it
     while (off < end) {
         uint64_t n = std::min(chunk, end - off);
         (void)hipSetDevice(device);
@@ -482,6 +659,7 @@ uint64_t StateKeyValue::serviceChunkIpc(const std::string& dstHost,
     }
     if (onDevice) {
         (void)hipSetDevice(device);
+        sync(); // HBM must be current before shipping out of it
         return IpcSender::get().ship(dstHost, devPtr + offset, len);
     }
     std::lock_guard<std::mutex> lock(kvMx);
@@ -499,8 +677,10 @@ void StateKeyValue::serviceSetIpc(const std::string& srcHost,
     }
     if (onDevice) {
         (void)hipSetDevice(device);
+        sync(); // order the landing write after pending mirror writes
         IpcReceiver::get().copyToDevice(srcHost, ipcOffset,
                                         devPtr + valOffset, len);
+        mirrorInvalidate(valOffset, len);
     } else {
         std::lock_guard<std::mutex> lock(kvMx);
         IpcReceiver::get().copyToHost(srcHost, ipcOffset,
@@ -640,6 +820,21 @@ size_t State::getKVCount()
 {
     std::lock_guard<std::mutex> lock(mx);
     return kvMap.size();
+}
+
+void State::syncAll()
+{
+    std::vector<std::shared_ptr<StateKeyValue>> kvs;
+    {
+        std::lock_guard<std::mutex> lock(mx);
+        kvs.reserve(kvMap.size());
+        for (auto& [k, kv] : kvMap) {
+            kvs.push_back(kv);
+        }
+    }
+    for (auto& kv : kvs) {
+        kv->sync();
+    }
 }
 
 void State::forceClearAll(bool global)

@@ -223,6 +223,40 @@ def test_state_kv_in_hbm(runtime):
     assert int(t[0]) == 7 and int(t[5000]) == 7
 
 
+@requires_gpu
+def test_state_kv_mirror_coherence(runtime):
+    """The pinned write-through mirror stays coherent with HBM across
+    direct device-pointer writes (data_ptr access invalidates the
+    mirror) and chunk writes land in HBM after sync()."""
+    n = 64 * 1024
+    kv = _core.state_get_kv_device("gpu", "mirrorkey", n)
+    kv.set(b"\x11" * n)
+    kv.sync()
+
+    # Bypass write: kernel adds 0x22 to every byte straight in HBM
+    add = torch.full((n,), 0x22, dtype=torch.uint8, device="cuda")
+    torch.cuda.synchronize()
+    _core.device_elementwise_op(kv.data_ptr, add.data_ptr(), n, 5, 0)
+    torch.cuda.synchronize()
+    # Chunk reads must observe the kernel's bytes, not a stale mirror
+    assert kv.get_chunk(0, 16) == bytes([0x33]) * 16
+    assert kv.get_chunk(n - 16, 16) == bytes([0x33]) * 16
+
+    # Partial write into an invalidated page keeps its neighbours
+    kv.set_chunk(100, b"\x44" * 8)
+    got = kv.get_chunk(96, 16)
+    assert got == bytes([0x33]) * 4 + bytes([0x44]) * 8 + bytes([0x33]) * 4
+
+    # After sync, HBM holds the merged content (read via a fresh kernel)
+    kv.sync()
+    out = torch.zeros(n, dtype=torch.uint8, device="cuda")
+    torch.cuda.synchronize()
+    _core.device_elementwise_op(out.data_ptr(), kv.data_ptr, n, 5, 0)
+    torch.cuda.synchronize()
+    assert int(out[100]) == 0x44 and int(out[99]) == 0x33
+    assert int(out[50000]) == 0x33
+
+
 def _gpu_thread_body(msg):
     # Flip one device page per thread: thread i XORs page (i) with 0x0F
     idx = msg.group_idx
