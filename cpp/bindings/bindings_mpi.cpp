@@ -425,6 +425,41 @@ void initMpiBindings(py::module_& m)
                            MpiBufferLoc::AUTO);
           });
 
+    m.def("mpi_isend_ptr",
+          [](int sendRank,
+             int recvRank,
+             uintptr_t ptr,
+             int count,
+             MpiDataType dtype) {
+              py::gil_scoped_release release;
+              return world().isend(sendRank,
+                                   recvRank,
+                                   (const uint8_t*)ptr,
+                                   dtype,
+                                   count,
+                                   MpiMessageType::NORMAL,
+                                   MpiBufferLoc::AUTO);
+          });
+    m.def("mpi_irecv_ptr",
+          [](int sendRank,
+             int recvRank,
+             uintptr_t ptr,
+             int count,
+             MpiDataType dtype) {
+              py::gil_scoped_release release;
+              return world().irecv(sendRank,
+                                   recvRank,
+                                   (uint8_t*)ptr,
+                                   dtype,
+                                   count,
+                                   MpiMessageType::NORMAL,
+                                   MpiBufferLoc::AUTO);
+          });
+    m.def("mpi_await", [](int requestId) {
+        py::gil_scoped_release release;
+        world().awaitAsyncRequest(requestId);
+    });
+
     m.def("mpi_msg_count_details", [](int rank) {
         return world().getMsgCountDetails(rank);
     });

@@ -93,6 +93,7 @@ class MpiWorld
     void initialiseRankFromMsg(Message& msg);
 
     std::string getHostForRank(int rank);
+    std::map<std::string, std::vector<int>> ranksPerHost();
     const std::string& getUser() const { return user; }
     const std::string& getFunction() const { return function; }
     int getId() const { return id; }
@@ -140,18 +141,24 @@ class MpiWorld
               int count,
               MpiMessageType messageType = MpiMessageType::NORMAL,
               MpiBufferLoc loc = MpiBufferLoc::AUTO);
+    // Async p2p. Host sends are buffered-eager (complete immediately,
+    // reference: src/mpi/MpiWorld.cpp:544-557); device sends/recvs on
+    // the RCCL plane enqueue on the rank stream with a hipEvent per
+    // request — awaitAsyncRequest waits on the event, isend never syncs
     int isend(int sendRank,
               int recvRank,
               const uint8_t* buffer,
               MpiDataType dataType,
               int count,
-              MpiMessageType messageType = MpiMessageType::NORMAL);
+              MpiMessageType messageType = MpiMessageType::NORMAL,
+              MpiBufferLoc loc = MpiBufferLoc::AUTO);
     int irecv(int sendRank,
               int recvRank,
               uint8_t* buffer,
               MpiDataType dataType,
               int count,
-              MpiMessageType messageType = MpiMessageType::NORMAL);
+              MpiMessageType messageType = MpiMessageType::NORMAL,
+              MpiBufferLoc loc = MpiBufferLoc::AUTO);
     void awaitAsyncRequest(int requestId);
     void sendRecv(const uint8_t* sendBuffer,
                   int sendCount,

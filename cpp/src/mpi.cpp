@@ -105,6 +105,7 @@ struct PendingRecv
     size_t bytes = 0;
     MpiMessageType type = MpiMessageType::NORMAL;
     bool done = false;
+    bool onDevice = false; // drain via the device plane (devRecv)
 };
 
 // Async request bookkeeping per rank-thread
@@ -112,6 +113,9 @@ struct PendingRecv
 struct MpiRankState
 {
     std::vector<PendingRecv> pendingRecvs;
+    // Stream-ordered RCCL requests: requestId → hipEvent_t recorded
+    // after the enqueued ncclSend/Recv (awaited, then destroyed)
+    std::map<int, hipEvent_t> deviceEvents;
     std::atomic<int> nextRequestId{ 1 };
 };
 
@@ -215,6 +219,19 @@ void MpiWorld::initialiseRankFromMsg(Message& msg)
         localRanks.end()) {
         localRanks.push_back(msg.mpiRank);
     }
+}
+
+// host → ranks on that host (rank-ascending). The first rank on each
+// host is its local leader (reference: initLocalRemoteLeaders,
+// src/mpi/MpiWorld.cpp:318-367). Every process derives the same map
+// from the group mappings, so leaders and batch layouts agree.
+std::map<std::string, std::vector<int>> MpiWorld::ranksPerHost()
+{
+    std::map<std::string, std::vector<int>> out;
+    for (int r = 0; r < size; r++) {
+        out[getHostForRank(r)].push_back(r);
+    }
+    return out;
 }
 
 std::string MpiWorld::getHostForRank(int rank)
@@ -755,8 +772,14 @@ void MpiWorld::recv(int sendRank,
     for (auto& p : rankState.pendingRecvs) {
         if (!p.done && p.sendRank == sendRank && p.recvRank == recvRank &&
             p.type == messageType) {
-            auto data = hostRecv(sendRank, recvRank, p.bytes, messageType);
-            std::memcpy(p.buffer, data.data(), data.size());
+            if (p.onDevice) {
+                devRecv(sendRank, recvRank, p.buffer, p.bytes,
+                        messageType);
+            } else {
+                auto data =
+                  hostRecv(sendRank, recvRank, p.bytes, messageType);
+                std::memcpy(p.buffer, data.data(), data.size());
+            }
             p.done = true;
         }
     }
@@ -769,11 +792,41 @@ int MpiWorld::isend(int sendRank,
                     const uint8_t* buffer,
                     MpiDataType dataType,
                     int count,
-                    MpiMessageType messageType)
+                    MpiMessageType messageType,
+                    MpiBufferLoc loc)
 {
+    int id = rankState.nextRequestId.fetch_add(1);
+    if (isDeviceBuffer(buffer, loc)) {
+        if (rcclUsable(sendRank)) {
+            // True async: enqueue on the rank stream, record an event,
+            // return. No sync inside isend (reference semantics:
+            // per-rank unacked buffers, src/mpi/MpiWorld.cpp:45-113)
+            ncclComm_t comm;
+            hipStream_t stream;
+            {
+                std::lock_guard<std::mutex> lock(rccl->mx);
+                comm = rccl->comms.at(sendRank);
+                stream = rccl->streams.at(sendRank);
+                HIP_CHECK(hipSetDevice(rccl->devices.at(sendRank)));
+            }
+            RCCL_CHECK(ncclSend(buffer, count, toNccl(dataType),
+                                recvRank, comm, stream));
+            hipEvent_t ev;
+            HIP_CHECK(
+              hipEventCreateWithFlags(&ev, hipEventDisableTiming));
+            HIP_CHECK(hipEventRecord(ev, stream));
+            rankState.deviceEvents[id] = ev;
+            return id;
+        }
+        // Fallback plane sends are buffered (staged copy + control
+        // message; the sender's buffer is immediately reusable)
+        devSend(sendRank, recvRank, buffer,
+                mpiTypeSize(dataType) * (size_t)count, messageType);
+        return id;
+    }
     // Buffered eager send: completes immediately
     send(sendRank, recvRank, buffer, dataType, count, messageType);
-    return rankState.nextRequestId.fetch_add(1);
+    return id;
 }
 
 int MpiWorld::irecv(int sendRank,
@@ -781,7 +834,8 @@ int MpiWorld::irecv(int sendRank,
                     uint8_t* buffer,
                     MpiDataType dataType,
                     int count,
-                    MpiMessageType messageType)
+                    MpiMessageType messageType,
+                    MpiBufferLoc loc)
 {
     PendingRecv p;
     p.requestId = rankState.nextRequestId.fetch_add(1);
@@ -790,12 +844,41 @@ int MpiWorld::irecv(int sendRank,
     p.buffer = buffer;
     p.bytes = mpiTypeSize(dataType) * (size_t)count;
     p.type = messageType;
+    if (isDeviceBuffer(buffer, loc)) {
+        if (rcclUsable(recvRank)) {
+            ncclComm_t comm;
+            hipStream_t stream;
+            {
+                std::lock_guard<std::mutex> lock(rccl->mx);
+                comm = rccl->comms.at(recvRank);
+                stream = rccl->streams.at(recvRank);
+                HIP_CHECK(hipSetDevice(rccl->devices.at(recvRank)));
+            }
+            RCCL_CHECK(ncclRecv(buffer, count, toNccl(dataType),
+                                sendRank, comm, stream));
+            hipEvent_t ev;
+            HIP_CHECK(
+              hipEventCreateWithFlags(&ev, hipEventDisableTiming));
+            HIP_CHECK(hipEventRecord(ev, stream));
+            rankState.deviceEvents[p.requestId] = ev;
+            return p.requestId;
+        }
+        p.onDevice = true; // drained via devRecv at await time
+    }
     rankState.pendingRecvs.push_back(p);
     return p.requestId;
 }
 
 void MpiWorld::awaitAsyncRequest(int requestId)
 {
+    // Stream-ordered RCCL request: wait on its event
+    auto evIt = rankState.deviceEvents.find(requestId);
+    if (evIt != rankState.deviceEvents.end()) {
+        HIP_CHECK(hipEventSynchronize(evIt->second));
+        (void)hipEventDestroy(evIt->second);
+        rankState.deviceEvents.erase(evIt);
+        return;
+    }
     auto& pending = rankState.pendingRecvs;
     auto target = std::find_if(
       pending.begin(), pending.end(), [&](const PendingRecv& p) {
@@ -812,8 +895,14 @@ void MpiWorld::awaitAsyncRequest(int requestId)
                 p.recvRank != target->recvRank || p.type != target->type) {
                 continue;
             }
-            auto data = hostRecv(p.sendRank, p.recvRank, p.bytes, p.type);
-            std::memcpy(p.buffer, data.data(), data.size());
+            if (p.onDevice) {
+                devRecv(p.sendRank, p.recvRank, p.buffer, p.bytes,
+                        p.type);
+            } else {
+                auto data =
+                  hostRecv(p.sendRank, p.recvRank, p.bytes, p.type);
+                std::memcpy(p.buffer, data.data(), data.size());
+            }
             p.done = true;
             if (p.requestId == requestId) {
                 break;
@@ -1003,15 +1092,37 @@ void MpiWorld::broadcast(int rootRank,
         HIP_CHECK(hipStreamSynchronize(stream));
         return;
     }
+    // Two-level host plane: the root sends once per remote host (to its
+    // local leader) plus to its own local ranks; leaders re-broadcast
+    // locally (reference: src/mpi/MpiWorld.cpp:786-854). Collapses to
+    // the flat fan-out when every rank has its own worker.
     size_t bytes = mpiTypeSize(dataType) * (size_t)count;
+    auto hostsMap = ranksPerHost();
+    const std::string& rootHost = getHostForRank(rootRank);
+    const std::string& myHost = getHostForRank(thisRank);
     if (thisRank == rootRank) {
-        for (int i = 0; i < size; i++) {
-            if (i != rootRank) {
-                hostSend(rootRank, i, buffer, bytes, messageType);
+        for (int r : hostsMap[rootHost]) {
+            if (r != rootRank) {
+                hostSend(rootRank, r, buffer, bytes, messageType);
+            }
+        }
+        for (auto& [h, ranks] : hostsMap) {
+            if (h != rootHost) {
+                hostSend(rootRank, ranks[0], buffer, bytes, messageType);
+            }
+        }
+    } else if (myHost != rootHost && thisRank == hostsMap[myHost][0]) {
+        // Local leader: take the root's copy, fan out on this host
+        auto data = hostRecv(rootRank, thisRank, bytes, messageType);
+        std::memcpy(buffer, data.data(), data.size());
+        for (int r : hostsMap[myHost]) {
+            if (r != thisRank) {
+                hostSend(thisRank, r, buffer, bytes, messageType);
             }
         }
     } else {
-        auto data = hostRecv(rootRank, thisRank, bytes, messageType);
+        int from = myHost == rootHost ? rootRank : hostsMap[myHost][0];
+        auto data = hostRecv(from, thisRank, bytes, messageType);
         std::memcpy(buffer, data.data(), data.size());
     }
 }
@@ -1050,22 +1161,60 @@ void MpiWorld::gather(int thisRank,
                       MpiDataType dataType,
                       int count)
 {
+    // Two-level host plane: leaders gather their host's chunks in rank
+    // order and send one batched buffer to the root, which unpacks by
+    // the shared ranksPerHost layout (reference: :917-1080)
     size_t bytes = mpiTypeSize(dataType) * (size_t)count;
+    auto hostsMap = ranksPerHost();
+    const std::string& rootHost = getHostForRank(rootRank);
+    const std::string& myHost = getHostForRank(thisRank);
     if (thisRank == rootRank) {
-        for (int i = 0; i < size; i++) {
-            uint8_t* dst = recvBuffer + (size_t)i * bytes;
-            if (i == rootRank) {
-                if (dst != sendBuffer) {
-                    std::memcpy(dst, sendBuffer, bytes);
-                }
-            } else {
-                auto data =
-                  hostRecv(i, rootRank, bytes, MpiMessageType::GATHER);
-                std::memcpy(dst, data.data(), data.size());
+        uint8_t* own = recvBuffer + (size_t)rootRank * bytes;
+        if (own != sendBuffer) {
+            std::memcpy(own, sendBuffer, bytes);
+        }
+        for (int r : hostsMap[rootHost]) {
+            if (r == rootRank) {
+                continue;
+            }
+            auto data =
+              hostRecv(r, rootRank, bytes, MpiMessageType::GATHER);
+            std::memcpy(recvBuffer + (size_t)r * bytes, data.data(),
+                        data.size());
+        }
+        for (auto& [h, ranks] : hostsMap) {
+            if (h == rootHost) {
+                continue;
+            }
+            auto batch = hostRecv(ranks[0], rootRank,
+                                  bytes * ranks.size(),
+                                  MpiMessageType::GATHER);
+            for (size_t i = 0; i < ranks.size(); i++) {
+                std::memcpy(recvBuffer + (size_t)ranks[i] * bytes,
+                            batch.data() + i * bytes,
+                            bytes);
             }
         }
-    } else {
+    } else if (myHost == rootHost) {
         hostSend(thisRank, rootRank, sendBuffer, bytes,
+                 MpiMessageType::GATHER);
+    } else if (thisRank == hostsMap[myHost][0]) {
+        auto& ranks = hostsMap[myHost];
+        std::vector<uint8_t> batch(bytes * ranks.size());
+        for (size_t i = 0; i < ranks.size(); i++) {
+            if (ranks[i] == thisRank) {
+                std::memcpy(batch.data() + i * bytes, sendBuffer, bytes);
+            } else {
+                auto data = hostRecv(ranks[i], thisRank, bytes,
+                                     MpiMessageType::GATHER);
+                std::memcpy(batch.data() + i * bytes, data.data(),
+                            data.size());
+            }
+        }
+        hostSend(thisRank, rootRank, batch.data(), batch.size(),
+                 MpiMessageType::GATHER);
+    } else {
+        hostSend(thisRank, hostsMap[myHost][0], sendBuffer, bytes,
                  MpiMessageType::GATHER);
     }
 }
@@ -1182,19 +1331,52 @@ void MpiWorld::reduce(int thisRank,
         HIP_CHECK(hipStreamSynchronize(stream));
         return;
     }
+    // Two-level host plane: remote hosts' ranks reduce at their local
+    // leader, leaders send one partial to the root (reference:
+    // src/mpi/MpiWorld.cpp:1127-1249)
     size_t bytes = mpiTypeSize(dataType) * (size_t)count;
+    auto hostsMap = ranksPerHost();
+    const std::string& rootHost = getHostForRank(rootRank);
+    const std::string& myHost = getHostForRank(thisRank);
     if (thisRank == rootRank) {
-        std::memcpy(recvBuffer, sendBuffer, bytes);
-        for (int i = 0; i < size; i++) {
-            if (i == rootRank) {
+        if (recvBuffer != sendBuffer) {
+            std::memcpy(recvBuffer, sendBuffer, bytes);
+        }
+        for (int r : hostsMap[rootHost]) {
+            if (r == rootRank) {
                 continue;
             }
             auto data =
-              hostRecv(i, rootRank, bytes, MpiMessageType::REDUCE);
+              hostRecv(r, rootRank, bytes, MpiMessageType::REDUCE);
             opReduceHost(op, dataType, count, data.data(), recvBuffer);
         }
-    } else {
+        for (auto& [h, ranks] : hostsMap) {
+            if (h == rootHost) {
+                continue;
+            }
+            auto data = hostRecv(ranks[0], rootRank, bytes,
+                                 MpiMessageType::REDUCE);
+            opReduceHost(op, dataType, count, data.data(), recvBuffer);
+        }
+    } else if (myHost == rootHost) {
         hostSend(thisRank, rootRank, sendBuffer, bytes,
+                 MpiMessageType::REDUCE);
+    } else if (thisRank == hostsMap[myHost][0]) {
+        // Local leader: fold this host's ranks, ship one partial
+        std::vector<uint8_t> partial(sendBuffer, sendBuffer + bytes);
+        for (int r : hostsMap[myHost]) {
+            if (r == thisRank) {
+                continue;
+            }
+            auto data =
+              hostRecv(r, thisRank, bytes, MpiMessageType::REDUCE);
+            opReduceHost(op, dataType, count, data.data(),
+                         partial.data());
+        }
+        hostSend(thisRank, rootRank, partial.data(), bytes,
+                 MpiMessageType::REDUCE);
+    } else {
+        hostSend(thisRank, hostsMap[myHost][0], sendBuffer, bytes,
                  MpiMessageType::REDUCE);
     }
 }
@@ -1531,7 +1713,8 @@ void MpiWorld::prepareMigration(int thisRank)
 {
     // No pending async requests may be in flight
     // (reference: src/mpi/MpiWorld.cpp:2095-2132)
-    if (!rankState.pendingRecvs.empty()) {
+    if (!rankState.pendingRecvs.empty() ||
+        !rankState.deviceEvents.empty()) {
         throw FaabricException(
           "cannot migrate with pending async MPI requests");
     }

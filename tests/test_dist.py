@@ -44,6 +44,54 @@ def _mpi_allreduce_fn(msg):
     return 0
 
 
+def _mpi_leader_collectives_fn(msg):
+    """Two-level host-plane collectives with non-zero roots: with 2
+    ranks per worker the local-leader trees (reference:
+    src/mpi/MpiWorld.cpp:318-367,786-854,1127-1249) take the
+    leader-batched paths, not the flat fan-out."""
+    import struct
+
+    from faabric_amd import _core
+
+    world_id, rank, size = _core.mpi_init()
+    _core.mpi_barrier(rank)
+
+    # Broadcast from root 3 (remote host's second rank)
+    got = _core.mpi_bcast_bytes(3, rank, struct.pack("<i", 777), 4)
+    (v,) = struct.unpack("<i", got)
+    if v != 777:
+        return 1
+
+    # Reduce to root 2 (a local leader)
+    out = _core.mpi_reduce_bytes(
+        rank, 2, struct.pack("<i", rank + 1),
+        _core.MpiDataType.INT32, _core.MpiOp.SUM,
+    )
+    if rank == 2:
+        (total,) = struct.unpack("<i", out)
+        if total != size * (size + 1) // 2:
+            return 2
+
+    # Gather to root 1 (non-leader on its host): leaders batch chunks
+    out = _core.mpi_gather_bytes(rank, 1, struct.pack("<i", rank * 5),
+                                 size)
+    if rank == 1:
+        vals = struct.unpack(f"<{size}i", out)
+        if list(vals) != [r * 5 for r in range(size)]:
+            return 3
+
+    # Allgather rides gather(0) + broadcast
+    out = _core.mpi_allgather_bytes(rank, struct.pack("<i", rank + 40),
+                                    size)
+    vals = struct.unpack(f"<{size}i", out)
+    if list(vals) != [r + 40 for r in range(size)]:
+        return 4
+
+    _core.mpi_barrier(rank)
+    msg.output_data = f"leader collectives ok rank {rank}"
+    return 0
+
+
 def _dist_thread_body(msg):
     import struct
 
@@ -146,6 +194,8 @@ def _worker_main(port_offset, stop_event, ready_event):
     _core.register_native_noop("dist", "noop")
     _core.register_native_sleep("dist", "sleep", 200)
     _core.register_function("dist", "mpi_allreduce", _mpi_allreduce_fn)
+    _core.register_function("dist", "leadercoll",
+                            _mpi_leader_collectives_fn)
     _core.register_function("dist", "threadbody", _dist_thread_body)
     _core.register_function("dist", "forkparent", _dist_fork_parent)
     _core.register_function("dist", "chainparent", _chain_parent_fn)
@@ -265,6 +315,27 @@ def test_mpi_world_across_hosts(cluster):
     assert len(results) == world_size
     for r in results:
         assert r.return_value == 0, r.output_data
+    hosts_used = {r.executed_host for r in results}
+    assert hosts_used == {"127.0.0.1@1000", "127.0.0.1@2000"}
+
+
+def test_mpi_leader_collectives_across_hosts(cluster):
+    """4-rank world over 2 workers (2 ranks each): broadcast/reduce/
+    gather with non-zero roots run through the local-leader trees."""
+    from faabric_amd import _core
+    from faabric_amd.runtime import wait_for_batch
+
+    world_size = 2 * WORKER_SLOTS
+    ber = _core.batch_exec_factory("dist", "leadercoll", 1)
+    msgs = ber.messages
+    msgs[0].is_mpi = True
+    msgs[0].mpi_world_size = world_size
+    ber.messages = msgs
+    _core.call_functions(ber)
+    results = wait_for_batch(ber.app_id, world_size, timeout_ms=60_000)
+    assert len(results) == world_size
+    for r in results:
+        assert r.return_value == 0, (r.mpi_rank, r.output_data)
     hosts_used = {r.executed_host for r in results}
     assert hosts_used == {"127.0.0.1@1000", "127.0.0.1@2000"}
 

@@ -130,6 +130,31 @@ def _device_collectives_fn(msg):
             msg.output_data = "reducescatter mismatch"
             return 1
 
+        # Async isend/irecv overlapping GPU compute: post the request,
+        # run a kernel on other data, then await and verify
+        ab = t.full((N,), float(rank + 200), device=dev)
+        ag = t.zeros(N, device=dev)
+        work = t.rand(1 << 22, device=dev)
+        t.cuda.synchronize()
+        if rank == 0:
+            req = _core.mpi_isend_ptr(0, 1, ab.data_ptr(), N, F)
+            for _ in range(4):
+                work = work * 1.0001 + 0.5  # overlapped compute
+            _core.mpi_await(req)
+            _core.mpi_recv_ptr(1, 0, ag.data_ptr(), N, F)
+        else:
+            req = _core.mpi_irecv_ptr(0, 1, ag.data_ptr(), N, F)
+            for _ in range(4):
+                work = work * 1.0001 + 0.5
+            _core.mpi_await(req)
+            _core.mpi_send_ptr(1, 0, ab.data_ptr(), N, F)
+        t.cuda.synchronize()
+        peer = 1 - rank
+        if not t.allclose(ag, t.full((N,), float(peer + 200),
+                                     device=dev)):
+            msg.output_data = "async isend/irecv mismatch"
+            return 1
+
         _core.mpi_barrier(rank)
         msg.output_data = f"device collectives ok rank {rank}"
         return 0
