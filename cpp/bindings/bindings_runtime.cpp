@@ -135,6 +135,7 @@ void initRuntimeBindings(py::module_& m)
       .def("start",
            &PlannerRuntime::start,
            py::arg("with_snapshot_server") = true,
+           py::arg("with_state_server") = false,
            py::call_guard<py::gil_scoped_release>())
       .def("shutdown",
            &PlannerRuntime::shutdown,
@@ -532,6 +533,24 @@ void initRuntimeBindings(py::module_& m)
         py::gil_scoped_release release;
         State::get().syncAll();
     });
+    m.def("set_state_mode", [](const std::string& mode) {
+        getSystemConfig().stateMode = mode;
+    });
+    m.def("state_acquire_lock",
+          [](const std::string& user, const std::string& key,
+             int expiryMs) {
+              py::gil_scoped_release release;
+              return State::get().acquireLock(user, key, expiryMs);
+          },
+          py::arg("user"),
+          py::arg("key"),
+          py::arg("expiry_ms") = 10000);
+    m.def("state_release_lock",
+          [](const std::string& user, const std::string& key,
+             uint64_t token) {
+              py::gil_scoped_release release;
+              State::get().releaseLock(user, key, token);
+          });
 
     m.def("state_get_kv_device",
           [](const std::string& user,

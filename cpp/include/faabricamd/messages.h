@@ -140,6 +140,9 @@ struct StatePart
     std::string key;     // 2
     uint64_t offset = 0; // 3
     std::vector<uint8_t> data; // 4
+    // Full value size so a push can establish the value at the right
+    // size on a host that has never seen the key (planner store mode)
+    uint64_t totalSize = 0; // 5
     std::string encode() const;
     static StatePart decode(const std::string& buf);
 };

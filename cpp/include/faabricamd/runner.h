@@ -40,12 +40,16 @@ class PlannerRuntime
   public:
     PlannerRuntime();
     ~PlannerRuntime();
-    void start(bool withSnapshotServer = true);
+    // withStateServer hosts the global KV + scripted-lock store for the
+    // "planner" state mode (the reference's Redis-service role)
+    void start(bool withSnapshotServer = true,
+               bool withStateServer = false);
     void shutdown();
 
   private:
     std::unique_ptr<PlannerServer> server;
     std::unique_ptr<SnapshotServer> snapshotServer;
+    std::unique_ptr<StateServer> stateServer;
     bool started = false;
 };
 

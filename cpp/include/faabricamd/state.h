@@ -38,6 +38,10 @@ enum class StateCalls : uint8_t
     // workers (hipipc.h): payload rides the arena, RPC carries segments
     PullIpc = 8,
     PushIpc = 9,
+    // Scripted-lock parity with the reference's Redis backend
+    // (reference: redis/Redis.h:154-168)
+    Lock = 10,
+    Unlock = 11,
 };
 
 class StateKeyValue
@@ -172,6 +176,23 @@ class State
     // Drain every KV's pending device writes (group-commit point)
     void syncAll();
 
+    // Distributed scripted locks (reference: Redis::acquireLock /
+    // releaseLock with delete-if-token-matches, redis/Redis.h:154-168,
+    // src/redis/Redis.cpp). The lock lives on the key's owner host (the
+    // planner in "planner" state mode). acquire returns 0 on contention.
+    uint64_t acquireLock(const std::string& user,
+                         const std::string& key,
+                         int expiryMs);
+    void releaseLock(const std::string& user,
+                     const std::string& key,
+                     uint64_t token);
+    uint64_t acquireLockLocal(const std::string& user,
+                              const std::string& key,
+                              int expiryMs);
+    bool releaseLockLocal(const std::string& user,
+                          const std::string& key,
+                          uint64_t token);
+
     // Owner directory: where a key lives. Defaults to this host on first
     // getKV; setMasterHost lets the deployment pin owner GPUs.
     void setMasterHost(const std::string& user,
@@ -184,6 +205,10 @@ class State
     std::mutex mx;
     std::map<std::string, std::shared_ptr<StateKeyValue>> kvMap;
     std::map<std::string, std::string> masterMap;
+
+    std::mutex locksMx;
+    // key → (token, expiry deadline ms)
+    std::map<std::string, std::pair<uint64_t, int64_t>> locks;
 };
 
 class StateServer : public MessageEndpointServer
@@ -208,7 +233,8 @@ class StateClient : public MessageEndpointClient
                    const std::string& key,
                    uint64_t offset,
                    const uint8_t* data,
-                   size_t len);
+                   size_t len,
+                   size_t totalSize = 0);
     // IPC bulk paths (payload in the arena; see StateCalls::PullIpc)
     IpcChunk pullChunkIpc(const std::string& user,
                           const std::string& key,
@@ -216,6 +242,12 @@ class StateClient : public MessageEndpointClient
                           size_t len,
                           const std::string& requesterHost);
     void pushChunkIpc(const IpcChunk& chunk);
+    uint64_t acquireLock(const std::string& user,
+                         const std::string& key,
+                         int expiryMs);
+    bool releaseLock(const std::string& user,
+                     const std::string& key,
+                     uint64_t token);
     size_t stateSize(const std::string& user, const std::string& key);
     void append(const std::string& user,
                 const std::string& key,

@@ -314,6 +314,9 @@ std::string StatePart::encode() const
     w.putString(2, key);
     w.putUInt64(3, offset);
     w.putBytes(4, data);
+    if (totalSize != 0) {
+        w.putUInt64(5, totalSize);
+    }
     return w.take();
 }
 
@@ -329,6 +332,7 @@ StatePart StatePart::decode(const std::string& buf)
             case 2: m.key = r.asString(); break;
             case 3: m.offset = r.asUInt64(); break;
             case 4: m.data = r.asBytes(); break;
+            case 5: m.totalSize = r.asUInt64(); break;
             default: r.skip(t);
         }
     }

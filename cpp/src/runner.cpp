@@ -98,7 +98,7 @@ PlannerRuntime::~PlannerRuntime()
     shutdown();
 }
 
-void PlannerRuntime::start(bool withSnapshotServer)
+void PlannerRuntime::start(bool withSnapshotServer, bool withStateServer)
 {
     if (started) {
         return;
@@ -108,6 +108,10 @@ void PlannerRuntime::start(bool withSnapshotServer)
     if (withSnapshotServer) {
         snapshotServer = std::make_unique<SnapshotServer>();
         snapshotServer->start();
+    }
+    if (withStateServer) {
+        stateServer = std::make_unique<StateServer>();
+        stateServer->start();
     }
     started = true;
     FAM_INFO("planner runtime started (port offset %d)", getPortOffset());
@@ -125,8 +129,12 @@ void PlannerRuntime::shutdown()
     if (snapshotServer) {
         snapshotServer->stop();
     }
+    if (stateServer) {
+        stateServer->stop();
+    }
     server.reset();
     snapshotServer.reset();
+    stateServer.reset();
 }
 
 } // namespace faabricamd

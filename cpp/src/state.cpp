@@ -460,7 +460,7 @@ void StateKeyValue::setChunk(uint64_t offset, const uint8_t* buffer,
             }
         }
         auto cli = getStateClient(masterHost);
-        cli->pushChunk(user, key, offset, buffer, len);
+        cli->pushChunk(user, key, offset, buffer, len, valueSize);
     }
 }
 
@@ -523,7 +523,8 @@ void StateKeyValue::pushFull()
         size_t len = std::min(STATE_STREAM_CHUNK_SIZE,
                               (size_t)(valueSize - off));
         readLocal(off, staging.data(), len);
-        cli->pushChunk(user, key, off, staging.data(), len);
+        cli->pushChunk(user, key, off, staging.data(), len,
+                       valueSize);
     }
     std::fill(dirtyChunks.begin(), dirtyChunks.end(), 0);
 }
@@ -590,7 +591,8 @@ void StateKeyValue::pushPartial()
                               (size_t)(valueSize - off));
         std::vector<uint8_t> staging(len);
         readLocal(off, staging.data(), len);
-        cli->pushChunk(user, key, off, staging.data(), len);
+        cli->pushChunk(user, key, off, staging.data(), len,
+                       valueSize);
         dirtyChunks[i] = 0;
     }
 }
@@ -718,6 +720,20 @@ State& State::get()
     return st;
 }
 
+// Default owner for a key with no pinned master. "inmemory" mode is the
+// reference's master-per-key backend (first-toucher owns); "planner"
+// mode is the second backend — the planner process hosts a global
+// StateServer and owns every key, playing the role of the reference's
+// Redis service (src/state/RedisStateKeyValue.cpp, redis/Redis.h).
+static std::string defaultMasterHost()
+{
+    auto& conf = getSystemConfig();
+    if (conf.stateMode == "planner") {
+        return conf.plannerHost;
+    }
+    return conf.endpointHost;
+}
+
 std::shared_ptr<StateKeyValue> State::getKV(const std::string& user,
                                             const std::string& key,
                                             size_t size)
@@ -733,7 +749,7 @@ std::shared_ptr<StateKeyValue> State::getKV(const std::string& user,
     if (mIt != masterMap.end()) {
         master = mIt->second;
     } else {
-        master = getSystemConfig().endpointHost;
+        master = defaultMasterHost();
         masterMap[k] = master;
     }
     auto kv = std::make_shared<StateKeyValue>(user, key, size, master);
@@ -757,7 +773,7 @@ std::shared_ptr<StateKeyValue> State::getKVDevice(const std::string& user,
     if (mIt != masterMap.end()) {
         master = mIt->second;
     } else {
-        master = getSystemConfig().endpointHost;
+        master = defaultMasterHost();
         masterMap[k] = master;
     }
     auto kv = std::make_shared<StateKeyValue>(
@@ -845,6 +861,70 @@ void State::forceClearAll(bool global)
     masterMap.clear();
 }
 
+uint64_t State::acquireLockLocal(const std::string& user,
+                                 const std::string& key,
+                                 int expiryMs)
+{
+    std::lock_guard<std::mutex> lock(locksMx);
+    std::string k = kvKeyOf(user, key);
+    int64_t now = getEpochMicros() / 1000;
+    auto it = locks.find(k);
+    if (it != locks.end() && it->second.second > now) {
+        return 0; // held and not expired
+    }
+    uint64_t token = ((uint64_t)generateGid() << 32) | generateGid();
+    if (token == 0) {
+        token = 1;
+    }
+    locks[k] = { token, now + expiryMs };
+    return token;
+}
+
+bool State::releaseLockLocal(const std::string& user,
+                             const std::string& key,
+                             uint64_t token)
+{
+    // Delete-if-token-matches, like the reference's scripted Redis
+    // release (a stale holder cannot free a re-acquired lock)
+    std::lock_guard<std::mutex> lock(locksMx);
+    std::string k = kvKeyOf(user, key);
+    auto it = locks.find(k);
+    if (it == locks.end() || it->second.first != token) {
+        return false;
+    }
+    locks.erase(it);
+    return true;
+}
+
+uint64_t State::acquireLock(const std::string& user,
+                            const std::string& key,
+                            int expiryMs)
+{
+    std::string owner = getMasterHost(user, key);
+    if (owner.empty()) {
+        owner = defaultMasterHost();
+    }
+    if (owner == getSystemConfig().endpointHost) {
+        return acquireLockLocal(user, key, expiryMs);
+    }
+    return getStateClient(owner)->acquireLock(user, key, expiryMs);
+}
+
+void State::releaseLock(const std::string& user,
+                        const std::string& key,
+                        uint64_t token)
+{
+    std::string owner = getMasterHost(user, key);
+    if (owner.empty()) {
+        owner = defaultMasterHost();
+    }
+    if (owner == getSystemConfig().endpointHost) {
+        releaseLockLocal(user, key, token);
+        return;
+    }
+    getStateClient(owner)->releaseLock(user, key, token);
+}
+
 void State::setMasterHost(const std::string& user,
                           const std::string& key,
                           const std::string& host)
@@ -893,10 +973,13 @@ std::string StateServer::doSyncRecv(uint8_t code, const std::string& body)
         }
         case StateCalls::Push: {
             auto part = StatePart::decode(body);
-            // Create on demand so a push can establish the value
-            auto kv = st.getKV(part.user,
-                               part.key,
-                               part.offset + part.data.size());
+            // Create on demand so a push can establish the value (sized
+            // by the pusher's full value, not the first chunk)
+            auto kv = st.getKV(
+              part.user,
+              part.key,
+              std::max<size_t>(part.totalSize,
+                               part.offset + part.data.size()));
             if (part.offset + part.data.size() > kv->size()) {
                 throw FaabricException("push beyond registered size");
             }
@@ -939,6 +1022,27 @@ std::string StateServer::doSyncRecv(uint8_t code, const std::string& body)
             kv->serviceSetIpc(req.srcHost, req.ipcOffset, req.valOffset,
                               req.len);
             return {};
+        }
+        case StateCalls::Lock: {
+            auto req = StateRequest::decode(body);
+            int32_t expiryMs = 10000;
+            if (req.data.size() >= sizeof(expiryMs)) {
+                std::memcpy(&expiryMs, req.data.data(), sizeof(expiryMs));
+            }
+            uint64_t token =
+              st.acquireLockLocal(req.user, req.key, expiryMs);
+            std::string out(sizeof(token), '\0');
+            std::memcpy(out.data(), &token, sizeof(token));
+            return out;
+        }
+        case StateCalls::Unlock: {
+            auto req = StateRequest::decode(body);
+            uint64_t token = 0;
+            if (req.data.size() >= sizeof(token)) {
+                std::memcpy(&token, req.data.data(), sizeof(token));
+            }
+            bool ok = st.releaseLockLocal(req.user, req.key, token);
+            return std::string(1, ok ? 1 : 0);
         }
         case StateCalls::Append: {
             auto req = StateRequest::decode(body);
@@ -1018,13 +1122,15 @@ void StateClient::pushChunk(const std::string& user,
                             const std::string& key,
                             uint64_t offset,
                             const uint8_t* data,
-                            size_t len)
+                            size_t len,
+                            size_t totalSize)
 {
     StatePart part;
     part.user = user;
     part.key = key;
     part.offset = offset;
     part.data.assign(data, data + len);
+    part.totalSize = totalSize;
     syncSend((uint8_t)StateCalls::Push, part.encode());
 }
 
@@ -1035,6 +1141,38 @@ size_t StateClient::stateSize(const std::string& user, const std::string& key)
     req.key = key;
     std::string resp = syncSend((uint8_t)StateCalls::Size, req.encode());
     return StateSizeResponse::decode(resp).stateSize;
+}
+
+uint64_t StateClient::acquireLock(const std::string& user,
+                                  const std::string& key,
+                                  int expiryMs)
+{
+    StateRequest req;
+    req.user = user;
+    req.key = key;
+    req.data.resize(sizeof(int32_t));
+    int32_t e = expiryMs;
+    std::memcpy(req.data.data(), &e, sizeof(e));
+    std::string resp = syncSend((uint8_t)StateCalls::Lock, req.encode());
+    uint64_t token = 0;
+    if (resp.size() >= sizeof(token)) {
+        std::memcpy(&token, resp.data(), sizeof(token));
+    }
+    return token;
+}
+
+bool StateClient::releaseLock(const std::string& user,
+                              const std::string& key,
+                              uint64_t token)
+{
+    StateRequest req;
+    req.user = user;
+    req.key = key;
+    req.data.resize(sizeof(token));
+    std::memcpy(req.data.data(), &token, sizeof(token));
+    std::string resp =
+      syncSend((uint8_t)StateCalls::Unlock, req.encode());
+    return !resp.empty() && resp[0] == 1;
 }
 
 void StateClient::append(const std::string& user,

@@ -51,10 +51,16 @@ class LocalRuntime:
         if self.slots is not None:
             _core.set_this_host_resources(self.slots, 0)
 
-    def start_planner(self, with_snapshot_server: bool = True):
+    def start_planner(
+        self,
+        with_snapshot_server: bool = True,
+        with_state_server: bool = False,
+    ):
+        """with_state_server hosts the global KV + scripted-lock store
+        for the "planner" state mode (the reference's Redis role)."""
         self._configure()
         self._planner = _core.PlannerRuntime()
-        self._planner.start(with_snapshot_server)
+        self._planner.start(with_snapshot_server, with_state_server)
         return self
 
     def start_worker(self):
