@@ -17,7 +17,7 @@ TARGET := faabric_amd/_core$(EXT_SUFFIX)
 CXXFLAGS := -O2 -g -std=c++20 -fPIC -Wall -Wno-unused-function \
             -Icpp/include -I$(PY_INC) -I$(PYBIND_INC) \
             --offload-arch=$(GPU_ARCH) -fvisibility=hidden -MMD -MP
-LDFLAGS := -shared -L/opt/rocm/lib -lrccl -lamdhip64 -lz -pthread
+LDFLAGS := -shared -L/opt/rocm/lib -lrccl -lamdhip64 -lz -l:libzstd.so.1 -pthread
 
 CPP_SRCS := $(wildcard cpp/src/*.cpp) $(wildcard cpp/bindings/*.cpp)
 HIP_SRCS := $(wildcard cpp/hip/*.hip)
@@ -56,7 +56,7 @@ asan-check:
 	$(HIPCC) -O1 -g -std=c++20 -fsanitize=address -fno-omit-frame-pointer \
 	    -Icpp/include --offload-arch=$(GPU_ARCH) \
 	    $(CPP_SRCS_CORE) cpp/hip/snapshot_kernels.hip examples/selftest.cpp \
-	    -L/opt/rocm/lib -lrccl -lamdhip64 -lz -pthread \
+	    -L/opt/rocm/lib -lrccl -lamdhip64 -lz -l:libzstd.so.1 -pthread \
 	    -o $(BUILD)/asan/selftest
 	FAABRIC_PORT_OFFSET=6600 $(BUILD)/asan/selftest
 
@@ -65,7 +65,7 @@ tsan-check:
 	$(HIPCC) -O1 -g -std=c++20 -fsanitize=thread \
 	    -Icpp/include --offload-arch=$(GPU_ARCH) \
 	    $(CPP_SRCS_CORE) cpp/hip/snapshot_kernels.hip examples/selftest.cpp \
-	    -L/opt/rocm/lib -lrccl -lamdhip64 -lz -pthread \
+	    -L/opt/rocm/lib -lrccl -lamdhip64 -lz -l:libzstd.so.1 -pthread \
 	    -o $(BUILD)/tsan/selftest
 	FAABRIC_PORT_OFFSET=6700 TSAN_OPTIONS="report_bugs=1" \
 	    $(BUILD)/tsan/selftest
@@ -74,7 +74,7 @@ tsan-check:
 
 $(BUILD)/%: $(BUILD)/examples/%.o $(CORE_OBJS)
 	$(HIPCC) $< $(CORE_OBJS) \
-	    -L/opt/rocm/lib -lrccl -lamdhip64 -lz -pthread -o $@
+	    -L/opt/rocm/lib -lrccl -lamdhip64 -lz -l:libzstd.so.1 -pthread -o $@
 
 $(BUILD)/examples/%.o: examples/%.cpp
 	@mkdir -p $(dir $@)

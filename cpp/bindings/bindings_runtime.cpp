@@ -694,6 +694,7 @@ void initRuntimeBindings(py::module_& m)
     m.def("clear_mocked_function_calls", &clearMockedFunctionCalls);
 
     // ---------------- util extras ------------------------------------------
+    m.def("delta_default_config", [] { return DeltaConfig{}.str(); });
     m.def("delta_encode",
           [](const py::bytes& oldData,
              const py::bytes& newData,

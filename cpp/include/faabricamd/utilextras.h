@@ -10,10 +10,9 @@
 namespace faabricamd {
 
 // ------------------------- delta codec --------------------------------------
-// Page-wise XOR-against-old + zlib compression. The reference's config
-// string is "pages=4096;xor;zstd=1" (src/util/config.cpp:27); this image
-// has no zstd dev package, so the codec compresses with zlib and the
-// config spelling is "pages=4096;xor;zlib=1". Stream format is
+// Page-wise XOR-against-old + zstd compression, config string
+// "pages=4096;xor;zstd=1" like the reference (src/util/config.cpp:27,
+// delta.cpp:15-57); "zlib=1" selects a zlib fallback. Stream format is
 // command-tagged like the reference's (delta.cpp:155-169).
 
 struct DeltaConfig
@@ -21,6 +20,9 @@ struct DeltaConfig
     size_t pageSize = 4096;
     bool xorWithOld = true;
     bool compress = true;
+    // zstd by default, matching the reference's "pages=4096;xor;zstd=1"
+    // (src/util/config.cpp:27); "zlib=1" selects the zlib fallback
+    bool useZstd = true;
 
     static DeltaConfig parse(const std::string& s);
     std::string str() const;

@@ -24,9 +24,28 @@ enum DeltaCmd : uint8_t
     DELTA_TOTAL_SIZE = 1,
     DELTA_XOR_PAGE = 2,
     DELTA_RAW_RANGE = 3,
-    DELTA_COMPRESSED = 4,
+    DELTA_COMPRESSED = 4,      // zlib stream
+    DELTA_COMPRESSED_ZSTD = 5, // zstd stream (reference's codec)
     DELTA_END = 0xff,
 };
+
+// The image ships libzstd.so.1 without headers; the C ABI below is
+// stable across zstd 1.x so we declare exactly what the codec needs
+// and link -l:libzstd.so.1 (matches the reference's zstd dependency,
+// /root/reference/src/util/delta.cpp:15-57)
+extern "C" {
+size_t ZSTD_compressBound(size_t srcSize);
+size_t ZSTD_compress(void* dst,
+                     size_t dstCapacity,
+                     const void* src,
+                     size_t srcSize,
+                     int level);
+size_t ZSTD_decompress(void* dst,
+                       size_t dstCapacity,
+                       const void* src,
+                       size_t srcSize);
+unsigned ZSTD_isError(size_t code);
+}
 static constexpr uint8_t DELTA_VERSION = 1;
 
 DeltaConfig DeltaConfig::parse(const std::string& s)
@@ -35,6 +54,7 @@ DeltaConfig DeltaConfig::parse(const std::string& s)
     conf.xorWithOld = s.find("xor") != std::string::npos;
     conf.compress = s.find("zlib=1") != std::string::npos ||
                     s.find("zstd=1") != std::string::npos;
+    conf.useZstd = s.find("zlib=1") == std::string::npos;
     auto pos = s.find("pages=");
     if (pos != std::string::npos) {
         conf.pageSize = (size_t)atoll(s.c_str() + pos + 6);
@@ -49,7 +69,7 @@ std::string DeltaConfig::str() const
         out += ";xor";
     }
     if (compress) {
-        out += ";zlib=1";
+        out += useZstd ? ";zstd=1" : ";zlib=1";
     }
     return out;
 }
@@ -141,7 +161,20 @@ std::vector<uint8_t> deltaEncode(const std::vector<uint8_t>& oldData,
 
     std::vector<uint8_t> out;
     out.push_back(DELTA_VERSION);
-    if (conf.compress) {
+    if (conf.compress && conf.useZstd) {
+        size_t bound = ZSTD_compressBound(cmds.size());
+        std::vector<uint8_t> compressed(bound);
+        size_t n = ZSTD_compress(compressed.data(), bound, cmds.data(),
+                                 cmds.size(), /*level=*/1);
+        if (ZSTD_isError(n)) {
+            throw FaabricException("zstd compress failed");
+        }
+        compressed.resize(n);
+        out.push_back(DELTA_COMPRESSED_ZSTD);
+        putU64(out, cmds.size());
+        putU64(out, compressed.size());
+        out.insert(out.end(), compressed.begin(), compressed.end());
+    } else if (conf.compress) {
         auto compressed = zlibCompress(cmds.data(), cmds.size());
         out.push_back(DELTA_COMPRESSED);
         putU64(out, cmds.size());
@@ -161,7 +194,16 @@ std::vector<uint8_t> deltaApply(const std::vector<uint8_t>& oldData,
     }
     std::vector<uint8_t> cmds;
     size_t pos = 1;
-    if (pos < delta.size() && delta[pos] == DELTA_COMPRESSED) {
+    if (pos < delta.size() && delta[pos] == DELTA_COMPRESSED_ZSTD) {
+        uint64_t rawLen = getU64(delta.data() + pos + 1);
+        uint64_t compLen = getU64(delta.data() + pos + 9);
+        cmds.resize(rawLen);
+        size_t n = ZSTD_decompress(cmds.data(), rawLen,
+                                   delta.data() + pos + 17, compLen);
+        if (ZSTD_isError(n) || n != rawLen) {
+            throw FaabricException("zstd decompress failed");
+        }
+    } else if (pos < delta.size() && delta[pos] == DELTA_COMPRESSED) {
         uint64_t rawLen = getU64(delta.data() + pos + 1);
         uint64_t compLen = getU64(delta.data() + pos + 9);
         cmds = zlibDecompress(delta.data() + pos + 17, compLen, rawLen);

@@ -25,6 +25,24 @@ def test_delta_roundtrip_xor_zlib():
     assert restored == new
 
 
+def test_delta_roundtrip_xor_zstd():
+    """Default codec is the reference's pages=4096;xor;zstd=1
+    (reference: src/util/delta.cpp:15-57, config.cpp:27)."""
+    random.seed(7)
+    old = bytes(random.getrandbits(8) for _ in range(64 * 1024))
+    new = bytearray(old)
+    for page in (1, 7, 13):
+        new[page * 4096 + 50] ^= 0xAA
+    new = bytes(new)
+
+    delta = _core.delta_encode(old, new, "pages=4096;xor;zstd=1")
+    assert len(delta) < len(new) // 4
+    assert _core.delta_apply(old, delta) == new
+
+    # Default config string spells zstd like the reference
+    assert "zstd=1" in _core.delta_default_config()
+
+
 def test_delta_no_compress_no_xor():
     old = b"a" * 8192
     new = b"a" * 4096 + b"b" * 4096
