@@ -181,8 +181,12 @@ def main():
         # Per-rank snapshot diff/apply GB/s (config 4; 0 on CPU)
         snap_diff = [parse_times(r.output_data, "snapdiff") for r in results]
         snap_apply = [parse_times(r.output_data, "snapapply") for r in results]
+        snap_region = [parse_times(r.output_data, "snapdiffregion")
+                       for r in results]
+        snap_pct = parse_times(results[0].output_data, "snapdirtypct")
         snap_diff_total = sum(v[0] for v in snap_diff if v)
         snap_apply_total = sum(v[0] for v in snap_apply if v)
+        snap_region_total = sum(v[0] for v in snap_region if v)
         k = min(len(s) for s in per_rank_step)
         step_ms = [max(s[i] for s in per_rank_step) for i in range(k)]
         ar_ms = [max(s[i] for s in per_rank_ar) for i in range(k)]
@@ -245,6 +249,11 @@ def main():
                 "snapshot_bytes_per_rank": args.snap_bytes if have_gpu else 0,
                 "snapshot_diff_gbps_total": round(snap_diff_total, 2),
                 "snapshot_apply_gbps_total": round(snap_apply_total, 2),
+                "snapshot_diff_region_gbps_total":
+                    round(snap_region_total, 2),
+                "snapshot_dirty": "random-scattered "
+                    + (str(int(snap_pct[0])) if snap_pct else "25")
+                    + "pct, random-byte region",
                 "gpu": have_gpu,
             },
         }

@@ -44,38 +44,32 @@ SnapBenchResult benchSnapshotPipelineImpl(size_t bytes,
     size_t nPages = bytes / DEVICE_PAGE;
     DeviceSnapshot snap(bytes, device);
 
-    // Updated view U: pseudo-random content tiled from a 16 MiB host seed
+    // Updated view U: whole-region pseudo-random bytes (device-side
+    // fill; every 8-byte word unique — no tiling the cache could love)
     uint8_t* updated = nullptr;
     if (hipMalloc(&updated, bytes) != hipSuccess) {
         throw FaabricException("hipMalloc updated failed");
     }
-    {
-        size_t seedBytes = std::min<size_t>(bytes, 16ull * 1024 * 1024);
-        std::vector<uint8_t> seed(seedBytes);
-        std::mt19937_64 gen(42);
-        for (size_t i = 0; i + 8 <= seedBytes; i += 8) {
-            uint64_t v = gen();
-            std::memcpy(seed.data() + i, &v, 8);
-        }
-        for (size_t off = 0; off < bytes; off += seedBytes) {
-            size_t n = std::min(seedBytes, bytes - off);
-            hipMemcpy(updated + off, seed.data(), n,
-                      hipMemcpyHostToDevice);
-        }
-    }
+    famFillRandom(updated, bytes, 42, nullptr);
+    hipDeviceSynchronize();
     snap.captureFromDevice(updated);
 
-    // Dirty a strided subset of pages in U by XOR-flipping them
+    // Dirty a RANDOMLY SCATTERED subset of pages in U by XOR-flipping
+    // them (BASELINE config 4 asks for randomized dirty distribution,
+    // not a stride or contiguous run)
     uint32_t nDirtyTarget = (uint32_t)((double)nPages * dirtyPct / 100.0);
     if (nDirtyTarget == 0 && dirtyPct > 0) {
         nDirtyTarget = 1;
     }
     if (nDirtyTarget > 0) {
-        std::vector<uint32_t> pages(nDirtyTarget);
-        double stride = (double)nPages / nDirtyTarget;
-        for (uint32_t i = 0; i < nDirtyTarget; i++) {
-            pages[i] = (uint32_t)(i * stride);
+        std::vector<uint32_t> all(nPages);
+        for (uint32_t i = 0; i < nPages; i++) {
+            all[i] = i;
         }
+        std::mt19937_64 gen(1234);
+        std::shuffle(all.begin(), all.end(), gen);
+        std::vector<uint32_t> pages(all.begin(),
+                                    all.begin() + nDirtyTarget);
         uint32_t* pagesDev = nullptr;
         uint8_t* flipDev = nullptr;
         hipMalloc(&pagesDev, nDirtyTarget * sizeof(uint32_t));
@@ -313,4 +307,43 @@ void initOpsBindings(py::module_& m)
           py::arg("warmup") = 2,
           py::arg("dirty_pct") = 25.0,
           py::arg("device") = 0);
+    m.def("fam_fill_random",
+          [](uintptr_t ptr, uint64_t bytes, uint64_t seed) {
+              py::gil_scoped_release release;
+              hipError_t rc =
+                famFillRandom((void*)ptr, bytes, seed, nullptr);
+              if (rc == hipSuccess) {
+                  rc = hipDeviceSynchronize();
+              }
+              if (rc != hipSuccess) {
+                  throw FaabricException("famFillRandom failed");
+              }
+          });
+    m.def("fam_touch_pages",
+          [](uintptr_t ptr, const std::vector<uint32_t>& pages,
+             uint64_t seed) {
+              py::gil_scoped_release release;
+              uint32_t* pagesDev = nullptr;
+              if (hipMalloc(&pagesDev,
+                            pages.size() * sizeof(uint32_t)) !=
+                  hipSuccess) {
+                  throw FaabricException("touch pages alloc failed");
+              }
+              hipError_t rc =
+                hipMemcpy(pagesDev, pages.data(),
+                          pages.size() * sizeof(uint32_t),
+                          hipMemcpyHostToDevice);
+              if (rc == hipSuccess) {
+                  rc = famTouchPages((void*)ptr, pagesDev,
+                                     (uint32_t)pages.size(), seed,
+                                     nullptr);
+              }
+              if (rc == hipSuccess) {
+                  rc = hipDeviceSynchronize();
+              }
+              (void)hipFree(pagesDev);
+              if (rc != hipSuccess) {
+                  throw FaabricException("famTouchPages failed");
+              }
+          });
 }

@@ -371,6 +371,50 @@ __global__ __launch_bounds__(FAM_KERNEL_BLOCK) void xorBytesKernel(
     }
 }
 
+__device__ inline u64 xorshift64(u64 x)
+{
+    x ^= x << 13;
+    x ^= x >> 7;
+    x ^= x << 17;
+    return x;
+}
+
+// Fill a buffer with deterministic pseudo-random bytes (bench init for
+// the "random-byte region" snapshot shapes — BASELINE.json config 4)
+__global__ __launch_bounds__(FAM_KERNEL_BLOCK) void fillRandomKernel(
+  u64* __restrict__ out,
+  u64 nWords,
+  u64 seed)
+{
+    for (u64 i = (u64)blockIdx.x * blockDim.x + threadIdx.x; i < nWords;
+         i += (u64)gridDim.x * blockDim.x) {
+        out[i] = xorshift64(seed ^ (i * 0x9e3779b97f4a7c15ULL) ^
+                            (i >> 3));
+    }
+}
+
+// Mutate a scattered set of pages in place (wave per page, every lane
+// perturbs its 64-byte slice) so the diff kernels face a randomized
+// dirty distribution, not a contiguous memset
+__global__ __launch_bounds__(FAM_KERNEL_BLOCK) void touchPagesKernel(
+  u64* __restrict__ buf,
+  const u32* __restrict__ pages,
+  u32 nPages,
+  u64 seed)
+{
+    const u32 wavesPerBlock = FAM_KERNEL_BLOCK / 64;
+    const u32 wave = threadIdx.x / 64;
+    const u32 lane = threadIdx.x % 64;
+    const u32 wordsPerPage = FAM_PAGE / 8; // 512
+    for (u32 p = blockIdx.x * wavesPerBlock + wave; p < nPages;
+         p += gridDim.x * wavesPerBlock) {
+        u64 base = (u64)pages[p] * wordsPerPage;
+        for (u32 w = lane; w < wordsPerPage; w += 64) {
+            buf[base + w] ^= xorshift64(seed ^ (base + w));
+        }
+    }
+}
+
 inline u32 gridFor(u64 items)
 {
     u64 blocks = (items + FAM_KERNEL_BLOCK - 1) / FAM_KERNEL_BLOCK;
@@ -411,6 +455,48 @@ hipError_t famDirtyPages(const void* snap,
                        (const uint4*)cur,
                        nPages,
                        flagsDev);
+    return hipGetLastError();
+}
+
+hipError_t famFillRandom(void* buf,
+                         uint64_t bytes,
+                         uint64_t seed,
+                         hipStream_t stream)
+{
+    if ((bytes % 8) != 0) {
+        return hipErrorInvalidValue;
+    }
+    u64 nWords = bytes / 8;
+    hipLaunchKernelGGL(fillRandomKernel,
+                       dim3(gridFor(nWords)),
+                       dim3(FAM_KERNEL_BLOCK),
+                       0,
+                       stream,
+                       (u64*)buf,
+                       nWords,
+                       seed);
+    return hipGetLastError();
+}
+
+hipError_t famTouchPages(void* buf,
+                         const uint32_t* pagesDev,
+                         uint32_t nPages,
+                         uint64_t seed,
+                         hipStream_t stream)
+{
+    u32 wavesPerBlock = FAM_KERNEL_BLOCK / 64;
+    u32 blocks = (nPages + wavesPerBlock - 1) / wavesPerBlock;
+    u32 grid = blocks < FAM_MAX_BLOCKS ? (blocks ? blocks : 1)
+                                       : FAM_MAX_BLOCKS;
+    hipLaunchKernelGGL(touchPagesKernel,
+                       dim3(grid),
+                       dim3(FAM_KERNEL_BLOCK),
+                       0,
+                       stream,
+                       (u64*)buf,
+                       pagesDev,
+                       nPages,
+                       seed);
     return hipGetLastError();
 }
 

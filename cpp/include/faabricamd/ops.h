@@ -35,6 +35,17 @@ hipError_t famCopyBuffer(const void* src,
                          void* dst,
                          uint64_t bytes,
                          hipStream_t stream);
+// Deterministic pseudo-random fill / scattered page mutation (bench
+// init for the random-byte snapshot shapes, BASELINE.json config 4)
+hipError_t famFillRandom(void* buf,
+                         uint64_t bytes,
+                         uint64_t seed,
+                         hipStream_t stream);
+hipError_t famTouchPages(void* buf,
+                         const uint32_t* pagesDev,
+                         uint32_t nPages,
+                         uint64_t seed,
+                         hipStream_t stream);
 
 hipError_t famXorBuffer(const void* a,
                         const void* b,

@@ -224,6 +224,37 @@ def test_state_kv_in_hbm(runtime):
 
 
 @requires_gpu
+def test_snapshot_scattered_random_diff(runtime):
+    """Random-byte region + randomly scattered dirty pages: the diff
+    kernels must find exactly the touched set and merge back to equal
+    buffers (the BASELINE config-4 shape, not a contiguous memset)."""
+    import random as pyrandom
+
+    n_pages = 512
+    nbytes = n_pages * 4096
+    buf = torch.empty(nbytes, dtype=torch.uint8, device="cuda")
+    _core.fam_fill_random(buf.data_ptr(), nbytes, 99)
+    snap = _core.DeviceSnapshot(nbytes)
+    snap.capture_from_ptr(buf.data_ptr())
+
+    pyrandom.seed(5)
+    dirty = sorted(pyrandom.sample(range(n_pages), 128))
+    _core.fam_touch_pages(buf.data_ptr(), dirty, 1234)
+
+    nd = snap.diff_xor(buf.data_ptr())
+    assert nd == 128
+    pages, payload = snap.gather_last_diff()
+    assert sorted(pages) == dirty
+
+    snap.apply_last_diff()
+    # Snapshot now equals the updated buffer: a second diff is empty
+    assert snap.diff_xor(buf.data_ptr()) == 0
+    got = snap.copy_out_host(nbytes)
+    expect = buf.cpu().numpy().tobytes()
+    assert got == expect
+
+
+@requires_gpu
 def test_state_kv_mirror_coherence(runtime):
     """The pinned write-through mirror stays coherent with HBM across
     direct device-pointer writes (data_ptr access invalidates the
