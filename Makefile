@@ -42,7 +42,8 @@ $(TARGET): $(OBJS)
 # C++ example binaries (reference examples/ + planner_server parity)
 CORE_OBJS = $(filter-out $(BUILD)/bindings/%,$(OBJS))
 EXAMPLE_BINS = $(BUILD)/check $(BUILD)/planner_server $(BUILD)/server \
-               $(BUILD)/selftest $(BUILD)/is_app_migratable
+               $(BUILD)/selftest $(BUILD)/is_app_migratable \
+               $(BUILD)/disttest
 examples: $(TARGET) $(EXAMPLE_BINS)
 
 # Sanitizer sweeps (reference CI parity: Address/Thread sanitised suites,
@@ -70,7 +71,30 @@ tsan-check:
 	FAABRIC_PORT_OFFSET=6700 TSAN_OPTIONS="report_bugs=1" \
 	    $(BUILD)/tsan/selftest
 
-.PHONY: asan-check tsan-check
+# Multi-process dist scenarios (cross-process transport, leader
+# collectives, live MPI migration) under each sanitizer — the reference
+# CI sanitizes its dist suite, not just unit tests
+asan-dist:
+	@mkdir -p $(BUILD)/asan
+	$(HIPCC) -O1 -g -std=c++20 -fsanitize=address -fno-omit-frame-pointer \
+	    -Icpp/include --offload-arch=$(GPU_ARCH) \
+	    $(CPP_SRCS_CORE) cpp/hip/snapshot_kernels.hip examples/disttest.cpp \
+	    -L/opt/rocm/lib -lrccl -lamdhip64 -lz -l:libzstd.so.1 -pthread \
+	    -o $(BUILD)/asan/disttest
+	DISTTEST_BASE_OFFSET=6200 DISTTEST_STOP_FILE=$(BUILD)/asan/stop \
+	    $(BUILD)/asan/disttest
+
+tsan-dist:
+	@mkdir -p $(BUILD)/tsan
+	$(HIPCC) -O1 -g -std=c++20 -fsanitize=thread \
+	    -Icpp/include --offload-arch=$(GPU_ARCH) \
+	    $(CPP_SRCS_CORE) cpp/hip/snapshot_kernels.hip examples/disttest.cpp \
+	    -L/opt/rocm/lib -lrccl -lamdhip64 -lz -l:libzstd.so.1 -pthread \
+	    -o $(BUILD)/tsan/disttest
+	DISTTEST_BASE_OFFSET=6400 DISTTEST_STOP_FILE=$(BUILD)/tsan/stop \
+	    TSAN_OPTIONS="report_bugs=1" $(BUILD)/tsan/disttest
+
+.PHONY: asan-check tsan-check asan-dist tsan-dist
 
 $(BUILD)/%: $(BUILD)/examples/%.o $(CORE_OBJS)
 	$(HIPCC) $< $(CORE_OBJS) \

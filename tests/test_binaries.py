@@ -173,6 +173,24 @@ def test_check_binary():
     assert "CHECK OK" in out.stdout
 
 
+def test_disttest_binary():
+    """Multi-process dist scenarios (two worker processes, leader
+    collectives, live MPI migration) — same binary the asan-dist /
+    tsan-dist targets run."""
+    disttest = os.path.join(BUILD, "disttest")
+    if not os.path.exists(disttest):
+        pytest.skip("disttest not built")
+    out = subprocess.run(
+        [disttest],
+        env={**os.environ, "DISTTEST_BASE_OFFSET": "7400",
+             "DISTTEST_STOP_FILE": "/tmp/disttest-pytest.stop",
+             "LOG_LEVEL": "error"},
+        capture_output=True, text=True, timeout=180,
+    )
+    assert out.returncode == 0, (out.stdout[-300:], out.stderr[-500:])
+    assert "DISTTEST OK" in out.stdout
+
+
 def test_selftest_binary():
     """The C++ integration sweep (batches, THREADS merge, MPI world,
     chaining, SPSC queue, snapshot semantics) — same binary the
