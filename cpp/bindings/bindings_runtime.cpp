@@ -5,6 +5,7 @@
 
 #include "faabricamd/endpoint.h"
 #include "faabricamd/executor.h"
+#include "faabricamd/flat.h"
 #include "faabricamd/hipipc.h"
 #include "faabricamd/planner.h"
 #include "faabricamd/ptp.h"
@@ -695,6 +696,75 @@ void initRuntimeBindings(py::module_& m)
 
     // ---------------- util extras ------------------------------------------
     m.def("delta_default_config", [] { return DeltaConfig{}.str(); });
+    // FlatBuffers snapshot wire format (tests pin the encoding)
+    m.def("flat_encode_push",
+          [](const std::string& key, uint64_t maxSize,
+             const py::bytes& contents,
+             const std::vector<std::tuple<int32_t, uint64_t, int32_t,
+                                          int32_t>>& regions) {
+              FlatSnapshotPush req;
+              req.key = key;
+              req.maxSize = maxSize;
+              std::string c = contents;
+              req.contents.assign(c.begin(), c.end());
+              for (auto& [off, len, dt, op] : regions) {
+                  FlatMergeRegion m;
+                  m.offset = off;
+                  m.length = len;
+                  m.dataType = dt;
+                  m.mergeOp = op;
+                  req.mergeRegions.push_back(m);
+              }
+              std::string out = req.encode();
+              return py::bytes(out);
+          });
+    m.def("flat_decode_push", [](const py::bytes& buf) {
+        std::string s = buf;
+        auto req = FlatSnapshotPush::decode(s);
+        py::list regions;
+        for (auto& m : req.mergeRegions) {
+            regions.append(py::make_tuple(m.offset, m.length, m.dataType,
+                                          m.mergeOp));
+        }
+        return py::make_tuple(
+          req.key, req.maxSize,
+          py::bytes((const char*)req.contents.data(),
+                    req.contents.size()),
+          regions);
+    });
+    m.def("flat_encode_thread_result",
+          [](int32_t appId, int32_t messageId, int32_t returnValue,
+             const std::string& key,
+             const std::vector<std::tuple<int32_t, int32_t, int32_t,
+                                          py::bytes>>& diffs) {
+              FlatThreadResult req;
+              req.appId = appId;
+              req.messageId = messageId;
+              req.returnValue = returnValue;
+              req.key = key;
+              for (auto& [off, dt, op, data] : diffs) {
+                  FlatSnapshotDiff d;
+                  d.offset = off;
+                  d.dataType = dt;
+                  d.mergeOp = op;
+                  std::string s = data;
+                  d.data.assign(s.begin(), s.end());
+                  req.diffs.push_back(std::move(d));
+              }
+              return py::bytes(req.encode());
+          });
+    m.def("flat_decode_thread_result", [](const py::bytes& buf) {
+        std::string s = buf;
+        auto req = FlatThreadResult::decode(s);
+        py::list diffs;
+        for (auto& d : req.diffs) {
+            diffs.append(py::make_tuple(
+              d.offset, d.dataType, d.mergeOp,
+              py::bytes((const char*)d.data.data(), d.data.size())));
+        }
+        return py::make_tuple(req.appId, req.messageId, req.returnValue,
+                              req.key, diffs, req.executedHost);
+    });
     m.def("delta_encode",
           [](const py::bytes& oldData,
              const py::bytes& newData,

@@ -215,6 +215,9 @@ enum class SnapshotCalls : uint8_t
     // Chunk of a device snapshot shipped via HIP IPC (same-node workers):
     // payload is already in our arena, body is an IpcChunk
     PushSnapshotIpc = 5,
+    // Device (HBM) snapshot as host bytes — our extension beyond the
+    // reference's flatbuffers surface, so it rides its own call
+    PushSnapshotDevice = 6,
 };
 
 class SnapshotServer : public MessageEndpointServer

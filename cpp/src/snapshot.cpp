@@ -4,6 +4,7 @@
 // device (HBM) path dispatches the same operations to gfx950 HIP kernels —
 // see cpp/hip/snapshot_kernels.hip and ops.cpp.
 #include "faabricamd/snapshot.h"
+#include "faabricamd/flat.h"
 #include "faabricamd/hipipc.h"
 #include "faabricamd/ops.h"
 #include "faabricamd/planner.h"
@@ -669,7 +670,8 @@ void SnapshotServer::doAsyncRecv(uint8_t code,
 {
     (void)seq;
     if ((SnapshotCalls)code == SnapshotCalls::DeleteSnapshot) {
-        auto req = SnapshotDeleteRequest::decode(body);
+        // FlatBuffers body (faabric.fbs SnapshotDeleteRequest)
+        auto req = FlatSnapshotDelete::decode(body);
         SnapshotRegistry::get().deleteSnapshot(req.key);
         // Device snapshots shipped here land in the device registry
         DeviceSnapshotRegistry::get().deleteSnapshot(req.key);
@@ -682,24 +684,8 @@ std::string SnapshotServer::doSyncRecv(uint8_t code, const std::string& body)
 {
     switch ((SnapshotCalls)code) {
         case SnapshotCalls::PushSnapshot: {
-            auto req = SnapshotPushRequest::decode(body);
-            if (req.onDevice) {
-                // HBM-resident snapshot: land it in this host's GPU
-                if (!gpuAvailable()) {
-                    throw FaabricException(
-                      "device snapshot pushed to GPU-less host");
-                }
-                size_t rounded =
-                  (req.contents.size() + DEVICE_PAGE - 1) / DEVICE_PAGE *
-                  DEVICE_PAGE;
-                auto dsnap =
-                  std::make_shared<DeviceSnapshot>(rounded, 0);
-                dsnap->copyInHost(req.contents.data(),
-                                  req.contents.size());
-                DeviceSnapshotRegistry::get().registerSnapshot(req.key,
-                                                               dsnap);
-                return {};
-            }
+            // FlatBuffers body (reference: faabric.fbs SnapshotPushRequest)
+            auto req = FlatSnapshotPush::decode(body);
             auto snap = std::make_shared<SnapshotData>(
               req.contents, std::max<size_t>(req.maxSize,
                                              req.contents.size()));
@@ -710,6 +696,22 @@ std::string SnapshotServer::doSyncRecv(uint8_t code, const std::string& body)
                                      (SnapshotMergeOperation)r.mergeOp);
             }
             SnapshotRegistry::get().registerSnapshot(req.key, snap);
+            return {};
+        }
+        case SnapshotCalls::PushSnapshotDevice: {
+            // HBM-resident snapshot as host bytes (protobuf body — our
+            // extension, kept off the flatbuffers surface)
+            auto req = SnapshotPushRequest::decode(body);
+            if (!gpuAvailable()) {
+                throw FaabricException(
+                  "device snapshot pushed to GPU-less host");
+            }
+            size_t rounded = (req.contents.size() + DEVICE_PAGE - 1) /
+                             DEVICE_PAGE * DEVICE_PAGE;
+            auto dsnap = std::make_shared<DeviceSnapshot>(rounded, 0);
+            dsnap->copyInHost(req.contents.data(), req.contents.size());
+            DeviceSnapshotRegistry::get().registerSnapshot(req.key,
+                                                           dsnap);
             return {};
         }
         case SnapshotCalls::PushSnapshotIpc: {
@@ -747,11 +749,17 @@ std::string SnapshotServer::doSyncRecv(uint8_t code, const std::string& body)
             return {};
         }
         case SnapshotCalls::PushSnapshotUpdate: {
-            auto req = SnapshotUpdateRequest::decode(body);
+            // FlatBuffers body (faabric.fbs SnapshotUpdateRequest)
+            auto req = FlatSnapshotUpdate::decode(body);
             auto snap = SnapshotRegistry::get().getSnapshot(req.key);
             std::vector<SnapshotDiff> diffs;
             for (const auto& d : req.diffs) {
-                diffs.push_back(SnapshotDiff::fromMsg(d));
+                SnapshotDiffMsg m;
+                m.offset = d.offset;
+                m.dataType = d.dataType;
+                m.mergeOp = d.mergeOp;
+                m.data = d.data;
+                diffs.push_back(SnapshotDiff::fromMsg(m));
             }
             snap->applyDiffs(diffs);
             for (const auto& r : req.mergeRegions) {
@@ -763,7 +771,22 @@ std::string SnapshotServer::doSyncRecv(uint8_t code, const std::string& body)
             return {};
         }
         case SnapshotCalls::ThreadResult: {
-            auto req = ThreadResultRequest::decode(body);
+            // FlatBuffers body (faabric.fbs ThreadResultRequest)
+            auto flat = FlatThreadResult::decode(body);
+            ThreadResultRequest req;
+            req.appId = flat.appId;
+            req.messageId = flat.messageId;
+            req.returnValue = flat.returnValue;
+            req.key = flat.key;
+            req.executedHost = flat.executedHost;
+            for (const auto& d : flat.diffs) {
+                SnapshotDiffMsg m;
+                m.offset = d.offset;
+                m.dataType = d.dataType;
+                m.mergeOp = d.mergeOp;
+                m.data = d.data;
+                req.diffs.push_back(m);
+            }
             // Queue the diffs onto the main-thread snapshot, then forward
             // the thread's result (reference: SnapshotServer.cpp:28-62)
             if (!req.key.empty() && !req.diffs.empty()) {
@@ -841,12 +864,14 @@ void SnapshotClient::pushSnapshot(const std::string& key, SnapshotData& data)
         mockedPushes.emplace_back(getHost(), key);
         return;
     }
-    SnapshotPushRequest req;
+    // FlatBuffers wire format (reference: faabric.fbs SnapshotPushRequest,
+    // SnapshotClient::pushSnapshot)
+    FlatSnapshotPush req;
     req.key = key;
     req.maxSize = data.getMaxSize();
     req.contents = data.getDataCopy();
     for (const auto& [off, r] : data.getMergeRegions()) {
-        SnapshotMergeRegionMsg m;
+        FlatMergeRegion m;
         m.offset = (int32_t)r.offset;
         m.length = r.length;
         m.dataType = (int32_t)r.dataType;
@@ -871,7 +896,7 @@ void SnapshotClient::pushDeviceSnapshot(const std::string& key,
     req.onDevice = true;
     req.contents.assign((const uint8_t*)hostCopy,
                         (const uint8_t*)hostCopy + size);
-    syncSend((uint8_t)SnapshotCalls::PushSnapshot, req.encode());
+    syncSend((uint8_t)SnapshotCalls::PushSnapshotDevice, req.encode());
 }
 
 void SnapshotClient::pushDeviceSnapshotFromDevice(const std::string& key,
@@ -925,13 +950,20 @@ void SnapshotClient::pushSnapshotUpdate(
   const std::vector<SnapshotDiff>& diffs,
   const std::vector<SnapshotMergeRegion>& regions)
 {
-    SnapshotUpdateRequest req;
+    // FlatBuffers wire format (faabric.fbs SnapshotUpdateRequest)
+    FlatSnapshotUpdate req;
     req.key = key;
     for (const auto& d : diffs) {
-        req.diffs.push_back(d.toMsg());
+        auto m = d.toMsg();
+        FlatSnapshotDiff fd;
+        fd.offset = m.offset;
+        fd.dataType = m.dataType;
+        fd.mergeOp = m.mergeOp;
+        fd.data = std::move(m.data);
+        req.diffs.push_back(std::move(fd));
     }
     for (const auto& r : regions) {
-        SnapshotMergeRegionMsg m;
+        FlatMergeRegion m;
         m.offset = (int32_t)r.offset;
         m.length = r.length;
         m.dataType = (int32_t)r.dataType;
@@ -961,12 +993,28 @@ void SnapshotClient::pushThreadResult(int32_t appId,
         mockedThreadResults.emplace_back(getHost(), req);
         return;
     }
-    syncSend((uint8_t)SnapshotCalls::ThreadResult, req.encode());
+    // FlatBuffers wire format (faabric.fbs ThreadResultRequest)
+    FlatThreadResult flat;
+    flat.appId = req.appId;
+    flat.messageId = req.messageId;
+    flat.returnValue = req.returnValue;
+    flat.key = req.key;
+    flat.executedHost = req.executedHost;
+    for (const auto& m : req.diffs) {
+        FlatSnapshotDiff fd;
+        fd.offset = m.offset;
+        fd.dataType = m.dataType;
+        fd.mergeOp = m.mergeOp;
+        fd.data = m.data;
+        flat.diffs.push_back(std::move(fd));
+    }
+    syncSend((uint8_t)SnapshotCalls::ThreadResult, flat.encode());
 }
 
 void SnapshotClient::deleteSnapshot(const std::string& key)
 {
-    SnapshotDeleteRequest req;
+    // FlatBuffers wire format (faabric.fbs SnapshotDeleteRequest)
+    FlatSnapshotDelete req;
     req.key = key;
     asyncSend((uint8_t)SnapshotCalls::DeleteSnapshot, req.encode());
 }
