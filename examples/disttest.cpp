@@ -67,6 +67,9 @@ static void awaitBatch(int32_t appId, int total, int timeoutMs)
 
 static int runWorker2()
 {
+    // Hard backstop: if the parent dies (aborted test run) or shutdown
+    // wedges, never outlive the scenario and squat on the ports
+    alarm(150);
     int off = basePortOffset() + 200;
     setPortOffset(off);
     getSystemConfig().endpointHost = "127.0.0.1@" + std::to_string(off);
@@ -86,6 +89,10 @@ static int runWorker2()
     for (int i = 0; i < 2400; i++) { // up to 2 min
         struct stat st;
         if (::stat(stop.c_str(), &st) == 0) {
+            break;
+        }
+        // Orphaned (parent died without writing the stop file)? Exit.
+        if (getppid() == 1) {
             break;
         }
         usleep(50000);
