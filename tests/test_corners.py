@@ -1,0 +1,243 @@
+"""Corner-depth parity: transport reconnect/timeout semantics, endpoint
+malformed-request handling, per-policy placement matrices, and decoder
+fuzz — the edges the reference's 114-file unit matrix covers
+(reference: tests/test/transport/*, tests/test/batch-scheduler/*,
+tests/test/planner/test_planner_endpoint.cpp).
+"""
+
+import json
+import multiprocessing as mp
+import os
+import random
+import sys
+import urllib.request
+
+import pytest
+
+from faabric_amd import _core
+
+REPO_ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+A = "10.0.0.1"
+B = "10.0.0.2"
+C = "10.0.0.3"
+EVICT = "E.VI.CT.ME"
+
+
+def decide(mode, hosts, n, **kw):
+    return _core.test_make_scheduling_decision(mode, hosts, n, **kw)
+
+
+# ---------------------------------------------------------------------------
+# Placement matrices (reference per-policy tables)
+# ---------------------------------------------------------------------------
+
+# Reference sort: free slots desc, total slots desc, then IP DESC
+# ("largest host alphabetically", BinPackScheduler.cpp getSortedHosts)
+BINPACK_TABLE = [
+    # (hosts [(ip, slots, used)], n, expected host sequence)
+    ([(A, 8, 0)], 8, [A] * 8),                      # exact fit one host
+    ([(A, 2, 0), (B, 2, 0), (C, 2, 0)], 6,
+     [C, C, B, B, A, A]),                            # all-tie: IP desc
+    ([(A, 4, 3), (B, 4, 1), (C, 4, 2)], 3,
+     [B, B, B]),                                     # most-free first
+    ([(A, 3, 1), (B, 3, 1)], 4, [B, B, A, A]),       # tie: IP desc
+    ([(A, 6, 5), (B, 6, 4), (C, 6, 3)], 5,
+     [C, C, C, B, B]),                               # descending free
+]
+
+
+@pytest.mark.parametrize("hosts,n,expect", BINPACK_TABLE)
+def test_binpack_placement_table(hosts, n, expect):
+    d = decide("bin-pack", hosts, n)
+    assert d.hosts == expect
+
+
+# Compact's NEW-decision sort equals bin-pack's (it differs on
+# DIST_CHANGE, where it maximises fully-free hosts — CompactScheduler
+# isFirstHostFuller + free-host comparison, covered in test_policies)
+COMPACT_TABLE = [
+    ([(A, 8, 0), (B, 8, 6)], 2, [A, A]),
+    ([(A, 4, 2), (B, 4, 1), (C, 4, 0)], 3, [C, C, C]),
+    ([(A, 4, 3), (B, 4, 3)], 2, [B, A]),
+]
+
+
+@pytest.mark.parametrize("hosts,n,expect", COMPACT_TABLE)
+def test_compact_placement_table(hosts, n, expect):
+    d = decide("compact", hosts, n)
+    assert d.hosts == expect
+
+
+def test_binpack_not_enough_slots_exact_boundary():
+    # n equals free+1: must refuse, not truncate
+    d = decide("bin-pack", [(A, 4, 2), (B, 4, 3)], 4)
+    assert d.app_id == _core.NOT_ENOUGH_SLOTS()
+    # n equals free exactly: must fit
+    d = decide("bin-pack", [(A, 4, 2), (B, 4, 3)], 3)
+    assert sorted(d.hosts) == [A, A, B]
+
+
+def test_spot_eviction_matrix():
+    # Doomed VM has ALL the capacity -> freeze, not half-placement
+    d = decide(
+        "spot",
+        [(EVICT, 8, 2), (A, 2, 2)],
+        2,
+        app_id=9,
+        migration=True,
+        in_flight=[(9, [EVICT, EVICT])],
+    )
+    assert d.app_id == _core.MUST_FREEZE()
+    # Non-migration scheduling refuses the doomed VM even when it is
+    # the only host with room
+    d = decide("spot", [(EVICT, 8, 0), (A, 2, 0)], 2)
+    assert d.hosts == [A, A]
+    # And reports NOT_ENOUGH_SLOTS rather than using it
+    d = decide("spot", [(EVICT, 8, 0), (A, 1, 0)], 2)
+    assert d.app_id == _core.NOT_ENOUGH_SLOTS()
+
+
+def test_compact_same_tenant_in_flight_does_not_distort_new():
+    # Another app's in-flight placement must not distort a NEW decision
+    # (the occupancy is already reflected in the host map's used slots)
+    d = decide(
+        "compact",
+        [(A, 4, 2), (B, 4, 0)],
+        2,
+        app_id=77,
+        in_flight=[(42, [A, A])],
+    )
+    assert d.hosts == [B, B]
+
+
+# ---------------------------------------------------------------------------
+# Transport: reconnect, timeout, remote-error propagation
+# ---------------------------------------------------------------------------
+
+RECON_OFF = 7800
+
+
+def _recon_server(stop, ready, restart):
+    sys.path.insert(0, REPO_ROOT)
+    from faabric_amd import _core as core
+
+    core.set_log_level("error")
+    core.set_port_offset(RECON_OFF)
+    core.set_endpoint_host(f"127.0.0.1@{RECON_OFF}")
+    kv = core.state_get_kv("corner", "reconnect", 4096)
+    kv.set(b"\x42" * 4096)
+    srv = core.StateServerHandle()
+    srv.start()
+    ready.set()
+    restart.wait(60)
+    # Drop every live connection, then come back on the same port
+    srv.stop()
+    srv2 = core.StateServerHandle()
+    srv2.start()
+    ready.set()
+    stop.wait(60)
+    srv2.stop()
+
+
+def test_sync_client_reconnects_after_server_restart():
+    """A sync client with a cached connection must transparently re-dial
+    after the server restarts (reference: transport reconnect tests)."""
+    ctx = mp.get_context("spawn")
+    stop = ctx.Event()
+    ready = ctx.Event()
+    restart = ctx.Event()
+    p = ctx.Process(target=_recon_server, args=(stop, ready, restart))
+    p.start()
+    try:
+        assert ready.wait(60)
+        prev_host = _core.get_endpoint_host()
+        _core.set_endpoint_host("127.0.0.1@7900")
+        _core.state_clear_all()
+        _core.state_set_master_host("corner", "reconnect",
+                                    f"127.0.0.1@{RECON_OFF}")
+        kv = _core.state_get_kv("corner", "reconnect", 4096)
+        assert kv.get_chunk(0, 16) == b"\x42" * 16  # connection cached
+        ready.clear()
+        restart.set()
+        assert ready.wait(60)  # server is back
+        # Same client object, dead socket: must reconnect and succeed
+        assert kv.get_chunk(16, 16) == b"\x42" * 16
+    finally:
+        stop.set()
+        p.join(timeout=30)
+        if p.is_alive():
+            p.terminate()
+        _core.state_clear_all()
+        _core.set_endpoint_host(prev_host)
+
+
+def test_remote_error_propagates_with_host():
+    """A server-side exception surfaces client-side as a remote error
+    naming the host, not a hang or a silent empty reply."""
+    ctx = mp.get_context("spawn")
+    stop = ctx.Event()
+    ready = ctx.Event()
+    restart = ctx.Event()
+    p = ctx.Process(target=_recon_server, args=(stop, ready, restart))
+    p.start()
+    try:
+        assert ready.wait(60)
+        prev_host = _core.get_endpoint_host()
+        _core.set_endpoint_host("127.0.0.1@7900")
+        _core.state_clear_all()
+        _core.state_set_master_host("corner", "reconnect",
+                                    f"127.0.0.1@{RECON_OFF}")
+        kv = _core.state_get_kv("corner", "reconnect", 1 << 20)
+        with pytest.raises(RuntimeError) as err:
+            kv.get_chunk((1 << 20) - 8, 8)  # out of the REAL 4 KiB value
+        assert "remote error" in str(err.value)
+        assert f"127.0.0.1@{RECON_OFF}" in str(err.value)
+    finally:
+        stop.set()
+        restart.set()
+        p.join(timeout=30)
+        if p.is_alive():
+            p.terminate()
+        _core.state_clear_all()
+        _core.set_endpoint_host(prev_host)
+
+
+def test_ptp_recv_times_out_cleanly():
+    with pytest.raises(RuntimeError) as err:
+        _core.ptp_recv(991234, 0, 1, False, 150)
+    assert "timeout" in str(err.value).lower()
+
+
+# ---------------------------------------------------------------------------
+# Decoder fuzz: random + truncated buffers must never crash
+# ---------------------------------------------------------------------------
+
+def test_wire_decoders_survive_fuzz():
+    random.seed(11)
+    good_flat = _core.flat_encode_push("k", 64, b"x" * 32, [(0, 16, 1, 2)])
+    good_msg = _core.Message().encode()
+    for trial in range(300):
+        n = random.randrange(0, 120)
+        buf = bytes(random.getrandbits(8) for _ in range(n))
+        for decoder in (_core.flat_decode_push,
+                        _core.flat_decode_thread_result):
+            try:
+                decoder(buf)
+            except Exception:
+                pass  # throwing is fine; crashing is not
+        try:
+            _core.Message.decode(buf)
+        except Exception:
+            pass
+        # Truncations of valid buffers
+        for good in (good_flat, good_msg):
+            cut = good[: random.randrange(0, max(1, len(good)))]
+            try:
+                _core.flat_decode_push(cut)
+            except Exception:
+                pass
+            try:
+                _core.Message.decode(cut)
+            except Exception:
+                pass
