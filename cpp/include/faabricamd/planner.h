@@ -29,6 +29,10 @@ enum class PlannerCalls : uint8_t
     RegisterHost = 3,
     RemoveHost = 4,
     SetMessageResult = 8,
+    // Natural batching of result reports: when many executors finish in
+    // a burst, their results coalesce into one RPC while the previous
+    // one is in flight (group commit; single results flush immediately)
+    SetMessageResultBatch = 15,
     GetMessageResult = 9,
     GetBatchResults = 10,
     GetSchedulingDecision = 11,
@@ -165,6 +169,9 @@ class PlannerClient
       std::shared_ptr<BatchExecuteRequest> req);
 
     void setMessageResult(std::shared_ptr<Message> msg);
+    // One RPC carrying many results (see PlannerCalls::SetMessageResultBatch)
+    void setMessageResultsBatch(
+      const std::vector<std::shared_ptr<Message>>& msgs);
     // Called by the worker's FunctionCallServer when the planner pushes a
     // result to this host
     void setMessageResultLocally(std::shared_ptr<Message> msg);

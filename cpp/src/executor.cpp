@@ -15,6 +15,9 @@
 
 namespace faabricamd {
 
+// Defined below: group-commit result reporting to the planner
+void resultBatcherEnqueue(std::shared_ptr<Message> msg);
+
 // ----------------------------- context -------------------------------------
 
 static thread_local std::shared_ptr<ExecutorContext> currentContext;
@@ -658,9 +661,87 @@ void Executor::handleTaskResult(Message& msg,
         return;
     }
 
-    // Report the result to the planner
-    auto resultMsg = std::make_shared<Message>(msg);
-    getPlannerClient().setMessageResult(resultMsg);
+    // Report the result to the planner through the result batcher:
+    // group commit — a lone result flushes immediately, a burst of
+    // finishing executors coalesces into one RPC while the previous
+    // send is in flight
+    resultBatcherEnqueue(std::make_shared<Message>(msg));
+}
+
+// ------------------------- result batcher -----------------------------------
+
+namespace {
+
+struct ResultBatcher
+{
+    std::mutex mx;
+    std::condition_variable cv;
+    std::vector<std::shared_ptr<Message>> queue;
+    std::thread worker;
+    bool stop = false;
+    bool started = false;
+
+    void ensureStarted()
+    {
+        std::lock_guard<std::mutex> lock(mx);
+        if (started) {
+            return;
+        }
+        started = true;
+        worker = std::thread([this] { run(); });
+    }
+
+    void run()
+    {
+        std::vector<std::shared_ptr<Message>> batch;
+        while (true) {
+            {
+                std::unique_lock<std::mutex> lock(mx);
+                cv.wait(lock, [&] { return stop || !queue.empty(); });
+                if (stop && queue.empty()) {
+                    return;
+                }
+                batch.swap(queue);
+            }
+            try {
+                getPlannerClient().setMessageResultsBatch(batch);
+            } catch (const std::exception& e) {
+                FAM_ERROR("result batch send failed: %s", e.what());
+            }
+            batch.clear();
+        }
+    }
+
+    ~ResultBatcher()
+    {
+        {
+            std::lock_guard<std::mutex> lock(mx);
+            stop = true;
+        }
+        cv.notify_all();
+        if (worker.joinable()) {
+            worker.join();
+        }
+    }
+};
+
+ResultBatcher& resultBatcher()
+{
+    static ResultBatcher instance;
+    return instance;
+}
+
+} // namespace
+
+void resultBatcherEnqueue(std::shared_ptr<Message> msg)
+{
+    auto& b = resultBatcher();
+    b.ensureStarted();
+    {
+        std::lock_guard<std::mutex> lock(b.mx);
+        b.queue.push_back(std::move(msg));
+    }
+    b.cv.notify_one();
 }
 
 // ----------------------------- chaining -------------------------------------

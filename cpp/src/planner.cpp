@@ -858,6 +858,27 @@ void PlannerServer::doAsyncRecv(uint8_t code,
         }
         return;
     }
+    if ((PlannerCalls)code == PlannerCalls::SetMessageResultBatch) {
+        // One RPC, many results: split and feed the same ingestion path
+        PbReader r(body);
+        uint32_t f;
+        WireType t;
+        while (r.next(f, t)) {
+            if (f != 1) {
+                r.skip(t);
+                continue;
+            }
+            std::string sub = r.asString();
+            if (resultWorkers.empty()) {
+                auto msg =
+                  std::make_shared<Message>(Message::decode(sub));
+                Planner::get().setMessageResult(msg);
+            } else {
+                resultQueue.enqueue(std::move(sub));
+            }
+        }
+        return;
+    }
     FAM_ERROR("planner server: bad async code %d", (int)code);
 }
 
@@ -1033,6 +1054,23 @@ void PlannerClient::setMessageResult(std::shared_ptr<Message> msg)
         return;
     }
     rpc.asyncSend((uint8_t)PlannerCalls::SetMessageResult, msg->encode());
+}
+
+void PlannerClient::setMessageResultsBatch(
+  const std::vector<std::shared_ptr<Message>>& msgs)
+{
+    if (isMockMode() || msgs.empty()) {
+        return;
+    }
+    if (msgs.size() == 1) {
+        setMessageResult(msgs[0]);
+        return;
+    }
+    PbWriter w;
+    for (const auto& m : msgs) {
+        w.putMessage(1, m->encode());
+    }
+    rpc.asyncSend((uint8_t)PlannerCalls::SetMessageResultBatch, w.take());
 }
 
 void PlannerClient::setMessageResultLocally(std::shared_ptr<Message> msg)
