@@ -67,10 +67,15 @@ def main(cycles=25):
         rs = wait_for_batch(ber.app_id, 1, 60_000)
         assert rs[0].return_value == 0, rs[0].output_data
 
-        # 2. HBM KV churn
+        # 2. HBM KV churn (pinned-mirror write-through path: partial
+        # writes, mirror fills, group-commit sync)
         kv = _core.state_get_kv_device("gsoak", f"k{i % 3}", 1 << 20)
         kv.set(bytes([i % 251]) * (1 << 20))
         assert kv.get_chunk(0, 1) == bytes([i % 251])
+        kv.set_chunk(777, bytes([(i + 1) % 251]) * 100)
+        kv.sync()
+        got = kv.get_chunk(770, 10)
+        assert got[:7] == bytes([i % 251]) * 7
 
         # 3. RCCL world churn: a fresh device-plane world per cycle; its
         # communicator and stream must be reclaimed when the rank finishes
@@ -83,24 +88,29 @@ def main(cycles=25):
         rs = wait_for_batch(ber.app_id, 1, 60_000)
         assert rs[0].return_value == 0, rs[0].output_data
 
-        # 4. Raw snapshot cycle
+        # 4. Raw snapshot cycle with the scattered-random kernels
         snap = _core.DeviceSnapshot(1 << 22, 0)
-        t = torch.full((1 << 22,), i % 127, dtype=torch.uint8,
-                       device="cuda")
-        torch.cuda.synchronize()
+        t = torch.empty(1 << 22, dtype=torch.uint8, device="cuda")
+        _core.fam_fill_random(t.data_ptr(), 1 << 22, i)
         snap.capture_from_ptr(t.data_ptr())
-        t[1234] ^= 0xFF
-        torch.cuda.synchronize()
+        _core.fam_touch_pages(t.data_ptr(), [3, 77, 500, 1001], i)
         nd = snap.diff_xor(t.data_ptr())
-        assert nd == 1, nd
+        assert nd == 4, nd
         del snap, t
 
         if (i + 1) % 10 == 0:
             torch.cuda.synchronize()
             free_now, _ = torch.cuda.mem_get_info()
             drift = (free0 - free_now) / (1 << 20)
-            print(f"cycle {i+1}/{cycles}: HBM drift {drift:+.1f} MiB",
-                  flush=True)
+            rss = 0
+            try:
+                import psutil
+
+                rss = psutil.Process().memory_info().rss / (1 << 20)
+            except Exception:
+                pass
+            print(f"cycle {i+1}/{cycles}: HBM drift {drift:+.1f} MiB, "
+                  f"RSS {rss:.0f} MiB", flush=True)
 
     torch.cuda.synchronize()
     free1, _ = torch.cuda.mem_get_info()
