@@ -91,7 +91,8 @@ inline constexpr size_t DEVICE_PAGE = 4096;
 class DeviceSnapshot
 {
   public:
-    DeviceSnapshot(size_t bytes, int device = 0);
+    // device -1 = this worker's configured GPU (FAABRIC_GPU_DEVICE)
+    DeviceSnapshot(size_t bytes, int device = -1);
     ~DeviceSnapshot();
     DeviceSnapshot(const DeviceSnapshot&) = delete;
 

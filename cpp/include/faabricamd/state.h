@@ -162,10 +162,11 @@ class State
                                          size_t size);
     // HBM-resident value on this host's GPU (north star: distributed
     // state lives in the 288 GB HBM3E per GPU)
+    // device -1 = this worker's configured GPU (FAABRIC_GPU_DEVICE)
     std::shared_ptr<StateKeyValue> getKVDevice(const std::string& user,
                                                const std::string& key,
                                                size_t size,
-                                               int device = 0);
+                                               int device = -1);
     std::shared_ptr<StateKeyValue> getKV(const std::string& user,
                                          const std::string& key);
     size_t getStateSize(const std::string& user, const std::string& key);

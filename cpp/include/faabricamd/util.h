@@ -97,6 +97,11 @@ struct SystemConfig
     std::string batchSchedulerMode; // bin-pack | compact | spot
     int overrideCpuCount = 0;
     int overrideGpuCount = -1; // -1 = probe HIP
+    // Which HIP device this worker's HBM lives on. Default deployments
+    // pin visibility per worker (HIP_VISIBLE_DEVICES=rank, device 0);
+    // all-visible deployments set FAABRIC_GPU_DEVICE=rank instead so
+    // same-node IPC maps peer GPUs over xGMI.
+    int gpuDevice = 0;
     bool useGpu = true;        // slots are GPUs when available
 
     // Timeouts (seconds, matching reference defaults

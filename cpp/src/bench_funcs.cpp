@@ -106,7 +106,7 @@ static int32_t benchRankStep(Message& msg)
     std::vector<uint8_t> hostA2aRecv;
 
     if (onGpu) {
-        if (hipSetDevice(0) != hipSuccess ||
+        if (hipSetDevice(getSystemConfig().gpuDevice) != hipSuccess ||
             hipMalloc(&sendBuf, bytes) != hipSuccess ||
             hipMalloc(&recvBuf, bytes) != hipSuccess) {
             msg.outputData = "hip alloc failed";
@@ -151,7 +151,7 @@ static int32_t benchRankStep(Message& msg)
       params.count("snapdirty") ? params["snapdirty"] : 25;
     if (onGpu && snapBytes > 0) {
         try {
-            DeviceSnapshot snap((size_t)snapBytes, 0);
+            DeviceSnapshot snap((size_t)snapBytes);
             uint8_t* updatedBuf = nullptr;
             if (hipMalloc(&updatedBuf, snapBytes) == hipSuccess) {
                 (void)famFillRandom(updatedBuf, (uint64_t)snapBytes,

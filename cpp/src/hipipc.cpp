@@ -228,6 +228,7 @@ IpcArenaInfo IpcReceiver::arenaFor(const std::string& senderHost)
         return arena->info;
     }
     uint64_t bytes = arenaBytes();
+    (void)hipSetDevice(getEnvVarInt("FAABRIC_GPU_DEVICE", 0));
     if (hipMalloc(&arena->base, bytes) != hipSuccess) {
         arena->base = nullptr;
         return arena->info;

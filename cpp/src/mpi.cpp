@@ -516,7 +516,9 @@ void MpiWorld::ensureRcclComm(int rank)
           it == localRanks.end()
             ? 0
             : (int)std::distance(localRanks.begin(), it);
-        device = localIdx % nDevices;
+        // Base device from config (all-visible deployments set
+        // FAABRIC_GPU_DEVICE per worker); extra local ranks spread
+        device = (getSystemConfig().gpuDevice + localIdx) % nDevices;
     }
     HIP_CHECK(hipSetDevice(device));
 

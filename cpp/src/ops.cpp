@@ -38,6 +38,9 @@ DeviceSnapshot::DeviceSnapshot(size_t bytes, int device)
   : bytes_(bytes)
   , device_(device)
 {
+    if (device_ < 0) {
+        device_ = getSystemConfig().gpuDevice;
+    }
     if ((bytes % DEVICE_PAGE) != 0) {
         throw FaabricException("device snapshot size must be page-aligned");
     }

@@ -563,6 +563,7 @@ void* PointToPointBroker::sideStream()
 {
     std::lock_guard<std::mutex> lock(streamMx);
     if (sideStream_ == nullptr) {
+        (void)hipSetDevice(getSystemConfig().gpuDevice);
         hipStream_t s = nullptr;
         if (hipStreamCreateWithFlags(&s, hipStreamNonBlocking) !=
             hipSuccess) {

@@ -708,7 +708,7 @@ std::string SnapshotServer::doSyncRecv(uint8_t code, const std::string& body)
             }
             size_t rounded = (req.contents.size() + DEVICE_PAGE - 1) /
                              DEVICE_PAGE * DEVICE_PAGE;
-            auto dsnap = std::make_shared<DeviceSnapshot>(rounded, 0);
+            auto dsnap = std::make_shared<DeviceSnapshot>(rounded);
             dsnap->copyInHost(req.contents.data(), req.contents.size());
             DeviceSnapshotRegistry::get().registerSnapshot(req.key,
                                                            dsnap);
@@ -735,7 +735,7 @@ std::string SnapshotServer::doSyncRecv(uint8_t code, const std::string& body)
             if (!dsnap) {
                 size_t rounded = (c.totalSize + DEVICE_PAGE - 1) /
                                  DEVICE_PAGE * DEVICE_PAGE;
-                dsnap = std::make_shared<DeviceSnapshot>(rounded, 0);
+                dsnap = std::make_shared<DeviceSnapshot>(rounded);
                 reg.registerSnapshot(c.key, dsnap);
             }
             if (c.valOffset + c.len > dsnap->size()) {

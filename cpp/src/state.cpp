@@ -31,6 +31,9 @@ StateKeyValue::StateKeyValue(std::string userIn,
   , onDevice(onDeviceIn)
   , device(deviceIn)
 {
+    if (onDevice && device < 0) {
+        device = getSystemConfig().gpuDevice;
+    }
     if (onDevice) {
         if (!gpuAvailable()) {
             throw FaabricException("device state KV requires a GPU");

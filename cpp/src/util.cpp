@@ -134,6 +134,7 @@ void SystemConfig::initialise()
     batchSchedulerMode = getEnvVar("BATCH_SCHEDULER_MODE", "bin-pack");
     overrideCpuCount = getEnvVarInt("OVERRIDE_CPU_COUNT", 0);
     overrideGpuCount = getEnvVarInt("OVERRIDE_GPU_COUNT", -1);
+    gpuDevice = getEnvVarInt("FAABRIC_GPU_DEVICE", 0);
     useGpu = getEnvVarInt("FAABRIC_USE_GPU", 1) != 0;
 
     globalMessageTimeout = getEnvVarInt("GLOBAL_MESSAGE_TIMEOUT", 60000);
