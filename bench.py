@@ -262,6 +262,8 @@ def main():
         dist.barrier()
     if RANK == 0 and result is not None:
         print(json.dumps(result), flush=True)
+        if os.environ.get("FAABRIC_PROF"):
+            print(_core.prof_summary(), file=sys.stderr, flush=True)
     if dist is not None:
         dist.barrier()
         dist.destroy_process_group()

@@ -67,6 +67,11 @@ class Scheduler
     std::unordered_map<std::string,
                        std::vector<std::shared_ptr<Executor>>>
       executors;
+    // Rotating claim hint per function key: with N warm executors a
+    // first-fit scan costs O(N^2) probes per N-message batch under the
+    // global lock; starting each scan after the previous claim makes it
+    // ~O(1) amortised
+    std::unordered_map<std::string, size_t> claimHints;
     HostResources overriddenResources;
     bool resourcesOverridden = false;
 

@@ -119,20 +119,42 @@ class StateKeyValue
     uint8_t* devPtr = nullptr;
 
     // Pinned host mirror for device values (group-commit write-through):
-    // chunk writes memcpy into the mirror and enqueue an async H2D on
-    // the KV stream with NO per-op sync — the ~12 us host-visible
+    // chunk writes memcpy into the mirror and enqueue an async H2D on a
+    // stripe stream with NO per-op sync — the ~12 us host-visible
     // latency of a synchronous 4 KiB HIP copy was the measured floor of
     // the batch path (BASELINE.md config-5 history). Reads are served
     // from the mirror (filled D2H on first touch). sync() is the
     // durability point; direct devPtr uses flush/invalidate around it.
-    std::mutex mirrorMx;
+    //
+    // The mirror is STRIPED: the value is split into 64 KiB blocks
+    // round-robined over KV_STRIPES (mutex, HIP stream) pairs so
+    // concurrent executors touching different ranges never serialise on
+    // one lock or one stream (128 concurrent kvtouch functions measured
+    // ~228 us of thread-time each on the single-mutex design). Per-range
+    // ordering holds because every byte maps to exactly one stripe.
+  public:
+    static constexpr int KV_STRIPES = 16;
+
+  private:
+    static constexpr uint64_t KV_STRIPE_BLOCK = 64 * 1024;
+    struct MirrorStripe
+    {
+        std::mutex mx;
+        void* stream = nullptr;
+    };
+    std::unique_ptr<MirrorStripe[]> stripes; // [KV_STRIPES] when enabled
+    std::mutex mirrorMx; // guards mirror bring-up only
     uint8_t* mirror = nullptr;
-    std::vector<char> mirrorValid; // per 4 KiB page
-    void* kvStream = nullptr;
+    std::vector<char> mirrorValid; // per 4 KiB page (owned by its stripe)
     bool mirrorFailed = false;
     bool mirrorUsable();
-    void mirrorFill(uint64_t offset, size_t len); // D2H fill, holds mirrorMx
-    void mirrorFlushLocked();                     // drain pending H2D
+    // Runtime-capped stripe count (FAABRIC_KV_STRIPES, default 16)
+    int stripeOf(uint64_t offset) const;
+    // Split [offset, offset+len) at stripe-block boundaries and run fn
+    // per subrange with that stripe's lock held
+    template<typename Fn>
+    void forEachStripeRange(uint64_t offset, size_t len, Fn&& fn);
+    void mirrorFillLocked(int stripe, uint64_t offset, size_t len);
     void mirrorInvalidate(uint64_t offset, size_t len);
 
   public:

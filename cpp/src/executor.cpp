@@ -704,7 +704,9 @@ struct ResultBatcher
                 batch.swap(queue);
             }
             try {
+                PROF_START(result_batch_send)
                 getPlannerClient().setMessageResultsBatch(batch);
+                PROF_END(result_batch_send)
             } catch (const std::exception& e) {
                 FAM_ERROR("result batch send failed: %s", e.what());
             }

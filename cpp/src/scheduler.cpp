@@ -76,6 +76,7 @@ void Scheduler::executeBatch(std::shared_ptr<BatchExecuteRequest> req)
     if (req->messages.empty()) {
         return;
     }
+    PROF_START(sched_execute_batch)
     bool isThreads = req->type == BatchExecuteType::THREADS;
 
     if (isTestMode()) {
@@ -138,6 +139,7 @@ void Scheduler::executeBatch(std::shared_ptr<BatchExecuteRequest> req)
             }
         }
     }
+    PROF_END(sched_execute_batch)
 }
 
 std::shared_ptr<Executor> Scheduler::claimExecutor(Message& msg)
@@ -147,9 +149,16 @@ std::shared_ptr<Executor> Scheduler::claimExecutor(Message& msg)
     auto& warm = executors[key];
 
     // Warm executor reuse (reference: src/scheduler/Scheduler.cpp:339-387)
-    for (auto& e : warm) {
-        if (e->tryClaim()) {
-            return e;
+    // starting at the rotating hint (see scheduler.h)
+    size_t n = warm.size();
+    if (n > 0) {
+        size_t& hint = claimHints[key];
+        for (size_t k = 0; k < n; k++) {
+            size_t i = (hint + k) % n;
+            if (warm[i]->tryClaim()) {
+                hint = (i + 1) % n;
+                return warm[i];
+            }
         }
     }
     auto exec = getExecutorFactory()->createExecutor(msg);
@@ -288,6 +297,7 @@ void Scheduler::shutdown()
         }
     }
     executors.clear();
+    claimHints.clear();
 }
 
 void Scheduler::reset()

@@ -53,9 +53,14 @@ StateKeyValue::StateKeyValue(std::string userIn,
 
 StateKeyValue::~StateKeyValue()
 {
-    if (kvStream != nullptr) {
-        (void)hipStreamSynchronize((hipStream_t)kvStream);
-        (void)hipStreamDestroy((hipStream_t)kvStream);
+    if (stripes) {
+        for (int s = 0; s < KV_STRIPES; s++) {
+            if (stripes[s].stream != nullptr) {
+                (void)hipStreamSynchronize(
+                  (hipStream_t)stripes[s].stream);
+                (void)hipStreamDestroy((hipStream_t)stripes[s].stream);
+            }
+        }
     }
     if (mirror != nullptr) {
         (void)hipHostFree(mirror);
@@ -93,32 +98,80 @@ bool StateKeyValue::mirrorUsable()
         return false;
     }
     void* pinned = nullptr;
-    hipStream_t s = nullptr;
     if (hipSetDevice(device) != hipSuccess ||
         hipHostMalloc(&pinned, valueSize, hipHostMallocDefault) !=
           hipSuccess) {
         mirrorFailed = true;
         return false;
     }
-    if (hipStreamCreateWithFlags(&s, hipStreamNonBlocking) != hipSuccess) {
-        (void)hipHostFree(pinned);
-        mirrorFailed = true;
-        return false;
+    auto s = std::make_unique<MirrorStripe[]>(KV_STRIPES);
+    for (int i = 0; i < KV_STRIPES; i++) {
+        hipStream_t st = nullptr;
+        if (hipStreamCreateWithFlags(&st, hipStreamNonBlocking) !=
+            hipSuccess) {
+            for (int j = 0; j < i; j++) {
+                (void)hipStreamDestroy((hipStream_t)s[j].stream);
+            }
+            (void)hipHostFree(pinned);
+            mirrorFailed = true;
+            return false;
+        }
+        s[i].stream = (void*)st;
     }
     mirrorValid.assign((valueSize + MIRROR_PAGE - 1) / MIRROR_PAGE, 0);
-    kvStream = (void*)s;
+    stripes = std::move(s);
     mirror = (uint8_t*)pinned;
     return true;
 }
 
-void StateKeyValue::mirrorFill(uint64_t offset, size_t len)
+template<typename Fn>
+void StateKeyValue::forEachStripeRange(uint64_t offset, size_t len, Fn&& fn)
 {
-    // Caller holds mirrorMx. Fill whole pages covering [offset, +len)
-    // whose mirror copy is stale. The D2H rides the KV stream, so it is
-    // ordered after any pending write-through H2Ds.
+    uint64_t pos = offset;
+    uint64_t end = offset + len;
+    while (pos < end) {
+        uint64_t blockEnd =
+          (pos / KV_STRIPE_BLOCK + 1) * KV_STRIPE_BLOCK;
+        uint64_t n = std::min(end, blockEnd) - pos;
+        int s = stripeOf(pos);
+        std::lock_guard<std::mutex> lock(stripes[s].mx);
+        fn(s, pos, (size_t)n);
+        pos += n;
+    }
+}
+
+// Runtime stripe count (1..KV_STRIPES) for A/B measurement
+static int kvStripeCount()
+{
+    static const int v = []() {
+        int n = getEnvVarInt("FAABRIC_KV_STRIPES", StateKeyValue::KV_STRIPES);
+        if (n < 1) {
+            n = 1;
+        }
+        if (n > StateKeyValue::KV_STRIPES) {
+            n = StateKeyValue::KV_STRIPES;
+        }
+        return n;
+    }();
+    return v;
+}
+
+int StateKeyValue::stripeOf(uint64_t offset) const
+{
+    return (int)((offset / KV_STRIPE_BLOCK) % kvStripeCount());
+}
+
+void StateKeyValue::mirrorFillLocked(int stripe,
+                                     uint64_t offset,
+                                     size_t len)
+{
+    // Caller holds the stripe's mutex; [offset, +len) lies inside this
+    // stripe's block. Fill whole stale pages. The D2H rides the stripe
+    // stream, so it is ordered after that range's pending H2Ds.
     size_t firstPage = offset / MIRROR_PAGE;
     size_t lastPage = (offset + len - 1) / MIRROR_PAGE;
     size_t run = 0;
+    bool filled = false;
     for (size_t p = firstPage; p <= lastPage + 1; p++) {
         bool stale = p <= lastPage && mirrorValid[p] == 0;
         if (stale) {
@@ -130,20 +183,16 @@ void StateKeyValue::mirrorFill(uint64_t offset, size_t len)
             size_t n = std::min(run * MIRROR_PAGE, valueSize - start);
             (void)hipMemcpyAsync(mirror + start, devPtr + start, n,
                                  hipMemcpyDeviceToHost,
-                                 (hipStream_t)kvStream);
+                                 (hipStream_t)stripes[stripe].stream);
             for (size_t q = p - run; q < p; q++) {
                 mirrorValid[q] = 1;
             }
+            filled = true;
             run = 0;
         }
     }
-    (void)hipStreamSynchronize((hipStream_t)kvStream);
-}
-
-void StateKeyValue::mirrorFlushLocked()
-{
-    if (kvStream != nullptr) {
-        (void)hipStreamSynchronize((hipStream_t)kvStream);
+    if (filled) {
+        (void)hipStreamSynchronize((hipStream_t)stripes[stripe].stream);
     }
 }
 
@@ -152,13 +201,15 @@ void StateKeyValue::mirrorInvalidate(uint64_t offset, size_t len)
     if (mirror == nullptr || len == 0) {
         return;
     }
-    std::lock_guard<std::mutex> lock(mirrorMx);
-    size_t firstPage = offset / MIRROR_PAGE;
-    size_t lastPage = (offset + len - 1) / MIRROR_PAGE;
-    for (size_t p = firstPage; p <= lastPage && p < mirrorValid.size();
-         p++) {
-        mirrorValid[p] = 0;
-    }
+    forEachStripeRange(offset, len, [&](int s, uint64_t o, size_t n) {
+        (void)s;
+        size_t firstPage = o / MIRROR_PAGE;
+        size_t lastPage = (o + n - 1) / MIRROR_PAGE;
+        for (size_t p = firstPage;
+             p <= lastPage && p < mirrorValid.size(); p++) {
+            mirrorValid[p] = 0;
+        }
+    });
 }
 
 void StateKeyValue::sync()
@@ -166,8 +217,10 @@ void StateKeyValue::sync()
     if (mirror == nullptr) {
         return;
     }
-    std::lock_guard<std::mutex> lock(mirrorMx);
-    mirrorFlushLocked();
+    for (int s = 0; s < KV_STRIPES; s++) {
+        std::lock_guard<std::mutex> lock(stripes[s].mx);
+        (void)hipStreamSynchronize((hipStream_t)stripes[s].stream);
+    }
 }
 
 // Per-thread HIP stream so concurrent chunk ops from different executors
@@ -196,20 +249,12 @@ void StateKeyValue::readLocal(uint64_t offset, uint8_t* out, size_t len)
 {
     if (onDevice) {
         if (mirrorUsable()) {
-            std::lock_guard<std::mutex> lock(mirrorMx);
-            bool allValid = true;
-            for (size_t p = offset / MIRROR_PAGE;
-                 p <= (offset + len - 1) / MIRROR_PAGE; p++) {
-                if (mirrorValid[p] == 0) {
-                    allValid = false;
-                    break;
-                }
-            }
-            if (!allValid) {
-                (void)hipSetDevice(device);
-                mirrorFill(offset, len);
-            }
-            std::memcpy(out, mirror + offset, len);
+            (void)hipSetDevice(device);
+            forEachStripeRange(
+              offset, len, [&](int s, uint64_t o, size_t n) {
+                  mirrorFillLocked(s, o, n); // no-op when pages valid
+                  std::memcpy(out + (o - offset), mirror + o, n);
+              });
             return;
         }
         (void)hipSetDevice(device);
@@ -234,35 +279,40 @@ void StateKeyValue::writeLocal(uint64_t offset,
     if (onDevice) {
         if (mirrorUsable()) {
             // Group-commit write-through: memcpy into the pinned mirror,
-            // enqueue the H2D with no per-op sync (stream order keeps
-            // HBM newest-wins), mark the pages readable. sync() is the
-            // durability point.
-            std::lock_guard<std::mutex> lock(mirrorMx);
+            // enqueue the H2D on the stripe stream with no per-op sync
+            // (per-stripe stream order keeps HBM newest-wins), mark the
+            // pages readable. sync() is the durability point.
             (void)hipSetDevice(device);
-            // A partially-written invalid boundary page must be filled
-            // from HBM first, or its untouched bytes would read stale
-            size_t firstPage = offset / MIRROR_PAGE;
-            size_t lastPage = (offset + len - 1) / MIRROR_PAGE;
-            auto coversPage = [&](size_t p) {
-                uint64_t start = p * MIRROR_PAGE;
-                uint64_t end =
-                  std::min<uint64_t>(start + MIRROR_PAGE, valueSize);
-                return offset <= start && offset + len >= end;
-            };
-            if (mirrorValid[firstPage] == 0 && !coversPage(firstPage)) {
-                mirrorFill(firstPage * MIRROR_PAGE, 1);
-            }
-            if (lastPage != firstPage && mirrorValid[lastPage] == 0 &&
-                !coversPage(lastPage)) {
-                mirrorFill(lastPage * MIRROR_PAGE, 1);
-            }
-            std::memcpy(mirror + offset, data, len);
-            for (size_t p = firstPage; p <= lastPage; p++) {
-                mirrorValid[p] = 1;
-            }
-            (void)hipMemcpyAsync(devPtr + offset, mirror + offset, len,
-                                 hipMemcpyHostToDevice,
-                                 (hipStream_t)kvStream);
+            forEachStripeRange(
+              offset, len, [&](int s, uint64_t o, size_t n) {
+                  // A partially-written invalid boundary page must be
+                  // filled from HBM first, or its untouched bytes would
+                  // read stale
+                  size_t firstPage = o / MIRROR_PAGE;
+                  size_t lastPage = (o + n - 1) / MIRROR_PAGE;
+                  auto coversPage = [&](size_t p) {
+                      uint64_t start = p * MIRROR_PAGE;
+                      uint64_t end = std::min<uint64_t>(
+                        start + MIRROR_PAGE, valueSize);
+                      return o <= start && o + n >= end;
+                  };
+                  if (mirrorValid[firstPage] == 0 &&
+                      !coversPage(firstPage)) {
+                      mirrorFillLocked(s, firstPage * MIRROR_PAGE, 1);
+                  }
+                  if (lastPage != firstPage &&
+                      mirrorValid[lastPage] == 0 &&
+                      !coversPage(lastPage)) {
+                      mirrorFillLocked(s, lastPage * MIRROR_PAGE, 1);
+                  }
+                  std::memcpy(mirror + o, data + (o - offset), n);
+                  for (size_t p = firstPage; p <= lastPage; p++) {
+                      mirrorValid[p] = 1;
+                  }
+                  (void)hipMemcpyAsync(
+                    devPtr + o, mirror + o, n, hipMemcpyHostToDevice,
+                    (hipStream_t)stripes[s].stream);
+              });
             return;
         }
         (void)hipSetDevice(device);
