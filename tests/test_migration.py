@@ -215,6 +215,10 @@ def test_mpi_world_migration(cluster):
     msgs = ber.messages
     msgs[0].is_mpi = True
     msgs[0].mpi_world_size = n
+    # "slow" = 20 migration checks over ~5 s, so the window reliably
+    # covers the blockers' exit even on a loaded machine (a single
+    # check can fire before the blockers free the slots)
+    msgs[0].input_data = b"slow"
     ber.messages = msgs
     decision = _core.SchedulingDecision()
     decision.app_id = ber.app_id
