@@ -326,6 +326,7 @@ static int32_t exampleMigrate(Message& msg)
         // and re-enters with "resumed". The slow variant keeps checking
         // so an eviction notice arriving later still lands.
         int rounds = slow ? 20 : 1;
+        int32_t preGroup = msg.groupId;
         for (int r = 0; r < rounds; r++) {
             if (slow) {
                 usleep(250 * 1000);
@@ -334,6 +335,13 @@ static int32_t exampleMigrate(Message& msg)
               std::vector<uint8_t>{ 'r', 'e', 's', 'u', 'm', 'e', 'd' });
             if (rc != 0) {
                 return rc;
+            }
+            // A migrated peer re-enters PAST this loop and never checks
+            // again; staying ranks must stop checking too once the app
+            // has been re-placed (the group id changes with every
+            // scheduling event) or they'd wait forever for a verdict
+            if (msg.groupId != preGroup) {
+                break;
             }
         }
     }
