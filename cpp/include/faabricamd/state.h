@@ -137,12 +137,11 @@ class StateKeyValue
 
   private:
     static constexpr uint64_t KV_STRIPE_BLOCK = 64 * 1024;
-    struct MirrorStripe
-    {
-        std::mutex mx;
-        void* stream = nullptr;
-    };
-    std::unique_ptr<MirrorStripe[]> stripes; // [KV_STRIPES] when enabled
+    // Stripes (mutex + HIP stream) live in one PROCESS-GLOBAL pool
+    // shared by every KV: per-KV stream sets measured ~1 MiB+ of HBM
+    // settle per stream. Ordering still holds — every (kv, byte) maps
+    // to exactly one stripe.
+    bool stripesReady = false;
     std::mutex mirrorMx; // guards mirror bring-up only
     uint8_t* mirror = nullptr;
     std::vector<char> mirrorValid; // per 4 KiB page (owned by its stripe)

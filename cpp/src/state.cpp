@@ -18,6 +18,38 @@ static std::string kvKeyOf(const std::string& user, const std::string& key)
     return user + "/" + key;
 }
 
+struct MirrorStripe
+{
+    std::mutex mx;
+    void* stream = nullptr;
+};
+
+// Process-global stripe pool shared by all device KVs
+static MirrorStripe gStripes[StateKeyValue::KV_STRIPES];
+static std::mutex gStripesMx;
+static bool gStripesReady = false;
+
+static bool ensureGlobalStripes(int device)
+{
+    std::lock_guard<std::mutex> lock(gStripesMx);
+    if (gStripesReady) {
+        return true;
+    }
+    (void)hipSetDevice(device);
+    for (int i = 0; i < StateKeyValue::KV_STRIPES; i++) {
+        if (gStripes[i].stream == nullptr) {
+            hipStream_t st = nullptr;
+            if (hipStreamCreateWithFlags(&st, hipStreamNonBlocking) !=
+                hipSuccess) {
+                return false;
+            }
+            gStripes[i].stream = (void*)st;
+        }
+    }
+    gStripesReady = true;
+    return true;
+}
+
 StateKeyValue::StateKeyValue(std::string userIn,
                              std::string keyIn,
                              size_t sizeIn,
@@ -53,13 +85,12 @@ StateKeyValue::StateKeyValue(std::string userIn,
 
 StateKeyValue::~StateKeyValue()
 {
-    if (stripes) {
+    if (stripesReady) {
+        // Streams are process-global; just drain writes targeting our
+        // HBM before it is freed
         for (int s = 0; s < KV_STRIPES; s++) {
-            if (stripes[s].stream != nullptr) {
-                (void)hipStreamSynchronize(
-                  (hipStream_t)stripes[s].stream);
-                (void)hipStreamDestroy((hipStream_t)stripes[s].stream);
-            }
+            std::lock_guard<std::mutex> lock(gStripes[s].mx);
+            (void)hipStreamSynchronize((hipStream_t)gStripes[s].stream);
         }
     }
     if (mirror != nullptr) {
@@ -104,22 +135,13 @@ bool StateKeyValue::mirrorUsable()
         mirrorFailed = true;
         return false;
     }
-    auto s = std::make_unique<MirrorStripe[]>(KV_STRIPES);
-    for (int i = 0; i < KV_STRIPES; i++) {
-        hipStream_t st = nullptr;
-        if (hipStreamCreateWithFlags(&st, hipStreamNonBlocking) !=
-            hipSuccess) {
-            for (int j = 0; j < i; j++) {
-                (void)hipStreamDestroy((hipStream_t)s[j].stream);
-            }
-            (void)hipHostFree(pinned);
-            mirrorFailed = true;
-            return false;
-        }
-        s[i].stream = (void*)st;
+    if (!ensureGlobalStripes(device)) {
+        (void)hipHostFree(pinned);
+        mirrorFailed = true;
+        return false;
     }
     mirrorValid.assign((valueSize + MIRROR_PAGE - 1) / MIRROR_PAGE, 0);
-    stripes = std::move(s);
+    stripesReady = true;
     mirror = (uint8_t*)pinned;
     return true;
 }
@@ -134,7 +156,7 @@ void StateKeyValue::forEachStripeRange(uint64_t offset, size_t len, Fn&& fn)
           (pos / KV_STRIPE_BLOCK + 1) * KV_STRIPE_BLOCK;
         uint64_t n = std::min(end, blockEnd) - pos;
         int s = stripeOf(pos);
-        std::lock_guard<std::mutex> lock(stripes[s].mx);
+        std::lock_guard<std::mutex> lock(gStripes[s].mx);
         fn(s, pos, (size_t)n);
         pos += n;
     }
@@ -158,7 +180,10 @@ static int kvStripeCount()
 
 int StateKeyValue::stripeOf(uint64_t offset) const
 {
-    return (int)((offset / KV_STRIPE_BLOCK) % kvStripeCount());
+    // Salt by KV identity so concurrent KVs do not all contend on the
+    // same stripes of the global pool
+    uint64_t salt = (uint64_t)(uintptr_t)this >> 6;
+    return (int)((salt + offset / KV_STRIPE_BLOCK) % kvStripeCount());
 }
 
 void StateKeyValue::mirrorFillLocked(int stripe,
@@ -183,7 +208,7 @@ void StateKeyValue::mirrorFillLocked(int stripe,
             size_t n = std::min(run * MIRROR_PAGE, valueSize - start);
             (void)hipMemcpyAsync(mirror + start, devPtr + start, n,
                                  hipMemcpyDeviceToHost,
-                                 (hipStream_t)stripes[stripe].stream);
+                                 (hipStream_t)gStripes[stripe].stream);
             for (size_t q = p - run; q < p; q++) {
                 mirrorValid[q] = 1;
             }
@@ -192,7 +217,7 @@ void StateKeyValue::mirrorFillLocked(int stripe,
         }
     }
     if (filled) {
-        (void)hipStreamSynchronize((hipStream_t)stripes[stripe].stream);
+        (void)hipStreamSynchronize((hipStream_t)gStripes[stripe].stream);
     }
 }
 
@@ -218,8 +243,8 @@ void StateKeyValue::sync()
         return;
     }
     for (int s = 0; s < KV_STRIPES; s++) {
-        std::lock_guard<std::mutex> lock(stripes[s].mx);
-        (void)hipStreamSynchronize((hipStream_t)stripes[s].stream);
+        std::lock_guard<std::mutex> lock(gStripes[s].mx);
+        (void)hipStreamSynchronize((hipStream_t)gStripes[s].stream);
     }
 }
 
@@ -311,7 +336,7 @@ void StateKeyValue::writeLocal(uint64_t offset,
                   }
                   (void)hipMemcpyAsync(
                     devPtr + o, mirror + o, n, hipMemcpyHostToDevice,
-                    (hipStream_t)stripes[s].stream);
+                    (hipStream_t)gStripes[s].stream);
               });
             return;
         }
