@@ -116,6 +116,73 @@ def submit_mpi_batch(user, func, world_size, timeout_ms=60_000):
     return wait_for_batch(ber.app_id, world_size, timeout_ms)
 
 
+def _dtype_matrix_payload(msg):
+    """Allreduce every datatype x op against a numpy model (reference:
+    op_reduce executes max/min/sum over int, uint64, double, long long —
+    src/mpi/MpiWorld.cpp:1266-1389; PROD comes free on this rebuild)."""
+    import struct as st
+
+    import numpy as np
+
+    from faabric_amd import _core as c
+
+    world_id, rank, size = c.mpi_init()
+    c.mpi_barrier(rank)
+
+    cases = [
+        (c.MpiDataType.INT32, "<4i", np.int32,
+         [rank + 1, -rank, rank * 7, 2]),
+        (c.MpiDataType.INT64, "<4q", np.int64,
+         [2**40 + rank, -(rank + 1), rank, 3]),
+        (c.MpiDataType.UINT64, "<4Q", np.uint64,
+         [2**50 + rank, rank + 1, 5, rank * 11]),
+        (c.MpiDataType.DOUBLE, "<4d", np.float64,
+         [rank + 0.5, -1.25 * rank, 3.0, rank * 0.125]),
+        (c.MpiDataType.FLOAT, "<4f", np.float32,
+         [rank + 0.5, 2.0, -float(rank), 1.5]),
+    ]
+    ops = [(c.MpiOp.SUM, np.add), (c.MpiOp.MAX, np.maximum),
+           (c.MpiOp.MIN, np.minimum), (c.MpiOp.PROD, np.multiply)]
+
+    for dtype, fmt, npt, mine in cases:
+        for op, npop in ops:
+            got = c.mpi_allreduce_bytes(
+                rank, st.pack(fmt, *[npt(v).item() for v in mine]),
+                dtype, op)
+            got_vals = np.array(st.unpack(fmt, got), dtype=npt)
+            # Model: fold every rank's contribution
+            acc = None
+            for r in range(size):
+                contrib = np.array(
+                    [npt(v).item() for v in [
+                        {0: r + 1, 1: -r, 2: r * 7, 3: 2}[i]
+                        if dtype == c.MpiDataType.INT32 else
+                        {0: 2**40 + r, 1: -(r + 1), 2: r, 3: 3}[i]
+                        if dtype == c.MpiDataType.INT64 else
+                        {0: 2**50 + r, 1: r + 1, 2: 5, 3: r * 11}[i]
+                        if dtype == c.MpiDataType.UINT64 else
+                        {0: r + 0.5, 1: -1.25 * r, 2: 3.0,
+                         3: r * 0.125}[i]
+                        if dtype == c.MpiDataType.DOUBLE else
+                        {0: r + 0.5, 1: 2.0, 2: -float(r), 3: 1.5}[i]
+                        for i in range(4)
+                    ]], dtype=npt)
+                acc = contrib if acc is None else npop(acc, contrib)
+            if not np.allclose(got_vals.astype(np.float64),
+                               acc.astype(np.float64)):
+                return 10 + int(dtype)
+    c.mpi_barrier(rank)
+    return 0
+
+
+def test_mpi_datatype_op_matrix(runtime):
+    _core.register_function("mpi", "dtypematrix", _dtype_matrix_payload)
+    results = submit_mpi_batch("mpi", "dtypematrix", WORLD_SIZE)
+    assert all(r.return_value == 0 for r in results), [
+        (r.mpi_rank, r.return_value) for r in results
+    ]
+
+
 def test_mpi_world_all_collectives(runtime):
     results = submit_mpi_batch("mpi", "alltests", WORLD_SIZE)
     assert len(results) == WORLD_SIZE
