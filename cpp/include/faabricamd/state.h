@@ -169,6 +169,9 @@ class StateKeyValue
     void pullRangeIpc(uint64_t offset, size_t len);
     void pushRangeIpc(uint64_t offset, size_t len);
     std::vector<char> dirtyChunks; // one flag per STATE_STREAM_CHUNK_SIZE
+    // Chunks already fetched from the master (reference: pulledMask) —
+    // a chunk is pulled at most once; local writes count as fresh
+    std::vector<char> pulledChunks;
     bool fullyPulled = false;
     std::vector<std::vector<uint8_t>> appendedValues;
 };

@@ -483,7 +483,40 @@ void StateKeyValue::getChunk(uint64_t offset, uint8_t* buffer, size_t len)
     if (offset + len > valueSize) {
         throw FaabricException("state chunk read out of bounds");
     }
+    // Lazy per-chunk pull: each streaming chunk is fetched from the
+    // master at most once (reference: StateKeyValue pulledMask,
+    // state/StateKeyValue.h:86-101); local writes keep chunks fresh
+    // via flagChunkDirty/pulledChunks marking below
     if (!isMaster()) {
+        bool needPull = false;
+        {
+            std::lock_guard<std::mutex> lock(kvMx);
+            if (fullyPulled) {
+                needPull = false;
+            } else {
+                if (pulledChunks.size() != dirtyChunks.size()) {
+                    pulledChunks.assign(dirtyChunks.size(), 0);
+                }
+                size_t first = offset / STATE_STREAM_CHUNK_SIZE;
+                size_t last =
+                  (offset + len - 1) / STATE_STREAM_CHUNK_SIZE;
+                for (size_t c = first; c <= last; c++) {
+                    if (pulledChunks[c] == 0 && dirtyChunks[c] == 0) {
+                        needPull = true;
+                    }
+                    pulledChunks[c] = 1;
+                }
+            }
+        }
+        if (!needPull) {
+            if (onDevice) {
+                readLocal(offset, buffer, len);
+            } else {
+                std::lock_guard<std::mutex> lock(kvMx);
+                readLocal(offset, buffer, len);
+            }
+            return;
+        }
         if (useIpcToMaster()) {
             try {
                 pullRangeIpc(offset, len);
@@ -527,6 +560,17 @@ void StateKeyValue::setChunk(uint64_t offset, const uint8_t* buffer,
         writeLocal(offset, buffer, len);
     }
     flagChunkDirty(offset, len);
+    {
+        // Locally-written chunks are fresh: never re-pull them
+        std::lock_guard<std::mutex> lock(kvMx);
+        if (pulledChunks.size() != dirtyChunks.size()) {
+            pulledChunks.assign(dirtyChunks.size(), 0);
+        }
+        for (size_t c = offset / STATE_STREAM_CHUNK_SIZE;
+             c <= (offset + len - 1) / STATE_STREAM_CHUNK_SIZE; c++) {
+            pulledChunks[c] = 1;
+        }
+    }
     if (!isMaster()) {
         if (useIpcToMaster()) {
             try {

@@ -125,8 +125,8 @@ def _recon_server(stop, ready, restart):
     core.set_log_level("error")
     core.set_port_offset(RECON_OFF)
     core.set_endpoint_host(f"127.0.0.1@{RECON_OFF}")
-    kv = core.state_get_kv("corner", "reconnect", 4096)
-    kv.set(b"\x42" * 4096)
+    kv = core.state_get_kv("corner", "reconnect", 256 * 1024)
+    kv.set(b"\x42" * (256 * 1024))
     srv = core.StateServerHandle()
     srv.start()
     ready.set()
@@ -156,13 +156,15 @@ def test_sync_client_reconnects_after_server_restart():
         _core.state_clear_all()
         _core.state_set_master_host("corner", "reconnect",
                                     f"127.0.0.1@{RECON_OFF}")
-        kv = _core.state_get_kv("corner", "reconnect", 4096)
+        kv = _core.state_get_kv("corner", "reconnect", 256 * 1024)
         assert kv.get_chunk(0, 16) == b"\x42" * 16  # connection cached
         ready.clear()
         restart.set()
         assert ready.wait(60)  # server is back
-        # Same client object, dead socket: must reconnect and succeed
-        assert kv.get_chunk(16, 16) == b"\x42" * 16
+        # Same client object, dead socket: must reconnect and succeed.
+        # Read a chunk NOT yet pulled (lazy pulledMask serves repeated
+        # reads of chunk 0 from cache) so an RPC actually happens.
+        assert kv.get_chunk(128 * 1024, 16) == b"\x42" * 16
     finally:
         stop.set()
         p.join(timeout=30)
@@ -190,7 +192,7 @@ def test_remote_error_propagates_with_host():
                                     f"127.0.0.1@{RECON_OFF}")
         kv = _core.state_get_kv("corner", "reconnect", 1 << 20)
         with pytest.raises(RuntimeError) as err:
-            kv.get_chunk((1 << 20) - 8, 8)  # out of the REAL 4 KiB value
+            kv.get_chunk((1 << 20) - 8, 8)  # out of the REAL 256 KiB value
         assert "remote error" in str(err.value)
         assert f"127.0.0.1@{RECON_OFF}" in str(err.value)
     finally:
