@@ -103,8 +103,12 @@ void Scheduler::executeBatch(std::shared_ptr<BatchExecuteRequest> req)
         size_t n = req->messages.size();
         auto dispatchOne = [&](size_t i) {
             try {
+                PROF_START(dispatch_claim)
                 auto exec = claimExecutor(req->messages[i]);
+                PROF_END(dispatch_claim)
+                PROF_START(dispatch_enqueue)
                 exec->executeTasks({ (int)i }, req);
+                PROF_END(dispatch_enqueue)
             } catch (const std::exception& e) {
                 // Failures set an error result instead of crashing the host
                 // (reference: src/scheduler/Scheduler.cpp:304-322)

@@ -268,13 +268,29 @@ thread_local std::map<std::string, std::chrono::steady_clock::time_point>
   profStarts;
 } // namespace
 
+// PROF aggregation is opt-in (FAABRIC_PROF=1) — the reference gates its
+// PROF macros behind TRACE_ALL for the same reason: map + mutex work on
+// every call is measurable in hot paths (3 pairs per message in the
+// batch path)
+static bool profEnabled()
+{
+    static const bool v = getEnvVarInt("FAABRIC_PROF", 0) != 0;
+    return v;
+}
+
 void profStart(const std::string& name)
 {
+    if (!profEnabled()) {
+        return;
+    }
     profStarts[name] = std::chrono::steady_clock::now();
 }
 
 void profEnd(const std::string& name)
 {
+    if (!profEnabled()) {
+        return;
+    }
     auto it = profStarts.find(name);
     if (it == profStarts.end()) {
         return;
