@@ -101,11 +101,21 @@ int32_t SchedulingDecision::removeMessage(int32_t messageId)
     size_t idx = (size_t)std::distance(messageIds.begin(), it);
     int32_t port = mpiPorts[idx];
     nFunctions--;
-    hosts.erase(hosts.begin() + idx);
-    messageIds.erase(messageIds.begin() + idx);
-    appIdxs.erase(appIdxs.begin() + idx);
-    groupIdxs.erase(groupIdxs.begin() + idx);
-    mpiPorts.erase(mpiPorts.begin() + idx);
+    // Swap-and-pop: results arrive per message and vector erases shift
+    // every later entry; position alignment with the in-flight request
+    // is preserved because Planner::setMessageResult removes the SAME
+    // message from both structures the same way
+    size_t last = messageIds.size() - 1;
+    hosts[idx] = std::move(hosts[last]);
+    messageIds[idx] = messageIds[last];
+    appIdxs[idx] = appIdxs[last];
+    groupIdxs[idx] = groupIdxs[last];
+    mpiPorts[idx] = mpiPorts[last];
+    hosts.pop_back();
+    messageIds.pop_back();
+    appIdxs.pop_back();
+    groupIdxs.pop_back();
+    mpiPorts.pop_back();
     return port;
 }
 

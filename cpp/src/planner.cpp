@@ -579,7 +579,12 @@ void Planner::setMessageResult(std::shared_ptr<Message> msg)
               req->messages.end(),
               [&](const Message& m) { return m.id == msgId; });
             if (msgIt != req->messages.end()) {
-                req->messages.erase(msgIt);
+                // Swap-and-pop (Message is fat; a mid-vector erase moves
+                // every later message — O(n^2) per completing batch).
+                // Alignment with the decision is preserved: its
+                // removeMessage swap-and-pops the same position.
+                *msgIt = std::move(req->messages.back());
+                req->messages.pop_back();
                 int32_t freedPort = decision->removeMessage(msgId);
                 if (hostIt != state.hostMap.end()) {
                     releaseHostMpiPort(hostIt->second, freedPort);
