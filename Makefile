@@ -26,7 +26,9 @@ CPP_OBJS := $(patsubst cpp/%.cpp,$(BUILD)/%.o,$(CPP_SRCS))
 HIP_OBJS := $(patsubst cpp/%.hip,$(BUILD)/%.o,$(HIP_SRCS))
 OBJS := $(CPP_OBJS) $(HIP_OBJS)
 
-all: $(TARGET)
+# Build the module AND the example binaries: the test suite runs the
+# binaries, and a stale build/ after a core change fails confusingly
+all: $(TARGET) examples
 
 $(BUILD)/%.o: cpp/%.cpp
 	@mkdir -p $(dir $@)

@@ -51,7 +51,7 @@ Scheduler& Scheduler::get()
 HostResources Scheduler::getThisHostResources()
 {
     {
-        std::lock_guard<std::mutex> lock(schedMx);
+        std::unique_lock<std::shared_mutex> lock(schedMx);
         if (resourcesOverridden) {
             return overriddenResources;
         }
@@ -66,7 +66,7 @@ HostResources Scheduler::getThisHostResources()
 
 void Scheduler::setThisHostResources(const HostResources& res)
 {
-    std::lock_guard<std::mutex> lock(schedMx);
+    std::unique_lock<std::shared_mutex> lock(schedMx);
     overriddenResources = res;
     resourcesOverridden = true;
 }
@@ -80,7 +80,7 @@ void Scheduler::executeBatch(std::shared_ptr<BatchExecuteRequest> req)
     bool isThreads = req->type == BatchExecuteType::THREADS;
 
     if (isTestMode()) {
-        std::lock_guard<std::mutex> lock(schedMx);
+        std::unique_lock<std::shared_mutex> lock(schedMx);
         for (const auto& msg : req->messages) {
             recordedMessages.push_back(msg);
         }
@@ -148,38 +148,61 @@ void Scheduler::executeBatch(std::shared_ptr<BatchExecuteRequest> req)
 
 std::shared_ptr<Executor> Scheduler::claimExecutor(Message& msg)
 {
-    std::lock_guard<std::mutex> lock(schedMx);
     std::string key = msg.user + "/" + msg.function;
-    auto& warm = executors[key];
 
-    // Warm executor reuse (reference: src/scheduler/Scheduler.cpp:339-387)
-    // starting at the rotating hint (see scheduler.h)
-    size_t n = warm.size();
-    if (n > 0) {
-        size_t& hint = claimHints[key];
-        for (size_t k = 0; k < n; k++) {
-            size_t i = (hint + k) % n;
-            if (warm[i]->tryClaim()) {
-                hint = (i + 1) % n;
-                return warm[i];
+    // Fast path: warm reuse under the SHARED lock — tryClaim is a CAS,
+    // so concurrent dispatchers scan in parallel
+    // (reference: src/scheduler/Scheduler.cpp:339-387)
+    {
+        std::shared_lock<std::shared_mutex> rlock(schedMx);
+        auto it = executors.find(key);
+        if (it != executors.end()) {
+            auto& pool = *it->second;
+            size_t n = pool.list.size();
+            if (n > 0) {
+                size_t start = pool.hint.load(std::memory_order_relaxed);
+                for (size_t k = 0; k < n; k++) {
+                    size_t i = (start + k) % n;
+                    if (pool.list[i]->tryClaim()) {
+                        pool.hint.store((i + 1) % n,
+                                        std::memory_order_relaxed);
+                        return pool.list[i];
+                    }
+                }
             }
+        }
+    }
+
+    // Slow path: grow the pool under the unique lock (re-scan first —
+    // another dispatcher may have released an executor meanwhile)
+    std::unique_lock<std::shared_mutex> lock(schedMx);
+    auto& poolPtr = executors[key];
+    if (!poolPtr) {
+        poolPtr = std::make_unique<WarmPool>();
+    }
+    auto& pool = *poolPtr;
+    for (auto& e : pool.list) {
+        if (e->tryClaim()) {
+            return e;
         }
     }
     auto exec = getExecutorFactory()->createExecutor(msg);
     exec->claim();
     if (getNumGpus() > 0) {
-        exec->gpuDevice = (int)((warm.size()) % (size_t)getNumGpus());
+        exec->gpuDevice =
+          (int)((pool.list.size()) % (size_t)getNumGpus());
     }
-    warm.push_back(exec);
+    pool.list.push_back(exec);
     return exec;
 }
 
 int Scheduler::reapStaleExecutors()
 {
-    std::lock_guard<std::mutex> lock(schedMx);
+    std::unique_lock<std::shared_mutex> lock(schedMx);
     const auto& conf = getSystemConfig();
     int reaped = 0;
-    for (auto& [key, list] : executors) {
+    for (auto& [key, pool] : executors) {
+        auto& list = pool->list;
         for (auto it = list.begin(); it != list.end();) {
             auto& exec = *it;
             if (!exec->isClaimed() &&
@@ -281,9 +304,9 @@ std::shared_ptr<PendingMigration> Scheduler::checkForMigrationOpportunities(
 
 void Scheduler::flushLocally()
 {
-    std::lock_guard<std::mutex> lock(schedMx);
-    for (auto& [key, list] : executors) {
-        for (auto& e : list) {
+    std::unique_lock<std::shared_mutex> lock(schedMx);
+    for (auto& [key, pool] : executors) {
+        for (auto& e : pool->list) {
             e->flush();
         }
     }
@@ -294,42 +317,41 @@ void Scheduler::flushLocally()
 void Scheduler::shutdown()
 {
     stopReaper();
-    std::lock_guard<std::mutex> lock(schedMx);
-    for (auto& [key, list] : executors) {
-        for (auto& e : list) {
+    std::unique_lock<std::shared_mutex> lock(schedMx);
+    for (auto& [key, pool] : executors) {
+        for (auto& e : pool->list) {
             e->shutdown();
         }
     }
     executors.clear();
-    claimHints.clear();
 }
 
 void Scheduler::reset()
 {
     shutdown();
-    std::lock_guard<std::mutex> lock(schedMx);
+    std::unique_lock<std::shared_mutex> lock(schedMx);
     recordedMessages.clear();
     resourcesOverridden = false;
 }
 
 std::vector<Message> Scheduler::getRecordedMessages()
 {
-    std::lock_guard<std::mutex> lock(schedMx);
+    std::unique_lock<std::shared_mutex> lock(schedMx);
     return recordedMessages;
 }
 
 void Scheduler::clearRecordedMessages()
 {
-    std::lock_guard<std::mutex> lock(schedMx);
+    std::unique_lock<std::shared_mutex> lock(schedMx);
     recordedMessages.clear();
 }
 
 size_t Scheduler::getExecutorCount()
 {
-    std::lock_guard<std::mutex> lock(schedMx);
+    std::unique_lock<std::shared_mutex> lock(schedMx);
     size_t n = 0;
-    for (auto& [key, list] : executors) {
-        n += list.size();
+    for (auto& [key, pool] : executors) {
+        n += pool->list.size();
     }
     return n;
 }
