@@ -274,8 +274,9 @@ void IpcReceiver::copyToDevice(const std::string& senderHost,
                                size_t size)
 {
     auto arena = find(senderHost);
-    if (offset + size > arena->size) {
-        // Offsets arrive off the wire; never read past the arena
+    // Offsets arrive off the wire; guard the sum against wrap too
+    if (offset > arena->size || size > arena->size ||
+        offset + size > arena->size) {
         throw FaabricException("ipc: segment out of arena bounds");
     }
     std::lock_guard<std::mutex> lock(arena->copyMx);
@@ -295,7 +296,9 @@ void IpcReceiver::copyToHost(const std::string& senderHost,
                              size_t size)
 {
     auto arena = find(senderHost);
-    if (offset + size > arena->size) {
+    // Offsets arrive off the wire; guard the sum against wrap too
+    if (offset > arena->size || size > arena->size ||
+        offset + size > arena->size) {
         throw FaabricException("ipc: segment out of arena bounds");
     }
     std::lock_guard<std::mutex> lock(arena->copyMx);

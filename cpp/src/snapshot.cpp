@@ -738,7 +738,8 @@ std::string SnapshotServer::doSyncRecv(uint8_t code, const std::string& body)
                 dsnap = std::make_shared<DeviceSnapshot>(rounded);
                 reg.registerSnapshot(c.key, dsnap);
             }
-            if (c.valOffset + c.len > dsnap->size()) {
+            if (c.valOffset > dsnap->size() || c.len > dsnap->size() ||
+                c.valOffset + c.len > dsnap->size()) {
                 throw FaabricException("snapshot ipc chunk out of range");
             }
             IpcReceiver::get().copyToDevice(

@@ -778,7 +778,8 @@ uint64_t StateKeyValue::serviceChunkIpc(const std::string& dstHost,
                                         uint64_t offset,
                                         size_t len)
 {
-    if (offset + len > valueSize) {
+    if (offset > valueSize || len > valueSize ||
+        offset + len > valueSize) {
         throw FaabricException("state ipc chunk out of bounds");
     }
     if (onDevice) {
@@ -796,7 +797,8 @@ void StateKeyValue::serviceSetIpc(const std::string& srcHost,
                                   uint64_t valOffset,
                                   size_t len)
 {
-    if (valOffset + len > valueSize) {
+    if (valOffset > valueSize || len > valueSize ||
+        valOffset + len > valueSize) {
         throw FaabricException("state ipc set out of bounds");
     }
     if (onDevice) {
