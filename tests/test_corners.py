@@ -212,6 +212,89 @@ def test_ptp_recv_times_out_cleanly():
 
 
 # ---------------------------------------------------------------------------
+# Failure detection: a worker that dies without deregistering expires
+# (reference: Planner host keep-alive + isHostExpired, Planner.cpp:166)
+# ---------------------------------------------------------------------------
+
+EXP_OFF = 8400
+
+
+def _exp_planner(stop, ready):
+    sys.path.insert(0, REPO_ROOT)
+    os.environ["PLANNER_HOST_KEEPALIVE_MS"] = "700"
+    from faabric_amd import _core as core
+    from faabric_amd.runtime import LocalRuntime
+
+    core.set_log_level("error")
+    rt = LocalRuntime(port_offset=EXP_OFF)
+    rt.start_planner(with_snapshot_server=False)
+    ready.set()
+    stop.wait(120)
+    rt.stop()
+
+
+def _exp_worker(ready):
+    sys.path.insert(0, REPO_ROOT)
+    from faabric_amd import _core as core
+    from faabric_amd.runtime import LocalRuntime
+
+    core.set_log_level("error")
+    rt = LocalRuntime(port_offset=EXP_OFF + 100,
+                      planner_port_offset=EXP_OFF, slots=2)
+    rt.start_worker()
+    ready.set()
+    import time as t
+
+    t.sleep(120)  # killed by the parent before this elapses
+
+
+def test_dead_worker_expires_from_membership():
+    import time
+
+    ctx = mp.get_context("spawn")
+    stop = ctx.Event()
+    p_ready = ctx.Event()
+    w_ready = ctx.Event()
+    planner = ctx.Process(target=_exp_planner, args=(stop, p_ready))
+    planner.start()
+    worker = None
+    try:
+        assert p_ready.wait(60)
+        prev = _core.get_endpoint_host()
+        _core.set_planner_host(f"127.0.0.1@{EXP_OFF}")
+        worker = ctx.Process(target=_exp_worker, args=(w_ready,))
+        worker.start()
+        assert w_ready.wait(60)
+        deadline = time.monotonic() + 10
+        while time.monotonic() < deadline:
+            if len(_core.get_available_hosts()) == 1:
+                break
+            time.sleep(0.05)
+        assert len(_core.get_available_hosts()) == 1
+
+        # Kill the worker without any deregistration
+        worker.kill()
+        worker.join(timeout=10)
+
+        # The planner must expire it once the keep-alive window lapses
+        deadline = time.monotonic() + 15
+        while time.monotonic() < deadline:
+            if len(_core.get_available_hosts()) == 0:
+                break
+            time.sleep(0.1)
+        assert len(_core.get_available_hosts()) == 0
+    finally:
+        stop.set()
+        planner.join(timeout=30)
+        if planner.is_alive():
+            planner.terminate()
+        if worker is not None and worker.is_alive():
+            worker.terminate()
+        _core.set_planner_host("127.0.0.1")
+        _core.set_endpoint_host(prev)
+
+
+# ---------------------------------------------------------------------------
 # Decoder fuzz: random + truncated buffers must never crash
 # ---------------------------------------------------------------------------
 

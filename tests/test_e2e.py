@@ -3,6 +3,7 @@ through the full RPC path (the reference's examples/check.cpp +
 tests/dist/scheduler/test_funcs.cpp coverage, single-host part)."""
 
 import threading
+import time
 
 import pytest
 
@@ -152,6 +153,48 @@ def test_ptp_group_barrier_and_lock(runtime):
         t.join(timeout=15)
     assert sorted(passed) == [0, 1, 2]
     assert counter["v"] == 3
+
+
+def test_ptp_group_recursive_lock(runtime):
+    """Recursive group lock: re-acquisition by the holder nests; a
+    waiter gets it only after full unwind (reference:
+    PointToPointBroker LOCK_GROUP_RECURSIVE handling)."""
+    decision = _core.SchedulingDecision()
+    decision.app_id = 556000
+    decision.group_id = 556001
+    decision.hosts = [runtime.identity] * 2
+    decision.message_ids = [1, 2]
+    decision.app_idxs = [0, 1]
+    decision.group_idxs = [0, 1]
+    decision.mpi_ports = [0, 0]
+    decision.n_functions = 2
+    _core.ptp_setup_local_mappings(decision)
+
+    order = []
+
+    def holder():
+        _core.ptp_group_lock(556001, 0, True)
+        _core.ptp_group_lock(556001, 0, True)  # nested re-acquire
+        order.append("held")
+        time.sleep(0.3)
+        _core.ptp_group_unlock(556001, 0, True)
+        order.append("partial-unwind")
+        time.sleep(0.2)
+        _core.ptp_group_unlock(556001, 0, True)
+
+    def waiter():
+        time.sleep(0.1)
+        _core.ptp_group_lock(556001, 1, True)
+        order.append("waiter-got-it")
+        _core.ptp_group_unlock(556001, 1, True)
+
+    t1 = threading.Thread(target=holder)
+    t2 = threading.Thread(target=waiter)
+    t1.start()
+    t2.start()
+    t1.join(timeout=15)
+    t2.join(timeout=15)
+    assert order == ["held", "partial-unwind", "waiter-got-it"]
 
 
 def test_state_local(runtime):
