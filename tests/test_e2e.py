@@ -171,19 +171,24 @@ def test_ptp_group_recursive_lock(runtime):
     _core.ptp_setup_local_mappings(decision)
 
     order = []
+    held = threading.Event()
+    waiting = threading.Event()
 
     def holder():
         _core.ptp_group_lock(556001, 0, True)
         _core.ptp_group_lock(556001, 0, True)  # nested re-acquire
         order.append("held")
-        time.sleep(0.3)
+        held.set()
+        waiting.wait(10)  # let the waiter block on the lock
+        time.sleep(0.2)
         _core.ptp_group_unlock(556001, 0, True)
         order.append("partial-unwind")
         time.sleep(0.2)
         _core.ptp_group_unlock(556001, 0, True)
 
     def waiter():
-        time.sleep(0.1)
+        held.wait(10)
+        waiting.set()
         _core.ptp_group_lock(556001, 1, True)
         order.append("waiter-got-it")
         _core.ptp_group_unlock(556001, 1, True)

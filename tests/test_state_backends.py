@@ -143,10 +143,12 @@ def test_lock_expiry(backend):
     from faabric_amd import _core
 
     user, key = "bk", f"exp-{backend}"
-    token = _core.state_acquire_lock(user, key, 150)
+    token = _core.state_acquire_lock(user, key, 400)
     assert token != 0
-    assert _core.state_acquire_lock(user, key, 150) == 0
-    time.sleep(0.3)
+    # Immediately contended (the 400 ms window gives slack on a loaded
+    # machine between these two calls)
+    assert _core.state_acquire_lock(user, key, 400) == 0
+    time.sleep(1.0)
     # Expired: a new holder can take it (reference Redis expiry semantics)
     token2 = _core.state_acquire_lock(user, key, 10_000)
     assert token2 != 0
