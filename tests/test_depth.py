@@ -255,6 +255,59 @@ def test_malformed_bodies(ep_runtime):
     assert status == 200
 
 
+def test_decision_cache_populates_and_clears(ep_runtime):
+    """NEW decisions land in the DecisionCache keyed (user, function,
+    size) — the reference's CACHED topology-hint store (reference:
+    src/batch-scheduler/DecisionCache.cpp)."""
+    _core.decision_cache_clear()
+    assert _core.decision_cache_size() == 0
+    ber = _core.batch_exec_factory("depth", "sleep", 2)
+    _core.call_functions(ber)
+    wait_for_batch(ber.app_id, 2, timeout_ms=30_000)
+    assert _core.decision_cache_size() == 1
+    # Same shape re-uses the same key; a different size adds an entry
+    ber2 = _core.batch_exec_factory("depth", "sleep", 2)
+    _core.call_functions(ber2)
+    wait_for_batch(ber2.app_id, 2, timeout_ms=30_000)
+    assert _core.decision_cache_size() == 1
+    ber3 = _core.batch_exec_factory("depth", "sleep", 3)
+    _core.call_functions(ber3)
+    wait_for_batch(ber3.app_id, 3, timeout_ms=30_000)
+    assert _core.decision_cache_size() == 2
+    _core.decision_cache_clear()
+    assert _core.decision_cache_size() == 0
+
+
+FLUSH_AVAILABLE_HOSTS = 2
+FLUSH_EXECUTORS = 3
+
+
+def test_flush_executors_op(ep_runtime):
+    """FLUSH_EXECUTORS drains every worker's warm pool (reference:
+    planner FLUSH_EXECUTORS -> FunctionCallServer flush)."""
+    ber = _core.batch_exec_factory("depth", "sleep", 2)
+    _core.call_functions(ber)
+    wait_for_batch(ber.app_id, 2, timeout_ms=30_000)
+    assert _core.get_executor_count() >= 2
+    status, _ = post(FLUSH_EXECUTORS)
+    assert status == 200
+    import time
+
+    deadline = time.monotonic() + 10
+    while time.monotonic() < deadline:
+        if _core.get_executor_count() == 0:
+            break
+        time.sleep(0.1)
+    assert _core.get_executor_count() == 0
+    # Flush also wipes loaded functions (reference flush semantics);
+    # re-register and confirm the host schedules again
+    _core.register_native_sleep("depth", "sleep", 400)
+    ber = _core.batch_exec_factory("depth", "sleep", 1)
+    _core.call_functions(ber)
+    results = wait_for_batch(ber.app_id, 1, timeout_ms=30_000)
+    assert results[0].return_value == 0
+
+
 def test_reset_clears_planner_state(ep_runtime):
     ber = _core.batch_exec_factory("depth", "sleep", 1)
     _core.call_functions(ber)
