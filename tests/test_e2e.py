@@ -67,6 +67,15 @@ def test_failing_function(runtime):
     assert "boom" in results[0].output_data
 
 
+def test_processes_batch_type(runtime):
+    """PROCESSES batches take the same path as FUNCTIONS (reference:
+    proto BatchExecuteType PROCESSES, src/proto/faabric.proto:35)."""
+    results = execute_batch("demo", "echo", 2, input_data=b"proc",
+                            batch_type=_core.BatchExecuteType.PROCESSES)
+    assert len(results) == 2
+    assert all(r.output_data == "proc" for r in results)
+
+
 def test_not_enough_slots(runtime):
     ber = _core.batch_exec_factory("demo", "noop", SLOTS + 1)
     decision = _core.call_functions(ber)
