@@ -340,6 +340,67 @@ def _snap_sender_proc(ready, done, result_q):
         done.set()
 
 
+def _planner_store_proc(stop, ready):
+    sys.path.insert(0, REPO_ROOT)
+    from faabric_amd import _core
+
+    _core.set_log_level("error")
+    _core.set_port_offset(5700)
+    _core.set_endpoint_host("127.0.0.1@5700")
+    srv = _core.StateServerHandle()
+    srv.start()
+    ready.set()
+    stop.wait(120)
+    srv.stop()
+
+
+@requires_gpu
+def test_planner_backed_device_kv():
+    """STATE_MODE=planner with an HBM-resident value: the worker's
+    device KV pushes/pulls against the planner's global store (the
+    Redis-service analog holds host bytes, like Redis did)."""
+    ctx = mp.get_context("spawn")
+    stop = ctx.Event()
+    ready = ctx.Event()
+    p = ctx.Process(target=_planner_store_proc, args=(stop, ready))
+    p.start()
+    try:
+        assert ready.wait(60)
+        from faabric_amd import _core
+
+        prev = _core.get_endpoint_host()
+        _core.set_state_mode("planner")
+        _core.set_planner_host("127.0.0.1@5700")
+        _core.set_endpoint_host("127.0.0.1@5800")
+        _core.state_clear_all()
+        kv = _core.state_get_kv_device("gk", "devval", 256 * 1024)
+        assert kv.on_device and not kv.is_master
+        kv.set(b"\x77" * (256 * 1024))  # lands in HBM AND pushes
+        # Drop the local replica; re-pull from the planner store into HBM
+        _core.state_clear_all()
+        kv2 = _core.state_get_kv_device("gk", "devval", 256 * 1024)
+        kv2.pull()
+        assert kv2.get_chunk(0, 64) == b"\x77" * 64
+        assert kv2.get_chunk(256 * 1024 - 64, 64) == b"\x77" * 64
+        # Partial push of a modified HBM range
+        kv2.set_chunk(1000, b"\x88" * 500)
+        kv2.push_partial()
+        _core.state_clear_all()
+        kv3 = _core.state_get_kv_device("gk", "devval", 256 * 1024)
+        assert kv3.get_chunk(990, 20) == (
+            b"\x77" * 10 + b"\x88" * 10
+        )
+    finally:
+        stop.set()
+        p.join(timeout=30)
+        if p.is_alive():
+            p.terminate()
+        _core.state_clear_all()
+        _core.set_state_mode("inmemory")
+        _core.set_planner_host("127.0.0.1")
+        _core.set_endpoint_host(prev)
+
+
 @requires_gpu
 def test_device_snapshot_streams_over_ipc():
     """An 8 MiB HBM snapshot ships worker-to-worker through a 4 MiB IPC
