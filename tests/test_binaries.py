@@ -185,7 +185,7 @@ def test_disttest_binary():
         env={**os.environ, "DISTTEST_BASE_OFFSET": "7400",
              "DISTTEST_STOP_FILE": "/tmp/disttest-pytest.stop",
              "LOG_LEVEL": "error"},
-        capture_output=True, text=True, timeout=180,
+        capture_output=True, text=True, timeout=300,
     )
     assert out.returncode == 0, (out.stdout[-300:], out.stderr[-500:])
     assert "DISTTEST OK" in out.stdout
