@@ -489,25 +489,31 @@ void StateKeyValue::getChunk(uint64_t offset, uint8_t* buffer, size_t len)
     // via flagChunkDirty/pulledChunks marking below
     if (!isMaster()) {
         bool needPull = false;
+        size_t first = offset / STATE_STREAM_CHUNK_SIZE;
+        size_t last = (offset + len - 1) / STATE_STREAM_CHUNK_SIZE;
         {
             std::lock_guard<std::mutex> lock(kvMx);
-            if (fullyPulled) {
-                needPull = false;
-            } else {
+            if (!fullyPulled) {
                 if (pulledChunks.size() != dirtyChunks.size()) {
                     pulledChunks.assign(dirtyChunks.size(), 0);
                 }
-                size_t first = offset / STATE_STREAM_CHUNK_SIZE;
-                size_t last =
-                  (offset + len - 1) / STATE_STREAM_CHUNK_SIZE;
                 for (size_t c = first; c <= last; c++) {
                     if (pulledChunks[c] == 0 && dirtyChunks[c] == 0) {
                         needPull = true;
+                        break;
                     }
-                    pulledChunks[c] = 1;
                 }
             }
         }
+        // NOTE: chunks are marked pulled only after a SUCCESSFUL pull
+        // below — marking first would turn a failed pull into silent
+        // stale reads forever after
+        auto markPulled = [&] {
+            std::lock_guard<std::mutex> lock(kvMx);
+            for (size_t c = first; c <= last; c++) {
+                pulledChunks[c] = 1;
+            }
+        };
         if (!needPull) {
             if (onDevice) {
                 readLocal(offset, buffer, len);
@@ -521,6 +527,7 @@ void StateKeyValue::getChunk(uint64_t offset, uint8_t* buffer, size_t len)
             try {
                 pullRangeIpc(offset, len);
                 readLocal(offset, buffer, len);
+                markPulled();
                 return;
             } catch (const std::exception& e) {
                 FAM_WARN("state ipc chunk pull failed (%s); falling back",
@@ -537,6 +544,7 @@ void StateKeyValue::getChunk(uint64_t offset, uint8_t* buffer, size_t len)
             writeLocal(offset, data.data(), data.size());
         }
         std::memcpy(buffer, data.data(), len);
+        markPulled();
         return;
     }
     if (onDevice) {
