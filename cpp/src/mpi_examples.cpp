@@ -325,7 +325,7 @@ static int32_t exampleMigrate(Message& msg)
         // broadcasts the verdict); a migrating/freezing rank unwinds here
         // and re-enters with "resumed". The slow variant keeps checking
         // so an eviction notice arriving later still lands.
-        int rounds = slow ? 20 : 1;
+        int rounds = slow ? 40 : 1;
         int32_t preGroup = msg.groupId;
         for (int r = 0; r < rounds; r++) {
             if (slow) {
