@@ -24,7 +24,7 @@ WARMUP = 2
 
 
 def rank_proc(rank, q):
-    os.environ["FAABRIC_IPC_ARENA_MB"] = "256"
+    os.environ["FAABRIC_IPC_ARENA_MB"] = "768"
     from faabric_amd import _core
     from faabric_amd.runtime import LocalRuntime, wait_for_batch
 
