@@ -102,6 +102,9 @@ struct SystemConfig
     // all-visible deployments set FAABRIC_GPU_DEVICE=rank instead so
     // same-node IPC maps peer GPUs over xGMI.
     int gpuDevice = 0;
+    // True when FAABRIC_GPU_DEVICE was set explicitly: this worker is
+    // pinned to one GPU and executors must not spread across devices
+    bool gpuDevicePinned = false;
     bool useGpu = true;        // slots are GPUs when available
 
     // Timeouts (seconds, matching reference defaults

@@ -272,8 +272,13 @@ std::shared_ptr<Executor> Scheduler::claimExecutor(Message& msg)
     auto exec = getExecutorFactory()->createExecutor(msg);
     exec->claim();
     if (getNumGpus() > 0) {
+        const auto& conf = getSystemConfig();
+        // A worker pinned to one GPU (FAABRIC_GPU_DEVICE) keeps every
+        // executor there; a multi-GPU single process spreads them
         exec->gpuDevice =
-          (int)((pool.list.size()) % (size_t)getNumGpus());
+          conf.gpuDevicePinned
+            ? conf.gpuDevice
+            : (int)((pool.list.size()) % (size_t)getNumGpus());
     }
     pool.list.push_back(exec);
     return exec;

@@ -135,6 +135,7 @@ void SystemConfig::initialise()
     overrideCpuCount = getEnvVarInt("OVERRIDE_CPU_COUNT", 0);
     overrideGpuCount = getEnvVarInt("OVERRIDE_GPU_COUNT", -1);
     gpuDevice = getEnvVarInt("FAABRIC_GPU_DEVICE", 0);
+    gpuDevicePinned = !getEnvVar("FAABRIC_GPU_DEVICE", "").empty();
     useGpu = getEnvVarInt("FAABRIC_USE_GPU", 1) != 0;
 
     globalMessageTimeout = getEnvVarInt("GLOBAL_MESSAGE_TIMEOUT", 60000);
