@@ -224,8 +224,9 @@ class PointToPointBroker
     std::map<int32_t, std::set<int32_t>> groupIdxs;
     std::map<int32_t, std::shared_ptr<FlagWaiter>> groupFlags;
 
-    std::mutex channelsMx;
-    std::map<int64_t, std::unique_ptr<Channel>> channels;
+    // Sharded: every local MPI message hits this map, one global
+    // mutex here was the old hot-path serialization point
+    ConcurrentMap<int64_t, std::shared_ptr<Channel>> channels;
 
     std::mutex sendSeqMx;
     std::map<int64_t, uint32_t> sendSeqs;

@@ -9,6 +9,7 @@
 // std::condition_variable throughout, no third-party lock-free deps.
 #pragma once
 
+#include <array>
 #include <atomic>
 #include <chrono>
 #include <condition_variable>
@@ -18,6 +19,7 @@
 #include <mutex>
 #include <stdexcept>
 #include <thread>
+#include <unordered_map>
 #include <vector>
 
 #include "faabricamd/util.h"
@@ -156,6 +158,191 @@ class FixedCapacityQueue
     std::vector<T> ring;
     alignas(64) std::atomic<size_t> head{ 0 };
     alignas(64) std::atomic<size_t> tail{ 0 };
+};
+
+// Bounded MPMC spin queue: per-slot sequence tickets (Vyukov scheme),
+// producers/consumers claim slots with fetch_add and spin until their
+// slot's ticket matches. No mutex, no cv — the low-latency local MPI
+// delivery option the reference enables with FAABRIC_USE_SPINLOCK
+// (reference: include/faabric/util/queue.h:220 SpinLockQueue on
+// atomic_queue; selection src/mpi/MpiWorld.h:29-33). Capacity is
+// rounded up to a power of two.
+template<typename T>
+class SpinLockQueue
+{
+  public:
+    explicit SpinLockQueue(size_t capacityIn)
+    {
+        if (capacityIn == 0) {
+            throw FaabricException("SpinLockQueue capacity must be > 0");
+        }
+        size_t c = 1;
+        while (c < capacityIn) {
+            c <<= 1;
+        }
+        mask = c - 1;
+        slots = std::make_unique<Slot[]>(c);
+        for (size_t i = 0; i < c; i++) {
+            slots[i].seq.store(i, std::memory_order_relaxed);
+        }
+    }
+
+    void enqueue(T value, int timeoutMs = DEFAULT_QUEUE_TIMEOUT_MS)
+    {
+        size_t pos = tail.fetch_add(1, std::memory_order_relaxed);
+        Slot& s = slots[pos & mask];
+        auto deadline = std::chrono::steady_clock::now() +
+                        std::chrono::milliseconds(timeoutMs);
+        // Slot is writable when its ticket equals our position
+        while (s.seq.load(std::memory_order_acquire) != pos) {
+            if (std::chrono::steady_clock::now() > deadline) {
+                throw QueueTimeoutException("spin enqueue timed out (full)");
+            }
+            std::this_thread::yield();
+        }
+        s.value = std::move(value);
+        s.seq.store(pos + 1, std::memory_order_release);
+    }
+
+    T dequeue(int timeoutMs = DEFAULT_QUEUE_TIMEOUT_MS)
+    {
+        size_t pos = head.fetch_add(1, std::memory_order_relaxed);
+        Slot& s = slots[pos & mask];
+        auto deadline = std::chrono::steady_clock::now() +
+                        std::chrono::milliseconds(timeoutMs);
+        while (s.seq.load(std::memory_order_acquire) != pos + 1) {
+            if (std::chrono::steady_clock::now() > deadline) {
+                throw QueueTimeoutException("spin dequeue timed out (empty)");
+            }
+            std::this_thread::yield();
+        }
+        T v = std::move(s.value);
+        // Ticket advances a full lap so the slot is writable at pos+cap
+        s.seq.store(pos + mask + 1, std::memory_order_release);
+        return v;
+    }
+
+    size_t size() const
+    {
+        size_t t = tail.load(std::memory_order_acquire);
+        size_t h = head.load(std::memory_order_acquire);
+        return t > h ? t - h : 0;
+    }
+
+  private:
+    struct Slot
+    {
+        alignas(64) std::atomic<size_t> seq{ 0 };
+        T value;
+    };
+    std::unique_ptr<Slot[]> slots;
+    size_t mask = 0;
+    alignas(64) std::atomic<size_t> head{ 0 };
+    alignas(64) std::atomic<size_t> tail{ 0 };
+};
+
+// Sharded concurrent hash map: key hash picks one of NSHARDS
+// mutex+unordered_map shards, so hot-path lookups from different keys
+// don't serialize on one lock (reference: util/concurrent_map.h, an
+// abseil-backed map — re-designed as plain sharding, no third-party
+// dependency).
+template<typename K, typename V, size_t NSHARDS = 16>
+class ConcurrentMap
+{
+    static_assert((NSHARDS & (NSHARDS - 1)) == 0, "power-of-two shards");
+
+  public:
+    // Returns a copy (value semantics keep the lock scope minimal);
+    // V is expected to be cheap to copy — use shared_ptr for big values
+    bool tryGet(const K& key, V& out) const
+    {
+        const Shard& s = shardFor(key);
+        std::lock_guard<std::mutex> lock(s.mx);
+        auto it = s.map.find(key);
+        if (it == s.map.end()) {
+            return false;
+        }
+        out = it->second;
+        return true;
+    }
+
+    // Insert-or-assign
+    void set(const K& key, V value)
+    {
+        Shard& s = shardFor(key);
+        std::lock_guard<std::mutex> lock(s.mx);
+        s.map[key] = std::move(value);
+    }
+
+    // Returns the existing value, or inserts make() and returns that.
+    // make() runs under the shard lock — keep it cheap.
+    template<typename F>
+    V getOrCreate(const K& key, F&& make)
+    {
+        Shard& s = shardFor(key);
+        std::lock_guard<std::mutex> lock(s.mx);
+        auto it = s.map.find(key);
+        if (it != s.map.end()) {
+            return it->second;
+        }
+        V v = make();
+        s.map.emplace(key, v);
+        return v;
+    }
+
+    bool erase(const K& key)
+    {
+        Shard& s = shardFor(key);
+        std::lock_guard<std::mutex> lock(s.mx);
+        return s.map.erase(key) > 0;
+    }
+
+    size_t size() const
+    {
+        size_t n = 0;
+        for (const auto& s : shards) {
+            std::lock_guard<std::mutex> lock(s.mx);
+            n += s.map.size();
+        }
+        return n;
+    }
+
+    void clear()
+    {
+        for (auto& s : shards) {
+            std::lock_guard<std::mutex> lock(s.mx);
+            s.map.clear();
+        }
+    }
+
+    // Visits every entry, one shard locked at a time (no global snapshot)
+    template<typename F>
+    void forEach(F&& fn) const
+    {
+        for (const auto& s : shards) {
+            std::lock_guard<std::mutex> lock(s.mx);
+            for (const auto& [k, v] : s.map) {
+                fn(k, v);
+            }
+        }
+    }
+
+  private:
+    struct Shard
+    {
+        mutable std::mutex mx;
+        std::unordered_map<K, V> map;
+    };
+    std::array<Shard, NSHARDS> shards;
+
+    Shard& shardFor(const K& key)
+    {
+        return shards[std::hash<K>{}(key) & (NSHARDS - 1)];
+    }
+    const Shard& shardFor(const K& key) const
+    {
+        return shards[std::hash<K>{}(key) & (NSHARDS - 1)];
+    }
 };
 
 class TokenPool

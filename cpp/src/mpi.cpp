@@ -5,6 +5,7 @@
 // the MI355X re-design notes (host plane on the PTP broker, device plane
 // on RCCL over xGMI).
 #include "faabricamd/mpi.h"
+#include "faabricamd/utilextras.h"
 #include "faabricamd/ops.h"
 #include "faabricamd/executor.h"
 #include "faabricamd/planner.h"
@@ -214,10 +215,21 @@ void MpiWorld::initialiseFromMsg(Message& msg)
 
 void MpiWorld::initialiseRankFromMsg(Message& msg)
 {
-    std::lock_guard<std::mutex> lock(worldMx);
-    if (std::find(localRanks.begin(), localRanks.end(), msg.mpiRank) ==
-        localRanks.end()) {
-        localRanks.push_back(msg.mpiRank);
+    {
+        std::lock_guard<std::mutex> lock(worldMx);
+        if (std::find(localRanks.begin(), localRanks.end(), msg.mpiRank) ==
+            localRanks.end()) {
+            localRanks.push_back(msg.mpiRank);
+        }
+    }
+    // Spin-poll mode burns the rank's core on purpose; pin it so the
+    // spinner and the scheduler don't migrate each other around
+    // (reference: src/mpi/MpiWorld.cpp:292-296 pins under
+    // FAABRIC_USE_SPINLOCK)
+    static const bool useSpin =
+      getEnvVarInt("FAABRIC_USE_SPINLOCK", 0) != 0;
+    if (useSpin) {
+        pinThreadToFreeCpu();
     }
 }
 

@@ -362,11 +362,10 @@ PointToPointBroker::Channel& PointToPointBroker::getChannel(int32_t groupId,
                                                             int32_t sendIdx,
                                                             int32_t recvIdx)
 {
-    std::lock_guard<std::mutex> lock(channelsMx);
-    auto& ch = channels[chanKey(groupId, sendIdx, recvIdx)];
-    if (!ch) {
-        ch = std::make_unique<Channel>();
-    }
+    auto ch = channels.getOrCreate(chanKey(groupId, sendIdx, recvIdx),
+                                   [] { return std::make_shared<Channel>(); });
+    // shared_ptr stays owned by the map for the broker's lifetime, so
+    // the reference is stable
     return *ch;
 }
 
@@ -507,7 +506,27 @@ PointToPointBroker::PtpPayload PointToPointBroker::recvPayload(
         return !ch.unorderedMsgs.empty() || !ch.bufferedMsgs.empty();
     };
 
-    if (!ch.cv.wait_for(lock, std::chrono::milliseconds(timeoutMs), ready)) {
+    // FAABRIC_USE_SPINLOCK: spin-poll instead of cv sleep — trades a
+    // burned core for wake-up latency on same-worker rank exchanges
+    // (reference: SpinLockQueue local MPI queues, src/mpi/MpiWorld.h:29-33).
+    // Default off: the cv path frees the core for other executors.
+    static const bool useSpin =
+      getEnvVarInt("FAABRIC_USE_SPINLOCK", 0) != 0;
+    if (useSpin) {
+        auto deadline = std::chrono::steady_clock::now() +
+                        std::chrono::milliseconds(timeoutMs);
+        while (!ready()) {
+            lock.unlock();
+            if (std::chrono::steady_clock::now() > deadline) {
+                throw QueueTimeoutException(
+                  "ptp recv timeout group " + std::to_string(groupId) + " " +
+                  std::to_string(sendIdx) + "->" + std::to_string(recvIdx));
+            }
+            std::this_thread::yield();
+            lock.lock();
+        }
+    } else if (!ch.cv.wait_for(
+                 lock, std::chrono::milliseconds(timeoutMs), ready)) {
         throw QueueTimeoutException(
           "ptp recv timeout group " + std::to_string(groupId) + " " +
           std::to_string(sendIdx) + "->" + std::to_string(recvIdx));

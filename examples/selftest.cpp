@@ -190,6 +190,66 @@ int main()
         CHECK(threw);
     }
 
+    // 5b. MPMC SpinLockQueue: 4 producers x 4 consumers
+    {
+        SpinLockQueue<int64_t> q(128);
+        std::atomic<int64_t> total{ 0 };
+        std::vector<std::thread> ts;
+        for (int c = 0; c < 4; c++) {
+            ts.emplace_back([&] {
+                for (int i = 0; i < 2500; i++) {
+                    total += q.dequeue(10000);
+                }
+            });
+        }
+        for (int p = 0; p < 4; p++) {
+            ts.emplace_back([&, p] {
+                for (int i = 0; i < 2500; i++) {
+                    q.enqueue(p * 2500 + i, 10000);
+                }
+            });
+        }
+        for (auto& t : ts) {
+            t.join();
+        }
+        CHECK(total.load() == 10000LL * 9999 / 2);
+        CHECK(q.size() == 0);
+    }
+
+    // 5c. ConcurrentMap under concurrent writers + getOrCreate races
+    {
+        ConcurrentMap<int, std::shared_ptr<int>> m;
+        std::vector<std::thread> ts;
+        std::atomic<int> creations{ 0 };
+        for (int t = 0; t < 8; t++) {
+            ts.emplace_back([&] {
+                for (int k = 0; k < 256; k++) {
+                    auto v = m.getOrCreate(k, [&] {
+                        creations++;
+                        return std::make_shared<int>(k * 7);
+                    });
+                    CHECK(*v == k * 7);
+                }
+            });
+        }
+        for (auto& t : ts) {
+            t.join();
+        }
+        CHECK(m.size() == 256);
+        // getOrCreate must have created each key exactly once
+        CHECK(creations.load() == 256);
+        std::shared_ptr<int> got;
+        CHECK(m.tryGet(100, got) && *got == 700);
+        CHECK(m.erase(100));
+        CHECK(!m.tryGet(100, got));
+        int visited = 0;
+        m.forEach([&](int k, const std::shared_ptr<int>& v) {
+            CHECK(*v == k * 7);
+            visited++;
+        });
+        CHECK(visited == 255);
+    }
+
     // 6. Snapshot diff/apply semantics
     {
         std::vector<uint8_t> base(8192, 0);
