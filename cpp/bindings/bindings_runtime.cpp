@@ -841,6 +841,33 @@ void initRuntimeBindings(py::module_& m)
         return py::bool_(ok);
     });
 
+    // Same check for the soft-dirty PTE tracker; returns None (skip)
+    // when the kernel lacks CONFIG_MEM_SOFT_DIRTY
+    m.def("_selftest_softpte_tracker", []() -> py::object {
+        if (!SoftPTEDirtyTracker::isAvailable()) {
+            return py::none();
+        }
+        bool ok;
+        {
+            py::gil_scoped_release release;
+            SoftPTEDirtyTracker tracker;
+            PageAlignedBuffer buf;
+            buf.resize(16 * 4096);
+            // Fault pages in BEFORE the reset so the dirty set below is
+            // from writes, not first-touch population
+            std::memset(buf.data(), 1, buf.size());
+            tracker.startTracking(buf.data(), buf.size());
+            buf.data()[5 * 4096 + 3] = 42;
+            buf.data()[12 * 4096] = 7;
+            tracker.stopTracking(buf.data(), buf.size());
+            auto dirty = tracker.getDirtyPages(buf.data(), buf.size());
+            // Soft-dirty may over-report (kernel can flag extra pages,
+            // e.g. on THP boundaries) but must include the two writes
+            ok = dirty.size() == 16 && dirty[5] == 1 && dirty[12] == 1;
+        }
+        return py::bool_(ok);
+    });
+
     // Native benchmark payloads (cpp/src/bench_funcs.cpp)
     m.def("set_bound_timeout",
           [](int ms) { getSystemConfig().boundTimeout = ms; });

@@ -94,6 +94,34 @@ class UffdDirtyTracker : public DirtyTracker
     static bool isAvailable();
 };
 
+// Kernel soft-dirty PTE bit: "4" to /proc/self/clear_refs resets the
+// bits, /proc/self/pagemap bit 55 reads them back (reference:
+// util/dirty.h:12-16,58-90 SoftPTEDirtyTracker). No handlers and no
+// mprotect, but process-wide: clear_refs resets EVERY mapping, so two
+// overlapping tracked regions are fine while an unrelated concurrent
+// tracker is not. Thread-local channel is empty (resolution is
+// process-wide) — the reference behaves the same way.
+class SoftPTEDirtyTracker : public DirtyTracker
+{
+  public:
+    SoftPTEDirtyTracker(); // throws if pagemap/clear_refs unusable
+    ~SoftPTEDirtyTracker() override;
+    std::string getType() const override { return "softpte"; }
+    void startTracking(uint8_t* region, size_t size) override;
+    void stopTracking(uint8_t* region, size_t size) override;
+    std::vector<char> getDirtyPages(uint8_t* region, size_t size) override;
+    void startThreadLocalTracking(uint8_t* region, size_t size) override;
+    void stopThreadLocalTracking(uint8_t* region, size_t size) override;
+    std::vector<char> getThreadLocalDirtyPages(uint8_t* region,
+                                               size_t size) override;
+
+    static bool isAvailable();
+
+  private:
+    int clearRefsFd = -1;
+    int pagemapFd = -1;
+};
+
 std::shared_ptr<DirtyTracker> getDirtyTracker();
 void resetDirtyTracker();
 

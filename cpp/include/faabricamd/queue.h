@@ -315,6 +315,25 @@ class ConcurrentMap
         }
     }
 
+    // Removes entries matching pred, one shard locked at a time
+    template<typename F>
+    size_t eraseIf(F&& pred)
+    {
+        size_t n = 0;
+        for (auto& s : shards) {
+            std::lock_guard<std::mutex> lock(s.mx);
+            for (auto it = s.map.begin(); it != s.map.end();) {
+                if (pred(it->first, it->second)) {
+                    it = s.map.erase(it);
+                    n++;
+                } else {
+                    ++it;
+                }
+            }
+        }
+        return n;
+    }
+
     // Visits every entry, one shard locked at a time (no global snapshot)
     template<typename F>
     void forEach(F&& fn) const

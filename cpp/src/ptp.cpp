@@ -748,16 +748,9 @@ void PointToPointBroker::clearGroup(int32_t groupId)
         }
         groupFlags.erase(groupId);
     }
-    {
-        std::lock_guard<std::mutex> lock(channelsMx);
-        for (auto it = channels.begin(); it != channels.end();) {
-            if ((int32_t)(it->first >> 32) == groupId) {
-                it = channels.erase(it);
-            } else {
-                ++it;
-            }
-        }
-    }
+    channels.eraseIf([groupId](int64_t key, const auto&) {
+        return (int32_t)(key >> 32) == groupId;
+    });
     {
         std::lock_guard<std::mutex> lock(sendSeqMx);
         for (auto it = sendSeqs.begin(); it != sendSeqs.end();) {
@@ -780,10 +773,7 @@ void PointToPointBroker::clear()
         groupIdxs.clear();
         groupFlags.clear();
     }
-    {
-        std::lock_guard<std::mutex> lock(channelsMx);
-        channels.clear();
-    }
+    channels.clear();
     {
         std::lock_guard<std::mutex> lock(sendSeqMx);
         sendSeqs.clear();

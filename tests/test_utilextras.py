@@ -79,3 +79,13 @@ def test_uffd_dirty_tracker():
     if result is None:
         pytest.skip("kernel lacks uffd write-protect")
     assert result
+
+
+def test_softpte_dirty_tracker():
+    """Soft-dirty PTE tracking via /proc/self/clear_refs + pagemap bit 55
+    (reference: util/dirty.h:58-90 SoftPTEDirtyTracker); skipped where the
+    kernel lacks CONFIG_MEM_SOFT_DIRTY."""
+    result = _core._selftest_softpte_tracker()
+    if result is None:
+        pytest.skip("kernel lacks soft-dirty PTE support")
+    assert result
