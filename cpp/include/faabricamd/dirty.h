@@ -1,5 +1,5 @@
 // Dirty tracking for HOST memory regions (reference:
-// include/faabric/util/dirty.h:24-225, src/util/dirty.cpp). Three modes,
+// include/faabric/util/dirty.h:24-225, src/util/dirty.cpp). Modes,
 // selected by DIRTY_TRACKING_MODE:
 //  - "compare"  (default): no tracking; diffs compare against the snapshot
 //               baseline (the honest mode for HBM, where no mprotect
@@ -10,9 +10,11 @@
 //  - "uffd":    userfaultfd write-protect mode — kernel-async fault
 //               delivery to a poller thread, no SIGSEGV involvement
 //               (reference UffdDirtyTracker, wp sub-mode)
-// The reference's soft-PTE tracker is not reproduced: this kernel ships
-// without CONFIG_MEM_SOFT_DIRTY (probed: /proc/self/clear_refs accepts
-// writes but pagemap bit 55 never sets), so it cannot be validated.
+//  - "softpte": kernel soft-dirty PTE bit via clear_refs/pagemap
+//               (reference SoftPTEDirtyTracker); availability is probed
+//               functionally — kernels without CONFIG_MEM_SOFT_DIRTY
+//               accept the clear_refs write but never set pagemap bit
+//               55, in which case it falls back to segfault tracking.
 #pragma once
 
 #include <cstdint>
