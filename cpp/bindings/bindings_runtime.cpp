@@ -869,6 +869,13 @@ void initRuntimeBindings(py::module_& m)
     });
 
     // Native benchmark payloads (cpp/src/bench_funcs.cpp)
+    m.def("wait_batch_done",
+          [](int32_t appId, int timeoutMs) {
+              py::gil_scoped_release release;
+              return getPlannerClient().waitBatchDone(appId, timeoutMs);
+          },
+          py::arg("app_id"),
+          py::arg("timeout_ms") = 30000);
     m.def("set_bound_timeout",
           [](int ms) { getSystemConfig().boundTimeout = ms; });
     m.def("reap_stale_executors", [] {

@@ -243,6 +243,10 @@ class MpiWorld
     // Migration support (reference: :2095-2132)
     void prepareMigration(int thisRank);
 
+    // Adopt a newer group id carried by a (re-)joining rank's message
+    // (migrated ranks can re-enter before stayed ranks refresh)
+    void refreshGroupFromMsg(Message& msg);
+
     // The RCCL communicator for this world's local rank (GPU path);
     // created lazily on first device-buffer collective
     void* getRcclComm(int rank);

@@ -39,6 +39,11 @@ enum class PlannerCalls : uint8_t
     GetNumMigrations = 12,
     CallBatch = 13,
     PreloadSchedulingDecision = 14,
+    // Event-driven batch completion: the caller registers once and the
+    // planner pushes BATCH_DONE to its function-call server when the
+    // last result lands — replaces status polling (which contended with
+    // result ingestion on the planner lock)
+    WaitBatchDone = 16,
 };
 
 // Number of MPI data-plane ports in each host's pool
@@ -63,6 +68,8 @@ struct PlannerState
     // msgId → hosts waiting for the result push
     std::map<int32_t, std::vector<std::string>> appResultWaiters;
     std::map<int32_t, std::shared_ptr<BatchExecuteRequest>> evictedRequests;
+    // appId → hosts to push BATCH_DONE to when the app's last result lands
+    std::map<int32_t, std::vector<std::string>> batchDoneWaiters;
     std::set<std::string> nextEvictedHostIps;
     int numMigrations = 0;
 };
@@ -79,6 +86,10 @@ class Planner
     bool registerHost(const Host& hostIn, bool overwrite);
     void removeHost(const Host& hostIn);
     std::vector<Host> getAvailableHosts();
+
+    // Returns true if the app is already complete (or unknown); else
+    // records host for a BATCH_DONE push on completion
+    bool registerBatchDoneWaiter(int32_t appId, const std::string& host);
 
     // --- scheduling ---
     std::shared_ptr<SchedulingDecision> callBatch(
@@ -181,6 +192,12 @@ class PlannerClient
     BatchExecuteRequestStatus getBatchResults(int32_t appId);
     // Lightweight poll: finished flag + result count only
     std::pair<bool, int> getBatchStatusCounts(int32_t appId);
+
+    // Block until appId's batch fully completes. Event-driven: registers
+    // with the planner for a BATCH_DONE push and sleeps on a local flag;
+    // falls back to a coarse status poll so a lost push cannot hang the
+    // caller. Returns false on timeout.
+    bool waitBatchDone(int32_t appId, int timeoutMs);
 
     SchedulingDecision getSchedulingDecision(int32_t appId);
     void preloadSchedulingDecision(

@@ -512,6 +512,16 @@ class FlagWaiter
         return flag;
     }
 
+    // Bounded wait that reports the flag state instead of throwing
+    bool waitMs(int ms)
+    {
+        std::unique_lock<std::mutex> lock(mx);
+        cv.wait_for(lock, std::chrono::milliseconds(ms), [this] {
+            return flag;
+        });
+        return flag;
+    }
+
   private:
     int timeoutMs;
     bool flag = false;

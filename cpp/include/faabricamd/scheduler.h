@@ -12,6 +12,7 @@
 
 #include "faabricamd/executor.h"
 #include "faabricamd/messages.h"
+#include "faabricamd/queue.h"
 #include "faabricamd/transport.h"
 
 namespace faabricamd {
@@ -22,6 +23,7 @@ enum class FunctionCalls : uint8_t
     ExecuteFunctions = 1,
     Flush = 2,
     SetMessageResult = 3,
+    BatchDone = 4,
 };
 
 class Scheduler
@@ -103,10 +105,18 @@ class FunctionCallClient : public MessageEndpointClient
     void executeFunctions(const BatchExecuteRequest& req);
     void setMessageResult(const Message& msg);
     void sendFlush();
+    void batchDone(int32_t appId);
 };
 
 std::shared_ptr<FunctionCallClient> getFunctionCallClient(
   const std::string& host);
+
+// Local BATCH_DONE flag registry (planner pushes, PlannerClient::
+// waitBatchDone sleeps). prepare() before registering with the planner
+// so a push can never be missed; discard() cleans up after the wait.
+std::shared_ptr<FlagWaiter> batchDoneWaiterPrepare(int32_t appId);
+void batchDoneWaiterDiscard(int32_t appId);
+void signalBatchDone(int32_t appId);
 void clearFunctionCallClients();
 
 // Mock-mode recording (reference: src/scheduler/FunctionCallClient.cpp:14-99)

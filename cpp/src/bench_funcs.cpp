@@ -279,15 +279,10 @@ static int32_t benchRankStep(Message& msg)
                 msg.outputData = "batch bench: not enough slots";
                 break;
             }
-            // Poll for completion (count-only status, cheap at 1024 msgs)
-            while (true) {
-                auto [finished, n] =
-                  getPlannerClient().getBatchStatusCounts(ber->appId);
-                if (finished && n >= total) {
-                    break;
-                }
-                usleep(200);
-            }
+            // Event-driven completion: planner pushes BATCH_DONE when
+            // the last result lands (no status polls contending with
+            // result ingestion on the planner lock)
+            getPlannerClient().waitBatchDone(ber->appId, 120000);
         }
         world.barrier(rank); // every host's batch functions returned
         // Group-commit: drain this host's pending HBM KV writes inside

@@ -1723,6 +1723,24 @@ std::map<std::string, int32_t> MpiWorld::getMsgCountDetails(int rank)
     return msgCounts[rank];
 }
 
+void MpiWorld::refreshGroupFromMsg(Message& msg)
+{
+    if (msg.groupId == 0) {
+        return;
+    }
+    {
+        std::lock_guard<std::mutex> lock(worldMx);
+        if (msg.groupId == groupId) {
+            return;
+        }
+    }
+    // New scheduling event renamed the group: wait for the new group's
+    // mappings (the planner distributes them with the dispatch), then
+    // adopt the new placement exactly like a stayed rank does
+    getPointToPointBroker().waitForMappingsOnThisHost(msg.groupId);
+    prepareMigration(msg.mpiRank);
+}
+
 void MpiWorld::prepareMigration(int thisRank)
 {
     // No pending async requests may be in flight
@@ -1774,6 +1792,7 @@ MpiWorld& MpiWorldRegistry::createWorld(Message& msg, int worldId)
 MpiWorld& MpiWorldRegistry::getOrInitialiseWorld(Message& msg)
 {
     std::shared_ptr<MpiWorld> world;
+    bool existed = false;
     {
         std::lock_guard<std::mutex> lock(mx);
         auto it = worlds.find(msg.mpiWorldId);
@@ -1783,7 +1802,16 @@ MpiWorld& MpiWorldRegistry::getOrInitialiseWorld(Message& msg)
             world->initialiseFromMsg(msg);
         } else {
             world = it->second;
+            existed = true;
         }
+    }
+    if (existed) {
+        // A migrated rank can land on a host whose world still carries
+        // the pre-migration group id (stayed ranks adopt the new one on
+        // their own schedule); refresh before any collective touches
+        // stale channels — otherwise this rank blocks forever on the
+        // old group while its peers talk on the new one
+        world->refreshGroupFromMsg(msg);
     }
     world->initialiseRankFromMsg(msg);
     return *world;

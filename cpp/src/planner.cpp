@@ -534,6 +534,7 @@ void Planner::setMessageResult(std::shared_ptr<Message> msg)
     }
 
     std::vector<std::string> waiters;
+    std::vector<std::string> batchWaiters;
     {
         std::unique_lock lock(plannerMx);
 
@@ -592,6 +593,11 @@ void Planner::setMessageResult(std::shared_ptr<Message> msg)
                 if (req->messages.empty()) {
                     state.inFlightReqs.erase(appId);
                     state.preloadedSchedulingDecisions.erase(appId);
+                    auto bwIt = state.batchDoneWaiters.find(appId);
+                    if (bwIt != state.batchDoneWaiters.end()) {
+                        batchWaiters = std::move(bwIt->second);
+                        state.batchDoneWaiters.erase(bwIt);
+                    }
                 }
             }
         }
@@ -607,6 +613,17 @@ void Planner::setMessageResult(std::shared_ptr<Message> msg)
         }
     }
 
+    // Batch fully complete: push BATCH_DONE to registered hosts
+    // (outside the lock)
+    for (const auto& host : batchWaiters) {
+        try {
+            getFunctionCallClient(host)->batchDone(appId);
+        } catch (const std::exception& e) {
+            FAM_ERROR(
+              "batch-done push to %s failed: %s", host.c_str(), e.what());
+        }
+    }
+
     // Push the result to hosts waiting on it (outside the lock)
     for (const auto& host : waiters) {
         try {
@@ -615,6 +632,19 @@ void Planner::setMessageResult(std::shared_ptr<Message> msg)
             FAM_ERROR("result push to %s failed: %s", host.c_str(), e.what());
         }
     }
+}
+
+bool Planner::registerBatchDoneWaiter(int32_t appId,
+                                      const std::string& host)
+{
+    std::unique_lock lock(plannerMx);
+    if (state.inFlightReqs.count(appId) == 0) {
+        // Completed (results in appResults) or unknown — either way the
+        // caller should check status, not wait for a push
+        return true;
+    }
+    state.batchDoneWaiters[appId].push_back(host);
+    return false;
 }
 
 std::shared_ptr<Message> Planner::getMessageResult(const Message& msg)
@@ -914,6 +944,26 @@ std::string PlannerServer::doSyncRecv(uint8_t code, const std::string& body)
             planner.removeHost(req.host);
             return {};
         }
+        case PlannerCalls::WaitBatchDone: {
+            PbReader r(body);
+            int32_t appId = 0;
+            std::string host;
+            uint32_t f;
+            WireType t;
+            while (r.next(f, t)) {
+                if (f == 1) {
+                    appId = r.asInt32();
+                } else if (f == 2) {
+                    host = r.asString();
+                } else {
+                    r.skip(t);
+                }
+            }
+            bool done = planner.registerBatchDoneWaiter(appId, host);
+            PbWriter w;
+            w.putBool(1, done);
+            return w.take();
+        }
         case PlannerCalls::SetMessageResult: {
             auto msg = std::make_shared<Message>(Message::decode(body));
             planner.setMessageResult(msg);
@@ -1159,6 +1209,63 @@ std::pair<bool, int> PlannerClient::getBatchStatusCounts(int32_t appId)
         return { false, -1 };
     }
     return { status.finished, status.expectedNumMessages };
+}
+
+bool PlannerClient::waitBatchDone(int32_t appId, int timeoutMs)
+{
+    // Local flag FIRST: a push that lands between the register RPC and
+    // the wait below still sets it
+    auto waiter = batchDoneWaiterPrepare(appId);
+
+    PbWriter w;
+    w.putInt32(1, appId);
+    w.putString(2, getSystemConfig().endpointHost);
+    std::string resp =
+      rpc.syncSend((uint8_t)PlannerCalls::WaitBatchDone, w.take());
+    bool already = false;
+    {
+        PbReader r(resp);
+        uint32_t f;
+        WireType t;
+        while (r.next(f, t)) {
+            if (f == 1) {
+                already = r.asBool();
+            } else {
+                r.skip(t);
+            }
+        }
+    }
+    if (already) {
+        batchDoneWaiterDiscard(appId);
+        return true;
+    }
+
+    // Sleep on the flag; wake every 500 ms for a status re-check so a
+    // lost push (e.g. planner restart) cannot hang the caller
+    auto deadline = std::chrono::steady_clock::now() +
+                    std::chrono::milliseconds(timeoutMs);
+    bool done = false;
+    while (!done) {
+        if (waiter->isSet()) {
+            done = true;
+            break;
+        }
+        if (std::chrono::steady_clock::now() > deadline) {
+            break;
+        }
+        if (waiter->waitMs(500)) {
+            done = true;
+            break;
+        }
+        auto [finished, n] = getBatchStatusCounts(appId);
+        (void)n;
+        if (finished) {
+            done = true;
+            break;
+        }
+    }
+    batchDoneWaiterDiscard(appId);
+    return done;
 }
 
 SchedulingDecision PlannerClient::getSchedulingDecision(int32_t appId)

@@ -11,6 +11,7 @@
 #include "faabricamd/planner.h"
 #include "faabricamd/ptp.h"
 #include "faabricamd/util.h"
+#include "faabricamd/wire.h"
 
 #include <hip/hip_runtime.h>
 
@@ -452,6 +453,41 @@ FunctionCallServer::FunctionCallServer()
                           "function-call")
 {}
 
+// ------------------- local BATCH_DONE flag registry -------------------------
+
+static std::mutex batchDoneMx;
+static std::map<int32_t, std::shared_ptr<FlagWaiter>> batchDoneFlags;
+
+std::shared_ptr<FlagWaiter> batchDoneWaiterPrepare(int32_t appId)
+{
+    std::lock_guard<std::mutex> lock(batchDoneMx);
+    auto& w = batchDoneFlags[appId];
+    if (!w) {
+        w = std::make_shared<FlagWaiter>();
+    }
+    return w;
+}
+
+void batchDoneWaiterDiscard(int32_t appId)
+{
+    std::lock_guard<std::mutex> lock(batchDoneMx);
+    batchDoneFlags.erase(appId);
+}
+
+void signalBatchDone(int32_t appId)
+{
+    std::shared_ptr<FlagWaiter> w;
+    {
+        std::lock_guard<std::mutex> lock(batchDoneMx);
+        auto it = batchDoneFlags.find(appId);
+        if (it == batchDoneFlags.end()) {
+            return; // push for a wait that already gave up
+        }
+        w = it->second;
+    }
+    w->setFlag(true);
+}
+
 void FunctionCallServer::doAsyncRecv(uint8_t code,
                                      const std::string& body,
                                      uint32_t seq)
@@ -467,6 +503,21 @@ void FunctionCallServer::doAsyncRecv(uint8_t code,
         case FunctionCalls::SetMessageResult: {
             auto msg = std::make_shared<Message>(Message::decode(body));
             getPlannerClient().setMessageResultLocally(msg);
+            break;
+        }
+        case FunctionCalls::BatchDone: {
+            PbReader r(body);
+            int32_t appId = 0;
+            uint32_t f;
+            WireType t;
+            while (r.next(f, t)) {
+                if (f == 1) {
+                    appId = (int32_t)r.varint();
+                } else {
+                    r.skip(t);
+                }
+            }
+            signalBatchDone(appId);
             break;
         }
         default:
@@ -526,6 +577,14 @@ void FunctionCallClient::setMessageResult(const Message& msg)
         return;
     }
     asyncSend((uint8_t)FunctionCalls::SetMessageResult, msg.encode());
+}
+
+void FunctionCallClient::batchDone(int32_t appId)
+{
+    PbWriter w;
+    w.putInt32(1, appId);
+    std::string body = w.take();
+    asyncSend((uint8_t)FunctionCalls::BatchDone, body.data(), body.size());
 }
 
 void FunctionCallClient::sendFlush()

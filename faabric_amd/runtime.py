@@ -104,6 +104,10 @@ def execute_batch(
 
 def wait_for_batch(app_id: int, count: int, timeout_ms: int = DEFAULT_TIMEOUT_MS):
     deadline = time.monotonic() + timeout_ms / 1000.0
+    # Event-driven: the planner pushes BATCH_DONE when the last result
+    # lands; wait_batch_done sleeps on a local flag (with its own coarse
+    # fallback poll), then one status fetch collects the results
+    _core.wait_batch_done(app_id, timeout_ms)
     while time.monotonic() < deadline:
         status = _core.get_batch_results(app_id)
         if (
