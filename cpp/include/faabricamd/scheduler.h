@@ -88,10 +88,15 @@ class Scheduler
     std::shared_ptr<ReaperThread> reaper;
 };
 
+// True when this process runs a FunctionCallServer (i.e. BATCH_DONE
+// pushes from the planner can actually reach it)
+bool functionCallServerRunning();
+
 class FunctionCallServer : public MessageEndpointServer
 {
   public:
     FunctionCallServer();
+    ~FunctionCallServer() override;
     void doAsyncRecv(uint8_t code,
                      const std::string& body,
                      uint32_t seq) override;

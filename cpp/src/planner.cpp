@@ -1213,6 +1213,26 @@ std::pair<bool, int> PlannerClient::getBatchStatusCounts(int32_t appId)
 
 bool PlannerClient::waitBatchDone(int32_t appId, int timeoutMs)
 {
+    // A process without a FunctionCallServer cannot receive the push —
+    // registering would stall the planner's result workers on a dead
+    // connection. Poll with a gentle backoff instead (matches the
+    // pre-push behavior for driver/test processes).
+    if (!functionCallServerRunning()) {
+        auto deadline = std::chrono::steady_clock::now() +
+                        std::chrono::milliseconds(timeoutMs);
+        int sleepUs = 200;
+        while (std::chrono::steady_clock::now() < deadline) {
+            auto [finished, n] = getBatchStatusCounts(appId);
+            (void)n;
+            if (finished) {
+                return true;
+            }
+            std::this_thread::sleep_for(std::chrono::microseconds(sleepUs));
+            sleepUs = std::min(sleepUs * 2, 20000);
+        }
+        return false;
+    }
+
     // Local flag FIRST: a push that lands between the register RPC and
     // the wait below still sets it
     auto waiter = batchDoneWaiterPrepare(appId);

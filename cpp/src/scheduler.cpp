@@ -447,11 +447,25 @@ size_t Scheduler::getExecutorCount()
 
 // ------------------------- RPC server / client ------------------------------
 
+static std::atomic<int> fnServersUp{ 0 };
+
+bool functionCallServerRunning()
+{
+    return fnServersUp.load(std::memory_order_relaxed) > 0;
+}
+
 FunctionCallServer::FunctionCallServer()
   : MessageEndpointServer(FUNCTION_CALL_ASYNC_PORT,
                           FUNCTION_CALL_SYNC_PORT,
                           "function-call")
-{}
+{
+    fnServersUp.fetch_add(1, std::memory_order_relaxed);
+}
+
+FunctionCallServer::~FunctionCallServer()
+{
+    fnServersUp.fetch_sub(1, std::memory_order_relaxed);
+}
 
 // ------------------- local BATCH_DONE flag registry -------------------------
 
