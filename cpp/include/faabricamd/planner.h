@@ -104,6 +104,8 @@ class Planner
 
     // --- results ---
     void setMessageResult(std::shared_ptr<Message> msg);
+    // Burst ingestion: one planner-lock acquisition for many results
+    void setMessageResults(std::vector<std::shared_ptr<Message>>& msgs);
     // nullptr when not (yet) available; registers msg.mainHost as a waiter
     std::shared_ptr<Message> getMessageResult(const Message& msg);
     // nullptr when the app is unknown
@@ -124,6 +126,11 @@ class Planner
     int hostTimeoutMs = 5000;
 
   private:
+    void setMessageResultLocked(
+      const std::shared_ptr<Message>& msg,
+      std::vector<std::pair<std::string, std::shared_ptr<Message>>>& waiters,
+      std::vector<std::pair<std::string, int32_t>>& batchWaiters);
+
     Planner();
 
     std::shared_mutex plannerMx;
@@ -180,6 +187,8 @@ class PlannerClient
       std::shared_ptr<BatchExecuteRequest> req);
 
     void setMessageResult(std::shared_ptr<Message> msg);
+    // Burst ingestion: one planner-lock acquisition for many results
+    void setMessageResults(std::vector<std::shared_ptr<Message>>& msgs);
     // One RPC carrying many results (see PlannerCalls::SetMessageResultBatch)
     void setMessageResultsBatch(
       const std::vector<std::shared_ptr<Message>>& msgs);

@@ -12,6 +12,7 @@
 #include "faabricamd/util.h"
 #include "faabricamd/wire.h"
 
+#include <string_view>
 #include <algorithm>
 #include <cassert>
 
@@ -524,6 +525,53 @@ std::shared_ptr<SchedulingDecision> Planner::getSchedulingDecision(
 
 void Planner::setMessageResult(std::shared_ptr<Message> msg)
 {
+    std::vector<std::shared_ptr<Message>> one;
+    one.push_back(std::move(msg));
+    setMessageResults(one);
+}
+
+// Ingest a burst of results under ONE planner-lock acquisition (the
+// result batcher ships ~15 per RPC; taking the unique lock per result
+// was the dominant serialized cost of a completing batch)
+void Planner::setMessageResults(
+  std::vector<std::shared_ptr<Message>>& msgs)
+{
+    // (host, msg) result pushes and (host, appId) batch-done pushes
+    // collected under the lock, sent after
+    std::vector<std::pair<std::string, std::shared_ptr<Message>>> waiters;
+    std::vector<std::pair<std::string, int32_t>> batchWaiters;
+    {
+        std::unique_lock lock(plannerMx);
+        for (auto& msg : msgs) {
+            setMessageResultLocked(msg, waiters, batchWaiters);
+        }
+    }
+
+    // Batch fully complete: push BATCH_DONE to registered hosts
+    for (const auto& [host, appId] : batchWaiters) {
+        try {
+            getFunctionCallClient(host)->batchDone(appId);
+        } catch (const std::exception& e) {
+            FAM_ERROR(
+              "batch-done push to %s failed: %s", host.c_str(), e.what());
+        }
+    }
+
+    // Push results to hosts waiting on them
+    for (const auto& [host, msg] : waiters) {
+        try {
+            getFunctionCallClient(host)->setMessageResult(*msg);
+        } catch (const std::exception& e) {
+            FAM_ERROR("result push to %s failed: %s", host.c_str(), e.what());
+        }
+    }
+}
+
+void Planner::setMessageResultLocked(
+  const std::shared_ptr<Message>& msg,
+  std::vector<std::pair<std::string, std::shared_ptr<Message>>>& waiters,
+  std::vector<std::pair<std::string, int32_t>>& batchWaiters)
+{
     int32_t appId = msg->appId;
     int32_t msgId = msg->id;
 
@@ -532,11 +580,7 @@ void Planner::setMessageResult(std::shared_ptr<Message> msg)
     if (msg->returnValue == MIGRATED_FUNCTION_RETURN_VALUE) {
         return;
     }
-
-    std::vector<std::string> waiters;
-    std::vector<std::string> batchWaiters;
     {
-        std::unique_lock lock(plannerMx);
 
         bool isFrozenMsg = msg->returnValue == FROZEN_FUNCTION_RETURN_VALUE;
         if (isFrozenMsg) {
@@ -595,7 +639,9 @@ void Planner::setMessageResult(std::shared_ptr<Message> msg)
                     state.preloadedSchedulingDecisions.erase(appId);
                     auto bwIt = state.batchDoneWaiters.find(appId);
                     if (bwIt != state.batchDoneWaiters.end()) {
-                        batchWaiters = std::move(bwIt->second);
+                        for (auto& h : bwIt->second) {
+                            batchWaiters.emplace_back(std::move(h), appId);
+                        }
                         state.batchDoneWaiters.erase(bwIt);
                     }
                 }
@@ -608,28 +654,10 @@ void Planner::setMessageResult(std::shared_ptr<Message> msg)
 
         auto wIt = state.appResultWaiters.find(msgId);
         if (wIt != state.appResultWaiters.end()) {
-            waiters = wIt->second;
+            for (auto& h : wIt->second) {
+                waiters.emplace_back(std::move(h), msg);
+            }
             state.appResultWaiters.erase(wIt);
-        }
-    }
-
-    // Batch fully complete: push BATCH_DONE to registered hosts
-    // (outside the lock)
-    for (const auto& host : batchWaiters) {
-        try {
-            getFunctionCallClient(host)->batchDone(appId);
-        } catch (const std::exception& e) {
-            FAM_ERROR(
-              "batch-done push to %s failed: %s", host.c_str(), e.what());
-        }
-    }
-
-    // Push the result to hosts waiting on it (outside the lock)
-    for (const auto& host : waiters) {
-        try {
-            getFunctionCallClient(host)->setMessageResult(*msg);
-        } catch (const std::exception& e) {
-            FAM_ERROR("result push to %s failed: %s", host.c_str(), e.what());
         }
     }
 }
@@ -867,12 +895,39 @@ void PlannerServer::resultWorkerLoop()
             continue;
         }
         try {
-            PROF_START(result_decode)
-            auto msg = std::make_shared<Message>(Message::decode(body));
-            PROF_END(result_decode)
-            PROF_START(result_core)
-            Planner::get().setMessageResult(msg);
-            PROF_END(result_core)
+            // Tag byte: 'S' single result, 'B' batch of results (one
+            // lock acquisition for the whole burst)
+            char tag = body[0];
+            std::string_view payload(body.data() + 1, body.size() - 1);
+            if (tag == 'B') {
+                PROF_START(result_decode)
+                std::vector<std::shared_ptr<Message>> msgs;
+                std::string payloadStr(payload);
+                PbReader r(payloadStr);
+                uint32_t f;
+                WireType t;
+                while (r.next(f, t)) {
+                    if (f != 1) {
+                        r.skip(t);
+                        continue;
+                    }
+                    auto sub = r.asString();
+                    msgs.push_back(
+                      std::make_shared<Message>(Message::decode(sub)));
+                }
+                PROF_END(result_decode)
+                PROF_START(result_core)
+                Planner::get().setMessageResults(msgs);
+                PROF_END(result_core)
+            } else {
+                PROF_START(result_decode)
+                auto msg = std::make_shared<Message>(
+                  Message::decode(std::string(payload)));
+                PROF_END(result_decode)
+                PROF_START(result_core)
+                Planner::get().setMessageResult(msg);
+                PROF_END(result_core)
+            }
         } catch (const std::exception& e) {
             FAM_ERROR("result ingestion failed: %s", e.what());
         }
@@ -889,28 +944,29 @@ void PlannerServer::doAsyncRecv(uint8_t code,
             auto msg = std::make_shared<Message>(Message::decode(body));
             Planner::get().setMessageResult(msg);
         } else {
-            resultQueue.enqueue(body);
+            resultQueue.enqueue("S" + body);
         }
         return;
     }
     if ((PlannerCalls)code == PlannerCalls::SetMessageResultBatch) {
-        // One RPC, many results: split and feed the same ingestion path
-        PbReader r(body);
-        uint32_t f;
-        WireType t;
-        while (r.next(f, t)) {
-            if (f != 1) {
-                r.skip(t);
-                continue;
+        // One RPC, many results: hand the whole burst to one worker so
+        // it ingests under a single planner-lock acquisition
+        if (resultWorkers.empty()) {
+            std::vector<std::shared_ptr<Message>> msgs;
+            PbReader r(body);
+            uint32_t f;
+            WireType t;
+            while (r.next(f, t)) {
+                if (f != 1) {
+                    r.skip(t);
+                    continue;
+                }
+                msgs.push_back(std::make_shared<Message>(
+                  Message::decode(r.asString())));
             }
-            std::string sub = r.asString();
-            if (resultWorkers.empty()) {
-                auto msg =
-                  std::make_shared<Message>(Message::decode(sub));
-                Planner::get().setMessageResult(msg);
-            } else {
-                resultQueue.enqueue(std::move(sub));
-            }
+            Planner::get().setMessageResults(msgs);
+        } else {
+            resultQueue.enqueue("B" + body);
         }
         return;
     }
