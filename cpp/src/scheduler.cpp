@@ -595,6 +595,9 @@ void FunctionCallClient::setMessageResult(const Message& msg)
 
 void FunctionCallClient::batchDone(int32_t appId)
 {
+    if (isMockMode()) {
+        return;
+    }
     PbWriter w;
     w.putInt32(1, appId);
     std::string body = w.take();

@@ -250,3 +250,32 @@ def test_executor_reaper(runtime):
         assert _core.get_executor_count() == 0
     finally:
         _core.set_bound_timeout(30_000)
+
+
+def test_wait_batch_done_event_path(runtime):
+    """Event-driven batch completion: the planner pushes BATCH_DONE to
+    the registered host's function-call server when an app's last result
+    lands (replaces status polling; see PlannerCalls::WaitBatchDone)."""
+    import time
+
+    ber = _core.batch_exec_factory("demo", "noop", SLOTS)
+    _core.call_functions(ber)
+    t0 = time.perf_counter()
+    assert _core.wait_batch_done(ber.app_id, 10_000)
+    elapsed = time.perf_counter() - t0
+    # An event (or the immediate already-done answer) must beat the
+    # 500 ms fallback poll by a wide margin
+    assert elapsed < 0.45, f"wait_batch_done fell back to polling: {elapsed}"
+    status = _core.get_batch_results(ber.app_id)
+    assert status.finished
+
+    # Already-completed app: returns immediately with True
+    t0 = time.perf_counter()
+    assert _core.wait_batch_done(ber.app_id, 10_000)
+    assert time.perf_counter() - t0 < 0.2
+
+    # Unknown app: the planner answers "not in flight" (treated as done,
+    # caller verifies via status — which reports unknown)
+    assert _core.wait_batch_done(987654321, 2_000)
+    unknown = _core.get_batch_results(987654321)
+    assert unknown.expected_num_messages == -1
