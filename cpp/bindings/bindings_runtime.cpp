@@ -869,6 +869,18 @@ void initRuntimeBindings(py::module_& m)
     });
 
     // Native benchmark payloads (cpp/src/bench_funcs.cpp)
+    // Runtime map sizes for leak hunting / observability
+    m.def("_debug_runtime_sizes", [] {
+        py::dict d;
+        d["planner_app_results"] = Planner::get().debugAppResultsCount();
+        d["planner_done_apps"] = Planner::get().debugDoneAppsCount();
+        d["planner_in_flight"] = Planner::get().debugInFlightCount();
+        d["broker_mappings"] = getPointToPointBroker().debugMappingsCount();
+        d["broker_channels"] = getPointToPointBroker().debugChannelsCount();
+        d["broker_send_seqs"] = getPointToPointBroker().debugSendSeqsCount();
+        d["decision_cache"] = DecisionCache::get().size();
+        return d;
+    });
     m.def("wait_batch_done",
           [](int32_t appId, int timeoutMs) {
               py::gil_scoped_release release;

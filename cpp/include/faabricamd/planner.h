@@ -6,6 +6,7 @@
 // slot is one GPU.
 #pragma once
 
+#include <deque>
 #include <atomic>
 #include <map>
 #include <memory>
@@ -65,6 +66,11 @@ struct PlannerState
       preloadedSchedulingDecisions;
     // appId → msgId → result
     std::map<int32_t, std::map<int32_t, std::shared_ptr<Message>>> appResults;
+    // Completed apps in completion order (epoch ms, appId): appResults
+    // entries are purged after FAABRIC_RESULT_TTL_MS or beyond
+    // FAABRIC_MAX_DONE_APPS, whichever hits first — without this a
+    // long-running planner leaks every result ever produced
+    std::deque<std::pair<int64_t, int32_t>> doneApps;
     // msgId → hosts waiting for the result push
     std::map<int32_t, std::vector<std::string>> appResultWaiters;
     std::map<int32_t, std::shared_ptr<BatchExecuteRequest>> evictedRequests;
@@ -86,6 +92,10 @@ class Planner
     bool registerHost(const Host& hostIn, bool overwrite);
     void removeHost(const Host& hostIn);
     std::vector<Host> getAvailableHosts();
+
+    size_t debugAppResultsCount();
+    size_t debugDoneAppsCount();
+    size_t debugInFlightCount();
 
     // Returns true if the app is already complete (or unknown); else
     // records host for a BATCH_DONE push on completion
@@ -126,10 +136,12 @@ class Planner
     int hostTimeoutMs = 5000;
 
   private:
+    void purgeOldResultsLocked();
     void setMessageResultLocked(
       const std::shared_ptr<Message>& msg,
       std::vector<std::pair<std::string, std::shared_ptr<Message>>>& waiters,
-      std::vector<std::pair<std::string, int32_t>>& batchWaiters);
+      std::vector<std::pair<std::string, int32_t>>& batchWaiters,
+      std::vector<std::pair<std::string, int32_t>>& groupClears);
 
     Planner();
 

@@ -38,6 +38,9 @@ enum class PointToPointCall : uint8_t
     IPC_ARENA = 6,   // sync: body = sender host, reply = IpcArenaInfo
     MESSAGE_IPC = 7, // async: IpcPtpMessage (payload already in arena)
     IPC_ACK = 8,     // async: IpcAck (receiver freed a segment)
+    // Group lifecycle: the planner broadcasts this to involved hosts
+    // when an app completes, so mappings/seq state don't accumulate
+    GROUP_CLEAR = 9, // async: PointToPointMessage (groupId only)
 };
 
 inline constexpr int32_t POINT_TO_POINT_MAIN_IDX = 0;
@@ -167,6 +170,14 @@ class PointToPointBroker
 
     // Group cleanup after an app finishes / migrates
     void clearGroup(int32_t groupId);
+
+    // Clear locally, or tell `host` to clear (used by the planner when
+    // an app completes)
+    void sendGroupClear(const std::string& host, int32_t groupId);
+
+    size_t debugMappingsCount();
+    size_t debugChannelsCount();
+    size_t debugSendSeqsCount();
     void clear();
 
     void updateHostForIdx(int32_t groupId,

@@ -540,11 +540,19 @@ void Planner::setMessageResults(
     // collected under the lock, sent after
     std::vector<std::pair<std::string, std::shared_ptr<Message>>> waiters;
     std::vector<std::pair<std::string, int32_t>> batchWaiters;
+    std::vector<std::pair<std::string, int32_t>> groupClears;
     {
         std::unique_lock lock(plannerMx);
         for (auto& msg : msgs) {
-            setMessageResultLocked(msg, waiters, batchWaiters);
+            setMessageResultLocked(msg, waiters, batchWaiters, groupClears);
         }
+    }
+
+    // Completed apps: retire their PTP group on every involved host
+    // (otherwise mappings/seq state accumulate forever)
+    auto& broker = getPointToPointBroker();
+    for (const auto& [host, groupId] : groupClears) {
+        broker.sendGroupClear(host, groupId);
     }
 
     // Batch fully complete: push BATCH_DONE to registered hosts
@@ -567,10 +575,34 @@ void Planner::setMessageResults(
     }
 }
 
+// Caller holds plannerMx. Results of completed apps are kept for
+// late getMessageResult/getBatchResults fetches, then purged by age
+// and count so a long-lived planner's memory stays bounded (the
+// reference only clears results on an explicit flush)
+void Planner::purgeOldResultsLocked()
+{
+    static const int64_t ttlMs =
+      getEnvVarInt("FAABRIC_RESULT_TTL_MS", 5 * 60 * 1000);
+    static const size_t maxDone =
+      (size_t)getEnvVarInt("FAABRIC_MAX_DONE_APPS", 512);
+    int64_t now = getGlobalClockEpochMillis();
+    while (!state.doneApps.empty() &&
+           (state.doneApps.size() > maxDone ||
+            now - state.doneApps.front().first > ttlMs)) {
+        int32_t appId = state.doneApps.front().second;
+        state.doneApps.pop_front();
+        // An app id re-entering flight (un-freeze) keeps its results
+        if (state.inFlightReqs.count(appId) == 0) {
+            state.appResults.erase(appId);
+        }
+    }
+}
+
 void Planner::setMessageResultLocked(
   const std::shared_ptr<Message>& msg,
   std::vector<std::pair<std::string, std::shared_ptr<Message>>>& waiters,
-  std::vector<std::pair<std::string, int32_t>>& batchWaiters)
+  std::vector<std::pair<std::string, int32_t>>& batchWaiters,
+  std::vector<std::pair<std::string, int32_t>>& groupClears)
 {
     int32_t appId = msg->appId;
     int32_t msgId = msg->id;
@@ -644,6 +676,18 @@ void Planner::setMessageResultLocked(
                         }
                         state.batchDoneWaiters.erase(bwIt);
                     }
+                    state.doneApps.emplace_back(
+                      getGlobalClockEpochMillis(), appId);
+                    purgeOldResultsLocked();
+                    if (decision->groupId != 0) {
+                        std::set<std::string> hosts(
+                          decision->hosts.begin(), decision->hosts.end());
+                        hosts.insert(getSystemConfig().endpointHost);
+                        for (const auto& h : hosts) {
+                            groupClears.emplace_back(h,
+                                                     decision->groupId);
+                        }
+                    }
                 }
             }
         }
@@ -673,6 +717,26 @@ bool Planner::registerBatchDoneWaiter(int32_t appId,
     }
     state.batchDoneWaiters[appId].push_back(host);
     return false;
+}
+
+size_t Planner::debugAppResultsCount()
+{
+    std::shared_lock lock(plannerMx);
+    size_t n = 0;
+    for (const auto& [a, m] : state.appResults) {
+        n += m.size();
+    }
+    return n;
+}
+size_t Planner::debugDoneAppsCount()
+{
+    std::shared_lock lock(plannerMx);
+    return state.doneApps.size();
+}
+size_t Planner::debugInFlightCount()
+{
+    std::shared_lock lock(plannerMx);
+    return state.inFlightReqs.size();
 }
 
 std::shared_ptr<Message> Planner::getMessageResult(const Message& msg)

@@ -734,6 +734,42 @@ size_t PointToPointBroker::recvMessageDevice(int32_t groupId,
     return p.host.size();
 }
 
+void PointToPointBroker::sendGroupClear(const std::string& host,
+                                        int32_t groupId)
+{
+    if (groupId == 0) {
+        return;
+    }
+    if (host == getSystemConfig().endpointHost) {
+        clearGroup(groupId);
+        return;
+    }
+    PointToPointMessage msg;
+    msg.groupId = groupId;
+    std::string body = msg.encode();
+    try {
+        getClient(host)->asyncSend(
+          (uint8_t)PointToPointCall::GROUP_CLEAR, body.data(), body.size());
+    } catch (const std::exception& e) {
+        FAM_ERROR("group clear to %s failed: %s", host.c_str(), e.what());
+    }
+}
+
+size_t PointToPointBroker::debugMappingsCount()
+{
+    std::lock_guard<std::mutex> lock(brokerMx);
+    return mappings.size();
+}
+size_t PointToPointBroker::debugChannelsCount()
+{
+    return channels.size();
+}
+size_t PointToPointBroker::debugSendSeqsCount()
+{
+    std::lock_guard<std::mutex> lock(sendSeqMx);
+    return sendSeqs.size();
+}
+
 void PointToPointBroker::clearGroup(int32_t groupId)
 {
     {
@@ -837,6 +873,11 @@ void PointToPointServer::doAsyncRecv(uint8_t code,
                                                       m.offset,
                                                       m.size,
                                                       seq);
+            break;
+        }
+        case PointToPointCall::GROUP_CLEAR: {
+            PointToPointMessage msg = PointToPointMessage::decode(body);
+            getPointToPointBroker().clearGroup(msg.groupId);
             break;
         }
         case PointToPointCall::IPC_ACK: {

@@ -279,3 +279,25 @@ def test_wait_batch_done_event_path(runtime):
     assert _core.wait_batch_done(987654321, 2_000)
     unknown = _core.get_batch_results(987654321)
     assert unknown.expected_num_messages == -1
+
+
+def test_runtime_state_bounded_after_batches(runtime):
+    """Completed apps must not accumulate runtime state: PTP group
+    mappings are retired by the planner's GROUP_CLEAR broadcast and
+    appResults is bounded by FAABRIC_RESULT_TTL_MS/FAABRIC_MAX_DONE_APPS
+    GC (a long-running planner previously leaked both)."""
+    before = _core._debug_runtime_sizes()
+    for _ in range(20):
+        ber = _core.batch_exec_factory("demo", "noop", SLOTS)
+        _core.call_functions(ber)
+        assert _core.wait_batch_done(ber.app_id, 10_000)
+    sizes = _core._debug_runtime_sizes()
+    assert sizes["planner_in_flight"] == 0
+    # Other tests in this module leave long-lived groups behind; the 20
+    # completed batches here must not add to the mapping/seq tables
+    assert sizes["broker_mappings"] <= before["broker_mappings"], (
+        before, sizes)
+    assert sizes["broker_send_seqs"] <= before["broker_send_seqs"], (
+        before, sizes)
+    # bounded, not necessarily zero (completed results kept for late fetch)
+    assert sizes["planner_done_apps"] <= 4096
