@@ -76,6 +76,12 @@ struct PlannerState
     std::map<int32_t, std::shared_ptr<BatchExecuteRequest>> evictedRequests;
     // appId → hosts to push BATCH_DONE to when the app's last result lands
     std::map<int32_t, std::vector<std::string>> batchDoneWaiters;
+    // appId → (groupId, hosts) of placements superseded by migrations;
+    // cleared (with their PTP state) when the app completes — mid-flight
+    // clearing could destroy unconsumed verdict messages
+    std::map<int32_t,
+             std::vector<std::pair<int32_t, std::vector<std::string>>>>
+      supersededGroups;
     std::set<std::string> nextEvictedHostIps;
     int numMigrations = 0;
 };

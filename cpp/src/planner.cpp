@@ -411,6 +411,13 @@ std::shared_ptr<SchedulingDecision> Planner::callBatch(
             }
 
             state.numMigrations += 1;
+            {
+                auto oldHosts = oldDec->uniqueHosts();
+                state.supersededGroups[appId].emplace_back(
+                  oldDec->groupId,
+                  std::vector<std::string>(oldHosts.begin(),
+                                           oldHosts.end()));
+            }
             updateBatchExecGroupId(*oldReq, newGroupId);
             state.inFlightReqs[appId] = { oldReq, decision };
             broker.setAndSendMappingsFromSchedulingDecision(*decision);
@@ -687,6 +694,20 @@ void Planner::setMessageResultLocked(
                             groupClears.emplace_back(h,
                                                      decision->groupId);
                         }
+                    }
+                    // Placements superseded by migrations are quiescent
+                    // once the app is done: retire them too
+                    auto sgIt = state.supersededGroups.find(appId);
+                    if (sgIt != state.supersededGroups.end()) {
+                        for (auto& [gid, hosts] : sgIt->second) {
+                            for (auto& h : hosts) {
+                                groupClears.emplace_back(std::move(h),
+                                                         gid);
+                            }
+                            groupClears.emplace_back(
+                              getSystemConfig().endpointHost, gid);
+                        }
+                        state.supersededGroups.erase(sgIt);
                     }
                 }
             }
