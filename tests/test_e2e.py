@@ -301,3 +301,39 @@ def test_runtime_state_bounded_after_batches(runtime):
         before, sizes)
     # bounded, not necessarily zero (completed results kept for late fetch)
     assert sizes["planner_done_apps"] <= 4096
+
+
+def test_concurrent_apps_interleaved(runtime):
+    """Many apps in flight at once: concurrent submitters + event waits
+    must not deadlock the dispatch pool (fan-outs serialize per batch)
+    or cross-wire results between apps."""
+    import threading
+
+    errors = []
+
+    def submitter(tid):
+        try:
+            for r in range(10):
+                # 4 submitters x 2 messages = SLOTS: fits capacity, so
+                # NOT_ENOUGH_SLOTS cannot occur
+                ber = _core.batch_exec_factory("demo", "echo", 2)
+                msgs = ber.messages
+                for m in msgs:
+                    m.input_data = f"t{tid}r{r}".encode()
+                ber.messages = msgs
+                d = _core.call_functions(ber)
+                assert d.app_id == ber.app_id, f"not scheduled: {d.app_id}"
+                assert _core.wait_batch_done(ber.app_id, 30_000)
+                status = _core.get_batch_results(ber.app_id)
+                assert status.finished
+                outs = {m.output_data for m in status.message_results}
+                assert outs == {f"t{tid}r{r}"}, outs
+        except Exception as e:  # pragma: no cover
+            errors.append(f"t{tid}: {e!r}")
+
+    threads = [threading.Thread(target=submitter, args=(t,)) for t in range(4)]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join(timeout=120)
+    assert not errors, errors
