@@ -297,6 +297,10 @@ enum class HttpMessageType : int32_t
     SET_POLICY = 13,
     GET_POLICY = 14,
     SET_NEXT_EVICTED_VM = 15,
+    // Runtime table sizes (results, in-flight, PTP state) for
+    // observability and leak monitoring — extension beyond the
+    // reference's op set
+    GET_RUNTIME_METRICS = 16,
 };
 
 struct InFlightAppEntry

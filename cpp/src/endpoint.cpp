@@ -3,6 +3,7 @@
 #include "faabricamd/endpoint.h"
 #include "faabricamd/json.h"
 #include "faabricamd/planner.h"
+#include "faabricamd/ptp.h"
 #include "faabricamd/scheduling.h"
 #include "faabricamd/util.h"
 
@@ -293,6 +294,24 @@ std::pair<int, std::string> PlannerEndpoint::handle(
             case HttpMessageType::SET_POLICY: {
                 planner.setPolicy(payload);
                 return { 200, "Policy set correctly" };
+            }
+            case HttpMessageType::GET_RUNTIME_METRICS: {
+                Json out = Json::object();
+                out["appResults"] =
+                  Json((int64_t)planner.debugAppResultsCount());
+                out["doneApps"] = Json((int64_t)planner.debugDoneAppsCount());
+                out["inFlightApps"] =
+                  Json((int64_t)planner.debugInFlightCount());
+                auto& broker = getPointToPointBroker();
+                out["ptpMappings"] =
+                  Json((int64_t)broker.debugMappingsCount());
+                out["ptpChannels"] =
+                  Json((int64_t)broker.debugChannelsCount());
+                out["ptpSendSeqs"] =
+                  Json((int64_t)broker.debugSendSeqsCount());
+                out["decisionCache"] =
+                  Json((int64_t)DecisionCache::get().size());
+                return { 200, out.dump() };
             }
             case HttpMessageType::GET_POLICY: {
                 return { 200, planner.getPolicy() };

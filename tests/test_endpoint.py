@@ -22,6 +22,7 @@ EXECUTE_BATCH = 10
 EXECUTE_BATCH_STATUS = 11
 SET_POLICY = 13
 GET_POLICY = 14
+GET_RUNTIME_METRICS = 16
 SET_NEXT_EVICTED_VM = 15
 
 
@@ -151,3 +152,16 @@ def test_chaining_and_exec_graph(runtime):
 def test_bad_request(runtime):
     status, _ = post(99)
     assert status == 400
+
+
+def test_get_runtime_metrics(runtime):
+    """GET_RUNTIME_METRICS exposes planner/broker table sizes (leak
+    monitoring; extension beyond the reference op set)."""
+    import json
+
+    status, body = post(GET_RUNTIME_METRICS)
+    assert status == 200
+    m = json.loads(body)
+    for key in ("appResults", "doneApps", "inFlightApps", "ptpMappings",
+                "ptpChannels", "ptpSendSeqs", "decisionCache"):
+        assert key in m and m[key] >= 0, m
