@@ -217,7 +217,16 @@ void Executor::executeTasks(std::vector<int> msgIdxs,
     } else if (!req->messages.empty() &&
                !req->messages[msgIdxs[0]].snapshotKey.empty() &&
                req->type != BatchExecuteType::THREADS) {
-        restore(req->messages[msgIdxs[0]].snapshotKey);
+        const std::string& key = req->messages[msgIdxs[0]].snapshotKey;
+        restore(key);
+        // Migration/freeze snapshots are single-use: drop them after
+        // landing in the arena (one leaked arena per migration
+        // otherwise). Fork (THREADS) snapshots are reused and cleaned
+        // by their own lifecycle.
+        if (startsWith(key, "migration_")) {
+            SnapshotRegistry::get().deleteSnapshot(key);
+            DeviceSnapshotRegistry::get().deleteSnapshot(key);
+        }
     }
 
     // Fault-driven dirty tracking of the restored arena (segfault mode;

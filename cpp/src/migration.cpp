@@ -52,13 +52,15 @@ int32_t migrationPoint(const std::vector<uint8_t>& reentryInput)
             auto [base, size] = exec->getMemoryView();
             auto snap = std::make_shared<SnapshotData>(
               std::vector<uint8_t>(base, base + size));
-            SnapshotRegistry::get().registerSnapshot(snapKey, snap);
             try {
                 getSnapshotClient(getSystemConfig().plannerHost)
                   ->pushSnapshot(snapKey, *snap);
             } catch (const std::exception& e) {
                 FAM_ERROR("freeze snapshot push failed: %s", e.what());
             }
+            // The planner owns the frozen state now; on thaw it pushes
+            // to whichever host is chosen (keeping a local copy here
+            // would leak one arena per freeze)
         }
         if (call.isMpi &&
             MpiWorldRegistry::get().worldExists(call.mpiWorldId)) {
@@ -120,10 +122,11 @@ int32_t migrationPoint(const std::vector<uint8_t>& reentryInput)
         if (size > 0) {
             auto snap = std::make_shared<SnapshotData>(
               std::vector<uint8_t>(base, base + size));
-            SnapshotRegistry::get().registerSnapshot(snapKey, snap);
             getSnapshotClient(migration->dstHost)
               ->pushSnapshot(snapKey, *snap);
             msg.snapshotKey = snapKey;
+            // No local registration: the destination restores and
+            // deletes it (single-use)
         }
     }
 

@@ -7,6 +7,7 @@
 #include "faabricamd/utilextras.h"
 #include "faabricamd/ptp.h"
 #include "faabricamd/scheduler.h"
+#include "faabricamd/ops.h"
 #include "faabricamd/snapshot.h"
 #include "faabricamd/state.h"
 #include "faabricamd/util.h"
@@ -560,6 +561,17 @@ void Planner::setMessageResults(
     auto& broker = getPointToPointBroker();
     for (const auto& [host, groupId] : groupClears) {
         broker.sendGroupClear(host, groupId);
+    }
+
+    // A completed (non-frozen) message consumed its freeze/migration
+    // snapshot: drop the planner's staged copy (one arena per freeze
+    // otherwise). Frozen results keep theirs for the thaw.
+    for (const auto& msg : msgs) {
+        if (msg->returnValue != FROZEN_FUNCTION_RETURN_VALUE &&
+            startsWith(msg->snapshotKey, "migration_")) {
+            SnapshotRegistry::get().deleteSnapshot(msg->snapshotKey);
+            DeviceSnapshotRegistry::get().deleteSnapshot(msg->snapshotKey);
+        }
     }
 
     // Batch fully complete: push BATCH_DONE to registered hosts
