@@ -226,7 +226,9 @@ class PointToPointBroker
                            int timeoutMs);
     void* sideStream(); // lazily-created HIP stream for D2D staging
 
-    Channel& getChannel(int32_t groupId, int32_t sendIdx, int32_t recvIdx);
+    std::shared_ptr<Channel> getChannel(int32_t groupId,
+                                        int32_t sendIdx,
+                                        int32_t recvIdx);
 
     std::mutex brokerMx;
     // (groupId, groupIdx) → host; (groupId, groupIdx) → mpi port

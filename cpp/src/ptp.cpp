@@ -358,15 +358,17 @@ void PointToPointBroker::updateHostForIdx(int32_t groupId,
     mappings[idxKey(groupId, groupIdx)] = newHost;
 }
 
-PointToPointBroker::Channel& PointToPointBroker::getChannel(int32_t groupId,
-                                                            int32_t sendIdx,
-                                                            int32_t recvIdx)
+std::shared_ptr<PointToPointBroker::Channel> PointToPointBroker::getChannel(
+  int32_t groupId,
+  int32_t sendIdx,
+  int32_t recvIdx)
 {
-    auto ch = channels.getOrCreate(chanKey(groupId, sendIdx, recvIdx),
-                                   [] { return std::make_shared<Channel>(); });
-    // shared_ptr stays owned by the map for the broker's lifetime, so
-    // the reference is stable
-    return *ch;
+    // Callers hold the shared_ptr across their wait: a concurrent
+    // clearGroup (app completion / migration retirement) may erase the
+    // map entry while a receiver still sleeps on the channel's cv —
+    // ownership keeps that a clean timeout instead of a use-after-free
+    return channels.getOrCreate(chanKey(groupId, sendIdx, recvIdx),
+                                [] { return std::make_shared<Channel>(); });
 }
 
 std::shared_ptr<MessageEndpointClient> PointToPointBroker::getClient(
@@ -431,7 +433,8 @@ void PointToPointBroker::deliverPayload(int32_t groupId,
                                         PtpPayload payload,
                                         uint32_t seq)
 {
-    Channel& ch = getChannel(groupId, sendIdx, recvIdx);
+    auto chPtr = getChannel(groupId, sendIdx, recvIdx);
+    Channel& ch = *chPtr;
     {
         std::lock_guard<std::mutex> lock(ch.mx);
         if (seq == NO_SEQ) {
@@ -496,7 +499,8 @@ PointToPointBroker::PtpPayload PointToPointBroker::recvPayload(
   bool mustOrderMsgs,
   int timeoutMs)
 {
-    Channel& ch = getChannel(groupId, sendIdx, recvIdx);
+    auto chPtr = getChannel(groupId, sendIdx, recvIdx);
+    Channel& ch = *chPtr;
     std::unique_lock<std::mutex> lock(ch.mx);
 
     auto ready = [&]() {
