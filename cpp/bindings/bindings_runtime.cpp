@@ -384,6 +384,26 @@ void initRuntimeBindings(py::module_& m)
         getPointToPointBroker().setUpLocalMappingsFromSchedulingDecision(d);
     });
     m.def("ptp_clear", [] { getPointToPointBroker().clear(); });
+    // Inject a message with an explicit sequence number, as the network
+    // would deliver it — lets tests exercise the out-of-order
+    // resequencing buffer directly
+    m.def("_test_ptp_deliver_seq",
+          [](int32_t appId,
+             int32_t groupId,
+             int32_t sendIdx,
+             int32_t recvIdx,
+             const py::bytes& data,
+             uint32_t seq) {
+              PointToPointMessage msg;
+              msg.appId = appId;
+              msg.groupId = groupId;
+              msg.sendIdx = sendIdx;
+              msg.recvIdx = recvIdx;
+              std::string s = data;
+              msg.data.assign(s.begin(), s.end());
+              py::gil_scoped_release release;
+              getPointToPointBroker().deliverRemoteMessage(msg, seq);
+          });
     // Standalone servers (multi-process PTP/IPC tests run these without
     // the rest of the worker fabric)
     py::class_<PointToPointServer>(m, "PointToPointServerHandle")

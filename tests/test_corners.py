@@ -326,3 +326,26 @@ def test_wire_decoders_survive_fuzz():
                 _core.Message.decode(cut)
             except Exception:
                 pass
+
+
+def test_ptp_out_of_order_resequencing():
+    """Ordered PTP delivery holds early-arriving sequence numbers in a
+    buffer and releases them in order (reference: PointToPointBroker
+    out-of-order buffer, src/transport/PointToPointBroker.cpp:778-859).
+    Injects seqs 2,0,1 as the network might deliver them."""
+    from faabric_amd import _core
+
+    g, s, r = 990001, 3, 4
+    for seq, payload in ((2, b"third"), (0, b"first"), (1, b"second")):
+        _core._test_ptp_deliver_seq(1, g, s, r, payload, seq)
+    got = [_core.ptp_recv(g, s, r, ordered=True, timeout_ms=2000)
+           for _ in range(3)]
+    assert got == [b"first", b"second", b"third"], got
+
+    # Unordered channel: arrival order wins, no buffering
+    g2 = 990002
+    for seq, payload in ((0, b"a"), (1, b"b")):
+        _core._test_ptp_deliver_seq(1, g2, s, r, payload, 0xFFFFFFFF)
+    got2 = {_core.ptp_recv(g2, s, r, ordered=False, timeout_ms=2000)
+            for _ in range(2)}
+    assert got2 == {b"a", b"b"}
