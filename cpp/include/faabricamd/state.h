@@ -9,6 +9,7 @@
 // dirty-mask partial push are kept.
 #pragma once
 
+#include <atomic>
 #include <map>
 #include <memory>
 #include <mutex>
@@ -145,6 +146,12 @@ class StateKeyValue
     std::mutex mirrorMx; // guards mirror bring-up only
     uint8_t* mirror = nullptr;
     std::vector<char> mirrorValid; // per 4 KiB page (owned by its stripe)
+    // Write-back dirty pages: writes memcpy into the mirror and mark
+    // here; sync() coalesces dirty runs into few H2D enqueues (per-op
+    // write-through enqueues convoyed on the HIP driver lock under
+    // 128-thread batches)
+    std::vector<char> mirrorDirty;
+    std::atomic<bool> mirrorAnyDirty{ false };
     bool mirrorFailed = false;
     bool mirrorUsable();
     // Runtime-capped stripe count (FAABRIC_KV_STRIPES, default 16)

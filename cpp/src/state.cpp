@@ -141,6 +141,7 @@ bool StateKeyValue::mirrorUsable()
         return false;
     }
     mirrorValid.assign((valueSize + MIRROR_PAGE - 1) / MIRROR_PAGE, 0);
+    mirrorDirty.assign(mirrorValid.size(), 0);
     stripesReady = true;
     mirror = (uint8_t*)pinned;
     return true;
@@ -233,6 +234,9 @@ void StateKeyValue::mirrorInvalidate(uint64_t offset, size_t len)
         for (size_t p = firstPage;
              p <= lastPage && p < mirrorValid.size(); p++) {
             mirrorValid[p] = 0;
+            // Direct-HBM landings are ordered after sync(); any dirty
+            // bit left here belongs to a write the landing overwrote
+            mirrorDirty[p] = 0;
         }
     });
 }
@@ -241,6 +245,38 @@ void StateKeyValue::sync()
 {
     if (mirror == nullptr) {
         return;
+    }
+    if (mirrorAnyDirty.exchange(false, std::memory_order_acq_rel)) {
+        (void)hipSetDevice(device);
+        // Coalesce dirty pages into maximal runs; each run is enqueued
+        // per stripe block under that stripe's lock (mutual exclusion
+        // with writers keeps the page content consistent). Pages
+        // dirtied concurrently re-set the flag and are committed by
+        // their writer's own sync.
+        size_t nPages = mirrorDirty.size();
+        size_t p = 0;
+        while (p < nPages) {
+            if (mirrorDirty[p] == 0) {
+                p++;
+                continue;
+            }
+            size_t q = p;
+            while (q < nPages && mirrorDirty[q] != 0) {
+                mirrorDirty[q] = 0;
+                q++;
+            }
+            uint64_t start = (uint64_t)p * MIRROR_PAGE;
+            size_t len =
+              std::min<uint64_t>((uint64_t)q * MIRROR_PAGE, valueSize) -
+              start;
+            forEachStripeRange(start, len, [&](int s, uint64_t o,
+                                               size_t n) {
+                (void)hipMemcpyAsync(devPtr + o, mirror + o, n,
+                                     hipMemcpyHostToDevice,
+                                     (hipStream_t)gStripes[s].stream);
+            });
+            p = q;
+        }
     }
     for (int s = 0; s < KV_STRIPES; s++) {
         std::lock_guard<std::mutex> lock(gStripes[s].mx);
@@ -262,6 +298,16 @@ static hipStream_t threadCopyStream()
         return s;
     }();
     return stream;
+}
+
+// Write-back (default): writes mark mirror pages dirty and sync()
+// coalesces the H2Ds. 0 = write-through (per-op H2D enqueue), kept for
+// A/B and as a safety valve.
+static bool kvWriteBack()
+{
+    static const bool v =
+      getEnvVarInt("FAABRIC_KV_WRITEBACK", 1) != 0;
+    return v;
 }
 
 static bool kvAsyncCopy()
@@ -303,10 +349,10 @@ void StateKeyValue::writeLocal(uint64_t offset,
 {
     if (onDevice) {
         if (mirrorUsable()) {
-            // Group-commit write-through: memcpy into the pinned mirror,
-            // enqueue the H2D on the stripe stream with no per-op sync
-            // (per-stripe stream order keeps HBM newest-wins), mark the
-            // pages readable. sync() is the durability point.
+            // Group-commit write-back: memcpy into the pinned mirror and
+            // mark the pages dirty; sync() (the durability point)
+            // coalesces dirty runs into few H2D enqueues. Reads serve
+            // from the mirror, so they always see the newest bytes.
             (void)hipSetDevice(device);
             forEachStripeRange(
               offset, len, [&](int s, uint64_t o, size_t n) {
@@ -331,13 +377,24 @@ void StateKeyValue::writeLocal(uint64_t offset,
                       mirrorFillLocked(s, lastPage * MIRROR_PAGE, 1);
                   }
                   std::memcpy(mirror + o, data + (o - offset), n);
-                  for (size_t p = firstPage; p <= lastPage; p++) {
-                      mirrorValid[p] = 1;
+                  if (kvWriteBack()) {
+                      for (size_t p = firstPage; p <= lastPage; p++) {
+                          mirrorValid[p] = 1;
+                          mirrorDirty[p] = 1;
+                      }
+                  } else {
+                      for (size_t p = firstPage; p <= lastPage; p++) {
+                          mirrorValid[p] = 1;
+                      }
+                      (void)hipMemcpyAsync(
+                        devPtr + o, mirror + o, n,
+                        hipMemcpyHostToDevice,
+                        (hipStream_t)gStripes[s].stream);
                   }
-                  (void)hipMemcpyAsync(
-                    devPtr + o, mirror + o, n, hipMemcpyHostToDevice,
-                    (hipStream_t)gStripes[s].stream);
               });
+            if (kvWriteBack()) {
+                mirrorAnyDirty.store(true, std::memory_order_release);
+            }
             return;
         }
         (void)hipSetDevice(device);
