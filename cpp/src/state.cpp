@@ -300,13 +300,17 @@ static hipStream_t threadCopyStream()
     return stream;
 }
 
-// Write-back (default): writes mark mirror pages dirty and sync()
-// coalesces the H2Ds. 0 = write-through (per-op H2D enqueue), kept for
-// A/B and as a safety valve.
+// Write policy, measured same-box (profiles/kv_writeback_ab.json):
+// write-through (default) enqueues one H2D per op from the worker
+// thread that did the write — the enqueues overlap batch execution.
+// Write-back (=1) marks pages dirty and coalesces H2Ds at sync(), but
+// that serializes the whole flush on the sync caller's critical path:
+// 23% slower at 128-function batches. Kept as a knob because the
+// trade flips for workloads with many tiny writes and rare syncs.
 static bool kvWriteBack()
 {
     static const bool v =
-      getEnvVarInt("FAABRIC_KV_WRITEBACK", 1) != 0;
+      getEnvVarInt("FAABRIC_KV_WRITEBACK", 0) != 0;
     return v;
 }
 
