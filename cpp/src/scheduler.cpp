@@ -244,12 +244,14 @@ std::shared_ptr<Executor> Scheduler::claimExecutor(Message& msg)
             auto& pool = *it->second;
             size_t n = pool.list.size();
             if (n > 0) {
-                size_t start = pool.hint.load(std::memory_order_relaxed);
+                // fetch_add hands every concurrent dispatcher a distinct
+                // starting slot; a shared load+store made 8 lanes chase
+                // the same executor and re-scan after CAS losses
+                size_t start =
+                  pool.hint.fetch_add(1, std::memory_order_relaxed);
                 for (size_t k = 0; k < n; k++) {
                     size_t i = (start + k) % n;
                     if (pool.list[i]->tryClaim()) {
-                        pool.hint.store((i + 1) % n,
-                                        std::memory_order_relaxed);
                         return pool.list[i];
                     }
                 }
