@@ -273,6 +273,14 @@ bool TcpConnection::recvFrame(WireHeader& hdr, std::string& body)
         need -= (size_t)n;
     }
 
+    // Sanity-cap the frame before allocating: a corrupt or hostile
+    // header must drop the connection, not OOM the process
+    static const uint64_t maxFrame =
+      (uint64_t)getEnvVarInt("FAABRIC_MAX_FRAME_MB", 1024) * 1024 * 1024;
+    if (hdr.size > maxFrame) {
+        throw SocketClosedException(
+          "frame size " + std::to_string(hdr.size) + " exceeds cap");
+    }
     body.resize(hdr.size);
     if (hdr.size > 0) {
         recvAll(body.data(), hdr.size);

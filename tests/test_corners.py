@@ -376,3 +376,55 @@ def test_ptp_large_payload_roundtrip():
     got = _core.ptp_recv(g, s, r, ordered=True, timeout_ms=10_000)
     assert len(got) == len(payload)
     assert hashlib.sha256(got).hexdigest() == hashlib.sha256(payload).hexdigest()
+
+
+def test_servers_survive_garbage_frames():
+    """Hostile/corrupt input at live service ports: random op codes,
+    random bodies, oversized length headers (the frame cap drops the
+    connection instead of allocating), truncated frames. Servers must
+    keep serving valid RPCs afterwards."""
+    import socket
+    import struct
+
+    from faabric_amd import _core
+    from faabric_amd.runtime import LocalRuntime
+
+    off = 8700
+    rt = LocalRuntime(port_offset=off, planner_port_offset=off, slots=2)
+    rt.start_planner(with_snapshot_server=False)
+    rt.start_worker()
+    try:
+        random.seed(13)
+        # planner async 8011+2? ports: planner sync/async + function pair
+        for port in (8011 + off, 8012 + off, 8005 + off, 8006 + off):
+            for trial in range(12):
+                s = socket.create_connection(("127.0.0.1", port), timeout=5)
+                try:
+                    kind = trial % 4
+                    if kind == 0:  # random junk, not even a header
+                        s.sendall(bytes(random.getrandbits(8)
+                                        for _ in range(random.randrange(1, 40))))
+                    elif kind == 1:  # valid header, random code + body
+                        body = bytes(random.getrandbits(8)
+                                     for _ in range(random.randrange(0, 64)))
+                        s.sendall(struct.pack("<B3xIQ",
+                                              random.randrange(0, 255),
+                                              0, len(body)) + body)
+                    elif kind == 2:  # hostile size field (would be 2^60 B)
+                        s.sendall(struct.pack("<B3xIQ", 1, 0, 1 << 60))
+                    else:  # truncated: header promises more than sent
+                        s.sendall(struct.pack("<B3xIQ", 1, 0, 1 << 20))
+                        s.sendall(b"short")
+                finally:
+                    s.close()
+        # All servers still answer real traffic
+        hosts = _core.get_available_hosts()
+        assert len(hosts) == 1
+        ber = _core.batch_exec_factory("corner", "noop2", 1)
+        _core.register_native_noop("corner", "noop2")
+        _core.call_functions(ber)
+        from faabric_amd.runtime import wait_for_batch
+        rs = wait_for_batch(ber.app_id, 1, 20_000)
+        assert rs[0].return_value == 0
+    finally:
+        rt.stop()
