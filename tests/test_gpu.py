@@ -433,3 +433,47 @@ def test_device_scan_single_rank(runtime):
     results = submit_mpi("gpu", "scan1", 1)
     assert results[0].return_value == 0, results[0].output_data
     assert results[0].output_data == "scan ok"
+
+
+@pytest.mark.gpu
+def test_state_kv_mirror_coherence_writeback():
+    """Same mirror-coherence semantics under FAABRIC_KV_WRITEBACK=1
+    (dirty pages flushed coalesced at sync); run in a subprocess because
+    the policy is a process-wide static."""
+    import os
+    import subprocess
+    import sys as _sys
+
+    REPO_ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    code = """
+import sys
+sys.path.insert(0, %r)
+from faabric_amd import _core
+import torch
+
+_core.set_log_level("error")
+n = 64 * 1024
+kv = _core.state_get_kv_device("gpu", "wbkey", n)
+kv.set(b"\\x11" * n)
+kv.sync()
+add = torch.full((n,), 0x22, dtype=torch.uint8, device="cuda")
+torch.cuda.synchronize()
+_core.device_elementwise_op(kv.data_ptr, add.data_ptr(), n, 5, 0)
+torch.cuda.synchronize()
+assert kv.get_chunk(0, 16) == bytes([0x33]) * 16
+kv.set_chunk(100, b"\\x44" * 8)
+got = kv.get_chunk(96, 16)
+assert got == bytes([0x33]) * 4 + bytes([0x44]) * 8 + bytes([0x33]) * 4
+kv.sync()
+out = torch.zeros(n, dtype=torch.uint8, device="cuda")
+torch.cuda.synchronize()
+_core.device_elementwise_op(out.data_ptr(), kv.data_ptr, n, 5, 0)
+torch.cuda.synchronize()
+assert int(out[100]) == 0x44 and int(out[99]) == 0x33 and int(out[50000]) == 0x33
+print("WB_COHERENCE_OK")
+""" % REPO_ROOT
+    env = dict(os.environ, FAABRIC_KV_WRITEBACK="1")
+    r = subprocess.run([_sys.executable, "-c", code], env=env,
+                       capture_output=True, text=True, timeout=180)
+    assert r.returncode == 0, r.stdout + r.stderr
+    assert "WB_COHERENCE_OK" in r.stdout
