@@ -4,7 +4,7 @@ pinned mirror, asserting steady-state RSS and bounded runtime tables
 (result GC + GROUP_CLEAR retirement). Catches per-batch leaks the
 functional tests can't see.
 
-Run on an MI355X box: python tools/gpu_batch_soak.py [seconds]
+Run on an MI355X box: python tools/gpu_batch_soak.py [seconds] [batch]
 """
 
 import os
@@ -24,13 +24,15 @@ def rss_mb():
 
 def main():
     seconds = int(sys.argv[1]) if len(sys.argv) > 1 else 120
+    batch_n = int(sys.argv[2]) if len(sys.argv) > 2 else 128
     _core.set_log_level("error")
-    rt = LocalRuntime(port_offset=7700, planner_port_offset=7700, slots=129)
+    rt = LocalRuntime(port_offset=7700, planner_port_offset=7700,
+                      slots=batch_n + 1)
     rt.start_planner(with_snapshot_server=False)
     rt.start_worker()
     _core.register_bench_functions()
 
-    batch = 128
+    batch = batch_n
     for _ in range(10):
         ber = _core.batch_exec_factory("bench", "kvtouch", batch)
         _core.call_functions(ber)
