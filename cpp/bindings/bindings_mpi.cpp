@@ -266,6 +266,25 @@ void initMpiBindings(py::module_& m)
               return py::bytes((const char*)out.data(), out.size());
           });
 
+    m.def("mpi_reducescatter_bytes",
+          [](int rank, const py::bytes& data, MpiDataType dtype,
+             int recvCount, MpiOp op) {
+              std::string s = data;
+              std::vector<uint8_t> out((size_t)recvCount *
+                                       mpiTypeSize(dtype));
+              {
+                  py::gil_scoped_release release;
+                  world().reduceScatter(rank,
+                                        (const uint8_t*)s.data(),
+                                        out.data(),
+                                        dtype,
+                                        recvCount,
+                                        op,
+                                        MpiBufferLoc::HOST);
+              }
+              return py::bytes((const char*)out.data(), out.size());
+          });
+
     m.def("mpi_sendrecv_bytes",
           [](int rank, int sendTo, int recvFrom, const py::bytes& data) {
               std::string s = data;

@@ -84,6 +84,16 @@ def mpi_payload(msg):
     )
     assert got == [sum(range(1, rank + 2))]
 
+    # reduce_scatter: each rank contributes [r*size+j for j], gets the
+    # SUM over ranks of its own slice
+    rs_send = ints(*[rank * size + j for j in range(size)])
+    got = unints(
+        _core.mpi_reducescatter_bytes(
+            rank, rs_send, _core.MpiDataType.INT32, 1, _core.MpiOp.SUM
+        )
+    )
+    assert got == [sum(r * size + rank for r in range(size))], got
+
     # sendrecv around the ring
     got = _core.mpi_sendrecv_bytes(
         rank, (rank + 1) % size, (rank - 1 + size) % size, ints(rank)
